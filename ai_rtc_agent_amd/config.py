@@ -1,0 +1,173 @@
+"""Configuration surface.
+
+Mirrors the reference's three config mechanisms:
+CLI flags (reference agent.py:441-455), environment variables
+(reference docs/environment.md:3-25, Dockerfile:54-56, lib/tracks.py:17-18),
+and the runtime API (POST /config + datachannel JSON — reference
+agent.py:398-412). Pipeline hyperparameter defaults follow the reference's
+canonical config (reference lib/pipeline.py:11-14, lib/wrapper.py:46-65).
+"""
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass, field
+from typing import List, Optional, Sequence
+
+
+def _env_bool(name: str, default: bool = False) -> bool:
+    v = os.environ.get(name)
+    if v is None:
+        return default
+    return v.strip().lower() not in ("", "0", "false", "no", "off")
+
+
+def _env_int(name: str, default: int) -> int:
+    v = os.environ.get(name)
+    return int(v) if v not in (None, "") else default
+
+
+def _env_float(name: str, default: float) -> float:
+    v = os.environ.get(name)
+    return float(v) if v not in (None, "") else default
+
+
+# ---------------------------------------------------------------------------
+# Environment variable surface (parity list)
+# ---------------------------------------------------------------------------
+# AUTH_TOKEN / WEBHOOK_URL           -> webhook events (utils/events.py)
+# TURN_* (replaces TWILIO_*)         -> ICE/TURN provisioning (media/ice.py)
+# WARMUP_FRAMES / DROP_FRAMES        -> track adapter (media/tracks.py)
+# ENGINES_CACHE (was TRT_ENGINES_CACHE) -> kernel-plan/graph cache dir
+# CIVITAI_CACHE / HF_HOME / HF_HUB_CACHE -> model asset caches
+# VCN_ENC / VCN_DEC (was NVENC/NVDEC) -> hardware codec toggles
+# VCN_ENC_PRESET / VCN_ENC_TUNING_INFO / VCN_ENC_{DEFAULT,MIN,MAX}_BITRATE
+#                                     -> encoder tunables (5-knob parity with
+#                                        NVENC_* in reference docs/environment.md:17-25)
+
+
+def engines_cache_dir() -> str:
+    return os.environ.get(
+        "ENGINES_CACHE", os.environ.get("TRT_ENGINES_CACHE", "./models/engines")
+    )
+
+
+def civitai_cache_dir() -> str:
+    # reference lib/utils.py:6-10
+    return os.environ.get("CIVITAI_CACHE", "./models/civitai")
+
+
+def warmup_frames() -> int:
+    # reference lib/tracks.py:17 (default 10)
+    return _env_int("WARMUP_FRAMES", 10)
+
+
+def drop_frames() -> int:
+    # reference lib/tracks.py:18 (default 0)
+    return _env_int("DROP_FRAMES", 0)
+
+
+def hw_encode_enabled() -> bool:
+    # reference Dockerfile:54 NVENC=true; we accept both spellings
+    return _env_bool("VCN_ENC", _env_bool("NVENC", False))
+
+
+def hw_decode_enabled() -> bool:
+    return _env_bool("VCN_DEC", _env_bool("NVDEC", False))
+
+
+@dataclass
+class EncoderConfig:
+    """VCN encoder knobs — parity with reference NVENC_* env surface."""
+
+    preset: str = field(default_factory=lambda: os.environ.get("VCN_ENC_PRESET", os.environ.get("NVENC_PRESET", "P3")))
+    tuning_info: str = field(default_factory=lambda: os.environ.get("VCN_ENC_TUNING_INFO", os.environ.get("NVENC_TUNING_INFO", "low_latency")))
+    default_bitrate: int = field(default_factory=lambda: _env_int("VCN_ENC_DEFAULT_BITRATE", _env_int("NVENC_DEFAULT_BITRATE", 4_000_000)))
+    min_bitrate: int = field(default_factory=lambda: _env_int("VCN_ENC_MIN_BITRATE", _env_int("NVENC_MIN_BITRATE", 1_000_000)))
+    max_bitrate: int = field(default_factory=lambda: _env_int("VCN_ENC_MAX_BITRATE", _env_int("NVENC_MAX_BITRATE", 8_000_000)))
+
+
+@dataclass
+class SimilarityFilterConfig:
+    """Stochastic similarity filter (reference lib/wrapper.py:57-59,192-195)."""
+
+    enabled: bool = False
+    threshold: float = 0.98
+    max_skip_frame: int = 10
+
+
+@dataclass
+class EngineConfig:
+    """Full pipeline configuration.
+
+    Defaults reproduce the reference's production config
+    (reference lib/pipeline.py:11-14,23-36; lib/wrapper.py:46-65).
+    """
+
+    model_id: str = "lykon/dreamshaper-8"
+    model_family: str = "sd15"  # sd15 | sd21 (sd-turbo) | sdxl
+    width: int = 512
+    height: int = 512
+    t_index_list: List[int] = field(default_factory=lambda: [18, 26, 35, 45])
+    num_inference_steps: int = 50
+    guidance_scale: float = 0.0
+    cfg_type: str = "self"  # none | full | self | initialize
+    delta: float = 1.0
+    frame_buffer_size: int = 1
+    use_denoising_batch: bool = True
+    use_lcm_lora: bool = True
+    use_tiny_vae: bool = True
+    do_add_noise: bool = True
+    dtype: str = "float16"
+    seed: int = 2
+    mode: str = "img2img"  # img2img | txt2img
+    prompt: str = "fireworks in the night sky"  # reference lib/pipeline.py:11
+    negative_prompt: str = ""
+    lora_dict: Optional[dict] = None
+    lcm_lora_id: Optional[str] = None
+    vae_id: Optional[str] = None
+    similarity_filter: SimilarityFilterConfig = field(default_factory=SimilarityFilterConfig)
+    encoder: EncoderConfig = field(default_factory=EncoderConfig)
+    device: str = "cuda"
+    # engine acceleration: "hip" (hand-written kernels + hipGraph) or "eager"
+    acceleration: str = "hip"
+    use_hip_graph: bool = True
+
+    @property
+    def denoising_steps(self) -> int:
+        return len(self.t_index_list)
+
+    @property
+    def unet_batch(self) -> int:
+        """The stream-batch law (reference lib/wrapper.py:159-163)."""
+        if self.use_denoising_batch:
+            b = self.denoising_steps * self.frame_buffer_size
+            if self.cfg_type == "initialize":
+                b += self.frame_buffer_size
+            elif self.cfg_type == "full":
+                b *= 2
+            return b
+        return self.frame_buffer_size
+
+    @property
+    def latent_height(self) -> int:
+        return self.height // 8
+
+    @property
+    def latent_width(self) -> int:
+        return self.width // 8
+
+
+def sd_turbo_config(**kw) -> EngineConfig:
+    """BASELINE.json headline config: SD-Turbo 512x512 1-step img2img."""
+    defaults = dict(
+        model_id="stabilityai/sd-turbo",
+        model_family="sd21",
+        t_index_list=[0],
+        num_inference_steps=1,
+        guidance_scale=0.0,
+        cfg_type="none",
+        use_lcm_lora=False,
+        use_tiny_vae=True,
+    )
+    defaults.update(kw)
+    return EngineConfig(**defaults)

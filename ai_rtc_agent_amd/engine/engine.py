@@ -1,0 +1,331 @@
+"""StreamDiffusionEngine — the per-frame diffusion engine.
+
+From-scratch replacement for reference L4+L5 (lib/wrapper.py +
+the external StreamDiffusion package). Public surface mirrors the
+reference wrapper contract:
+
+- ctor option surface            (reference lib/wrapper.py:34-132)
+- prepare(prompt, steps, g)      (reference lib/wrapper.py:197-234)
+- __call__/img2img/txt2img       (reference lib/wrapper.py:236-343)
+- update_prompt                  (reference lib/pipeline.py:44-45)
+- update_t_index_list            (reference lib/wrapper.py:389-407)
+- stream-batch law B = len(t_index)*frame_buffer (lib/wrapper.py:159-163)
+
+MI355X-native execution: the whole per-frame step (VAE encode -> stream-batch
+UNet -> scheduler -> VAE decode) runs over static device buffers so it can be
+captured ONCE into a hipGraph (torch.cuda.CUDAGraph is hipGraph on ROCm) and
+replayed per frame — this replaces the reference's TensorRT engine loading
+(lib/wrapper.py:409-512). Prompt and t_index updates write into
+graph-external buffers in place; the graph is never re-captured
+(SURVEY.md §7 hard part #2).
+"""
+from __future__ import annotations
+
+import time
+from typing import List, Optional, Sequence
+
+import torch
+
+from ..config import EngineConfig
+from ..models import TinyVAE, TextEncoder, UNet2DCondition, UNetConfig
+from ..models.lora import fuse_lora_state_dict, load_lora_file, make_random_lora
+from ..utils.timers import StageTimers
+from .. import ops
+from .rcfg import ResidualCFG
+from .scheduler import StreamScheduler
+from .similarity import StochasticSimilarityFilter
+
+
+def _unet_config_for(family: str) -> UNetConfig:
+    return {
+        "sd15": UNetConfig.sd15,
+        "sd21": UNetConfig.sd21,
+        "sdxl": UNetConfig.sdxl,
+        "tiny": UNetConfig.tiny,
+    }[family]()
+
+
+class StreamDiffusionEngine:
+    def __init__(
+        self,
+        cfg: EngineConfig,
+        unet: Optional[UNet2DCondition] = None,
+        vae: Optional[TinyVAE] = None,
+        text_encoder: Optional[TextEncoder] = None,
+    ) -> None:
+        self.cfg = cfg
+        self.device = torch.device(cfg.device if torch.cuda.is_available() or cfg.device == "cpu" else "cpu")
+        self.dtype = getattr(torch, cfg.dtype) if self.device.type == "cuda" else torch.float32
+        torch.manual_seed(cfg.seed)
+
+        self.scheduler = StreamScheduler(num_inference_steps=cfg.num_inference_steps)
+        self.rcfg = ResidualCFG(cfg.cfg_type, cfg.guidance_scale, cfg.delta)
+        self.sim_filter: Optional[StochasticSimilarityFilter] = None
+        if cfg.similarity_filter.enabled:
+            gen = torch.Generator().manual_seed(cfg.seed)
+            self.sim_filter = StochasticSimilarityFilter(
+                cfg.similarity_filter.threshold, cfg.similarity_filter.max_skip_frame, gen
+            )
+
+        ucfg = _unet_config_for(cfg.model_family)
+        self.unet = unet if unet is not None else UNet2DCondition(ucfg)
+        self.vae = vae if vae is not None else TinyVAE()
+        self.text_encoder = text_encoder if text_encoder is not None else TextEncoder(
+            hidden=ucfg.cross_attention_dim, layers=2 if cfg.model_family == "tiny" else 12
+        )
+        self.ctx_dim = ucfg.cross_attention_dim
+
+        # LoRA fusion happens BEFORE device placement / graph capture
+        # (reference fuses before TRT compile, lib/wrapper.py:645-697).
+        if cfg.use_lcm_lora:
+            sd = (
+                load_lora_file(cfg.lcm_lora_id)
+                if cfg.lcm_lora_id
+                else make_random_lora(self.unet, rank=4, seed=cfg.seed)
+            )
+            fuse_lora_state_dict(self.unet, sd, scale=1.0)
+        if cfg.lora_dict:
+            for path, scale in cfg.lora_dict.items():
+                fuse_lora_state_dict(self.unet, load_lora_file(path), scale=scale)
+
+        self.unet = self.unet.to(self.device, self.dtype).eval()
+        self.vae = self.vae.to(self.device, self.dtype).eval()
+        self.text_encoder = self.text_encoder.to(self.device).eval()
+
+        self.timers = StageTimers(use_cuda=self.device.type == "cuda")
+        self._graph: Optional[torch.cuda.CUDAGraph] = None
+        self._prepared = False
+        self.prompt: str = cfg.prompt
+
+    # ------------------------------------------------------------------
+    # prepare
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def prepare(
+        self,
+        prompt: Optional[str] = None,
+        num_inference_steps: Optional[int] = None,
+        guidance_scale: Optional[float] = None,
+        t_index_list: Optional[Sequence[int]] = None,
+    ) -> None:
+        cfg = self.cfg
+        if prompt is not None:
+            self.prompt = prompt
+        if num_inference_steps is not None:
+            cfg.num_inference_steps = num_inference_steps
+            self.scheduler = StreamScheduler(num_inference_steps=num_inference_steps)
+        if guidance_scale is not None:
+            cfg.guidance_scale = guidance_scale
+            self.rcfg = ResidualCFG(cfg.cfg_type, guidance_scale, cfg.delta)
+        if t_index_list is not None:
+            cfg.t_index_list = list(t_index_list)
+
+        n = cfg.denoising_steps
+        fbs = cfg.frame_buffer_size
+        B = n * fbs
+        lh, lw = cfg.latent_height, cfg.latent_width
+        dev, dt = self.device, self.dtype
+
+        self._coeff = self.scheduler.coefficients(cfg.t_index_list, fbs, dev, torch.float32)
+        # NHWC coefficient views: (B,1,1,1) already broadcast over (B,h,w,c)
+        self._embeds = self.text_encoder.encode(self.prompt, dev, dt)
+        self._embeds_batch = self._embeds.expand(B, -1, -1).contiguous()
+        if cfg.cfg_type == "full":
+            neg = self.text_encoder.encode(cfg.negative_prompt, dev, dt)
+            self._embeds_full = torch.cat([neg.expand(B, -1, -1), self._embeds_batch], dim=0).contiguous()
+
+        g = torch.Generator(device="cpu").manual_seed(cfg.seed)
+        self._init_noise = torch.randn((B, lh, lw, 4), generator=g).to(dev, dt)
+        self._x_t_buffer = torch.zeros((max(0, B - fbs), lh, lw, 4), device=dev, dtype=dt)
+        self.rcfg.reset(self._init_noise)
+
+        # static I/O buffers (graph-stable addresses)
+        self._frame_in = torch.zeros((fbs, cfg.height, cfg.width, 3), device=dev, dtype=torch.uint8)
+        self._img_in = torch.zeros((fbs, cfg.height, cfg.width, 3), device=dev, dtype=dt)
+        self._ts_batch = self._coeff["sub_timesteps_tensor"].to(dev)
+        self._prev_out: Optional[torch.Tensor] = None
+        self._graph = None
+        self._prepared = True
+
+    # ------------------------------------------------------------------
+    # runtime config updates (POST /config + datachannel)
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def update_prompt(self, prompt: str) -> None:
+        """Re-encode the prompt and overwrite the cached embeddings IN PLACE
+        (graph-external buffer update; reference lib/pipeline.py:44-45)."""
+        self.prompt = prompt
+        emb = self.text_encoder.encode(prompt, self.device, self.dtype)
+        self._embeds.copy_(emb)
+        self._embeds_batch.copy_(emb.expand_as(self._embeds_batch))
+        if self.cfg.cfg_type == "full":
+            B = self._embeds_batch.shape[0]
+            self._embeds_full[B:].copy_(self._embeds_batch)
+
+    @torch.no_grad()
+    def update_t_index_list(self, t_index_list: Sequence[int]) -> None:
+        """Contract of reference lib/wrapper.py:389-407: no-op when
+        unchanged; length changes require prepare() (batch shape changes)."""
+        t_index_list = list(t_index_list)
+        if t_index_list == self.cfg.t_index_list:
+            return
+        if len(t_index_list) != len(self.cfg.t_index_list):
+            self.cfg.t_index_list = t_index_list
+            self.prepare()
+            return
+        self.cfg.t_index_list = t_index_list
+        new = self.scheduler.coefficients(
+            t_index_list, self.cfg.frame_buffer_size, self.device, torch.float32
+        )
+        for k in ("alpha_prod_t_sqrt", "beta_prod_t_sqrt", "c_skip", "c_out"):
+            self._coeff[k].copy_(new[k])
+        self._coeff["sub_timesteps"] = new["sub_timesteps"]
+        self._ts_batch.copy_(new["sub_timesteps_tensor"].to(self.device))
+
+    # ------------------------------------------------------------------
+    # core step (graph-capturable: static shapes, static buffers)
+    # ------------------------------------------------------------------
+    def _unet_batch_input(self, x_t: torch.Tensor) -> torch.Tensor:
+        cfg = self.cfg
+        if cfg.cfg_type == "full" and self.rcfg.active:
+            return torch.cat([x_t, x_t], dim=0)
+        if cfg.cfg_type == "initialize" and self.rcfg.active:
+            return torch.cat([x_t[: cfg.frame_buffer_size], x_t], dim=0)
+        return x_t
+
+    def _unet_batch_embeds(self) -> torch.Tensor:
+        cfg = self.cfg
+        if cfg.cfg_type == "full" and self.rcfg.active:
+            return self._embeds_full
+        if cfg.cfg_type == "initialize" and self.rcfg.active:
+            return torch.cat(
+                [self._embeds_batch[: cfg.frame_buffer_size], self._embeds_batch], dim=0
+            )
+        return self._embeds_batch
+
+    def _unet_batch_timesteps(self) -> torch.Tensor:
+        cfg = self.cfg
+        if cfg.cfg_type == "full" and self.rcfg.active:
+            return torch.cat([self._ts_batch, self._ts_batch], dim=0)
+        if cfg.cfg_type == "initialize" and self.rcfg.active:
+            return torch.cat([self._ts_batch[: cfg.frame_buffer_size], self._ts_batch], dim=0)
+        return self._ts_batch
+
+    @torch.no_grad()
+    def _step_core(self) -> torch.Tensor:
+        """One stream-batch denoise round over static buffers.
+
+        _img_in (fbs,H,W,3 dtype) -> output image (fbs,H,W,3 dtype).
+        """
+        cfg = self.cfg
+        fbs = cfg.frame_buffer_size
+        co = self._coeff
+
+        if cfg.mode == "img2img":
+            x0 = self.vae.encode(self._img_in)
+            x_t0 = self.scheduler.add_noise(
+                x0,
+                self._init_noise[:fbs],
+                co["alpha_prod_t_sqrt"][:fbs],
+                co["beta_prod_t_sqrt"][:fbs],
+            ) if cfg.do_add_noise else x0
+        else:  # txt2img: stage-0 input is pure noise
+            x_t0 = self._init_noise[:fbs]
+
+        x_t = torch.cat([x_t0, self._x_t_buffer], dim=0) if self._x_t_buffer.shape[0] else x_t0
+
+        eps = self.unet(
+            self._unet_batch_input(x_t),
+            self._unet_batch_timesteps(),
+            self._unet_batch_embeds(),
+        )
+        eps = self.rcfg.apply(eps, fbs)
+        denoised = self.scheduler.step_batch(eps, x_t, co)
+
+        if denoised.shape[0] > fbs:
+            # shift: stage i output -> stage i+1 input at tau_{i+1}
+            nxt = self.scheduler.add_noise(
+                denoised[:-fbs],
+                self._init_noise[fbs:],
+                co["alpha_prod_t_sqrt"][fbs:],
+                co["beta_prod_t_sqrt"][fbs:],
+            ) if cfg.do_add_noise else denoised[:-fbs]
+            self._x_t_buffer.copy_(nxt)
+
+        return self.vae.decode(denoised[-fbs:])
+
+    def _maybe_capture(self) -> None:
+        if (
+            self._graph is not None
+            or self.device.type != "cuda"
+            or not self.cfg.use_hip_graph
+        ):
+            return
+        # warmup on a side stream, then capture (hipGraph on ROCm)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._graph_out = self._step_core()
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._graph_out = self._step_core()
+        self._graph = g
+
+    # ------------------------------------------------------------------
+    # public frame API
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def __call__(self, frame_u8: torch.Tensor) -> torch.Tensor:
+        """frame_u8: (H,W,3) or (fbs,H,W,3) uint8 RGB on any device.
+        Returns stylised (H,W,3) / (fbs,H,W,3) uint8 RGB on self.device."""
+        assert self._prepared, "call prepare() first"
+        squeeze = frame_u8.dim() == 3
+        if squeeze:
+            frame_u8 = frame_u8.unsqueeze(0)
+        t_in = time.perf_counter()
+
+        with self.timers.stage("preprocess"):
+            frame_u8 = frame_u8.to(self.device, non_blocking=True)
+            img = ops.preprocess_from_u8(frame_u8, self.dtype)
+
+        if self.sim_filter is not None and self._prev_out is not None:
+            if self.sim_filter.should_skip(img):
+                out = self._prev_out
+                self.timers.frame_done()
+                return out[0] if squeeze else out
+
+        with self.timers.stage("diffusion"):
+            self._img_in.copy_(img)
+            if self.device.type == "cuda" and self.cfg.use_hip_graph:
+                self._maybe_capture()
+                self._graph.replay()
+                decoded = self._graph_out
+            else:
+                decoded = self._step_core()
+
+        with self.timers.stage("postprocess"):
+            out = ops.postprocess_to_u8(decoded)
+
+        self._prev_out = out
+        self.timers.frame_done()
+        return out[0] if squeeze else out
+
+    # reference wrapper exposes explicit img2img/txt2img entry points
+    @torch.no_grad()
+    def img2img(self, frame_u8: torch.Tensor) -> torch.Tensor:
+        assert self.cfg.mode == "img2img"
+        return self(frame_u8)
+
+    @torch.no_grad()
+    def txt2img(self) -> torch.Tensor:
+        assert self.cfg.mode == "txt2img"
+        fbs = self.cfg.frame_buffer_size
+        dummy = torch.zeros(
+            (fbs, self.cfg.height, self.cfg.width, 3), dtype=torch.uint8, device=self.device
+        )
+        return self(dummy)
+
+    def stats(self) -> dict:
+        return self.timers.snapshot()

@@ -1,0 +1,211 @@
+"""Op dispatch: hand-written HIP/CDNA4 kernels on GPU, torch reference on CPU.
+
+The GPU activation layout is NHWC throughout (implicit-GEMM friendly on MFMA:
+the K = (r,s,c) gather reads 16B-contiguous channel runs — see
+ops/csrc/conv2d_mfma.hip). The torch reference path permutes to NCHW for
+F.conv2d and back; it exists for CPU tests and as the numerics golden that
+GPU tests compare the HIP kernels against (SURVEY.md §4 item b).
+
+Replaces (MI355X-natively) reference components N3-N7 of SURVEY.md §2.2:
+CV-CUDA convertto/reformat (lib/pipeline.py:61-63) -> fused pre/post kernels;
+TensorRT UNet/VAE engines (lib/wrapper.py:409-512) -> these kernels + hipGraph.
+
+Policy: on a CUDA/HIP device the HIP extension is REQUIRED — ops raise if it
+is missing rather than silently falling back to eager torch.
+"""
+from __future__ import annotations
+
+import os
+
+import torch
+import torch.nn.functional as F
+
+_EXT = None
+_EXT_ERR: str | None = None
+
+
+def hip_ext():
+    """Load the in-tree HIP extension (built by setup.py / __graft_entry__.build)."""
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        from . import _load_ext
+
+        _EXT = _load_ext.load()
+    except Exception as e:  # pragma: no cover - exercised only on GPU boxes
+        _EXT_ERR = f"{type(e).__name__}: {e}"
+        _EXT = None
+    return _EXT
+
+
+def hip_available() -> bool:
+    return torch.cuda.is_available() and hip_ext() is not None
+
+
+def _require_ext():
+    ext = hip_ext()
+    if ext is None:
+        raise RuntimeError(
+            "HIP extension not available on a GPU device "
+            f"(load error: {_EXT_ERR}). Build it with "
+            "`python setup.py build_ext --inplace` (PYTORCH_ROCM_ARCH=gfx950); "
+            "silent eager fallback on GPU is disabled by design."
+        )
+    return ext
+
+
+def _use_hip(x: torch.Tensor) -> bool:
+    if not x.is_cuda:
+        return False
+    if os.environ.get("AIRTC_FORCE_EAGER") == "1":
+        return False
+    return True
+
+
+# ---------------------------------------------------------------------------
+# conv2d (NHWC activations, OIHW weights)
+# ---------------------------------------------------------------------------
+
+def conv2d_nhwc(
+    x: torch.Tensor,
+    weight: torch.Tensor,
+    bias: torch.Tensor | None = None,
+    stride: int = 1,
+    padding: int = 1,
+    fuse_silu: bool = False,
+) -> torch.Tensor:
+    """x: (B,H,W,C) contiguous; weight: (O,I,R,S) [torch layout]; out (B,H',W',O).
+
+    GPU path: implicit-GEMM on MFMA with on-the-fly im2col gather
+    (ops/csrc/conv2d_mfma.hip); weights are pre-transformed to (R,S,I,O) by
+    models.plan.PlannedConv at engine-build time.
+    """
+    if _use_hip(x):
+        ext = _require_ext()
+        return ext.conv2d_nhwc(x, weight, bias, stride, padding, fuse_silu)
+    xc = x.permute(0, 3, 1, 2)
+    y = F.conv2d(xc, weight, bias, stride=stride, padding=padding)
+    y = y.permute(0, 2, 3, 1).contiguous()
+    return F.silu(y) if fuse_silu else y
+
+
+# ---------------------------------------------------------------------------
+# normalisations
+# ---------------------------------------------------------------------------
+
+def group_norm_silu_nhwc(
+    x: torch.Tensor,
+    num_groups: int,
+    gamma: torch.Tensor,
+    beta: torch.Tensor,
+    eps: float = 1e-5,
+    silu: bool = True,
+) -> torch.Tensor:
+    """Fused GroupNorm(+SiLU) on NHWC. GPU: single kernel, LDS-staged stats."""
+    if _use_hip(x):
+        ext = _require_ext()
+        return ext.group_norm_silu_nhwc(x, num_groups, gamma, beta, eps, silu)
+    b, h, w, c = x.shape
+    xc = x.permute(0, 3, 1, 2).float()
+    y = F.group_norm(xc, num_groups, gamma.float(), beta.float(), eps)
+    y = y.permute(0, 2, 3, 1)
+    if silu:
+        y = F.silu(y)
+    return y.to(x.dtype).contiguous()
+
+
+def layer_norm(
+    x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor, eps: float = 1e-5
+) -> torch.Tensor:
+    if _use_hip(x):
+        ext = _require_ext()
+        return ext.layer_norm(x, gamma, beta, eps)
+    return F.layer_norm(x.float(), (x.shape[-1],), gamma.float(), beta.float(), eps).to(x.dtype)
+
+
+# ---------------------------------------------------------------------------
+# attention / MLP
+# ---------------------------------------------------------------------------
+
+def attention(
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, num_heads: int
+) -> torch.Tensor:
+    """q: (B, Lq, C); k,v: (B, Lk, C). Returns (B, Lq, C).
+
+    GPU: flash-style fused kernel (online softmax, MFMA QK^T and PV,
+    LDS-tiled K/V) — ops/csrc/attention_mfma.hip.
+    """
+    if _use_hip(q):
+        ext = _require_ext()
+        return ext.attention(q, k, v, num_heads)
+    b, lq, c = q.shape
+    d = c // num_heads
+    qh = q.view(b, lq, num_heads, d).permute(0, 2, 1, 3).float()
+    kh = k.view(b, -1, num_heads, d).permute(0, 2, 1, 3).float()
+    vh = v.view(b, -1, num_heads, d).permute(0, 2, 1, 3).float()
+    o = F.scaled_dot_product_attention(qh, kh, vh)
+    return o.permute(0, 2, 1, 3).reshape(b, lq, c).to(q.dtype)
+
+
+def linear(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor | None = None) -> torch.Tensor:
+    """Plain library GEMM (hipBLASLt via torch on ROCm) — allowed per design
+    notes: hand-written kernels are reserved for fused/hot ops."""
+    return F.linear(x, weight, bias)
+
+
+def geglu(x: torch.Tensor) -> torch.Tensor:
+    """GEGLU activation: split last dim, a * gelu(b). GPU: fused kernel."""
+    if _use_hip(x):
+        ext = _require_ext()
+        return ext.geglu(x)
+    a, b = x.chunk(2, dim=-1)
+    return a * F.gelu(b.float()).to(x.dtype)
+
+
+def silu(x: torch.Tensor) -> torch.Tensor:
+    if _use_hip(x):
+        ext = _require_ext()
+        return ext.silu(x)
+    return F.silu(x)
+
+
+# ---------------------------------------------------------------------------
+# resampling
+# ---------------------------------------------------------------------------
+
+def upsample_nearest2x_nhwc(x: torch.Tensor) -> torch.Tensor:
+    if _use_hip(x):
+        ext = _require_ext()
+        return ext.upsample_nearest2x_nhwc(x)
+    xc = x.permute(0, 3, 1, 2)
+    y = F.interpolate(xc, scale_factor=2, mode="nearest")
+    return y.permute(0, 2, 3, 1).contiguous()
+
+
+# ---------------------------------------------------------------------------
+# frame pre/post-processing (replaces CV-CUDA convertto+reformat, N3+N4)
+# ---------------------------------------------------------------------------
+
+def preprocess_from_u8(frame_u8: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
+    """u8 (H,W,3) or (B,H,W,3) RGB -> dtype (B,H,W,3) in [-1, 1].
+
+    Fuses the reference's cvcuda.convertto (u8->f32 x 1/255,
+    lib/pipeline.py:61) + normalisation; layout stays NHWC (the GPU path
+    never needs the reference's NHWC->NCHW reformat, lib/pipeline.py:63).
+    """
+    if frame_u8.dim() == 3:
+        frame_u8 = frame_u8.unsqueeze(0)
+    if _use_hip(frame_u8):
+        ext = _require_ext()
+        return ext.preprocess_u8(frame_u8, dtype == torch.float16)
+    return (frame_u8.to(torch.float32) / 255.0 * 2.0 - 1.0).to(dtype)
+
+
+def postprocess_to_u8(img: torch.Tensor) -> torch.Tensor:
+    """dtype (B,H,W,3) in [-1,1] -> u8 (B,H,W,3). Reference lib/pipeline.py:72-74."""
+    if _use_hip(img):
+        ext = _require_ext()
+        return ext.postprocess_u8(img)
+    x = (img.float() + 1.0) * 127.5
+    return x.round().clamp(0, 255).to(torch.uint8)
