@@ -1,0 +1,202 @@
+// Flash-style fused attention (forward) on MFMA — CDNA4 16x16x32 f16.
+//
+// MI355X-native replacement for the self/cross-attention inside the
+// reference's TRT UNet engine. Shapes served:
+//   self-attn : Lq = Lk = HW (4096/1024/256/64), d = head_dim (padded to 32)
+//   cross-attn: Lk = 77 text tokens
+//
+// Structure (guide Appendix B "fused attention prefill"):
+//  - workgroup = 4 waves = one 64-row Q tile; wave owns 16 q-rows
+//  - K/V streamed through LDS in 64-key tiles; V stored TRANSPOSED ([d][kv])
+//    at staging so the PV B-fragment is a contiguous ds_read_b128
+//  - online softmax: running (m, l) per q-row, kept lane-local (each lane
+//    owns the same 4 q-rows its C-fragments do), 4x shfl_xor row reduce
+//  - P goes through a per-wave LDS tile to convert the C-fragment layout
+//    into the A-fragment layout for PV (layout bridge; same-wave, ordered
+//    by an lgkmcnt(0) wait, no barrier)
+//  - Lk tail masked with -1e30 before the max (cross-attn 77 of 128)
+//  - kernel is TEMPLATED on head dim D so every accumulator index is
+//    compile-time constant (guide §5.4 rule 20: runtime-indexed vector
+//    arrays go to scratch)
+//
+// Addressing: q/k/v/out are (b, h)-strided views (no host-side copies for
+// d%32==0 models; ops/interface.py pads other head sizes).
+
+#include "common.h"
+
+#define KVT 64  // kv tile
+#define QT 64   // q rows per workgroup
+#define PPITCH (KVT + 8)
+
+template <int D>
+__global__ __launch_bounds__(256) void attention_kernel(
+    const f16* __restrict__ q, const f16* __restrict__ k,
+    const f16* __restrict__ v, f16* __restrict__ out, int H, int Lq, int Lk,
+    long q_sb, long q_sh, long q_row, long k_sb, long k_sh, long k_row,
+    long o_sb, long o_sh, long o_row, float scale) {
+  constexpr int KPITCH = D + 8;
+  constexpr int VPITCH = KVT + 8;
+  constexpr int D8 = D / 8, D32 = D / 32, D16 = D / 16;
+  __shared__ f16 ldsK[KVT * KPITCH];
+  __shared__ f16 ldsV[D * VPITCH];
+  __shared__ f16 ldsP[4 * 16 * PPITCH];
+
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int q0 = blockIdx.x * QT;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+
+  const f16* qb = q + b * q_sb + h * q_sh;
+  const f16* kb = k + b * k_sb + h * k_sh;
+  const f16* vb = v + b * k_sb + h * k_sh;  // v shares k's layout
+  f16* ob = out + b * o_sb + h * o_sh;
+
+  // ---- Q fragments in registers (one load, reused every kv tile) ----
+  f16x8 aq[D32];
+  {
+    int qrow = q0 + wid * 16 + (lane & 15);
+    if (qrow >= Lq) qrow = Lq - 1;
+    const f16* qr = qb + (long)qrow * q_row + (lane >> 4) * 8;
+#pragma unroll
+    for (int i = 0; i < D32; ++i)
+      aq[i] = *reinterpret_cast<const f16x8*>(qr + i * 32);
+  }
+
+  f32x4 o_acc[D16];
+#pragma unroll
+  for (int i = 0; i < D16; ++i) o_acc[i] = {0.f, 0.f, 0.f, 0.f};
+  float m_i[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  float l_i[4] = {0.f, 0.f, 0.f, 0.f};
+
+  f16* myP = &ldsP[wid * 16 * PPITCH];
+  const int fcol = (lane >> 4) * 8;
+
+  for (int t0 = 0; t0 < Lk; t0 += KVT) {
+    // ---- stage K tile [64][D] and V tile transposed [D][64] ----
+    if (t0) __syncthreads();  // previous tile's reads complete
+    for (int i = tid; i < KVT * D8; i += 256) {
+      const int row = i / D8, c8 = (i - row * D8) * 8;
+      int krow = t0 + row;
+      if (krow >= Lk) krow = Lk - 1;  // masked later
+      f16x8 k8 = *reinterpret_cast<const f16x8*>(&kb[(long)krow * k_row + c8]);
+      *reinterpret_cast<f16x8*>(&ldsK[row * KPITCH + c8]) = k8;
+      f16x8 v8 = *reinterpret_cast<const f16x8*>(&vb[(long)krow * k_row + c8]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ldsV[(c8 + j) * VPITCH + row] = v8[j];
+    }
+    __syncthreads();
+
+    // ---- S = scale * Q K^T  (4 col fragments of 16) ----
+    f32x4 sfrag[4];
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      sfrag[nf] = {0.f, 0.f, 0.f, 0.f};
+      const f16* kr = &ldsK[(nf * 16 + (lane & 15)) * KPITCH + fcol];
+#pragma unroll
+      for (int ds = 0; ds < D32; ++ds) {
+        f16x8 bfrag = *reinterpret_cast<const f16x8*>(kr + ds * 32);
+        sfrag[nf] = mfma16x16x32(aq[ds], bfrag, sfrag[nf]);
+      }
+    }
+
+    // ---- online softmax update (rows lane-local) ----
+    float p[4][4];  // [nf][reg j]
+    float mnew[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) mnew[j] = -1e30f;
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int kcol = t0 + nf * 16 + (lane & 15);
+      const bool valid = kcol < Lk;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float val = valid ? sfrag[nf][j] * scale : -1e30f;
+        p[nf][j] = val;
+        mnew[j] = fmaxf(mnew[j], val);
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        mnew[j] = fmaxf(mnew[j], __shfl_xor(mnew[j], off, 64));
+      const float mn = fmaxf(m_i[j], mnew[j]);
+      const float alpha = __expf(m_i[j] - mn);
+      float rs = 0.f;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        p[nf][j] = __expf(p[nf][j] - mn);
+        rs += p[nf][j];
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) rs += __shfl_xor(rs, off, 64);
+      l_i[j] = l_i[j] * alpha + rs;
+      m_i[j] = mn;
+#pragma unroll
+      for (int f = 0; f < D16; ++f) o_acc[f][j] *= alpha;
+    }
+
+    // ---- P -> LDS (C-frag layout -> A-frag layout bridge) ----
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        myP[((lane >> 4) * 4 + j) * PPITCH + nf * 16 + (lane & 15)] =
+            (f16)p[nf][j];
+    // same-wave LDS write->read: wait for the writes, keep reads below
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+    // ---- O += P V ----
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      f16x8 afrag = *reinterpret_cast<const f16x8*>(
+          &myP[(lane & 15) * PPITCH + ks * 32 + fcol]);
+#pragma unroll
+      for (int f = 0; f < D16; ++f) {
+        f16x8 bfrag = *reinterpret_cast<const f16x8*>(
+            &ldsV[(f * 16 + (lane & 15)) * VPITCH + ks * 32 + fcol]);
+        o_acc[f] = mfma16x16x32(afrag, bfrag, o_acc[f]);
+      }
+    }
+  }
+
+  // ---- epilogue: O /= l, masked stores ----
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int qrow = q0 + wid * 16 + (lane >> 4) * 4 + j;
+    if (qrow >= Lq) continue;
+    const float inv_l = 1.0f / l_i[j];
+#pragma unroll
+    for (int f = 0; f < D16; ++f)
+      ob[(long)qrow * o_row + f * 16 + (lane & 15)] =
+          (f16)(o_acc[f][j] * inv_l);
+  }
+}
+
+extern "C" void airtc_attention(const uint16_t* q, const uint16_t* k,
+                                const uint16_t* v, uint16_t* out, int B, int H,
+                                int Lq, int Lk, int d, long q_sb, long q_sh,
+                                long q_row, long k_sb, long k_sh, long k_row,
+                                long o_sb, long o_sh, long o_row, float scale,
+                                hipStream_t s) {
+  dim3 grid(ceil_div(Lq, QT), B * H);
+  const f16* qp = reinterpret_cast<const f16*>(q);
+  const f16* kp = reinterpret_cast<const f16*>(k);
+  const f16* vp = reinterpret_cast<const f16*>(v);
+  f16* op = reinterpret_cast<f16*>(out);
+#define LAUNCH(D)                                                           \
+  hipLaunchKernelGGL(attention_kernel<D>, grid, dim3(256), 0, s, qp, kp,    \
+                     vp, op, H, Lq, Lk, q_sb, q_sh, q_row, k_sb, k_sh,      \
+                     k_row, o_sb, o_sh, o_row, scale)
+  switch (d) {
+    case 32: LAUNCH(32); break;
+    case 64: LAUNCH(64); break;
+    case 96: LAUNCH(96); break;
+    case 128: LAUNCH(128); break;
+    case 160: LAUNCH(160); break;
+    default: break;  // host wrapper guarantees one of the above
+  }
+#undef LAUNCH
+}

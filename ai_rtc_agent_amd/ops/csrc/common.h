@@ -1,0 +1,64 @@
+// Common device-side helpers for the MI355X (gfx950 / CDNA4) kernels.
+//
+// Conventions:
+//  - wavefront = 64 lanes (CDNA), blocks are multiples of 64 threads
+//  - f16 storage, f32 accumulation; MFMA shape 16x16x32 (f16 in, f32 out)
+//  - activations NHWC; channel runs are the fast (contiguous) dimension
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#define WAVE 64
+
+typedef _Float16 f16;
+typedef __attribute__((__vector_size__(8 * sizeof(_Float16)))) _Float16 f16x8;
+typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4;
+typedef __attribute__((__vector_size__(2 * sizeof(float)))) float f32x2;
+typedef __attribute__((__vector_size__(4 * sizeof(uint32_t)))) uint32_t u32x4;
+
+// D = A(16x32) @ B(32x16) + C  — per-wave MFMA, f16 inputs, f32 accum.
+// A fragment: lane l holds A[l%16][(l/16)*8 + j], j=0..7  (one ds_read_b128)
+// B fragment: lane l holds B[(l/16)*8 + j][l%16]
+// C/D       : lane l holds D[(l/16)*4 + r][l%16], r=0..3
+__device__ __forceinline__ f32x4 mfma16x16x32(f16x8 a, f16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+}
+
+__device__ __forceinline__ float sigmoidf_dev(float x) {
+  return 1.0f / (1.0f + __expf(-x));
+}
+
+__device__ __forceinline__ float siluf(float x) { return x * sigmoidf_dev(x); }
+
+__device__ __forceinline__ float geluf(float x) {
+  // erf-based GELU (matches torch default)
+  return 0.5f * x * (1.0f + erff(x * 0.70710678118654752f));
+}
+
+// activation codes shared with the host side
+enum ActCode { ACT_NONE = 0, ACT_SILU = 1, ACT_RELU = 2 };
+
+__device__ __forceinline__ float apply_act(float v, int act) {
+  if (act == ACT_SILU) return siluf(v);
+  if (act == ACT_RELU) return v > 0.0f ? v : 0.0f;
+  return v;
+}
+
+// butterfly reduce over all 64 lanes of a wave
+template <typename Op>
+__device__ __forceinline__ float wave_reduce(float v, Op op) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = op(v, __shfl_xor(v, off, 64));
+  return v;
+}
+
+struct SumOp {
+  __device__ float operator()(float a, float b) const { return a + b; }
+};
+struct MaxOp {
+  __device__ float operator()(float a, float b) const { return fmaxf(a, b); }
+};
+
+__device__ __forceinline__ int ceil_div_dev(int a, int b) { return (a + b - 1) / b; }
+static inline int ceil_div(int a, int b) { return (a + b - 1) / b; }

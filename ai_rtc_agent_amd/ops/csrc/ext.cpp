@@ -1,0 +1,161 @@
+// Torch bindings for the MI355X HIP kernels (host-side translation unit).
+//
+// Thin layer: validates tensors, allocates outputs through the torch caching
+// allocator (hipGraph-capture safe), and calls the extern "C" launchers from
+// kernels.h on the current stream. No compute logic lives here.
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "kernels.h"
+
+#define CHECK_IN(x)                                                     \
+  TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                       \
+  TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+
+namespace {
+
+hipStream_t cur_stream() {
+  return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+}
+
+const uint16_t* h_ptr(const torch::Tensor& t) {
+  return reinterpret_cast<const uint16_t*>(t.data_ptr<at::Half>());
+}
+uint16_t* h_ptr_mut(torch::Tensor& t) {
+  return reinterpret_cast<uint16_t*>(t.data_ptr<at::Half>());
+}
+
+torch::Tensor conv2d(torch::Tensor x_pad, torch::Tensor w_perm,
+                     c10::optional<torch::Tensor> bias, int64_t HO, int64_t WO,
+                     int64_t R, int64_t S, int64_t stride, int64_t act,
+                     bool use_mfma) {
+  CHECK_IN(x_pad);
+  CHECK_IN(w_perm);
+  TORCH_CHECK(x_pad.dtype() == torch::kHalf && w_perm.dtype() == torch::kHalf);
+  const int B = x_pad.size(0), Hp = x_pad.size(1), Wp = x_pad.size(2),
+            IC = x_pad.size(3);
+  const int OC = w_perm.size(0);
+  auto out = torch::empty({B, HO, WO, OC}, x_pad.options());
+  const float* bp = nullptr;
+  if (bias.has_value()) {
+    TORCH_CHECK(bias->dtype() == torch::kFloat && bias->is_contiguous());
+    bp = bias->data_ptr<float>();
+  }
+  auto fn = use_mfma ? airtc_conv2d_mfma : airtc_conv2d_direct;
+  fn(h_ptr(x_pad), h_ptr(w_perm), bp, h_ptr_mut(out), B, Hp, Wp, IC, (int)HO,
+     (int)WO, OC, (int)R, (int)S, (int)stride, (int)act, cur_stream());
+  return out;
+}
+
+torch::Tensor group_norm_silu(torch::Tensor x, int64_t groups,
+                              torch::Tensor gamma, torch::Tensor beta,
+                              double eps, int64_t act) {
+  CHECK_IN(x);
+  const int B = x.size(0);
+  const int C = x.size(-1);
+  const long HW = x.numel() / ((long)B * C);
+  auto out = torch::empty_like(x);
+  airtc_group_norm_silu(h_ptr(x), gamma.data_ptr<float>(),
+                        beta.data_ptr<float>(), h_ptr_mut(out), B, (int)HW, C,
+                        (int)groups, (float)eps, (int)act, cur_stream());
+  return out;
+}
+
+torch::Tensor layer_norm(torch::Tensor x, torch::Tensor gamma,
+                         torch::Tensor beta, double eps) {
+  CHECK_IN(x);
+  const int C = x.size(-1);
+  const long rows = x.numel() / C;
+  auto out = torch::empty_like(x);
+  airtc_layer_norm(h_ptr(x), gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                   h_ptr_mut(out), rows, C, (float)eps, cur_stream());
+  return out;
+}
+
+torch::Tensor attention_bhlc(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                             int64_t heads, double scale) {
+  CHECK_IN(q);
+  CHECK_IN(k);
+  CHECK_IN(v);
+  const int B = q.size(0), Lq = q.size(1), C = q.size(2);
+  const int Lk = k.size(1), Ck = k.size(2);
+  const int d = C / (int)heads;
+  TORCH_CHECK(Ck == C, "q/k channel mismatch");
+  TORCH_CHECK(d % 32 == 0 && d <= 160, "head_dim must be padded to one of 32/64/96/128/160");
+  auto out = torch::empty_like(q);
+  airtc_attention(h_ptr(q), h_ptr(k), h_ptr(v), h_ptr_mut(out), B, (int)heads,
+                  Lq, Lk, d, (long)Lq * C, d, C, (long)Lk * C, d, C,
+                  (long)Lq * C, d, C, (float)scale, cur_stream());
+  return out;
+}
+
+torch::Tensor silu(torch::Tensor x) {
+  CHECK_IN(x);
+  auto out = torch::empty_like(x);
+  airtc_silu_f16(h_ptr(x), h_ptr_mut(out), x.numel(), cur_stream());
+  return out;
+}
+
+torch::Tensor geglu(torch::Tensor x) {
+  CHECK_IN(x);
+  const long inner = x.size(-1) / 2;
+  const long rows = x.numel() / (2 * inner);
+  auto sizes = x.sizes().vec();
+  sizes.back() = inner;
+  auto out = torch::empty(sizes, x.options());
+  airtc_geglu_f16(h_ptr(x), h_ptr_mut(out), rows, inner, cur_stream());
+  return out;
+}
+
+torch::Tensor add_act(torch::Tensor a, torch::Tensor b, int64_t act) {
+  CHECK_IN(a);
+  CHECK_IN(b);
+  auto out = torch::empty_like(a);
+  airtc_add_act_f16(h_ptr(a), h_ptr(b), h_ptr_mut(out), a.numel(), (int)act,
+                    cur_stream());
+  return out;
+}
+
+torch::Tensor upsample2x(torch::Tensor x) {
+  CHECK_IN(x);
+  const int B = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  auto out = torch::empty({B, 2 * H, 2 * W, C}, x.options());
+  airtc_upsample2x_f16(h_ptr(x), h_ptr_mut(out), B, H, W, C, cur_stream());
+  return out;
+}
+
+torch::Tensor preprocess_u8(torch::Tensor frame) {
+  CHECK_IN(frame);
+  TORCH_CHECK(frame.dtype() == torch::kUInt8);
+  auto out = torch::empty(frame.sizes(),
+                          frame.options().dtype(torch::kHalf));
+  airtc_preprocess_u8(frame.data_ptr<uint8_t>(),
+                      reinterpret_cast<uint16_t*>(out.data_ptr<at::Half>()),
+                      frame.numel(), cur_stream());
+  return out;
+}
+
+torch::Tensor postprocess_u8(torch::Tensor img) {
+  CHECK_IN(img);
+  TORCH_CHECK(img.dtype() == torch::kHalf);
+  auto out = torch::empty(img.sizes(), img.options().dtype(torch::kUInt8));
+  airtc_postprocess_u8(h_ptr(img), out.data_ptr<uint8_t>(), img.numel(),
+                       cur_stream());
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("conv2d", &conv2d, "implicit-GEMM MFMA conv2d (NHWC)");
+  m.def("group_norm_silu", &group_norm_silu);
+  m.def("layer_norm", &layer_norm);
+  m.def("attention_bhlc", &attention_bhlc);
+  m.def("silu", &silu);
+  m.def("geglu", &geglu);
+  m.def("add_act", &add_act);
+  m.def("upsample2x", &upsample2x);
+  m.def("preprocess_u8", &preprocess_u8);
+  m.def("postprocess_u8", &postprocess_u8);
+}

@@ -2,7 +2,7 @@
 
 The GPU activation layout is NHWC throughout (implicit-GEMM friendly on MFMA:
 the K = (r,s,c) gather reads 16B-contiguous channel runs — see
-ops/csrc/conv2d_mfma.hip). The torch reference path permutes to NCHW for
+ops/csrc/conv2d.hip). The torch reference path permutes to NCHW for
 F.conv2d and back; it exists for CPU tests and as the numerics golden that
 GPU tests compare the HIP kernels against (SURVEY.md §4 item b).
 
@@ -10,11 +10,14 @@ Replaces (MI355X-natively) reference components N3-N7 of SURVEY.md §2.2:
 CV-CUDA convertto/reformat (lib/pipeline.py:61-63) -> fused pre/post kernels;
 TensorRT UNet/VAE engines (lib/wrapper.py:409-512) -> these kernels + hipGraph.
 
-Policy: on a CUDA/HIP device the HIP extension is REQUIRED — ops raise if it
-is missing rather than silently falling back to eager torch.
+Dispatch policy: fp16 tensors on a CUDA/HIP device take the HIP kernels and
+RAISE if the extension is missing (no silent eager fallback on GPU). fp32 or
+CPU tensors take the torch reference path (that is what GPU numerics tests
+compare against).
 """
 from __future__ import annotations
 
+import math
 import os
 
 import torch
@@ -22,6 +25,8 @@ import torch.nn.functional as F
 
 _EXT = None
 _EXT_ERR: str | None = None
+
+ACT_NONE, ACT_SILU, ACT_RELU = 0, 1, 2
 
 
 def hip_ext():
@@ -56,11 +61,19 @@ def _require_ext():
 
 
 def _use_hip(x: torch.Tensor) -> bool:
-    if not x.is_cuda:
+    if not x.is_cuda or x.dtype != torch.float16:
         return False
     if os.environ.get("AIRTC_FORCE_EAGER") == "1":
         return False
     return True
+
+
+def _cached(t: torch.Tensor, key: str, build):
+    c = getattr(t, key, None)
+    if c is None:
+        c = build()
+        setattr(t, key, c)
+    return c
 
 
 # ---------------------------------------------------------------------------
@@ -74,20 +87,44 @@ def conv2d_nhwc(
     stride: int = 1,
     padding: int = 1,
     fuse_silu: bool = False,
+    act: int | None = None,
 ) -> torch.Tensor:
     """x: (B,H,W,C) contiguous; weight: (O,I,R,S) [torch layout]; out (B,H',W',O).
 
     GPU path: implicit-GEMM on MFMA with on-the-fly im2col gather
-    (ops/csrc/conv2d_mfma.hip); weights are pre-transformed to (R,S,I,O) by
-    models.plan.PlannedConv at engine-build time.
+    (ops/csrc/conv2d.hip). Weights are lazily pre-transformed to the
+    (O, R*S*I) GEMM layout and cached on the parameter (the AOT `build`
+    step warms every cache before hipGraph capture).
     """
+    if act is None:
+        act = ACT_SILU if fuse_silu else ACT_NONE
     if _use_hip(x):
         ext = _require_ext()
-        return ext.conv2d_nhwc(x, weight, bias, stride, padding, fuse_silu)
+        O, I, R, S = weight.shape
+        w_perm = _cached(
+            weight,
+            "_airtc_wperm",
+            lambda: weight.detach().permute(0, 2, 3, 1).reshape(O, -1).contiguous().half(),
+        )
+        b32 = None
+        if bias is not None:
+            b32 = _cached(bias, "_airtc_b32", lambda: bias.detach().float().contiguous())
+        x_pad = F.pad(x, (0, 0, padding, padding, padding, padding)) if padding else x
+        B, Hp, Wp, _ = x_pad.shape
+        HO = (Hp - R) // stride + 1
+        WO = (Wp - S) // stride + 1
+        use_mfma = (I % 32 == 0)
+        return ext.conv2d(x_pad, w_perm, b32, HO, WO, R, S, stride, act, use_mfma)
+
     xc = x.permute(0, 3, 1, 2)
-    y = F.conv2d(xc, weight, bias, stride=stride, padding=padding)
-    y = y.permute(0, 2, 3, 1).contiguous()
-    return F.silu(y) if fuse_silu else y
+    y = F.conv2d(xc.float(), weight.float(), None if bias is None else bias.float(),
+                 stride=stride, padding=padding)
+    y = y.permute(0, 2, 3, 1)
+    if act == ACT_SILU:
+        y = F.silu(y)
+    elif act == ACT_RELU:
+        y = F.relu(y)
+    return y.to(x.dtype).contiguous()
 
 
 # ---------------------------------------------------------------------------
@@ -102,10 +139,12 @@ def group_norm_silu_nhwc(
     eps: float = 1e-5,
     silu: bool = True,
 ) -> torch.Tensor:
-    """Fused GroupNorm(+SiLU) on NHWC. GPU: single kernel, LDS-staged stats."""
+    """Fused GroupNorm(+SiLU) on NHWC. GPU: single kernel, wave-reduced stats."""
     if _use_hip(x):
         ext = _require_ext()
-        return ext.group_norm_silu_nhwc(x, num_groups, gamma, beta, eps, silu)
+        g32 = _cached(gamma, "_airtc_g32", lambda: gamma.detach().float().contiguous())
+        b32 = _cached(beta, "_airtc_b32", lambda: beta.detach().float().contiguous())
+        return ext.group_norm_silu(x, num_groups, g32, b32, eps, ACT_SILU if silu else ACT_NONE)
     b, h, w, c = x.shape
     xc = x.permute(0, 3, 1, 2).float()
     y = F.group_norm(xc, num_groups, gamma.float(), beta.float(), eps)
@@ -120,7 +159,9 @@ def layer_norm(
 ) -> torch.Tensor:
     if _use_hip(x):
         ext = _require_ext()
-        return ext.layer_norm(x, gamma, beta, eps)
+        g32 = _cached(gamma, "_airtc_g32", lambda: gamma.detach().float().contiguous())
+        b32 = _cached(beta, "_airtc_b32", lambda: beta.detach().float().contiguous())
+        return ext.layer_norm(x, g32, b32, eps)
     return F.layer_norm(x.float(), (x.shape[-1],), gamma.float(), beta.float(), eps).to(x.dtype)
 
 
@@ -128,17 +169,38 @@ def layer_norm(
 # attention / MLP
 # ---------------------------------------------------------------------------
 
+_ATTN_DIMS = (32, 64, 96, 128, 160)
+
+
 def attention(
     q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, num_heads: int
 ) -> torch.Tensor:
     """q: (B, Lq, C); k,v: (B, Lk, C). Returns (B, Lq, C).
 
     GPU: flash-style fused kernel (online softmax, MFMA QK^T and PV,
-    LDS-tiled K/V) — ops/csrc/attention_mfma.hip.
+    LDS-tiled K/V) — ops/csrc/attention.hip. head_dim % 32 == 0 runs
+    zero-copy on (b,h)-strided views; other head dims (sd15's 40/80) are
+    zero-padded to the next supported size (softmax-invariant).
     """
     if _use_hip(q):
         ext = _require_ext()
-        return ext.attention(q, k, v, num_heads)
+        b, lq, c = q.shape
+        lk = k.shape[1]
+        d = c // num_heads
+        scale = 1.0 / math.sqrt(d)
+        if d in _ATTN_DIMS:
+            return ext.attention_bhlc(q, k, v, num_heads, scale)
+        dpad = min(x for x in _ATTN_DIMS if x >= d)
+
+        def rearrange(t, L):
+            th = t.view(b, L, num_heads, d)
+            th = F.pad(th, (0, dpad - d))
+            return th.permute(0, 2, 1, 3).reshape(b * num_heads, L, dpad).contiguous()
+
+        o = ext.attention_bhlc(rearrange(q, lq), rearrange(k, lk), rearrange(v, lk), 1, scale)
+        o = o.view(b, num_heads, lq, dpad)[..., :d]
+        return o.permute(0, 2, 1, 3).reshape(b, lq, c).contiguous()
+
     b, lq, c = q.shape
     d = c // num_heads
     qh = q.view(b, lq, num_heads, d).permute(0, 2, 1, 3).float()
@@ -157,17 +219,27 @@ def linear(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor | None = No
 def geglu(x: torch.Tensor) -> torch.Tensor:
     """GEGLU activation: split last dim, a * gelu(b). GPU: fused kernel."""
     if _use_hip(x):
-        ext = _require_ext()
-        return ext.geglu(x)
+        return _require_ext().geglu(x)
     a, b = x.chunk(2, dim=-1)
-    return a * F.gelu(b.float()).to(x.dtype)
+    return (a.float() * F.gelu(b.float())).to(x.dtype)
 
 
 def silu(x: torch.Tensor) -> torch.Tensor:
-    if _use_hip(x):
-        ext = _require_ext()
-        return ext.silu(x)
-    return F.silu(x)
+    if _use_hip(x) and x.numel() % 8 == 0:
+        return _require_ext().silu(x)
+    return F.silu(x.float()).to(x.dtype)
+
+
+def add_act(a: torch.Tensor, b: torch.Tensor, act: int = ACT_NONE) -> torch.Tensor:
+    """Fused residual add + optional activation (TAESD add+relu, resnet skips)."""
+    if _use_hip(a) and a.numel() % 8 == 0:
+        return _require_ext().add_act(a, b, act)
+    y = a.float() + b.float()
+    if act == ACT_SILU:
+        y = F.silu(y)
+    elif act == ACT_RELU:
+        y = F.relu(y)
+    return y.to(a.dtype)
 
 
 # ---------------------------------------------------------------------------
@@ -175,9 +247,8 @@ def silu(x: torch.Tensor) -> torch.Tensor:
 # ---------------------------------------------------------------------------
 
 def upsample_nearest2x_nhwc(x: torch.Tensor) -> torch.Tensor:
-    if _use_hip(x):
-        ext = _require_ext()
-        return ext.upsample_nearest2x_nhwc(x)
+    if _use_hip(x) and x.shape[-1] % 8 == 0:
+        return _require_ext().upsample2x(x)
     xc = x.permute(0, 3, 1, 2)
     y = F.interpolate(xc, scale_factor=2, mode="nearest")
     return y.permute(0, 2, 3, 1).contiguous()
@@ -196,16 +267,15 @@ def preprocess_from_u8(frame_u8: torch.Tensor, dtype: torch.dtype) -> torch.Tens
     """
     if frame_u8.dim() == 3:
         frame_u8 = frame_u8.unsqueeze(0)
-    if _use_hip(frame_u8):
-        ext = _require_ext()
-        return ext.preprocess_u8(frame_u8, dtype == torch.float16)
-    return (frame_u8.to(torch.float32) / 255.0 * 2.0 - 1.0).to(dtype)
+    if frame_u8.is_cuda and dtype == torch.float16 and hip_ext() is not None \
+            and os.environ.get("AIRTC_FORCE_EAGER") != "1":
+        return _require_ext().preprocess_u8(frame_u8.contiguous())
+    return (frame_u8.to(torch.float32) / 127.5 - 1.0).to(dtype)
 
 
 def postprocess_to_u8(img: torch.Tensor) -> torch.Tensor:
     """dtype (B,H,W,3) in [-1,1] -> u8 (B,H,W,3). Reference lib/pipeline.py:72-74."""
     if _use_hip(img):
-        ext = _require_ext()
-        return ext.postprocess_u8(img)
+        return _require_ext().postprocess_u8(img.contiguous())
     x = (img.float() + 1.0) * 127.5
     return x.round().clamp(0, 255).to(torch.uint8)

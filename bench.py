@@ -1,0 +1,151 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: img2img FPS + p50 glass-to-glass latency,
+SD-Turbo 512x512 1-step stream-batch img2img (BASELINE.json headline config).
+
+    python bench.py --gpus N --steps K --warmup W
+
+Multi-GPU (launched by the driver via torch.distributed.run, one rank per
+GPU over RCCL): frame-level data parallelism — each rank runs its own
+pipeline replica on its own synthetic stream (weak scaling; this is the
+8-concurrent-peers serving model of SURVEY.md §5.8), after an RCCL weight
+broadcast from rank 0 over xGMI.
+
+Synthetic data: random uint8 RGB frames; random-init SD-Turbo-architecture
+weights (no network in this environment — BASELINE.md notes the same).
+Rank 0 prints ONE JSON line with the whole-job aggregate FPS.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from ai_rtc_agent_amd.config import sd_turbo_config
+from ai_rtc_agent_amd.engine import StreamDiffusionEngine
+from ai_rtc_agent_amd.parallel import broadcast_engine_weights, init_distributed
+
+import torch.distributed as dist
+
+
+def main() -> None:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=60)
+    p.add_argument("--warmup", type=int, default=20)
+    p.add_argument("--width", type=int, default=512)
+    p.add_argument("--no-graph", action="store_true")
+    p.add_argument("--latency-frames", type=int, default=32)
+    args = p.parse_args()
+
+    rank, world, local = init_distributed()
+    use_cuda = torch.cuda.is_available()
+    if not use_cuda and "--steps" not in sys.argv:
+        # no-GPU smoke only: the full SD-Turbo UNet on CPU is ~0.5 s/frame
+        args.steps, args.warmup, args.latency_frames = 4, 1, 4
+    device = f"cuda:{local}" if use_cuda else "cpu"
+
+    cfg = sd_turbo_config(
+        device=device,
+        width=args.width,
+        height=args.width,
+        use_hip_graph=not args.no_graph and use_cuda,
+    )
+    eng = StreamDiffusionEngine(cfg)
+    broadcast_engine_weights(eng)  # RCCL over xGMI; no-op at world=1
+    eng.prepare()
+
+    # synthetic stream: a small ring of random frames, resident on device
+    g = torch.Generator().manual_seed(1234 + rank)
+    frames = [
+        torch.randint(0, 256, (args.width, args.width, 3), generator=g, dtype=torch.uint8).to(device)
+        for _ in range(4)
+    ]
+
+    def sync():
+        if use_cuda:
+            torch.cuda.synchronize()
+        if world > 1:
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        eng(frames[i % len(frames)])
+
+    # ---- timed region: exactly K steps, barrier+sync bracketed ----
+    sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        eng(frames[i % len(frames)])
+    if use_cuda:
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    sync()
+    elapsed = t1 - t0
+
+    # MAX over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if world > 1:
+        if use_cuda:
+            t = t.to(device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    # ---- p50 glass-to-glass: per-frame sync'd submit->output latency ----
+    lat = []
+    for i in range(args.latency_frames):
+        if use_cuda:
+            torch.cuda.synchronize()
+        s = time.perf_counter()
+        out = eng(frames[i % len(frames)])
+        if use_cuda:
+            torch.cuda.synchronize()
+        lat.append((time.perf_counter() - s) * 1000.0)
+    lat.sort()
+    p50 = lat[len(lat) // 2]
+    if world > 1:
+        lt = torch.tensor([p50], dtype=torch.float64)
+        if use_cuda:
+            lt = lt.to(device)
+        dist.all_reduce(lt, op=dist.ReduceOp.MAX)
+        p50 = float(lt.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    fps_total = world * args.steps / elapsed
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "img2img FPS (SD-Turbo 512x512 1-step)",
+            "value": round(fps_total, 2),
+            "unit": "frames/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "p50_glass_to_glass_ms": round(p50, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp16" if use_cuda else "fp32",
+            "data": "synthetic (random frames, random-init SD-Turbo-arch weights)",
+            "config": {
+                "model": "sd-turbo (SD2.1-base UNet geometry, TAESD, 1-step)",
+                "global_batch": world * cfg.frame_buffer_size,
+                "resolution": f"{args.width}x{args.width}",
+                "t_index_list": cfg.t_index_list,
+                "parallelism": f"frame-level dp{world}",
+                "hip_graph": cfg.use_hip_graph,
+            },
+        }))
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

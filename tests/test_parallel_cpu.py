@@ -1,0 +1,84 @@
+"""Multi-process distributed tests on CPU (gloo, world_size=2).
+
+Covers the RCCL-over-xGMI code path structure (same torch.distributed calls;
+backend swaps to nccl==RCCL on the GPU box) — SURVEY.md §4 item (e).
+"""
+import os
+import tempfile
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+
+def _worker_broadcast(rank, world, rdv, q):
+    try:
+        dist.init_process_group(
+            "gloo", init_method=f"file://{rdv}", rank=rank, world_size=world
+        )
+        torch.manual_seed(100 + rank)  # deliberately different weights
+        m = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.Linear(16, 4))
+        from ai_rtc_agent_amd.parallel import broadcast_module
+
+        broadcast_module(m, src=0)
+        flat = torch.cat([p.data.flatten() for p in m.parameters()])
+        q.put((rank, flat))
+        dist.barrier()
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, e))
+
+
+@pytest.mark.timeout(120)
+def test_broadcast_module_syncs_weights():
+    rdv = tempfile.mktemp(prefix="airtc_rdv_")
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_broadcast, args=(r, 2, rdv, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, val = q.get(timeout=110)
+        assert not isinstance(val, Exception), f"rank {rank}: {val}"
+        results[rank] = val
+    for p in procs:
+        p.join(timeout=30)
+    assert torch.equal(results[0], results[1]), "weights must match after broadcast"
+
+
+def _worker_bench_style(rank, world, rdv, q):
+    try:
+        os.environ["RANK"] = str(rank)
+        os.environ["WORLD_SIZE"] = str(world)
+        os.environ["LOCAL_RANK"] = str(rank)
+        dist.init_process_group(
+            "gloo", init_method=f"file://{rdv}", rank=rank, world_size=world
+        )
+        # MAX-over-ranks reduction law used by bench.py
+        t = torch.tensor([1.0 + rank], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        q.put((rank, float(t.item())))
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, e))
+    finally:
+        for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK"):
+            os.environ.pop(k, None)
+
+
+@pytest.mark.timeout(120)
+def test_max_over_ranks_reduction():
+    rdv = tempfile.mktemp(prefix="airtc_rdv2_")
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_bench_style, args=(r, 2, rdv, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for _ in range(2):
+        rank, val = q.get(timeout=110)
+        assert not isinstance(val, Exception), f"rank {rank}: {val}"
+        assert val == 2.0
+    for p in procs:
+        p.join(timeout=30)
