@@ -126,7 +126,9 @@ class StreamDiffusionEngine:
         lh, lw = cfg.latent_height, cfg.latent_width
         dev, dt = self.device, self.dtype
 
-        self._coeff = self.scheduler.coefficients(cfg.t_index_list, fbs, dev, torch.float32)
+        # coefficients live in the engine compute dtype: scheduler tensor math
+        # must not promote the latent out of f16 (the HIP dispatch dtype)
+        self._coeff = self.scheduler.coefficients(cfg.t_index_list, fbs, dev, self.dtype)
         # NHWC coefficient views: (B,1,1,1) already broadcast over (B,h,w,c)
         self._embeds = self.text_encoder.encode(self.prompt, dev, dt)
         self._embeds_batch = self._embeds.expand(B, -1, -1).contiguous()
@@ -175,7 +177,7 @@ class StreamDiffusionEngine:
             return
         self.cfg.t_index_list = t_index_list
         new = self.scheduler.coefficients(
-            t_index_list, self.cfg.frame_buffer_size, self.device, torch.float32
+            t_index_list, self.cfg.frame_buffer_size, self.device, self.dtype
         )
         for k in ("alpha_prod_t_sqrt", "beta_prod_t_sqrt", "c_skip", "c_out"):
             self._coeff[k].copy_(new[k])

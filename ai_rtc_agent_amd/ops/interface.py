@@ -141,6 +141,11 @@ def group_norm_silu_nhwc(
 ) -> torch.Tensor:
     """Fused GroupNorm(+SiLU) on NHWC. GPU: single kernel, wave-reduced stats."""
     if _use_hip(x):
+        if (x.shape[-1] // num_groups) % 2 != 0:
+            raise ValueError(
+                f"group_norm kernel needs even channels-per-group, got "
+                f"C={x.shape[-1]} groups={num_groups}"
+            )
         ext = _require_ext()
         g32 = _cached(gamma, "_airtc_g32", lambda: gamma.detach().float().contiguous())
         b32 = _cached(beta, "_airtc_b32", lambda: beta.detach().float().contiguous())

@@ -84,6 +84,7 @@ def test_layer_norm_vs_torch():
     (1024, 77, 640, 10),  # cross-attn vs text tokens, d=64
     (256, 256, 320, 8),   # sd15 d=40 (padded path)
     (100, 77, 128, 2),    # ragged Lq tail
+    (64, 64, 32, 2),      # d=16 (tiny config padded path)
 ])
 def test_attention_vs_sdpa(lq, lk, c, heads):
     q = rnd(2, lq, c, seed=lq)
