@@ -36,10 +36,11 @@ class _Block(nn.Module):
         self.c3 = Conv2d(c, c, 3)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        h = torch.relu(self.c1(x))
-        h = torch.relu(self.c2(h))
-        h = self.c3(h)
-        return torch.relu(h + x)
+        # relu fused into each conv epilogue; the skip add is fused into c3
+        # (epilogue order: relu(conv + bias + residual))
+        h = self.c1(x, act=ops.ACT_RELU)
+        h = self.c2(h, act=ops.ACT_RELU)
+        return self.c3(h, residual=x, act=ops.ACT_RELU)
 
 
 class TAESDEncoder(nn.Module):
@@ -91,7 +92,7 @@ class TAESDDecoder(nn.Module):
     def forward(self, z: torch.Tensor) -> torch.Tensor:
         # tanh clamp keeps extreme latents in the trained range (TAESD-style)
         z = torch.tanh(z / 3.0) * 3.0
-        h = torch.relu(self.conv_in(z))
+        h = self.conv_in(z, act=ops.ACT_RELU)
         for b in self.stage1:
             h = b(h)
         h = self.up1(ops.upsample_nearest2x_nhwc(h))

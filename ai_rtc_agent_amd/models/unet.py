@@ -114,8 +114,18 @@ class Conv2d(nn.Module):
         self.weight = nn.Parameter(torch.randn(cout, cin, k, k) * (1.0 / math.sqrt(cin * k * k)))
         self.bias = nn.Parameter(torch.zeros(cout)) if bias else None
 
-    def forward(self, x: torch.Tensor, fuse_silu: bool = False) -> torch.Tensor:
-        return ops.conv2d_nhwc(x, self.weight, self.bias, self.stride, self.padding, fuse_silu)
+    def forward(
+        self,
+        x: torch.Tensor,
+        fuse_silu: bool = False,
+        act: int | None = None,
+        residual: torch.Tensor | None = None,
+        channel_bias: torch.Tensor | None = None,
+    ) -> torch.Tensor:
+        return ops.conv2d_nhwc(
+            x, self.weight, self.bias, self.stride, self.padding, fuse_silu,
+            act=act, residual=residual, channel_bias=channel_bias,
+        )
 
 
 class GroupNormSiLU(nn.Module):
@@ -233,11 +243,11 @@ class ResnetBlock(nn.Module):
         self.shortcut = Conv2d(cin, cout, 1) if cin != cout else None
 
     def forward(self, x: torch.Tensor, temb: torch.Tensor) -> torch.Tensor:
-        h = self.conv1(self.norm1(x))
-        h = h + self.time_emb_proj(ops.silu(temb))[:, None, None, :]
-        h = self.conv2(self.norm2(h))
+        # time-emb add fused into conv1's epilogue; skip add fused into conv2's
+        temb_b = self.time_emb_proj(ops.silu(temb))
+        h = self.conv1(self.norm1(x), channel_bias=temb_b)
         skip = self.shortcut(x) if self.shortcut is not None else x
-        return h + skip
+        return self.conv2(self.norm2(h), residual=skip)
 
 
 class Downsample(nn.Module):

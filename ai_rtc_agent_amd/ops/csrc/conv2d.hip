@@ -1,161 +1,237 @@
 // Conv2d NHWC — implicit-GEMM on MFMA (CDNA4 16x16x32 f16 tiles).
 //
-// This is the MI355X-native replacement for the reference's TensorRT
-// UNet/VAE conv engines (reference
-// lib/wrapper.py:409-512). Not a port: designed for gfx950 per
-// /opt/skills/guides/cdna_hip_programming.md —
-//  - GEMM view M=(B·HO·WO) pixels, N=OC, K=R·S·IC with the gather
-//    on-the-fly (im2col never materialised)
-//  - inputs are zero-PADDED NHWC so the K-gather has no bounds checks;
-//    channel runs are 16B-contiguous (IC%32==0 on this path), so each lane
-//    stages 8 f16 per load (guide G13)
-//  - LDS tiles padded by one 16B access width (guide G4) for
-//    conflict-reduced ds_read_b128 fragment reads
-//  - T14 async-stage split: next K-tile's global loads issue before the
-//    current tile's MFMAs so HBM latency hides under compute
-//  - bias + activation fused in the epilogue
+// MI355X-native replacement for the reference's TensorRT UNet/VAE conv
+// engines (reference lib/wrapper.py:409-512).
+// Designed for gfx950 per /opt/skills/guides/cdna_hip_programming.md:
 //
-// Tile: BM=128 pixels x BN=64 out-channels x BK=32, 4 waves (2x2), each wave
-// a 64x32 sub-tile = 4x2 fragments of 16x16, 8 MFMA per K-step.
+//  - GEMM view M=(HO·WO) pixels x N=OC x K=R·S·IC; the im2col gather runs
+//    on the fly with INLINE zero-padding (per-load bounds predicate — no
+//    padded input copy, no aten F.pad fill+copy kernels)
+//  - channel runs are 16B-contiguous (IC%32==0 on the MFMA path): each lane
+//    stages 8 f16 per load (guide G13)
+//  - LDS tiles padded by one 16B access width (guide G4) for ds_read_b128
+//  - T14 async-stage split: next K-tile's global loads issue before the
+//    current tile's MFMAs
+//  - EPILOGUE FUSION: bias + per-(batch,channel) bias (time-embedding add)
+//    + residual add + activation, all in the conv store
+//  - two tile geometries + SPLIT-K: the UNet's small-spatial wide-channel
+//    layers (8²x1280: M=64, K=11520) would otherwise launch 20 workgroups
+//    on a 256-CU chip (measured 235us each, 56% of frame time); split-K
+//    over the K loop with f32 slab partials + a finalize pass fills the
+//    chip (profiles/ has the before/after)
+//
+// Geometry A (large M): BM=128 BN=64, 4 waves (2x2), wave=64x32, 8 MFMA/step
+// Geometry B (small M): BM=64  BN=64, 4 waves (2x2), wave=32x32, 4 MFMA/step,
+//                       optional split-K over blockIdx.z
 
 #include "common.h"
 
-#define BM 128
 #define BN 64
 #define BK 32
-#define APITCH (BK + 8)  // f16 elements per LDS row (+16B pad)
+#define KPITCH (BK + 8)  // f16 elements per LDS row (+16B pad)
 
+struct KPos {
+  int r, s, ic0;
+};
+
+__device__ __forceinline__ KPos kpos_at(int k0, int IC, int S) {
+  const int rs = k0 / IC;
+  return {rs / S, rs - (rs / S) * S, k0 - rs * IC};
+}
+
+// one staged 16B A-load with inline zero-padding
+__device__ __forceinline__ f16x8 load_a(const f16* xb, int ho_s, int wo_s,
+                                        int r, int s, int pad, int H, int W,
+                                        int IC, int ic) {
+  const int hi = ho_s + r - pad;
+  const int wi = wo_s + s - pad;
+  const bool ok = (unsigned)hi < (unsigned)H && (unsigned)wi < (unsigned)W;
+  const int hc = ok ? hi : 0, wc = ok ? wi : 0;
+  f16x8 v = *reinterpret_cast<const f16x8*>(&xb[((long)hc * W + wc) * IC + ic]);
+  if (!ok) v = f16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  return v;
+}
+
+// fused epilogue value: acc + bias + cbias, + residual, then act
+__device__ __forceinline__ f16 epilogue(float acc, const float* bias,
+                                        const f16* cbias, long cb_off,
+                                        const f16* residual, long idx, int oc,
+                                        int act) {
+  float v = acc;
+  if (bias) v += bias[oc];
+  if (cbias) v += (float)cbias[cb_off + oc];
+  if (residual) v += (float)residual[idx];
+  return (f16)apply_act(v, act);
+}
+
+// ---------------------------------------------------------------------------
+// templated MFMA conv: MFRAG = M-fragments per wave (4 -> BM=128, 2 -> BM=64)
+// SPLITK > 1 only used with the small geometry; partials go to ws (f32).
+// ---------------------------------------------------------------------------
+template <int MFRAG>
 __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
     const f16* __restrict__ x, const f16* __restrict__ w,
-    const float* __restrict__ bias, f16* __restrict__ out, int Hp, int Wp,
-    int IC, int HO, int WO, int OC, int R, int S, int stride, int act,
-    int K) {
-  __shared__ f16 ldsA[BM * APITCH];
-  __shared__ f16 ldsB[BN * APITCH];
+    const float* __restrict__ bias, const f16* __restrict__ cbias,
+    const f16* __restrict__ residual, f16* __restrict__ out,
+    float* __restrict__ ws, int H, int W, int IC, int HO, int WO, int OC,
+    int R, int S, int stride, int pad, int act, int K, int splitk) {
+  constexpr int BM = MFRAG * 32;          // 128 or 64
+  constexpr int ALOADS = MFRAG / 2;       // staged 16B A-loads per thread
+  __shared__ f16 ldsA[BM * KPITCH];
+  __shared__ f16 ldsB[BN * KPITCH];
 
   const int M = HO * WO;
   const int m0 = blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
-  const long b = blockIdx.z;
-  const f16* xb = x + b * (long)Hp * Wp * IC;
+  const int b = blockIdx.z / splitk;
+  const int split = blockIdx.z - b * splitk;
+  const f16* xb = x + (long)b * H * W * IC;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const int wm = wid & 1;   // wave row (2 waves over M)
-  const int wn = wid >> 1;  // wave col (2 waves over N)
+  const int wm = wid & 1;
+  const int wn = wid >> 1;
 
-  // --- staging assignments (per K-step) ---
-  // A: 512 x 16B loads; thread t does flats {t, t+256}
-  int a_row[2], a_hi[2], a_wi[2], a_k8[2];
+  // staging map
+  int a_row[ALOADS], a_ho[ALOADS], a_wo[ALOADS], a_k8[ALOADS];
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
-    int flat = tid + i * 256;
+  for (int i = 0; i < ALOADS; ++i) {
+    int flat = tid + i * 256;           // [0, BM*4)
     a_row[i] = flat >> 2;
     a_k8[i] = (flat & 3) * 8;
-    int m = m0 + a_row[i];
-    if (m >= M) m = M - 1;  // clamp: duplicate loads, stores masked
-    a_hi[i] = (m / WO) * stride;
-    a_wi[i] = (m % WO) * stride;
+    int m = min(m0 + a_row[i], M - 1);
+    a_ho[i] = (m / WO) * stride;
+    a_wo[i] = (m % WO) * stride;
   }
-  // B: 256 x 16B loads
   const int b_row = tid >> 2;
   const int b_k8 = (tid & 3) * 8;
   const int b_oc = min(n0 + b_row, OC - 1);
   const f16* wrow = w + (long)b_oc * K + b_k8;
 
-  f32x4 acc[4][2];
+  f32x4 acc[MFRAG][2];
 #pragma unroll
-  for (int mi = 0; mi < 4; ++mi)
+  for (int mi = 0; mi < MFRAG; ++mi)
 #pragma unroll
     for (int ni = 0; ni < 2; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
 
+  // K-step range for this split
   const int nk = K / BK;
+  const int per = (nk + splitk - 1) / splitk;
+  const int k_lo = split * per;
+  const int k_hi = min(nk, k_lo + per);
+  // NOTE: an empty split (uneven tail) still stores its zero slab below —
+  // the finalize pass reads every slab.
 
-  // prologue: load K-tile 0 into registers
-  f16x8 regA[2], regB;
-  {
-    const int rs = 0, ic0 = 0, r = 0, sc = 0;
-    (void)rs; (void)ic0; (void)r; (void)sc;
+  // prologue: stage first K-tile of this split into registers
+  f16x8 regA[ALOADS], regB;
+  if (k_lo < k_hi) {
+    KPos p = kpos_at(k_lo * BK, IC, S);
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
-      regA[i] = *reinterpret_cast<const f16x8*>(
-          &xb[((long)a_hi[i] * Wp + a_wi[i]) * IC + a_k8[i]]);
-    regB = *reinterpret_cast<const f16x8*>(wrow);
+    for (int i = 0; i < ALOADS; ++i)
+      regA[i] = load_a(xb, a_ho[i], a_wo[i], p.r, p.s, pad, H, W, IC,
+                       p.ic0 + a_k8[i]);
+    regB = *reinterpret_cast<const f16x8*>(wrow + k_lo * BK);
   }
 
-  for (int kt = 0; kt < nk; ++kt) {
-    __syncthreads();  // previous tile's fragment reads done
-    // write staged registers to LDS
+  for (int kt = k_lo; kt < k_hi; ++kt) {
+    __syncthreads();
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
-      *reinterpret_cast<f16x8*>(&ldsA[a_row[i] * APITCH + a_k8[i]]) = regA[i];
-    *reinterpret_cast<f16x8*>(&ldsB[b_row * APITCH + b_k8]) = regB;
+    for (int i = 0; i < ALOADS; ++i)
+      *reinterpret_cast<f16x8*>(&ldsA[a_row[i] * KPITCH + a_k8[i]]) = regA[i];
+    *reinterpret_cast<f16x8*>(&ldsB[b_row * KPITCH + b_k8]) = regB;
     __syncthreads();
 
-    // T14: issue NEXT tile's global loads before this tile's MFMAs
-    if (kt + 1 < nk) {
-      const int k0 = (kt + 1) * BK;
-      const int rs = k0 / IC;  // BK | IC, so one (r,s) per K-step
-      const int ic0 = k0 - rs * IC;
-      const int r = rs / S;
-      const int sc = rs - r * S;
+    if (kt + 1 < k_hi) {  // T14: next tile's loads before MFMAs
+      KPos p = kpos_at((kt + 1) * BK, IC, S);
 #pragma unroll
-      for (int i = 0; i < 2; ++i)
-        regA[i] = *reinterpret_cast<const f16x8*>(
-            &xb[((long)(a_hi[i] + r) * Wp + (a_wi[i] + sc)) * IC + ic0 +
-                a_k8[i]]);
-      regB = *reinterpret_cast<const f16x8*>(wrow + k0);
+      for (int i = 0; i < ALOADS; ++i)
+        regA[i] = load_a(xb, a_ho[i], a_wo[i], p.r, p.s, pad, H, W, IC,
+                         p.ic0 + a_k8[i]);
+      regB = *reinterpret_cast<const f16x8*>(wrow + (kt + 1) * BK);
     }
 
-    // fragments + MFMA
-    const int arow_base = wm * 64 + (lane & 15);
+    const int arow_base = wm * (MFRAG * 16) + (lane & 15);
     const int fcol = (lane >> 4) * 8;
     f16x8 bfrag[2];
 #pragma unroll
     for (int ni = 0; ni < 2; ++ni)
       bfrag[ni] = *reinterpret_cast<const f16x8*>(
-          &ldsB[(wn * 32 + ni * 16 + (lane & 15)) * APITCH + fcol]);
+          &ldsB[(wn * 32 + ni * 16 + (lane & 15)) * KPITCH + fcol]);
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi) {
+    for (int mi = 0; mi < MFRAG; ++mi) {
       f16x8 afrag = *reinterpret_cast<const f16x8*>(
-          &ldsA[(arow_base + mi * 16) * APITCH + fcol]);
+          &ldsA[(arow_base + mi * 16) * KPITCH + fcol]);
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni)
         acc[mi][ni] = mfma16x16x32(afrag, bfrag[ni], acc[mi][ni]);
     }
   }
 
-  // epilogue: bias + activation + masked f16 stores
-  f16* ob = out + b * (long)M * OC;
+  if (splitk == 1) {
+    f16* ob = out + (long)b * M * OC;
+    const long cb_off = (long)b * OC;
 #pragma unroll
-  for (int ni = 0; ni < 2; ++ni) {
-    const int col = n0 + wn * 32 + ni * 16 + (lane & 15);
-    if (col >= OC) continue;
-    const float bv = bias ? bias[col] : 0.0f;
+    for (int ni = 0; ni < 2; ++ni) {
+      const int col = n0 + wn * 32 + ni * 16 + (lane & 15);
+      if (col >= OC) continue;
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi) {
+      for (int mi = 0; mi < MFRAG; ++mi)
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const int m = m0 + wm * 64 + mi * 16 + (lane >> 4) * 4 + j;
-        if (m < M)
-          ob[(long)m * OC + col] = (f16)apply_act(acc[mi][ni][j] + bv, act);
-      }
+        for (int j = 0; j < 4; ++j) {
+          const int m = m0 + wm * (MFRAG * 16) + mi * 16 + (lane >> 4) * 4 + j;
+          if (m < M) {
+            const long idx = (long)b * M * OC + (long)m * OC + col;
+            ob[(long)m * OC + col] = epilogue(acc[mi][ni][j], bias, cbias,
+                                              cb_off, residual, idx, col, act);
+          }
+        }
+    }
+  } else {
+    // plain f32 slab store; finalize kernel reduces + applies the epilogue
+    float* wsb = ws + ((long)b * splitk + split) * M * OC;
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int col = n0 + wn * 32 + ni * 16 + (lane & 15);
+      if (col >= OC) continue;
+#pragma unroll
+      for (int mi = 0; mi < MFRAG; ++mi)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int m = m0 + wm * (MFRAG * 16) + mi * 16 + (lane >> 4) * 4 + j;
+          if (m < M) wsb[(long)m * OC + col] = acc[mi][ni][j];
+        }
     }
   }
 }
 
-// ---------------------------------------------------------------------------
-// Direct conv for small/ragged IC (conv_in with IC=3/4, final RGB convs):
-// K is tiny (<= 9*31), one thread per output element, f32 accumulate.
-// ---------------------------------------------------------------------------
-__global__ void conv2d_direct_kernel(const f16* __restrict__ x,
-                                     const f16* __restrict__ w,
+__global__ void conv_splitk_finalize(const float* __restrict__ ws,
                                      const float* __restrict__ bias,
-                                     f16* __restrict__ out, int Hp, int Wp,
-                                     int IC, int HO, int WO, int OC, int R,
-                                     int S, int stride, int act, int K,
-                                     long total) {
+                                     const f16* __restrict__ cbias,
+                                     const f16* __restrict__ residual,
+                                     f16* __restrict__ out, int M, int OC,
+                                     int splitk, int act, long total) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * (long)blockDim.x) {
+    const int oc = i % OC;
+    const long bm = i / OC;
+    const long b = bm / M;
+    const float* p = ws + (long)b * splitk * M * OC + (bm - b * M) * OC + oc;
+    float a = 0.f;
+    for (int s = 0; s < splitk; ++s) a += p[(long)s * M * OC];
+    out[i] = epilogue(a, bias, cbias, b * OC, residual, i, oc, act);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Direct conv for small/ragged IC (conv_in with IC=3/4, final RGB convs)
+// ---------------------------------------------------------------------------
+__global__ void conv2d_direct_kernel(
+    const f16* __restrict__ x, const f16* __restrict__ w,
+    const float* __restrict__ bias, const f16* __restrict__ cbias,
+    const f16* __restrict__ residual, f16* __restrict__ out, int H, int W,
+    int IC, int HO, int WO, int OC, int R, int S, int stride, int pad, int act,
+    int K, long total) {
   const int M = HO * WO;
   for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
        i += gridDim.x * (long)blockDim.x) {
@@ -164,45 +240,91 @@ __global__ void conv2d_direct_kernel(const f16* __restrict__ x,
     const int m = rest % M;
     const long b = rest / M;
     const int ho = m / WO, wo = m % WO;
-    const f16* xb = x + b * (long)Hp * Wp * IC;
+    const f16* xb = x + b * (long)H * W * IC;
     const f16* wk = w + (long)oc * K;
-    float a = bias ? bias[oc] : 0.0f;
-    for (int r = 0; r < R; ++r)
+    float a = 0.f;
+    for (int r = 0; r < R; ++r) {
+      const int hi = ho * stride + r - pad;
+      if ((unsigned)hi >= (unsigned)H) continue;
       for (int s = 0; s < S; ++s) {
-        const f16* xr =
-            &xb[((long)(ho * stride + r) * Wp + (wo * stride + s)) * IC];
+        const int wi = wo * stride + s - pad;
+        if ((unsigned)wi >= (unsigned)W) continue;
+        const f16* xr = &xb[((long)hi * W + wi) * IC];
         const f16* wr = &wk[(r * S + s) * IC];
         for (int c = 0; c < IC; ++c) a += (float)xr[c] * (float)wr[c];
       }
-    out[i] = (f16)apply_act(a, act);
+    }
+    out[i] = epilogue(a, bias, cbias, b * OC, residual, i, oc, act);
   }
 }
 
-extern "C" void airtc_conv2d_mfma(const uint16_t* x_pad, const uint16_t* w,
-                                  const float* bias, uint16_t* out, int B,
-                                  int Hp, int Wp, int IC, int HO, int WO,
-                                  int OC, int R, int S, int stride, int act,
-                                  hipStream_t s) {
-  const int K = R * S * IC;
-  dim3 grid(ceil_div(HO * WO, BM), ceil_div(OC, BN), B);
-  hipLaunchKernelGGL(conv2d_mfma_kernel, grid, dim3(256), 0, s,
-                     reinterpret_cast<const f16*>(x_pad),
-                     reinterpret_cast<const f16*>(w), bias,
-                     reinterpret_cast<f16*>(out), Hp, Wp, IC, HO, WO, OC, R, S,
-                     stride, act, K);
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+extern "C" int airtc_conv2d_splitk_for(int B, int HO, int WO, int OC, int IC) {
+  // path/geometry decision, exported so the host can size the workspace:
+  // returns 0 = direct, 1 = BM128, -k = BM64 with split-K k
+  if (IC % 32 != 0) return 0;
+  const int M = HO * WO;
+  const long blocks128 = (long)ceil_div(M, 128) * ceil_div(OC, BN) * B;
+  if (blocks128 >= 120) return 1;
+  const long blocks64 = (long)ceil_div(M, 64) * ceil_div(OC, BN) * B;
+  int k = (int)((256 + blocks64 - 1) / blocks64);
+  if (k < 1) k = 1;
+  if (k > 16) k = 16;
+  return -k;
 }
 
-extern "C" void airtc_conv2d_direct(const uint16_t* x_pad, const uint16_t* w,
-                                    const float* bias, uint16_t* out, int B,
-                                    int Hp, int Wp, int IC, int HO, int WO,
-                                    int OC, int R, int S, int stride, int act,
-                                    hipStream_t s) {
+extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
+                                  const float* bias, const uint16_t* cbias,
+                                  const uint16_t* residual, uint16_t* out,
+                                  float* ws, int B, int H, int W, int IC,
+                                  int HO, int WO, int OC, int R, int S,
+                                  int stride, int pad, int act, int path,
+                                  hipStream_t s) {
+  const int K = R * S * IC;
+  const f16* xp = reinterpret_cast<const f16*>(x);
+  const f16* wp = reinterpret_cast<const f16*>(w);
+  const f16* cb = reinterpret_cast<const f16*>(cbias);
+  const f16* res = reinterpret_cast<const f16*>(residual);
+  f16* op = reinterpret_cast<f16*>(out);
+  const int M = HO * WO;
+
+  if (path == 1) {
+    dim3 grid(ceil_div(M, 128), ceil_div(OC, BN), B);
+    hipLaunchKernelGGL(conv2d_mfma_kernel<4>, grid, dim3(256), 0, s, xp, wp,
+                       bias, cb, res, op, ws, H, W, IC, HO, WO, OC, R, S,
+                       stride, pad, act, K, 1);
+    return;
+  }
+  const int splitk = -path;
+  dim3 grid(ceil_div(M, 64), ceil_div(OC, BN), B * splitk);
+  hipLaunchKernelGGL(conv2d_mfma_kernel<2>, grid, dim3(256), 0, s, xp, wp,
+                     splitk == 1 ? bias : nullptr, splitk == 1 ? cb : nullptr,
+                     splitk == 1 ? res : nullptr, op, ws, H, W, IC, HO, WO, OC,
+                     R, S, stride, pad, act, K, splitk);
+  if (splitk > 1) {
+    long total = (long)B * M * OC;
+    int blocks = (int)min((long)2048, (total + 255) / 256);
+    hipLaunchKernelGGL(conv_splitk_finalize, dim3(blocks), dim3(256), 0, s, ws,
+                       bias, cb, res, op, M, OC, splitk, act, total);
+  }
+}
+
+extern "C" void airtc_conv2d_direct(const uint16_t* x, const uint16_t* w,
+                                    const float* bias, const uint16_t* cbias,
+                                    const uint16_t* residual, uint16_t* out,
+                                    int B, int H, int W, int IC, int HO,
+                                    int WO, int OC, int R, int S, int stride,
+                                    int pad, int act, hipStream_t s) {
   const int K = R * S * IC;
   long total = (long)B * HO * WO * OC;
   int blocks = (int)min((long)4096, (total + 255) / 256);
   hipLaunchKernelGGL(conv2d_direct_kernel, dim3(blocks), dim3(256), 0, s,
-                     reinterpret_cast<const f16*>(x_pad),
+                     reinterpret_cast<const f16*>(x),
                      reinterpret_cast<const f16*>(w), bias,
-                     reinterpret_cast<f16*>(out), Hp, Wp, IC, HO, WO, OC, R, S,
-                     stride, act, K, total);
+                     reinterpret_cast<const f16*>(cbias),
+                     reinterpret_cast<const f16*>(residual),
+                     reinterpret_cast<f16*>(out), H, W, IC, HO, WO, OC, R, S,
+                     stride, pad, act, K, total);
 }

@@ -26,25 +26,54 @@ uint16_t* h_ptr_mut(torch::Tensor& t) {
   return reinterpret_cast<uint16_t*>(t.data_ptr<at::Half>());
 }
 
-torch::Tensor conv2d(torch::Tensor x_pad, torch::Tensor w_perm,
-                     c10::optional<torch::Tensor> bias, int64_t HO, int64_t WO,
-                     int64_t R, int64_t S, int64_t stride, int64_t act,
-                     bool use_mfma) {
-  CHECK_IN(x_pad);
+torch::Tensor conv2d(torch::Tensor x, torch::Tensor w_perm,
+                     c10::optional<torch::Tensor> bias,
+                     c10::optional<torch::Tensor> cbias,
+                     c10::optional<torch::Tensor> residual, int64_t R,
+                     int64_t S, int64_t stride, int64_t pad, int64_t act) {
+  CHECK_IN(x);
   CHECK_IN(w_perm);
-  TORCH_CHECK(x_pad.dtype() == torch::kHalf && w_perm.dtype() == torch::kHalf);
-  const int B = x_pad.size(0), Hp = x_pad.size(1), Wp = x_pad.size(2),
-            IC = x_pad.size(3);
+  TORCH_CHECK(x.dtype() == torch::kHalf && w_perm.dtype() == torch::kHalf);
+  const int B = x.size(0), H = x.size(1), W = x.size(2), IC = x.size(3);
   const int OC = w_perm.size(0);
-  auto out = torch::empty({B, HO, WO, OC}, x_pad.options());
+  const int HO = (H + 2 * (int)pad - (int)R) / (int)stride + 1;
+  const int WO = (W + 2 * (int)pad - (int)S) / (int)stride + 1;
+  auto out = torch::empty({B, HO, WO, OC}, x.options());
   const float* bp = nullptr;
   if (bias.has_value()) {
     TORCH_CHECK(bias->dtype() == torch::kFloat && bias->is_contiguous());
     bp = bias->data_ptr<float>();
   }
-  auto fn = use_mfma ? airtc_conv2d_mfma : airtc_conv2d_direct;
-  fn(h_ptr(x_pad), h_ptr(w_perm), bp, h_ptr_mut(out), B, Hp, Wp, IC, (int)HO,
-     (int)WO, OC, (int)R, (int)S, (int)stride, (int)act, cur_stream());
+  const uint16_t* cb = nullptr;
+  if (cbias.has_value()) {
+    CHECK_IN((*cbias));
+    TORCH_CHECK(cbias->dtype() == torch::kHalf);
+    cb = h_ptr(*cbias);
+  }
+  const uint16_t* res = nullptr;
+  if (residual.has_value()) {
+    CHECK_IN((*residual));
+    TORCH_CHECK(residual->dtype() == torch::kHalf);
+    TORCH_CHECK(residual->numel() == out.numel(), "residual shape mismatch");
+    res = h_ptr(*residual);
+  }
+  const int path = airtc_conv2d_splitk_for(B, HO, WO, OC, IC);
+  if (path == 0) {
+    airtc_conv2d_direct(h_ptr(x), h_ptr(w_perm), bp, cb, res, h_ptr_mut(out),
+                        B, H, W, IC, HO, WO, OC, (int)R, (int)S, (int)stride,
+                        (int)pad, (int)act, cur_stream());
+    return out;
+  }
+  float* wsp = nullptr;
+  torch::Tensor ws;
+  if (path < -1) {
+    ws = torch::empty({(long)B * (-path), (long)HO * WO, OC},
+                      x.options().dtype(torch::kFloat));
+    wsp = ws.data_ptr<float>();
+  }
+  airtc_conv2d_mfma(h_ptr(x), h_ptr(w_perm), bp, cb, res, h_ptr_mut(out), wsp,
+                    B, H, W, IC, HO, WO, OC, (int)R, (int)S, (int)stride,
+                    (int)pad, (int)act, path, cur_stream());
   return out;
 }
 
@@ -56,9 +85,13 @@ torch::Tensor group_norm_silu(torch::Tensor x, int64_t groups,
   const int C = x.size(-1);
   const long HW = x.numel() / ((long)B * C);
   auto out = torch::empty_like(x);
+  const int nchunk = airtc_group_norm_nchunk(B, (int)groups);
+  auto ws = torch::empty({(long)B * groups * nchunk * 2},
+                         x.options().dtype(torch::kFloat));
   airtc_group_norm_silu(h_ptr(x), gamma.data_ptr<float>(),
-                        beta.data_ptr<float>(), h_ptr_mut(out), B, (int)HW, C,
-                        (int)groups, (float)eps, (int)act, cur_stream());
+                        beta.data_ptr<float>(), h_ptr_mut(out),
+                        ws.data_ptr<float>(), B, (int)HW, C, (int)groups,
+                        (float)eps, (int)act, cur_stream());
   return out;
 }
 

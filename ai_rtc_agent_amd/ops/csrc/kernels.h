@@ -20,25 +20,34 @@ void airtc_upsample2x_f16(const uint16_t* in, uint16_t* out, int B, int H,
                           int W, int C, hipStream_t s);
 
 // norms ---------------------------------------------------------------------
+int airtc_group_norm_nchunk(int B, int G);
 void airtc_group_norm_silu(const uint16_t* x, const float* gamma,
-                           const float* beta, uint16_t* out, int B, int HW,
-                           int C, int G, float eps, int act, hipStream_t s);
+                           const float* beta, uint16_t* out, float* ws, int B,
+                           int HW, int C, int G, float eps, int act,
+                           hipStream_t s);
 void airtc_layer_norm(const uint16_t* x, const float* gamma, const float* beta,
                       uint16_t* out, long rows, int C, float eps,
                       hipStream_t s);
 
 // conv ----------------------------------------------------------------------
-// x_pad: (B, Hp, Wp, IC) NHWC f16 (already zero-padded when padding=1)
-// w    : (OC, R*S*IC) f16, k order = (r, s, ic)
-// out  : (B, HO, WO, OC) f16
-void airtc_conv2d_mfma(const uint16_t* x_pad, const uint16_t* w,
-                       const float* bias, uint16_t* out, int B, int Hp, int Wp,
-                       int IC, int HO, int WO, int OC, int R, int S,
-                       int stride, int act, hipStream_t s);
-void airtc_conv2d_direct(const uint16_t* x_pad, const uint16_t* w,
-                         const float* bias, uint16_t* out, int B, int Hp,
-                         int Wp, int IC, int HO, int WO, int OC, int R, int S,
-                         int stride, int act, hipStream_t s);
+// x   : (B, H, W, IC) NHWC f16 (zero-padding handled inline)
+// w   : (OC, R*S*IC) f16, k order = (r, s, ic)
+// cbias: optional per-(batch, out-channel) f16 bias (B, OC) — time-emb add
+// residual: optional f16 tensor with out's shape, added pre-activation
+// out : (B, HO, WO, OC) f16
+// path: from airtc_conv2d_splitk_for (0=direct, 1=BM128, -k=BM64 splitk k);
+// ws  : f32 workspace (B*k, HO*WO, OC) required when path < -1
+int airtc_conv2d_splitk_for(int B, int HO, int WO, int OC, int IC);
+void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w, const float* bias,
+                       const uint16_t* cbias, const uint16_t* residual,
+                       uint16_t* out, float* ws, int B, int H, int W, int IC,
+                       int HO, int WO, int OC, int R, int S, int stride,
+                       int pad, int act, int path, hipStream_t s);
+void airtc_conv2d_direct(const uint16_t* x, const uint16_t* w,
+                         const float* bias, const uint16_t* cbias,
+                         const uint16_t* residual, uint16_t* out, int B, int H,
+                         int W, int IC, int HO, int WO, int OC, int R, int S,
+                         int stride, int pad, int act, hipStream_t s);
 
 // attention -----------------------------------------------------------------
 // q: base+strides address (B,H) heads; row stride in elements.

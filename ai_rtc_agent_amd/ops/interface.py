@@ -88,13 +88,19 @@ def conv2d_nhwc(
     padding: int = 1,
     fuse_silu: bool = False,
     act: int | None = None,
+    residual: torch.Tensor | None = None,
+    channel_bias: torch.Tensor | None = None,
 ) -> torch.Tensor:
     """x: (B,H,W,C) contiguous; weight: (O,I,R,S) [torch layout]; out (B,H',W',O).
 
-    GPU path: implicit-GEMM on MFMA with on-the-fly im2col gather
-    (ops/csrc/conv2d.hip). Weights are lazily pre-transformed to the
-    (O, R*S*I) GEMM layout and cached on the parameter (the AOT `build`
-    step warms every cache before hipGraph capture).
+    GPU path: implicit-GEMM on MFMA with on-the-fly im2col gather and inline
+    zero-padding (ops/csrc/conv2d.hip). Weights are lazily pre-transformed
+    to the (O, R*S*I) GEMM layout and cached on the parameter (the AOT
+    `build` step warms every cache before hipGraph capture).
+
+    Epilogue fusion (order): y = act(conv + bias + channel_bias + residual)
+      channel_bias: (B, O) — the resnet time-embedding add
+      residual:     broadcast-free tensor of the output shape — skip adds
     """
     if act is None:
         act = ACT_SILU if fuse_silu else ACT_NONE
@@ -109,17 +115,27 @@ def conv2d_nhwc(
         b32 = None
         if bias is not None:
             b32 = _cached(bias, "_airtc_b32", lambda: bias.detach().float().contiguous())
-        x_pad = F.pad(x, (0, 0, padding, padding, padding, padding)) if padding else x
-        B, Hp, Wp, _ = x_pad.shape
-        HO = (Hp - R) // stride + 1
-        WO = (Wp - S) // stride + 1
-        use_mfma = (I % 32 == 0)
-        return ext.conv2d(x_pad, w_perm, b32, HO, WO, R, S, stride, act, use_mfma)
+        return ext.conv2d(
+            x,
+            w_perm,
+            b32,
+            None if channel_bias is None else channel_bias.contiguous(),
+            None if residual is None else residual.contiguous(),
+            R,
+            S,
+            stride,
+            padding,
+            act,
+        )
 
     xc = x.permute(0, 3, 1, 2)
     y = F.conv2d(xc.float(), weight.float(), None if bias is None else bias.float(),
                  stride=stride, padding=padding)
     y = y.permute(0, 2, 3, 1)
+    if channel_bias is not None:
+        y = y + channel_bias.float()[:, None, None, :]
+    if residual is not None:
+        y = y + residual.float()
     if act == ACT_SILU:
         y = F.silu(y)
     elif act == ACT_RELU:

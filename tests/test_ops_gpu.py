@@ -40,6 +40,8 @@ def test_ext_loads():
     (320, 640, 8, 1, 0, 1),     # 1x1 projection
     (4, 32, 16, 1, 1, 3),       # small-IC direct path (VAE conv_in)
     (96, 40, 12, 1, 1, 3),      # ragged OC, non-pow2 spatial
+    (1280, 1280, 8, 1, 1, 3),   # 8x8 wide UNet layer -> split-K path
+    (640, 1280, 16, 2, 1, 3),   # stride-2 into split-K territory
 ])
 def test_conv2d_vs_torch(ic, oc, h, stride, pad, r):
     x = rnd(2, h, h, ic, seed=ic + oc)
@@ -50,6 +52,28 @@ def test_conv2d_vs_torch(ic, oc, h, stride, pad, r):
                    stride=stride, padding=pad).permute(0, 2, 3, 1)
     assert y.shape == ref.shape
     assert_close(y, ref, 2e-2 * max(1.0, math.sqrt(ic * r * r) / 8), "conv2d")
+
+
+def test_conv2d_fused_residual_and_channel_bias():
+    # epilogue law: y = act(conv + bias + channel_bias + residual)
+    x = rnd(2, 8, 8, 64, seed=20)
+    w = rnd(32, 64, 3, 3, seed=21, scale=0.04)
+    cb = rnd(2, 32, seed=22)
+    res = rnd(2, 8, 8, 32, seed=23)
+    y = ops.conv2d_nhwc(x, w, None, residual=res, channel_bias=cb, act=ops.ACT_RELU)
+    base = F.conv2d(x.permute(0, 3, 1, 2).float(), w.float(), padding=1).permute(0, 2, 3, 1)
+    ref = F.relu(base + cb.float()[:, None, None, :] + res.float())
+    assert_close(y, ref, 2e-2, "conv fused epilogue")
+
+
+def test_conv2d_splitk_fused_epilogue():
+    # split-K path (small spatial, wide channels) must apply the same epilogue
+    x = rnd(1, 8, 8, 1280, seed=30, scale=0.2)
+    w = rnd(1280, 1280, 3, 3, seed=31, scale=0.01)
+    res = rnd(1, 8, 8, 1280, seed=32)
+    y = ops.conv2d_nhwc(x, w, None, residual=res)
+    base = F.conv2d(x.permute(0, 3, 1, 2).float(), w.float(), padding=1).permute(0, 2, 3, 1)
+    assert_close(y, base + res.float(), 5e-2, "conv splitk epilogue")
 
 
 def test_conv2d_fused_silu():
