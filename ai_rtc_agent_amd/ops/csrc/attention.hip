@@ -7,8 +7,10 @@
 //
 // Structure (guide Appendix B "fused attention prefill"):
 //  - workgroup = 4 waves = one 64-row Q tile; wave owns 16 q-rows
-//  - K/V streamed through LDS in 64-key tiles; V stored TRANSPOSED ([d][kv])
-//    at staging so the PV B-fragment is a contiguous ds_read_b128
+//  - K/V streamed through LDS in 64-key tiles, BOTH row-major (cheap vec8
+//    staging); the PV B-fragment comes out of the row-major V tile via the
+//    gfx950 hardware transpose read ds_read_b64_tr_b16 (guide T10) — this
+//    replaced a 16-scalar-ds_write-per-thread V transpose at staging
 //  - online softmax: running (m, l) per q-row, kept lane-local (each lane
 //    owns the same 4 q-rows its C-fragments do), 4x shfl_xor row reduce
 //  - P goes through a per-wave LDS tile to convert the C-fragment layout
@@ -28,6 +30,44 @@
 #define QT 64   // q rows per workgroup
 #define PPITCH (KVT + 8)
 
+typedef __attribute__((__vector_size__(2 * sizeof(unsigned)))) unsigned u32x2;
+typedef const __attribute__((address_space(3))) f16* lds_cptr;
+
+__device__ __forceinline__ unsigned lds_addr(const f16* p) {
+  return (unsigned)(unsigned long long)(lds_cptr)p;
+}
+
+// B-fragment (8 keys x 16 cols) from a ROW-major [kv][pitch] f16 tile via two
+// hardware transpose reads. Per 16-lane group g (= keys kbase+8g..+7): lane
+// il supplies the address of its 4 contiguous f16 (row kbase+(il>>2),
+// col c0+(il&3)*4); the read hands lane il column c0+il over those rows.
+// Alignment: addresses are 8B-aligned (pitch even in f16) — required, a
+// misaligned tr read returns wrong data silently (guide G17).
+__device__ __forceinline__ f16x8 tr_bfrag(const f16* tile, int pitch,
+                                          int kbase, int c0, int lane) {
+  const int il = lane & 15;
+  const int g = lane >> 4;
+  const f16* p0 = tile + (kbase + g * 8 + (il >> 2)) * pitch + c0 + (il & 3) * 4;
+  const unsigned a0 = lds_addr(p0);
+  const unsigned a1 = a0 + 4 * (unsigned)pitch * 2;  // keys +4..+7
+  u32x2 lo, hi;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(lo), "=&v"(hi)
+      : "v"(a0), "v"(a1)
+      : "memory");
+  __builtin_amdgcn_sched_barrier(0);  // rule 18: keep the MFMA below the wait
+  union {
+    u32x2 u2[2];
+    f16x8 f8;
+  } cvt;
+  cvt.u2[0] = lo;
+  cvt.u2[1] = hi;
+  return cvt.f8;
+}
+
 template <int D>
 __global__ __launch_bounds__(256) void attention_kernel(
     const f16* __restrict__ q, const f16* __restrict__ k,
@@ -35,10 +75,9 @@ __global__ __launch_bounds__(256) void attention_kernel(
     long q_sb, long q_sh, long q_row, long k_sb, long k_sh, long k_row,
     long o_sb, long o_sh, long o_row, float scale) {
   constexpr int KPITCH = D + 8;
-  constexpr int VPITCH = KVT + 8;
   constexpr int D8 = D / 8, D32 = D / 32, D16 = D / 16;
   __shared__ f16 ldsK[KVT * KPITCH];
-  __shared__ f16 ldsV[D * VPITCH];
+  __shared__ f16 ldsV[KVT * KPITCH];  // row-major like K; PV reads via tr_b16
   __shared__ f16 ldsP[4 * 16 * PPITCH];
 
   const int bh = blockIdx.y;
@@ -83,8 +122,7 @@ __global__ __launch_bounds__(256) void attention_kernel(
       f16x8 k8 = *reinterpret_cast<const f16x8*>(&kb[(long)krow * k_row + c8]);
       *reinterpret_cast<f16x8*>(&ldsK[row * KPITCH + c8]) = k8;
       f16x8 v8 = *reinterpret_cast<const f16x8*>(&vb[(long)krow * k_row + c8]);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) ldsV[(c8 + j) * VPITCH + row] = v8[j];
+      *reinterpret_cast<f16x8*>(&ldsV[row * KPITCH + c8]) = v8;
     }
     __syncthreads();
 
@@ -148,15 +186,14 @@ __global__ __launch_bounds__(256) void attention_kernel(
     // same-wave LDS write->read: wait for the writes, keep reads below
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
-    // ---- O += P V ----
+    // ---- O += P V  (V B-fragments via hardware transpose read) ----
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       f16x8 afrag = *reinterpret_cast<const f16x8*>(
           &myP[(lane & 15) * PPITCH + ks * 32 + fcol]);
 #pragma unroll
       for (int f = 0; f < D16; ++f) {
-        f16x8 bfrag = *reinterpret_cast<const f16x8*>(
-            &ldsV[(f * 16 + (lane & 15)) * VPITCH + ks * 32 + fcol]);
+        f16x8 bfrag = tr_bfrag(ldsV, KPITCH, ks * 32, f * 16, lane);
         o_acc[f] = mfma16x16x32(afrag, bfrag, o_acc[f]);
       }
     }

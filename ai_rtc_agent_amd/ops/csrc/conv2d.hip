@@ -263,16 +263,24 @@ __global__ void conv2d_direct_kernel(
 // ---------------------------------------------------------------------------
 extern "C" int airtc_conv2d_splitk_for(int B, int HO, int WO, int OC, int IC) {
   // path/geometry decision, exported so the host can size the workspace:
-  // returns 0 = direct, 1 = BM128, -k = BM64 with split-K k
+  // returns 0 = direct, +k = BM128 split-K k, -k = BM64 split-K k.
+  // Target ~2 blocks/CU (>=480 workgroups): at 1 block/CU a wave can only
+  // hide latency with its own ILP (measured: 64x64x320 conv at 160 blocks
+  // ran 87us = 86 TF; split-K over the K loop fills the chip).
   if (IC % 32 != 0) return 0;
   const int M = HO * WO;
-  const long blocks128 = (long)ceil_div(M, 128) * ceil_div(OC, BN) * B;
-  if (blocks128 >= 120) return 1;
+  if (M >= 2048) {
+    const long blocks = (long)ceil_div(M, 128) * ceil_div(OC, BN) * B;
+    long k = (480 + blocks - 1) / blocks;
+    if (k < 1) k = 1;
+    if (k > 8) k = 8;
+    return (int)k;
+  }
   const long blocks64 = (long)ceil_div(M, 64) * ceil_div(OC, BN) * B;
-  int k = (int)((256 + blocks64 - 1) / blocks64);
+  long k = (512 + blocks64 - 1) / blocks64;
   if (k < 1) k = 1;
-  if (k > 16) k = 16;
-  return -k;
+  if (k > 32) k = 32;
+  return (int)(-k);
 }
 
 extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
@@ -290,19 +298,20 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
   f16* op = reinterpret_cast<f16*>(out);
   const int M = HO * WO;
 
-  if (path == 1) {
-    dim3 grid(ceil_div(M, 128), ceil_div(OC, BN), B);
+  const int splitk = path > 0 ? path : -path;
+  const int bm = path > 0 ? 128 : 64;
+  dim3 grid(ceil_div(M, bm), ceil_div(OC, BN), B * splitk);
+  const float* b1 = splitk == 1 ? bias : nullptr;
+  const f16* cb1 = splitk == 1 ? cb : nullptr;
+  const f16* res1 = splitk == 1 ? res : nullptr;
+  if (path > 0)
     hipLaunchKernelGGL(conv2d_mfma_kernel<4>, grid, dim3(256), 0, s, xp, wp,
-                       bias, cb, res, op, ws, H, W, IC, HO, WO, OC, R, S,
-                       stride, pad, act, K, 1);
-    return;
-  }
-  const int splitk = -path;
-  dim3 grid(ceil_div(M, 64), ceil_div(OC, BN), B * splitk);
-  hipLaunchKernelGGL(conv2d_mfma_kernel<2>, grid, dim3(256), 0, s, xp, wp,
-                     splitk == 1 ? bias : nullptr, splitk == 1 ? cb : nullptr,
-                     splitk == 1 ? res : nullptr, op, ws, H, W, IC, HO, WO, OC,
-                     R, S, stride, pad, act, K, splitk);
+                       b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC, R, S,
+                       stride, pad, act, K, splitk);
+  else
+    hipLaunchKernelGGL(conv2d_mfma_kernel<2>, grid, dim3(256), 0, s, xp, wp,
+                       b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC, R, S,
+                       stride, pad, act, K, splitk);
   if (splitk > 1) {
     long total = (long)B * M * OC;
     int blocks = (int)min((long)2048, (total + 255) / 256);

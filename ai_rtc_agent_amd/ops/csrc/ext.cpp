@@ -66,8 +66,9 @@ torch::Tensor conv2d(torch::Tensor x, torch::Tensor w_perm,
   }
   float* wsp = nullptr;
   torch::Tensor ws;
-  if (path < -1) {
-    ws = torch::empty({(long)B * (-path), (long)HO * WO, OC},
+  const int splitk = path > 0 ? path : -path;
+  if (splitk > 1) {
+    ws = torch::empty({(long)B * splitk, (long)HO * WO, OC},
                       x.options().dtype(torch::kFloat));
     wsp = ws.data_ptr<float>();
   }
@@ -108,18 +109,20 @@ torch::Tensor layer_norm(torch::Tensor x, torch::Tensor gamma,
 
 torch::Tensor attention_bhlc(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                              int64_t heads, double scale) {
-  CHECK_IN(q);
-  CHECK_IN(k);
-  CHECK_IN(v);
+  // q/k/v: (B, L, C) with unit channel stride; row/batch strides are free so
+  // chunk() views of a fused QKV projection run zero-copy.
+  TORCH_CHECK(q.is_cuda() && q.stride(2) == 1, "q: unit channel stride");
+  TORCH_CHECK(k.stride(2) == 1 && v.stride(2) == 1, "k/v: unit channel stride");
+  TORCH_CHECK(k.strides() == v.strides(), "k/v must share layout");
   const int B = q.size(0), Lq = q.size(1), C = q.size(2);
   const int Lk = k.size(1), Ck = k.size(2);
   const int d = C / (int)heads;
   TORCH_CHECK(Ck == C, "q/k channel mismatch");
   TORCH_CHECK(d % 32 == 0 && d <= 160, "head_dim must be padded to one of 32/64/96/128/160");
-  auto out = torch::empty_like(q);
+  auto out = torch::empty({B, Lq, C}, q.options());
   airtc_attention(h_ptr(q), h_ptr(k), h_ptr(v), h_ptr_mut(out), B, (int)heads,
-                  Lq, Lk, d, (long)Lq * C, d, C, (long)Lk * C, d, C,
-                  (long)Lq * C, d, C, (float)scale, cur_stream());
+                  Lq, Lk, d, q.stride(0), d, q.stride(1), k.stride(0), d,
+                  k.stride(1), (long)Lq * C, d, C, (float)scale, cur_stream());
   return out;
 }
 

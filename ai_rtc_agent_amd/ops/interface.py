@@ -214,7 +214,7 @@ def attention(
         dpad = min(x for x in _ATTN_DIMS if x >= d)
 
         def rearrange(t, L):
-            th = t.view(b, L, num_heads, d)
+            th = t.reshape(b, L, num_heads, d)
             th = F.pad(th, (0, dpad - d))
             return th.permute(0, 2, 1, 3).reshape(b * num_heads, L, dpad).contiguous()
 
@@ -224,9 +224,9 @@ def attention(
 
     b, lq, c = q.shape
     d = c // num_heads
-    qh = q.view(b, lq, num_heads, d).permute(0, 2, 1, 3).float()
-    kh = k.view(b, -1, num_heads, d).permute(0, 2, 1, 3).float()
-    vh = v.view(b, -1, num_heads, d).permute(0, 2, 1, 3).float()
+    qh = q.reshape(b, lq, num_heads, d).permute(0, 2, 1, 3).float()
+    kh = k.reshape(b, -1, num_heads, d).permute(0, 2, 1, 3).float()
+    vh = v.reshape(b, -1, num_heads, d).permute(0, 2, 1, 3).float()
     o = F.scaled_dot_product_attention(qh, kh, vh)
     return o.permute(0, 2, 1, 3).reshape(b, lq, c).to(q.dtype)
 
