@@ -166,10 +166,10 @@ class CrossAttention(nn.Module):
         self.to_v = Linear(ctx_dim, dim, bias=False)
         self.to_out = Linear(dim, dim)
 
-    def _fused_w(self, name: str, parts) -> torch.Tensor:
+    def _fused_w(self, name: str, parts, ref: torch.Tensor) -> torch.Tensor:
         w = getattr(self, name, None)
-        if w is None:
-            w = torch.cat([p.weight for p in parts], dim=0)
+        if w is None or w.device != ref.device or w.dtype != ref.dtype:
+            w = torch.cat([p.weight.detach().to(ref.device, ref.dtype) for p in parts], dim=0)
             setattr(self, name, w)
         return w
 
@@ -177,11 +177,11 @@ class CrossAttention(nn.Module):
         # fused projections: one GEMM for QKV (self) / KV (cross); the
         # attention kernel consumes the chunk views zero-copy (strided q/k/v)
         if ctx is None:
-            qkv = ops.linear(x, self._fused_w("_wqkv", (self.to_q, self.to_k, self.to_v)))
+            qkv = ops.linear(x, self._fused_w("_wqkv", (self.to_q, self.to_k, self.to_v), x))
             q, k, v = qkv.chunk(3, dim=-1)
         else:
             q = self.to_q(x)
-            kv = ops.linear(ctx, self._fused_w("_wkv", (self.to_k, self.to_v)))
+            kv = ops.linear(ctx, self._fused_w("_wkv", (self.to_k, self.to_v), x))
             k, v = kv.chunk(2, dim=-1)
         o = ops.attention(q, k, v, self.heads)
         return self.to_out(o)
