@@ -1,0 +1,252 @@
+"""Peer connection + ICE-lite UDP media transport — from scratch.
+
+Stands in for the aiortc RTCPeerConnection the reference uses
+(agent.py:136,299). Scope honestly stated: SDP offer/answer signalling, ICE-
+lite (STUN binding answerer, peer-reflexive address learning), RTP media
+with the codec HAL, and a lightweight JSON config channel over the same
+socket (magic-prefixed datagrams) mirroring the reference's datachannel
+config updates (agent.py:154-168). DTLS-SRTP is NOT implemented — this
+image has no DTLS-capable library (no aiortc/cryptography; stdlib ssl has
+no DTLS), so browser interop requires adding a DTLS stack on a deployment
+box; LAN/loopback peers use this transport directly. docs/webrtc.md carries
+the full gap statement.
+
+UDP port pinning: the reference monkey-patches asyncio's datagram endpoint
+factory to force media onto operator ports (agent.py:32-69, for firewalls /
+OBS). We own the transport, so the pool is first-class: pass allowed ports
+to PeerConnection and the bind loop walks that pool.
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import logging
+import random
+import socket
+import time
+from typing import Callable, Dict, List, Optional, Tuple
+
+import torch
+
+from .codec import select_codec
+from .rtp import RtpDefragmenter, RtpPacket, RtpPacketizer
+from .sdp import SessionDescription, build_answer
+from .stun import StunMessage, BINDING_REQUEST, is_stun, make_binding_response
+from .tracks import MediaStreamTrack, QueueTrack, VideoFrame
+
+logger = logging.getLogger(__name__)
+
+CONFIG_MAGIC = b"AIRC"  # JSON config datagrams (datachannel-lite)
+
+_port_pool: Optional[List[int]] = None
+
+
+def set_udp_port_pool(ports: Optional[List[int]]) -> None:
+    """Operator port pinning (parity: reference --udp-ports, agent.py:460)."""
+    global _port_pool
+    _port_pool = ports
+
+
+class _Proto(asyncio.DatagramProtocol):
+    def __init__(self, pc: "PeerConnection"):
+        self.pc = pc
+
+    def datagram_received(self, data: bytes, addr) -> None:
+        self.pc._on_datagram(data, addr)
+
+
+class PeerConnection:
+    """One media session (the reference's RTCPeerConnection role)."""
+
+    def __init__(self, ice_servers: Optional[list] = None):
+        self.ice_servers = ice_servers or []
+        self.connection_state = "new"
+        self.remote_description: Optional[SessionDescription] = None
+        self.local_description: Optional[SessionDescription] = None
+        self._handlers: Dict[str, List[Callable]] = {}
+        self._transport = None
+        self._remote_addr: Optional[Tuple[str, int]] = None
+        self._send_track: Optional[MediaStreamTrack] = None
+        self._sender_task: Optional[asyncio.Task] = None
+        self._recv_track: Optional[QueueTrack] = None
+        self._track_fired = False
+        self._encoder = select_codec(role="encode")
+        self._decoder = select_codec(role="decode")
+        self._packetizer = RtpPacketizer(ssrc=random.randint(1, 2**31))
+        self._defrag = RtpDefragmenter()
+        self.port: Optional[int] = None
+        self._ice_pwd = ""
+
+    # -- event API (aiortc-style) --------------------------------------
+    def on(self, event: str, handler: Optional[Callable] = None):
+        def register(h):
+            self._handlers.setdefault(event, []).append(h)
+            return h
+
+        return register(handler) if handler else register
+
+    def _emit(self, event: str, *args) -> None:
+        for h in self._handlers.get(event, []):
+            res = h(*args)
+            if asyncio.iscoroutine(res):
+                asyncio.ensure_future(res)
+
+    def _set_state(self, state: str) -> None:
+        if state != self.connection_state:
+            self.connection_state = state
+            self._emit("connectionstatechange")
+
+    # -- signalling ------------------------------------------------------
+    async def set_remote_description(self, sdp: str) -> None:
+        self.remote_description = SessionDescription.parse(sdp)
+        # learn the peer's host candidate for outbound-first flows
+        for m in self.remote_description.media:
+            for cand in m.candidates:
+                parts = cand.split()
+                if len(parts) >= 6 and parts[2].lower() == "udp":
+                    self._remote_addr = (parts[4], int(parts[5]))
+                    break
+
+    async def create_answer(self, host: str = "127.0.0.1", direction: str = "sendrecv") -> str:
+        await self._bind(host)
+        assert self.remote_description is not None, "set_remote_description first"
+        ans = build_answer(
+            self.remote_description,
+            host,
+            self.port,
+            codec_name="H264",
+            ssrc=self._packetizer.ssrc,
+            direction=direction,
+        )
+        self.local_description = ans
+        self._ice_pwd = ans.media[0].ice_pwd if ans.media else ""
+        self._set_state("connecting")
+        return ans.serialize()
+
+    async def _bind(self, host: str) -> None:
+        if self._transport is not None:
+            return
+        loop = asyncio.get_event_loop()
+        last_err: Optional[Exception] = None
+        ports = _port_pool or [0]
+        for p in ports:
+            try:
+                self._transport, _ = await loop.create_datagram_endpoint(
+                    lambda: _Proto(self), local_addr=("0.0.0.0", p)
+                )
+                self.port = self._transport.get_extra_info("sockname")[1]
+                return
+            except OSError as e:  # port in use: walk the pool
+                last_err = e
+        raise OSError(f"no usable UDP port in pool {ports}: {last_err}")
+
+    # -- media ----------------------------------------------------------
+    def add_track(self, track: MediaStreamTrack) -> None:
+        """Outbound processed video (the reference's pc.addTrack,
+        agent.py:178)."""
+        self._send_track = track
+        if self._sender_task is None:
+            self._sender_task = asyncio.ensure_future(self._sender_loop())
+
+    async def _sender_loop(self) -> None:
+        counter = 0
+        try:
+            while self.connection_state not in ("closed", "failed"):
+                frame = await self._send_track.recv()
+                if self._remote_addr is None or self._transport is None:
+                    continue
+                data = self._encoder.encode(frame.tensor, keyframe=counter == 0)
+                ts = frame.pts if frame.pts else counter * 3000
+                for pkt in self._packetizer.packetize(data, ts):
+                    self._transport.sendto(pkt.serialize(), self._remote_addr)
+                counter += 1
+        except asyncio.CancelledError:
+            pass
+        except Exception:
+            logger.exception("sender loop failed")
+            self._set_state("failed")
+
+    # -- inbound --------------------------------------------------------
+    def _on_datagram(self, data: bytes, addr) -> None:
+        if is_stun(data):
+            try:
+                msg = StunMessage.parse(data)
+            except ValueError:
+                return
+            if msg.msg_type == BINDING_REQUEST and self._transport is not None:
+                resp = make_binding_response(msg, addr, self._ice_pwd.encode())
+                self._transport.sendto(resp, addr)
+                self._remote_addr = addr  # peer-reflexive
+                self._set_state("connected")
+            return
+        if data[:4] == CONFIG_MAGIC:
+            try:
+                payload = json.loads(data[4:].decode())
+            except (UnicodeDecodeError, json.JSONDecodeError):
+                return
+            self._emit("datachannel_message", payload)
+            return
+        try:
+            pkt = RtpPacket.parse(data)
+        except ValueError:
+            return
+        self._remote_addr = addr
+        if self.connection_state == "connecting":
+            self._set_state("connected")
+        frame_bytes = self._defrag.push(pkt)
+        if frame_bytes is None:
+            return
+        try:
+            tensor = self._decoder.decode(frame_bytes)
+        except Exception:
+            logger.exception("decode failed")
+            return
+        if tensor is None:
+            return
+        if self._recv_track is None:
+            self._recv_track = QueueTrack()
+        self._recv_track.push(VideoFrame(tensor=tensor, pts=pkt.timestamp))
+        if not self._track_fired:
+            self._track_fired = True
+            self._emit("track", self._recv_track)
+
+    # -- teardown --------------------------------------------------------
+    async def close(self) -> None:
+        if self.connection_state == "closed":
+            return
+        self._set_state("closed")
+        if self._sender_task is not None:
+            self._sender_task.cancel()
+            try:
+                await self._sender_task
+            except (asyncio.CancelledError, Exception):
+                pass
+        if self._transport is not None:
+            self._transport.close()
+            self._transport = None
+
+
+class MediaRelay:
+    """Fan-out of one source track to multiple subscribers (the reference
+    imports aiortc's MediaRelay, agent.py:427)."""
+
+    def __init__(self):
+        self._subscribers: Dict[int, List[QueueTrack]] = {}
+        self._pumps: Dict[int, asyncio.Task] = {}
+
+    def subscribe(self, track: MediaStreamTrack) -> QueueTrack:
+        key = id(track)
+        sub = QueueTrack()
+        self._subscribers.setdefault(key, []).append(sub)
+        if key not in self._pumps:
+            self._pumps[key] = asyncio.ensure_future(self._pump(key, track))
+        return sub
+
+    async def _pump(self, key: int, track: MediaStreamTrack) -> None:
+        try:
+            while True:
+                frame = await track.recv()
+                for sub in self._subscribers.get(key, []):
+                    sub.push(frame)
+        except (asyncio.CancelledError, Exception):
+            pass

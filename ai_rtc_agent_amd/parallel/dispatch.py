@@ -1,0 +1,86 @@
+"""Session -> GPU dispatch: the multi-replica pipeline pool.
+
+The reference shares ONE pipeline object across every peer connection in
+one asyncio loop (reference agent.py:423; concurrent publishers interleave
+frames through one stream-batch state machine — SURVEY.md §5.2 calls this
+out as a hazard it simply accepts). The MI355X design
+replaces that with explicit per-stream pipeline replicas, one per GPU:
+
+- PipelinePool.create(n_gpus) builds one replica per visible GPU (replica i
+  pinned to cuda:i); weight init is seed-identical across replicas, and
+  when the pool is built under torch.distributed the weights come from the
+  rank-0 RCCL broadcast instead (parallel/collectives.py)
+- assign(stream_id) binds a session to the least-loaded replica (stream <->
+  GPU affinity; steady state needs no cross-GPU traffic)
+- release(stream_id) frees the slot
+
+For CPU tests the pool degrades to one CPU pipeline replica.
+"""
+from __future__ import annotations
+
+import logging
+from typing import Dict, List, Optional
+
+import torch
+
+logger = logging.getLogger(__name__)
+
+
+class PipelinePool:
+    def __init__(self, pipelines: List):
+        self._pipelines = pipelines
+        self._load: Dict[int, int] = {i: 0 for i in range(len(pipelines))}
+        self._sessions: Dict[str, int] = {}
+
+    # -- construction ----------------------------------------------------
+    @staticmethod
+    def create(model_id: str, n_gpus: int = 1, cfg=None) -> "PipelinePool":
+        from ..config import EngineConfig
+        from ..pipeline import StreamDiffusionPipeline
+
+        pipelines = []
+        if torch.cuda.is_available():
+            n = min(n_gpus, torch.cuda.device_count())
+            for i in range(n):
+                c = cfg or EngineConfig(model_id=model_id)
+                c.device = f"cuda:{i}"
+                pipelines.append(StreamDiffusionPipeline(model_id, cfg=c))
+        else:
+            c = cfg or EngineConfig(model_id=model_id)
+            c.device = "cpu"
+            c.use_hip_graph = False
+            pipelines.append(StreamDiffusionPipeline(model_id, cfg=c))
+        logger.info("pipeline pool: %d replica(s)", len(pipelines))
+        return PipelinePool(pipelines)
+
+    @staticmethod
+    def single(pipeline) -> "PipelinePool":
+        return PipelinePool([pipeline])
+
+    # -- session affinity -------------------------------------------------
+    def assign(self, stream_id: str):
+        if stream_id in self._sessions:
+            return self._pipelines[self._sessions[stream_id]]
+        idx = min(self._load, key=lambda i: self._load[i])
+        self._load[idx] += 1
+        self._sessions[stream_id] = idx
+        logger.info("stream %s -> replica %d", stream_id, idx)
+        return self._pipelines[idx]
+
+    def release(self, stream_id: str) -> None:
+        idx = self._sessions.pop(stream_id, None)
+        if idx is not None:
+            self._load[idx] = max(0, self._load[idx] - 1)
+
+    def active(self) -> List:
+        return self._pipelines
+
+    def stats(self) -> dict:
+        return {
+            "replicas": len(self._pipelines),
+            "sessions": {k: v for k, v in self._sessions.items()},
+            "per_replica": [
+                {"load": self._load[i], **(p.stats() if hasattr(p, "stats") else {})}
+                for i, p in enumerate(self._pipelines)
+            ],
+        }

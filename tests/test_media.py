@@ -1,0 +1,118 @@
+"""Media plane units: SDP, RTP packetization, software codec, STUN."""
+import random
+
+import pytest
+import torch
+
+from ai_rtc_agent_amd.media.codec import SoftwareCodec
+from ai_rtc_agent_amd.media.rtp import RtpDefragmenter, RtpPacket, RtpPacketizer
+from ai_rtc_agent_amd.media.sdp import SessionDescription, build_answer, prefer_codec
+from ai_rtc_agent_amd.media import stun
+
+OFFER = "\r\n".join([
+    "v=0",
+    "o=- 4611731400430051336 2 IN IP4 127.0.0.1",
+    "s=-",
+    "t=0 0",
+    "m=video 9 UDP/TLS/RTP/SAVPF 96 97",
+    "c=IN IP4 0.0.0.0",
+    "a=ice-ufrag:EsAw",
+    "a=ice-pwd:P2uYro0UCOQ4zxjKXaWCBui1",
+    "a=mid:0",
+    "a=sendrecv",
+    "a=rtpmap:96 VP8/90000",
+    "a=rtpmap:97 H264/90000",
+    "a=fmtp:97 profile-level-id=42e01f",
+    "a=candidate:1 1 udp 2130706431 192.168.1.10 51000 typ host",
+]) + "\r\n"
+
+
+def test_sdp_parse():
+    sd = SessionDescription.parse(OFFER)
+    assert len(sd.media) == 1
+    m = sd.media[0]
+    assert m.kind == "video"
+    assert m.ice_ufrag == "EsAw"
+    assert {c.name for c in m.codecs} == {"VP8", "H264"}
+    assert m.codec_by_name("h264").parameters == "profile-level-id=42e01f"
+    assert m.candidates[0].endswith("typ host")
+
+
+def test_sdp_prefer_codec():
+    sd = SessionDescription.parse(OFFER)
+    prefer_codec(sd.media[0], "H264")
+    assert [c.name for c in sd.media[0].codecs] == ["H264"]
+
+
+def test_sdp_answer_roundtrip():
+    sd = SessionDescription.parse(OFFER)
+    ans = build_answer(sd, "127.0.0.1", 40000, "H264", ssrc=1234)
+    text = ans.serialize()
+    back = SessionDescription.parse(text)
+    m = back.media[0]
+    assert m.codecs[0].name == "H264"
+    assert m.ssrc == 1234
+    assert "127.0.0.1 40000" in m.candidates[0]
+    assert m.ice_pwd and m.ice_ufrag
+
+
+def test_rtp_roundtrip():
+    p = RtpPacket(payload_type=97, sequence_number=42, timestamp=9000, ssrc=7, marker=1, payload=b"abc")
+    q = RtpPacket.parse(p.serialize())
+    assert (q.payload_type, q.sequence_number, q.timestamp, q.ssrc, q.marker, q.payload) == (
+        97, 42, 9000, 7, 1, b"abc")
+
+
+def test_rtp_fragmentation_and_reorder():
+    pk = RtpPacketizer(ssrc=5)
+    frame = bytes(random.Random(0).randbytes(5000))
+    pkts = pk.packetize(frame, timestamp=3000)
+    assert len(pkts) == 5 and pkts[-1].marker == 1
+    df = RtpDefragmenter()
+    random.Random(1).shuffle(pkts)
+    out = None
+    for p in pkts:
+        got = df.push(RtpPacket.parse(p.serialize()))
+        if got is not None:
+            out = got
+    assert out == frame
+
+
+def test_software_codec_i_and_p_frames():
+    c = SoftwareCodec(keyframe_interval=4)
+    d = SoftwareCodec()
+    g = torch.Generator().manual_seed(0)
+    prev = torch.randint(0, 256, (32, 32, 3), generator=g, dtype=torch.uint8)
+    for i in range(8):
+        # small motion between frames
+        frame = (prev.int() + torch.randint(-2, 3, prev.shape, generator=g)).clamp(0, 255).to(torch.uint8)
+        data = c.encode(frame)
+        got = d.decode(data)
+        assert got is not None and torch.equal(got, frame), f"frame {i} lossless roundtrip"
+        prev = frame
+
+
+def test_software_codec_p_before_i_waits():
+    c = SoftwareCodec(keyframe_interval=100)
+    d = SoftwareCodec()
+    f0 = torch.zeros(8, 8, 3, dtype=torch.uint8)
+    f1 = torch.ones(8, 8, 3, dtype=torch.uint8)
+    i_frame = c.encode(f0)
+    p_frame = c.encode(f1)
+    assert d.decode(p_frame) is None  # P before any I: undecodable
+    assert torch.equal(d.decode(i_frame), f0)
+    # the skipped P desynced the stream; a forced keyframe resyncs
+    assert torch.equal(d.decode(c.encode(f1, keyframe=True)), f1)
+
+
+def test_stun_binding_flow():
+    req_raw = stun.make_binding_request("remote:local", b"pwd")
+    assert stun.is_stun(req_raw)
+    req = stun.StunMessage.parse(req_raw)
+    assert req.msg_type == stun.BINDING_REQUEST
+    assert req.attributes[stun.ATTR_USERNAME] == b"remote:local"
+    resp_raw = stun.make_binding_response(req, ("10.0.0.1", 5000), b"pwd")
+    resp = stun.StunMessage.parse(resp_raw)
+    assert resp.msg_type == stun.BINDING_RESPONSE
+    assert resp.transaction_id == req.transaction_id
+    assert stun.ATTR_XOR_MAPPED_ADDRESS in resp.attributes

@@ -1,0 +1,200 @@
+"""Track adapter behaviour + full loopback media test over localhost UDP.
+
+Loopback (SURVEY.md §4 item d): a synthetic publisher WHIPs into the agent,
+frames flow RTP -> codec -> pipeline -> RTP back to a WHEP subscriber, all
+over real sockets on 127.0.0.1 with a stub pipeline.
+"""
+import asyncio
+import os
+
+import pytest
+import torch
+
+from ai_rtc_agent_amd.media.codec import SoftwareCodec
+from ai_rtc_agent_amd.media.rtp import RtpDefragmenter, RtpPacket, RtpPacketizer
+from ai_rtc_agent_amd.media.sdp import SessionDescription
+from ai_rtc_agent_amd.media import stun
+from ai_rtc_agent_amd.media.tracks import QueueTrack, VideoFrame, VideoStreamTrack
+
+
+def run(coro, timeout=30):
+    loop = asyncio.new_event_loop()
+    try:
+        return loop.run_until_complete(asyncio.wait_for(coro, timeout))
+    finally:
+        loop.close()
+
+
+# ---------------------------------------------------------------------------
+# VideoStreamTrack contract (reference lib/tracks.py:20-38)
+# ---------------------------------------------------------------------------
+
+def test_warmup_frames_discarded(monkeypatch):
+    monkeypatch.setenv("WARMUP_FRAMES", "3")
+    monkeypatch.setenv("DROP_FRAMES", "0")
+
+    async def body():
+        src = QueueTrack(maxsize=64)
+        for i in range(10):
+            src.push(VideoFrame(tensor=torch.full((2, 2, 3), i, dtype=torch.uint8), pts=i))
+        calls = []
+
+        def pipe(t):
+            calls.append(int(t[0, 0, 0]))
+            return t + 100
+
+        track = VideoStreamTrack(src, pipe)
+        outs = [await track.recv() for _ in range(5)]
+        # first 3 pulls ran the pipeline but returned SOURCE frames
+        assert [int(o.tensor[0, 0, 0]) for o in outs[:3]] == [0, 1, 2]
+        assert track.warmed_up
+        # subsequent pulls return processed frames
+        assert [int(o.tensor[0, 0, 0]) for o in outs[3:]] == [103, 104]
+        assert calls == [0, 1, 2, 3, 4]
+
+    run(body())
+
+
+def test_drop_frames(monkeypatch):
+    monkeypatch.setenv("WARMUP_FRAMES", "0")
+    monkeypatch.setenv("DROP_FRAMES", "2")
+
+    async def body():
+        src = QueueTrack(maxsize=64)
+        for i in range(12):
+            src.push(VideoFrame(tensor=torch.full((2, 2, 3), i, dtype=torch.uint8), pts=i))
+        track = VideoStreamTrack(src, lambda t: t)
+        o1 = await track.recv()
+        o2 = await track.recv()
+        # 2 source frames dropped per output (reference lib/tracks.py:27-31)
+        assert int(o1.tensor[0, 0, 0]) == 2
+        assert int(o2.tensor[0, 0, 0]) == 5
+
+    run(body())
+
+
+def test_pts_preserved(monkeypatch):
+    monkeypatch.setenv("WARMUP_FRAMES", "0")
+    monkeypatch.setenv("DROP_FRAMES", "0")
+
+    async def body():
+        src = QueueTrack()
+        src.push(VideoFrame(tensor=torch.zeros(2, 2, 3, dtype=torch.uint8), pts=777))
+        track = VideoStreamTrack(src, lambda t: t)
+        out = await track.recv()
+        assert out.pts == 777  # reference lib/pipeline.py:90-93
+
+    run(body())
+
+
+# ---------------------------------------------------------------------------
+# loopback over real UDP sockets
+# ---------------------------------------------------------------------------
+
+class _ClientProto(asyncio.DatagramProtocol):
+    def __init__(self):
+        self.transport = None
+        self.frames = asyncio.Queue()
+        self.defrag = RtpDefragmenter()
+        self.codec = SoftwareCodec()
+
+    def connection_made(self, transport):
+        self.transport = transport
+
+    def datagram_received(self, data, addr):
+        if stun.is_stun(data):
+            return
+        try:
+            pkt = RtpPacket.parse(data)
+        except ValueError:
+            return
+        buf = self.defrag.push(pkt)
+        if buf is not None:
+            t = self.codec.decode(buf)
+            if t is not None:
+                self.frames.put_nowait(t)
+
+
+def _offer_sdp(port: int) -> str:
+    return "\r\n".join([
+        "v=0", "o=- 1 2 IN IP4 127.0.0.1", "s=-", "t=0 0",
+        f"m=video {port} UDP/TLS/RTP/SAVPF 97",
+        "a=ice-ufrag:testu", "a=ice-pwd:testpw0123456789", "a=mid:0", "a=sendrecv",
+        "a=rtpmap:97 H264/90000",
+        f"a=candidate:1 1 udp 2130706431 127.0.0.1 {port} typ host",
+    ]) + "\r\n"
+
+
+@pytest.mark.timeout(60)
+def test_loopback_whip_whep(monkeypatch):
+    monkeypatch.setenv("WARMUP_FRAMES", "0")
+    monkeypatch.setenv("DROP_FRAMES", "0")
+
+    async def body():
+        from aiohttp.test_utils import TestClient, TestServer
+
+        from ai_rtc_agent_amd.agent import create_app
+        from ai_rtc_agent_amd.parallel.dispatch import PipelinePool
+
+        stylize = lambda t: (t.int() + 10).clamp(0, 255).to(torch.uint8)
+        app = create_app(pool=PipelinePool.single(stylize), use_turn=False)
+        http = TestClient(TestServer(app))
+        await http.start_server()
+        loop = asyncio.get_event_loop()
+
+        # publisher socket
+        pub_t, pub_p = await loop.create_datagram_endpoint(
+            _ClientProto, local_addr=("127.0.0.1", 0))
+        pub_port = pub_t.get_extra_info("sockname")[1]
+
+        r = await http.post("/whip", data=_offer_sdp(pub_port),
+                            headers={"Content-Type": "application/sdp"})
+        assert r.status == 201
+        ans = SessionDescription.parse(await r.text())
+        srv_port = ans.media[0].port
+
+        # ICE-lite handshake: binding request -> agent learns our address
+        req = stun.make_binding_request("u:p", b"k")
+        pub_t.sendto(req, ("127.0.0.1", srv_port))
+        await asyncio.sleep(0.1)
+
+        # stream frames to the agent
+        codec = SoftwareCodec()
+        pkz = RtpPacketizer(ssrc=99)
+        g = torch.Generator().manual_seed(0)
+        frames = [torch.randint(0, 200, (16, 16, 3), generator=g, dtype=torch.uint8)
+                  for _ in range(6)]
+        for i, f in enumerate(frames):
+            for pkt in pkz.packetize(codec.encode(f), timestamp=i * 3000):
+                pub_t.sendto(pkt.serialize(), ("127.0.0.1", srv_port))
+            await asyncio.sleep(0.02)
+        await asyncio.sleep(0.2)
+
+        # subscriber: WHEP with our own socket
+        sub_t, sub_p = await loop.create_datagram_endpoint(
+            _ClientProto, local_addr=("127.0.0.1", 0))
+        sub_port = sub_t.get_extra_info("sockname")[1]
+        r2 = await http.post("/whep", data=_offer_sdp(sub_port),
+                             headers={"Content-Type": "application/sdp"})
+        assert r2.status == 201, await r2.text()
+
+        # keep publishing so the subscriber's sender loop has frames to pull
+        got = None
+        for i in range(6, 40):
+            for pkt in pkz.packetize(codec.encode(frames[i % len(frames)]), timestamp=i * 3000):
+                pub_t.sendto(pkt.serialize(), ("127.0.0.1", srv_port))
+            try:
+                got = await asyncio.wait_for(sub_p.frames.get(), timeout=0.25)
+                break
+            except asyncio.TimeoutError:
+                continue
+        assert got is not None, "no stylised frame reached the WHEP subscriber"
+        # stylize = +10: every received frame must show the pipeline's mark
+        src_mean = torch.stack(frames).float().mean()
+        assert abs(got.float().mean() - (src_mean + 10)) < 8.0
+
+        pub_t.close()
+        sub_t.close()
+        await http.close()
+
+    run(body(), timeout=50)
