@@ -112,19 +112,41 @@ __global__ __launch_bounds__(256) void attention_kernel(
   f16* myP = &ldsP[wid * 16 * PPITCH];
   const int fcol = (lane >> 4) * 8;
 
-  for (int t0 = 0; t0 < Lk; t0 += KVT) {
-    // ---- stage K tile [64][D] and V tile transposed [D][64] ----
-    if (t0) __syncthreads();  // previous tile's reads complete
-    for (int i = tid; i < KVT * D8; i += 256) {
-      const int row = i / D8, c8 = (i - row * D8) * 8;
-      int krow = t0 + row;
+  // T14 async-stage split: each thread owns LOADS_PT row-chunks of the K and
+  // V tiles; tile t+1's global loads are issued right after tile t's LDS
+  // image is published, so HBM latency hides under t's MFMA/softmax work.
+  constexpr int LOADS_PT = (KVT * D8 + 255) / 256;
+  f16x8 regK[LOADS_PT], regV[LOADS_PT];
+  int st_row[LOADS_PT], st_c8[LOADS_PT];
+#pragma unroll
+  for (int i = 0; i < LOADS_PT; ++i) {
+    const int flat = tid + i * 256;
+    st_row[i] = flat / D8;
+    st_c8[i] = (flat - st_row[i] * D8) * 8;
+  }
+
+  auto stage_load = [&](int t0) {
+#pragma unroll
+    for (int i = 0; i < LOADS_PT; ++i) {
+      if (st_row[i] >= KVT) continue;
+      int krow = t0 + st_row[i];
       if (krow >= Lk) krow = Lk - 1;  // masked later
-      f16x8 k8 = *reinterpret_cast<const f16x8*>(&kb[(long)krow * k_row + c8]);
-      *reinterpret_cast<f16x8*>(&ldsK[row * KPITCH + c8]) = k8;
-      f16x8 v8 = *reinterpret_cast<const f16x8*>(&vb[(long)krow * k_row + c8]);
-      *reinterpret_cast<f16x8*>(&ldsV[row * KPITCH + c8]) = v8;
+      regK[i] = *reinterpret_cast<const f16x8*>(&kb[(long)krow * k_row + st_c8[i]]);
+      regV[i] = *reinterpret_cast<const f16x8*>(&vb[(long)krow * k_row + st_c8[i]]);
+    }
+  };
+
+  stage_load(0);
+  for (int t0 = 0; t0 < Lk; t0 += KVT) {
+    if (t0) __syncthreads();  // previous tile's reads complete
+#pragma unroll
+    for (int i = 0; i < LOADS_PT; ++i) {
+      if (st_row[i] >= KVT) continue;
+      *reinterpret_cast<f16x8*>(&ldsK[st_row[i] * KPITCH + st_c8[i]]) = regK[i];
+      *reinterpret_cast<f16x8*>(&ldsV[st_row[i] * KPITCH + st_c8[i]]) = regV[i];
     }
     __syncthreads();
+    if (t0 + KVT < Lk) stage_load(t0 + KVT);
 
     // ---- S = scale * Q K^T  (4 col fragments of 16) ----
     f32x4 sfrag[4];
