@@ -42,6 +42,7 @@ def _unet_config_for(family: str) -> UNetConfig:
         "sd21": UNetConfig.sd21,
         "sdxl": UNetConfig.sdxl,
         "tiny": UNetConfig.tiny,
+        "tiny_xl": UNetConfig.tiny_xl,
     }[family]()
 
 
@@ -71,9 +72,12 @@ class StreamDiffusionEngine:
         self.unet = unet if unet is not None else UNet2DCondition(ucfg)
         self.vae = vae if vae is not None else TinyVAE()
         self.text_encoder = text_encoder if text_encoder is not None else TextEncoder(
-            hidden=ucfg.cross_attention_dim, layers=2 if cfg.model_family == "tiny" else 12
+            hidden=ucfg.cross_attention_dim,
+            layers=2 if cfg.model_family == "tiny" else 12,
+            pooled_dim=1280 if ucfg.addition_embed_dim else None,
         )
         self.ctx_dim = ucfg.cross_attention_dim
+        self.addition_embed_dim = ucfg.addition_embed_dim
 
         # LoRA fusion happens BEFORE device placement / graph capture
         # (reference fuses before TRT compile, lib/wrapper.py:645-697).
@@ -136,6 +140,13 @@ class StreamDiffusionEngine:
             neg = self.text_encoder.encode(cfg.negative_prompt, dev, dt)
             self._embeds_full = torch.cat([neg.expand(B, -1, -1), self._embeds_batch], dim=0).contiguous()
 
+        # sdxl addition conditioning: pooled text (1280) + 6 sinusoidal
+        # time-id embeddings of 256 (orig h/w, crop t/l, target h/w) = 2816
+        self._added_cond = None
+        if self.addition_embed_dim:
+            self._added_cond = torch.zeros((B, self.addition_embed_dim), device=dev, dtype=dt)
+            self._refresh_added_cond()
+
         g = torch.Generator(device="cpu").manual_seed(cfg.seed)
         self._init_noise = torch.randn((B, lh, lw, 4), generator=g).to(dev, dt)
         self._x_t_buffer = torch.zeros((max(0, B - fbs), lh, lw, 4), device=dev, dtype=dt)
@@ -148,6 +159,21 @@ class StreamDiffusionEngine:
         self._prev_out: Optional[torch.Tensor] = None
         self._graph = None
         self._prepared = True
+
+    @torch.no_grad()
+    def _refresh_added_cond(self) -> None:
+        """(Re)compute the sdxl addition-embedding into its static buffer."""
+        from ..models.unet import timestep_embedding
+
+        cfg = self.cfg
+        pooled = self.text_encoder.pooled(self.prompt, self.device, self.dtype)
+        ids = torch.tensor(
+            [cfg.height, cfg.width, 0, 0, cfg.height, cfg.width],
+            dtype=torch.float32, device=self.device,
+        )
+        time_emb = timestep_embedding(ids, 256).flatten()[None].to(self.dtype)
+        vec = torch.cat([pooled, time_emb], dim=-1)
+        self._added_cond.copy_(vec.expand_as(self._added_cond))
 
     # ------------------------------------------------------------------
     # runtime config updates (POST /config + datachannel; SURVEY.md §3.5)
@@ -163,6 +189,8 @@ class StreamDiffusionEngine:
         if self.cfg.cfg_type == "full":
             B = self._embeds_batch.shape[0]
             self._embeds_full[B:].copy_(self._embeds_batch)
+        if self._added_cond is not None:
+            self._refresh_added_cond()
 
     @torch.no_grad()
     def update_t_index_list(self, t_index_list: Sequence[int]) -> None:
@@ -236,10 +264,15 @@ class StreamDiffusionEngine:
 
         x_t = torch.cat([x_t0, self._x_t_buffer], dim=0) if self._x_t_buffer.shape[0] else x_t0
 
+        added = self._added_cond
+        if added is not None and (cfg.cfg_type in ("full", "initialize") and self.rcfg.active):
+            extra = added.shape[0] if cfg.cfg_type == "full" else fbs
+            added = torch.cat([added[:extra], added], dim=0)
         eps = self.unet(
             self._unet_batch_input(x_t),
             self._unet_batch_timesteps(),
             self._unet_batch_embeds(),
+            added_cond=added,
         )
         eps = self.rcfg.apply(eps, fbs)
         denoised = self.scheduler.step_batch(eps, x_t, co)

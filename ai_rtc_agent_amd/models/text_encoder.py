@@ -51,10 +51,15 @@ class TextEncoder(nn.Module):
         vocab_size: int = 49408,
         max_length: int = 77,
         seed: int = 0,
+        pooled_dim: int | None = None,
     ):
         super().__init__()
         if heads is None:
             heads = hidden // 64 if hidden % 64 == 0 else 8
+        # sdxl: the addition-embedding consumes a 1280-dim pooled text vector
+        self.pooled_proj = (
+            nn.Linear(hidden, pooled_dim, bias=False) if pooled_dim else None
+        )
         torch.manual_seed(seed)
         self.tokenizer = HashTokenizer(vocab_size, max_length)
         self.hidden = hidden
@@ -85,6 +90,10 @@ class TextEncoder(nn.Module):
 
     @torch.no_grad()
     def pooled(self, prompt: str, device=None, dtype=torch.float32) -> torch.Tensor:
-        """EOS-token pooled embedding (sdxl addition-embed path)."""
-        emb = self.encode(prompt, device=device, dtype=dtype)
-        return emb[:, -1]
+        """EOS-token pooled embedding (sdxl addition-embed path);
+        projected to pooled_dim (1280 for sdxl) when configured."""
+        emb = self.encode(prompt, device=device, dtype=torch.float32)
+        p = emb[:, -1]
+        if self.pooled_proj is not None:
+            p = self.pooled_proj(p.to(self.pooled_proj.weight.dtype))
+        return p.to(dtype)

@@ -67,6 +67,13 @@ class UNetConfig:
         )
 
     @staticmethod
+    def tiny_xl(ctx: int = 64) -> "UNetConfig":
+        """Small sdxl-shaped config (addition-embedding path) for CPU tests."""
+        cfg = UNetConfig.tiny(ctx)
+        cfg.addition_embed_dim = 2816  # 1280 pooled + 6 x 256 time ids
+        return cfg
+
+    @staticmethod
     def tiny(ctx: int = 64) -> "UNetConfig":
         """Small config for CPU tests."""
         return UNetConfig(

@@ -26,7 +26,7 @@ import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
-from ai_rtc_agent_amd.config import sd_turbo_config
+from ai_rtc_agent_amd.config import EngineConfig, sd_turbo_config
 from ai_rtc_agent_amd.engine import StreamDiffusionEngine
 from ai_rtc_agent_amd.parallel import broadcast_engine_weights, init_distributed
 
@@ -41,7 +41,14 @@ def main() -> None:
     p.add_argument("--width", type=int, default=512)
     p.add_argument("--no-graph", action="store_true")
     p.add_argument("--latency-frames", type=int, default=32)
+    p.add_argument(
+        "--model", default="sd-turbo", choices=["sd-turbo", "sd15", "sdxl"],
+        help="sd-turbo 1-step (headline) | sd15 4-step LCM+RCFG+filter "
+             "(BASELINE config[2]) | sdxl-turbo 1-step (config[3], 1024px)",
+    )
     args = p.parse_args()
+    if args.model == "sdxl" and "--width" not in " ".join(sys.argv):
+        args.width = 1024
 
     rank, world, local = init_distributed()
     use_cuda = torch.cuda.is_available()
@@ -50,12 +57,21 @@ def main() -> None:
         args.steps, args.warmup, args.latency_frames = 4, 1, 4
     device = f"cuda:{local}" if use_cuda else "cpu"
 
-    cfg = sd_turbo_config(
-        device=device,
-        width=args.width,
-        height=args.width,
-        use_hip_graph=not args.no_graph and use_cuda,
-    )
+    graph = not args.no_graph and use_cuda
+    if args.model == "sd-turbo":
+        cfg = sd_turbo_config(device=device, width=args.width, height=args.width, use_hip_graph=graph)
+        model_desc = "sd-turbo (SD2.1-base UNet geometry, TAESD, 1-step)"
+    elif args.model == "sd15":
+        # BASELINE config[2]: SD1.5 + LCM-LoRA 4-step RCFG + similarity filter
+        cfg = EngineConfig(device=device, width=args.width, height=args.width,
+                           use_hip_graph=graph)
+        cfg.similarity_filter.enabled = True
+        model_desc = "sd15 (dreamshaper-8 arch, LCM-LoRA fused, 4-step RCFG self, sim filter)"
+    else:  # sdxl
+        cfg = sd_turbo_config(device=device, width=args.width, height=args.width,
+                              use_hip_graph=graph, model_family="sdxl",
+                              model_id="stabilityai/sdxl-turbo")
+        model_desc = "sdxl-turbo (1-step, addition-embedding path)"
     # test-only escape hatch: CPU contract tests swap in the tiny family so
     # the 2-rank gloo run finishes in seconds (never set on GPU benches)
     if os.environ.get("AIRTC_BENCH_FAMILY"):
@@ -125,7 +141,7 @@ def main() -> None:
 
     if rank == 0:
         print(json.dumps({
-            "metric": "img2img FPS (SD-Turbo 512x512 1-step)",
+            "metric": f"img2img FPS (SD-Turbo-class {args.model} {args.width}x{args.width} {len(cfg.t_index_list)}-step)",
             "value": round(fps_total, 2),
             "unit": "frames/s",
             "n_gpus": world,
@@ -139,7 +155,7 @@ def main() -> None:
             "dtype": "fp16" if use_cuda else "fp32",
             "data": "synthetic (random frames, random-init SD-Turbo-arch weights)",
             "config": {
-                "model": "sd-turbo (SD2.1-base UNet geometry, TAESD, 1-step)",
+                "model": model_desc,
                 "global_batch": world * cfg.frame_buffer_size,
                 "resolution": f"{args.width}x{args.width}",
                 "t_index_list": cfg.t_index_list,
