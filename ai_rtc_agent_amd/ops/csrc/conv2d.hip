@@ -27,8 +27,6 @@
 #include "common.h"
 
 #define BN 64
-#define BK 32
-#define KPITCH (BK + 8)  // f16 elements per LDS row (+16B pad)
 
 struct KPos {
   int r, s, ic0;
@@ -66,17 +64,23 @@ __device__ __forceinline__ f16 epilogue(float acc, const float* bias,
 
 // ---------------------------------------------------------------------------
 // templated MFMA conv: MFRAG = M-fragments per wave (4 -> BM=128, 2 -> BM=64)
-// SPLITK > 1 only used with the small geometry; partials go to ws (f32).
+// BK = K-tile depth: 64 when IC%64==0 (all SD/TAESD layers — halves the
+// barrier count per K element, 2 MFMA K-chunks per stage), else 32.
+// SPLITK > 1: partials go to ws (f32), finalize pass reduces.
 // ---------------------------------------------------------------------------
-template <int MFRAG>
+template <int MFRAG, int BK>
 __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
     const f16* __restrict__ x, const f16* __restrict__ w,
     const float* __restrict__ bias, const f16* __restrict__ cbias,
     const f16* __restrict__ residual, f16* __restrict__ out,
     float* __restrict__ ws, int H, int W, int IC, int HO, int WO, int OC,
     int R, int S, int stride, int pad, int act, int K, int splitk) {
-  constexpr int BM = MFRAG * 32;          // 128 or 64
-  constexpr int ALOADS = MFRAG / 2;       // staged 16B A-loads per thread
+  constexpr int BM = MFRAG * 32;              // 128 or 64
+  constexpr int KPITCH = BK + 8;              // +16B row pad (guide G4)
+  constexpr int ALOADS = MFRAG * BK / 64;     // staged 16B A-loads per thread
+  constexpr int BLOADS = BK / 32;             // staged 16B B-loads per thread
+  constexpr int KSH = BK == 64 ? 3 : 2;       // flat -> (row, k8) shifts
+  constexpr int KMSK = BK / 8 - 1;
   __shared__ f16 ldsA[BM * KPITCH];
   __shared__ f16 ldsB[BN * KPITCH];
 
@@ -97,17 +101,22 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
   int a_row[ALOADS], a_ho[ALOADS], a_wo[ALOADS], a_k8[ALOADS];
 #pragma unroll
   for (int i = 0; i < ALOADS; ++i) {
-    int flat = tid + i * 256;           // [0, BM*4)
-    a_row[i] = flat >> 2;
-    a_k8[i] = (flat & 3) * 8;
+    int flat = tid + i * 256;           // [0, BM*BK/8)
+    a_row[i] = flat >> KSH;
+    a_k8[i] = (flat & KMSK) * 8;
     int m = min(m0 + a_row[i], M - 1);
     a_ho[i] = (m / WO) * stride;
     a_wo[i] = (m % WO) * stride;
   }
-  const int b_row = tid >> 2;
-  const int b_k8 = (tid & 3) * 8;
-  const int b_oc = min(n0 + b_row, OC - 1);
-  const f16* wrow = w + (long)b_oc * K + b_k8;
+  int b_row[BLOADS], b_k8[BLOADS];
+  const f16* wrow[BLOADS];
+#pragma unroll
+  for (int i = 0; i < BLOADS; ++i) {
+    int flat = tid + i * 256;           // [0, BN*BK/8)
+    b_row[i] = flat >> KSH;
+    b_k8[i] = (flat & KMSK) * 8;
+    wrow[i] = w + (long)min(n0 + b_row[i], OC - 1) * K + b_k8[i];
+  }
 
   f32x4 acc[MFRAG][2];
 #pragma unroll
@@ -124,14 +133,16 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
   // the finalize pass reads every slab.
 
   // prologue: stage first K-tile of this split into registers
-  f16x8 regA[ALOADS], regB;
+  f16x8 regA[ALOADS], regB[BLOADS];
   if (k_lo < k_hi) {
     KPos p = kpos_at(k_lo * BK, IC, S);
 #pragma unroll
     for (int i = 0; i < ALOADS; ++i)
       regA[i] = load_a(xb, a_ho[i], a_wo[i], p.r, p.s, pad, H, W, IC,
                        p.ic0 + a_k8[i]);
-    regB = *reinterpret_cast<const f16x8*>(wrow + k_lo * BK);
+#pragma unroll
+    for (int i = 0; i < BLOADS; ++i)
+      regB[i] = *reinterpret_cast<const f16x8*>(wrow[i] + k_lo * BK);
   }
 
   for (int kt = k_lo; kt < k_hi; ++kt) {
@@ -139,7 +150,9 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
 #pragma unroll
     for (int i = 0; i < ALOADS; ++i)
       *reinterpret_cast<f16x8*>(&ldsA[a_row[i] * KPITCH + a_k8[i]]) = regA[i];
-    *reinterpret_cast<f16x8*>(&ldsB[b_row * KPITCH + b_k8]) = regB;
+#pragma unroll
+    for (int i = 0; i < BLOADS; ++i)
+      *reinterpret_cast<f16x8*>(&ldsB[b_row[i] * KPITCH + b_k8[i]]) = regB[i];
     __syncthreads();
 
     if (kt + 1 < k_hi) {  // T14: next tile's loads before MFMAs
@@ -148,23 +161,28 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
       for (int i = 0; i < ALOADS; ++i)
         regA[i] = load_a(xb, a_ho[i], a_wo[i], p.r, p.s, pad, H, W, IC,
                          p.ic0 + a_k8[i]);
-      regB = *reinterpret_cast<const f16x8*>(wrow + (kt + 1) * BK);
+#pragma unroll
+      for (int i = 0; i < BLOADS; ++i)
+        regB[i] = *reinterpret_cast<const f16x8*>(wrow[i] + (kt + 1) * BK);
     }
 
     const int arow_base = wm * (MFRAG * 16) + (lane & 15);
-    const int fcol = (lane >> 4) * 8;
-    f16x8 bfrag[2];
 #pragma unroll
-    for (int ni = 0; ni < 2; ++ni)
-      bfrag[ni] = *reinterpret_cast<const f16x8*>(
-          &ldsB[(wn * 32 + ni * 16 + (lane & 15)) * KPITCH + fcol]);
-#pragma unroll
-    for (int mi = 0; mi < MFRAG; ++mi) {
-      f16x8 afrag = *reinterpret_cast<const f16x8*>(
-          &ldsA[(arow_base + mi * 16) * KPITCH + fcol]);
+    for (int kk = 0; kk < BK / 32; ++kk) {
+      const int fcol = kk * 32 + (lane >> 4) * 8;
+      f16x8 bfrag[2];
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni)
-        acc[mi][ni] = mfma16x16x32(afrag, bfrag[ni], acc[mi][ni]);
+        bfrag[ni] = *reinterpret_cast<const f16x8*>(
+            &ldsB[(wn * 32 + ni * 16 + (lane & 15)) * KPITCH + fcol]);
+#pragma unroll
+      for (int mi = 0; mi < MFRAG; ++mi) {
+        f16x8 afrag = *reinterpret_cast<const f16x8*>(
+            &ldsA[(arow_base + mi * 16) * KPITCH + fcol]);
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = mfma16x16x32(afrag, bfrag[ni], acc[mi][ni]);
+      }
     }
   }
 
@@ -363,14 +381,17 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
   const float* b1 = splitk == 1 ? bias : nullptr;
   const f16* cb1 = splitk == 1 ? cb : nullptr;
   const f16* res1 = splitk == 1 ? res : nullptr;
-  if (path > 0)
-    hipLaunchKernelGGL(conv2d_mfma_kernel<4>, grid, dim3(256), 0, s, xp, wp,
-                       b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC, R, S,
-                       stride, pad, act, K, splitk);
-  else
-    hipLaunchKernelGGL(conv2d_mfma_kernel<2>, grid, dim3(256), 0, s, xp, wp,
-                       b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC, R, S,
-                       stride, pad, act, K, splitk);
+  const bool bk64 = (IC % 64 == 0);
+#define CONV_LAUNCH(MF, BKV)                                                 \
+  hipLaunchKernelGGL((conv2d_mfma_kernel<MF, BKV>), grid, dim3(256), 0, s,   \
+                     xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC, R, \
+                     S, stride, pad, act, K, splitk)
+  if (path > 0) {
+    if (bk64) CONV_LAUNCH(4, 64); else CONV_LAUNCH(4, 32);
+  } else {
+    if (bk64) CONV_LAUNCH(2, 64); else CONV_LAUNCH(2, 32);
+  }
+#undef CONV_LAUNCH
   if (splitk > 1) {
     long total = (long)B * M * OC;
     int blocks = (int)min((long)2048, (total + 255) / 256);
