@@ -79,6 +79,10 @@ class StreamDiffusionEngine:
         self.ctx_dim = ucfg.cross_attention_dim
         self.addition_embed_dim = ucfg.addition_embed_dim
 
+        # real weights when available: model_id as a local diffusers-style
+        # dir, or an HF-cache snapshot (random init otherwise — offline)
+        self._load_weights_if_present(cfg.model_id)
+
         # LoRA fusion happens BEFORE device placement / graph capture
         # (reference fuses before TRT compile, lib/wrapper.py:645-697).
         if cfg.use_lcm_lora:
@@ -100,6 +104,24 @@ class StreamDiffusionEngine:
         self._graph: Optional[torch.cuda.CUDAGraph] = None
         self._prepared = False
         self.prompt: str = cfg.prompt
+
+    def _load_weights_if_present(self, model_id: str) -> None:
+        import glob
+        import os
+
+        from ..models.load import load_model_dir
+
+        candidates = [model_id] if os.path.isdir(model_id) else []
+        hub = os.environ.get("HF_HUB_CACHE", os.path.expanduser("~/.cache/huggingface/hub"))
+        candidates += glob.glob(
+            os.path.join(hub, "models--" + model_id.replace("/", "--"), "snapshots", "*")
+        )
+        for c in candidates:
+            try:
+                if load_model_dir(self, c):
+                    return
+            except Exception:  # corrupt cache entry: keep random init
+                continue
 
     # ------------------------------------------------------------------
     # prepare
