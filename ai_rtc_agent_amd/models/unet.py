@@ -107,8 +107,8 @@ class Linear(nn.Module):
         self.weight = nn.Parameter(torch.randn(dout, din) * (1.0 / math.sqrt(din)))
         self.bias = nn.Parameter(torch.zeros(dout)) if bias else None
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return ops.linear(x, self.weight, self.bias)
+    def forward(self, x: torch.Tensor, residual: torch.Tensor | None = None) -> torch.Tensor:
+        return ops.linear(x, self.weight, self.bias, residual=residual)
 
 
 class Conv2d(nn.Module):
@@ -180,7 +180,12 @@ class CrossAttention(nn.Module):
             setattr(self, name, w)
         return w
 
-    def forward(self, x: torch.Tensor, ctx: torch.Tensor | None = None) -> torch.Tensor:
+    def forward(
+        self,
+        x: torch.Tensor,
+        ctx: torch.Tensor | None = None,
+        residual: torch.Tensor | None = None,
+    ) -> torch.Tensor:
         # fused projections: one GEMM for QKV (self) / KV (cross); the
         # attention kernel consumes the chunk views zero-copy (strided q/k/v)
         if ctx is None:
@@ -191,7 +196,8 @@ class CrossAttention(nn.Module):
             kv = ops.linear(ctx, self._fused_w("_wkv", (self.to_k, self.to_v), x))
             k, v = kv.chunk(2, dim=-1)
         o = ops.attention(q, k, v, self.heads)
-        return self.to_out(o)
+        # residual fused into the out-projection epilogue
+        return self.to_out(o, residual=residual)
 
 
 class FeedForwardGEGLU(nn.Module):
@@ -201,8 +207,8 @@ class FeedForwardGEGLU(nn.Module):
         self.proj = Linear(dim, inner * 2)
         self.out = Linear(inner, dim)
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.out(ops.geglu(self.proj(x)))
+    def forward(self, x: torch.Tensor, residual: torch.Tensor | None = None) -> torch.Tensor:
+        return self.out(ops.geglu(self.proj(x)), residual=residual)
 
 
 class BasicTransformerBlock(nn.Module):
@@ -216,9 +222,10 @@ class BasicTransformerBlock(nn.Module):
         self.ff = FeedForwardGEGLU(dim)
 
     def forward(self, x: torch.Tensor, ctx: torch.Tensor) -> torch.Tensor:
-        x = x + self.attn1(self.norm1(x))
-        x = x + self.attn2(self.norm2(x), ctx)
-        x = x + self.ff(self.norm3(x))
+        # every residual add is fused into the closing projection's epilogue
+        x = self.attn1(self.norm1(x), residual=x)
+        x = self.attn2(self.norm2(x), ctx, residual=x)
+        x = self.ff(self.norm3(x), residual=x)
         return x
 
 
@@ -248,10 +255,9 @@ class SpatialTransformer(nn.Module):
         for blk in self.blocks:
             x = blk(x, ctx)
         if self.linear_proj:
-            x = self.proj_out(x).view(b, h, w, c)
-        else:
-            x = self.proj_out(x.view(b, h, w, c))
-        return x + res
+            # spatial residual fused into proj_out's epilogue
+            return self.proj_out(x, residual=res.view(b, h * w, c)).view(b, h, w, c)
+        return self.proj_out(x.view(b, h, w, c), residual=res)
 
 
 class ResnetBlock(nn.Module):

@@ -346,6 +346,7 @@ extern "C" int airtc_conv2d_splitk_for(int B, int HO, int WO, int OC, int IC) {
   // ran 87us = 86 TF; split-K over the K loop fills the chip).
   if (IC % 32 != 0) return 0;
   const int M = HO * WO;
+  if ((long)M * B >= 65536 && OC <= BN) return 100;  // BM256 (TAESD hi-res)
   if (M >= 2048) {
     const long blocks = (long)ceil_div(M, 128) * ceil_div(OC, BN) * B;
     long k = (480 + blocks - 1) / blocks;
@@ -375,8 +376,9 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
   f16* op = reinterpret_cast<f16*>(out);
   const int M = HO * WO;
 
-  const int splitk = path > 0 ? path : -path;
-  const int bm = path > 0 ? 128 : 64;
+  const bool bm256 = (path == 100);
+  const int splitk = bm256 ? 1 : (path > 0 ? path : -path);
+  const int bm = bm256 ? 256 : (path > 0 ? 128 : 64);
   dim3 grid(ceil_div(M, bm), ceil_div(OC, BN), B * splitk);
   const float* b1 = splitk == 1 ? bias : nullptr;
   const f16* cb1 = splitk == 1 ? cb : nullptr;
@@ -386,7 +388,9 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
   hipLaunchKernelGGL((conv2d_mfma_kernel<MF, BKV>), grid, dim3(256), 0, s,   \
                      xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC, R, \
                      S, stride, pad, act, K, splitk)
-  if (path > 0) {
+  if (bm256) {
+    if (bk64) CONV_LAUNCH(8, 64); else CONV_LAUNCH(8, 32);
+  } else if (path > 0) {
     if (bk64) CONV_LAUNCH(4, 64); else CONV_LAUNCH(4, 32);
   } else {
     if (bk64) CONV_LAUNCH(2, 64); else CONV_LAUNCH(2, 32);

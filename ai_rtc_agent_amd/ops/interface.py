@@ -231,10 +231,34 @@ def attention(
     return o.permute(0, 2, 1, 3).reshape(b, lq, c).to(q.dtype)
 
 
-def linear(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor | None = None) -> torch.Tensor:
-    """Plain library GEMM (hipBLASLt via torch on ROCm) — allowed per design
-    notes: hand-written kernels are reserved for fused/hot ops."""
-    return F.linear(x, weight, bias)
+def linear(
+    x: torch.Tensor,
+    weight: torch.Tensor,
+    bias: torch.Tensor | None = None,
+    residual: torch.Tensor | None = None,
+) -> torch.Tensor:
+    """Linear projection. Plain GEMMs go to hipBLASLt (allowed: library
+    GEMMs); projections that carry a residual add (attention out-proj, FF
+    out, transformer proj_out) run through OUR MFMA kernel as a 1x1 conv
+    with the residual fused into the epilogue — one kernel instead of
+    GEMM + aten add."""
+    if residual is not None and _use_hip(x) and x.dim() == 3 \
+            and weight.shape[1] % 32 == 0:
+        ext = _require_ext()
+        b, l, c = x.shape
+        b32 = None
+        if bias is not None:
+            b32 = _cached(bias, "_airtc_b32", lambda: bias.detach().float().contiguous())
+        w16 = _cached(weight, "_airtc_w16", lambda: weight.detach().contiguous().half())
+        y = ext.conv2d(
+            x.reshape(b, l, 1, c), w16, b32, None,
+            residual.reshape(b, l, 1, -1).contiguous(), 1, 1, 1, 0, ACT_NONE,
+        )
+        return y.reshape(b, l, -1)
+    y = F.linear(x, weight, bias)
+    if residual is not None:
+        y = y + residual
+    return y
 
 
 def geglu(x: torch.Tensor) -> torch.Tensor:
