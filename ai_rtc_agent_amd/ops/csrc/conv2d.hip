@@ -346,7 +346,10 @@ extern "C" int airtc_conv2d_splitk_for(int B, int HO, int WO, int OC, int IC) {
   // ran 87us = 86 TF; split-K over the K loop fills the chip).
   if (IC % 32 != 0) return 0;
   const int M = HO * WO;
-  if ((long)M * B >= 65536 && OC <= BN) return 100;  // BM256 (TAESD hi-res)
+  // NOTE: a BM256 (MFRAG8) geometry measured SLOWER on the TAESD hi-res
+  // layers (24.5 -> 37.1us @256²x64ch: fewer blocks + bigger staging
+  // footprint cost more than the barrier amortisation bought) — path 100
+  // exists but is never selected.
   if (M >= 2048) {
     const long blocks = (long)ceil_div(M, 128) * ceil_div(OC, BN) * B;
     long k = (480 + blocks - 1) / blocks;
