@@ -24,6 +24,8 @@
 // Geometry B (small M): BM=64  BN=64, 4 waves (2x2), wave=32x32, 4 MFMA/step,
 //                       optional split-K over blockIdx.z
 
+#include <stdlib.h>
+
 #include "common.h"
 
 #define BN 64
@@ -68,7 +70,7 @@ __device__ __forceinline__ f16 epilogue(float acc, const float* bias,
 // barrier count per K element, 2 MFMA K-chunks per stage), else 32.
 // SPLITK > 1: partials go to ws (f32), finalize pass reduces.
 // ---------------------------------------------------------------------------
-template <int MFRAG, int BK>
+template <int MFRAG, int BK, bool DBUF>
 __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
     const f16* __restrict__ x, const f16* __restrict__ w,
     const float* __restrict__ bias, const f16* __restrict__ cbias,
@@ -81,8 +83,9 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
   constexpr int BLOADS = BK / 32;             // staged 16B B-loads per thread
   constexpr int KSH = BK == 64 ? 3 : 2;       // flat -> (row, k8) shifts
   constexpr int KMSK = BK / 8 - 1;
-  __shared__ f16 ldsA[BM * KPITCH];
-  __shared__ f16 ldsB[BN * KPITCH];
+  constexpr int NBUF = DBUF ? 2 : 1;
+  __shared__ f16 ldsA[NBUF * BM * KPITCH];
+  __shared__ f16 ldsB[NBUF * BN * KPITCH];
 
   const int M = HO * WO;
   const int m0 = blockIdx.x * BM;
@@ -132,41 +135,32 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
   // NOTE: an empty split (uneven tail) still stores its zero slab below —
   // the finalize pass reads every slab.
 
-  // prologue: stage first K-tile of this split into registers
+  // staging helpers
   f16x8 regA[ALOADS], regB[BLOADS];
-  if (k_lo < k_hi) {
-    KPos p = kpos_at(k_lo * BK, IC, S);
+  auto load_tile = [&](int kt) {
+    KPos p = kpos_at(kt * BK, IC, S);
 #pragma unroll
     for (int i = 0; i < ALOADS; ++i)
       regA[i] = load_a(xb, a_ho[i], a_wo[i], p.r, p.s, pad, H, W, IC,
                        p.ic0 + a_k8[i]);
 #pragma unroll
     for (int i = 0; i < BLOADS; ++i)
-      regB[i] = *reinterpret_cast<const f16x8*>(wrow[i] + k_lo * BK);
-  }
-
-  for (int kt = k_lo; kt < k_hi; ++kt) {
-    __syncthreads();
+      regB[i] = *reinterpret_cast<const f16x8*>(wrow[i] + kt * BK);
+  };
+  auto write_tile = [&](int buf) {
+    f16* la = &ldsA[buf * BM * KPITCH];
+    f16* lb = &ldsB[buf * BN * KPITCH];
 #pragma unroll
     for (int i = 0; i < ALOADS; ++i)
-      *reinterpret_cast<f16x8*>(&ldsA[a_row[i] * KPITCH + a_k8[i]]) = regA[i];
+      *reinterpret_cast<f16x8*>(&la[a_row[i] * KPITCH + a_k8[i]]) = regA[i];
 #pragma unroll
     for (int i = 0; i < BLOADS; ++i)
-      *reinterpret_cast<f16x8*>(&ldsB[b_row[i] * KPITCH + b_k8[i]]) = regB[i];
-    __syncthreads();
-
-    if (kt + 1 < k_hi) {  // T14: next tile's loads before MFMAs
-      KPos p = kpos_at((kt + 1) * BK, IC, S);
-#pragma unroll
-      for (int i = 0; i < ALOADS; ++i)
-        regA[i] = load_a(xb, a_ho[i], a_wo[i], p.r, p.s, pad, H, W, IC,
-                         p.ic0 + a_k8[i]);
-#pragma unroll
-      for (int i = 0; i < BLOADS; ++i)
-        regB[i] = *reinterpret_cast<const f16x8*>(wrow[i] + (kt + 1) * BK);
-    }
-
-    const int arow_base = wm * (MFRAG * 16) + (lane & 15);
+      *reinterpret_cast<f16x8*>(&lb[b_row[i] * KPITCH + b_k8[i]]) = regB[i];
+  };
+  const int arow_base = wm * (MFRAG * 16) + (lane & 15);
+  auto compute_tile = [&](int buf) {
+    const f16* la = &ldsA[buf * BM * KPITCH];
+    const f16* lb = &ldsB[buf * BN * KPITCH];
 #pragma unroll
     for (int kk = 0; kk < BK / 32; ++kk) {
       const int fcol = kk * 32 + (lane >> 4) * 8;
@@ -174,15 +168,45 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni)
         bfrag[ni] = *reinterpret_cast<const f16x8*>(
-            &ldsB[(wn * 32 + ni * 16 + (lane & 15)) * KPITCH + fcol]);
+            &lb[(wn * 32 + ni * 16 + (lane & 15)) * KPITCH + fcol]);
 #pragma unroll
       for (int mi = 0; mi < MFRAG; ++mi) {
         f16x8 afrag = *reinterpret_cast<const f16x8*>(
-            &ldsA[(arow_base + mi * 16) * KPITCH + fcol]);
+            &la[(arow_base + mi * 16) * KPITCH + fcol]);
 #pragma unroll
         for (int ni = 0; ni < 2; ++ni)
           acc[mi][ni] = mfma16x16x32(afrag, bfrag[ni], acc[mi][ni]);
       }
+    }
+  };
+
+  if constexpr (!DBUF) {
+    // 2-barrier single-buffer loop (T14 prefetch only)
+    if (k_lo < k_hi) load_tile(k_lo);
+    for (int kt = k_lo; kt < k_hi; ++kt) {
+      __syncthreads();
+      write_tile(0);
+      __syncthreads();
+      if (kt + 1 < k_hi) load_tile(kt + 1);
+      compute_tile(0);
+    }
+  } else {
+    // ONE barrier per K-step: double-buffered LDS, write tile t+1 AFTER the
+    // barrier, re-issue the loads for t+2 immediately (guide T14/G15 form)
+    if (k_lo < k_hi) {
+      load_tile(k_lo);
+      write_tile(0);
+      if (k_lo + 1 < k_hi) load_tile(k_lo + 1);
+    }
+    __syncthreads();
+    for (int kt = k_lo; kt < k_hi; ++kt) {
+      const int cur = (kt - k_lo) & 1;
+      if (kt + 1 < k_hi) {
+        write_tile(cur ^ 1);
+        if (kt + 2 < k_hi) load_tile(kt + 2);
+      }
+      compute_tile(cur);
+      __syncthreads();
     }
   }
 
@@ -379,24 +403,28 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
   f16* op = reinterpret_cast<f16*>(out);
   const int M = HO * WO;
 
-  const bool bm256 = (path == 100);
-  const int splitk = bm256 ? 1 : (path > 0 ? path : -path);
-  const int bm = bm256 ? 256 : (path > 0 ? 128 : 64);
+  const int splitk = path > 0 ? path : -path;
+  const int bm = path > 0 ? 128 : 64;
   dim3 grid(ceil_div(M, bm), ceil_div(OC, BN), B * splitk);
   const float* b1 = splitk == 1 ? bias : nullptr;
   const f16* cb1 = splitk == 1 ? cb : nullptr;
   const f16* res1 = splitk == 1 ? res : nullptr;
   const bool bk64 = (IC % 64 == 0);
-#define CONV_LAUNCH(MF, BKV)                                                 \
-  hipLaunchKernelGGL((conv2d_mfma_kernel<MF, BKV>), grid, dim3(256), 0, s,   \
-                     xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC, R, \
-                     S, stride, pad, act, K, splitk)
-  if (bm256) {
-    if (bk64) CONV_LAUNCH(8, 64); else CONV_LAUNCH(8, 32);
-  } else if (path > 0) {
-    if (bk64) CONV_LAUNCH(4, 64); else CONV_LAUNCH(4, 32);
+  static int dbuf = -1;
+  if (dbuf < 0) {  // A/B toggle: AIRTC_CONV_DBUF=0 -> 2-barrier loop
+    const char* e = getenv("AIRTC_CONV_DBUF");
+    dbuf = e ? atoi(e) : 1;
+  }
+#define CONV_LAUNCH(MF, BKV, DB)                                              \
+  hipLaunchKernelGGL((conv2d_mfma_kernel<MF, BKV, DB>), grid, dim3(256), 0,   \
+                     s, xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC,  \
+                     R, S, stride, pad, act, K, splitk)
+  if (path > 0) {
+    if (bk64) { if (dbuf) CONV_LAUNCH(4, 64, true); else CONV_LAUNCH(4, 64, false); }
+    else CONV_LAUNCH(4, 32, false);
   } else {
-    if (bk64) CONV_LAUNCH(2, 64); else CONV_LAUNCH(2, 32);
+    if (bk64) { if (dbuf) CONV_LAUNCH(2, 64, true); else CONV_LAUNCH(2, 64, false); }
+    else CONV_LAUNCH(2, 32, false);
   }
 #undef CONV_LAUNCH
   if (splitk > 1) {

@@ -42,6 +42,8 @@ def test_ext_loads():
     (96, 40, 12, 1, 1, 3),      # ragged OC, non-pow2 spatial
     (1280, 1280, 8, 1, 1, 3),   # 8x8 wide UNet layer -> split-K path
     (640, 1280, 16, 2, 1, 3),   # stride-2 into split-K territory
+    (64, 64, 64, 1, 1, 3),      # TAESD-shaped: BM128 large-spatial path
+    (320, 64, 64, 1, 0, 1),     # 1x1 conv on the BM128 path
 ])
 def test_conv2d_vs_torch(ic, oc, h, stride, pad, r):
     x = rnd(2, h, h, ic, seed=ic + oc)
