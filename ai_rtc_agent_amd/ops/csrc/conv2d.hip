@@ -411,9 +411,13 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
   const f16* res1 = splitk == 1 ? res : nullptr;
   const bool bk64 = (IC % 64 == 0);
   static int dbuf = -1;
-  if (dbuf < 0) {  // A/B toggle: AIRTC_CONV_DBUF=0 -> 2-barrier loop
+  if (dbuf < 0) {
+    // Measured A/B on MI355X: the double-buffer single-barrier schedule is
+    // SLOWER here (114.8 vs 118.6 fps end-to-end) — the 2x LDS footprint
+    // costs more occupancy than the barrier removal buys at these tiles
+    // (guide §5.5: the lever is regime-gated). Default stays 2-barrier.
     const char* e = getenv("AIRTC_CONV_DBUF");
-    dbuf = e ? atoi(e) : 1;
+    dbuf = e ? atoi(e) : 0;
   }
 #define CONV_LAUNCH(MF, BKV, DB)                                              \
   hipLaunchKernelGGL((conv2d_mfma_kernel<MF, BKV, DB>), grid, dim3(256), 0,   \
