@@ -61,6 +61,17 @@ class VcnH264Codec:
 
     @staticmethod
     def _probe():
+        # authoritative: the native probe (ops/csrc/vcn.cpp) walks
+        # dlopen -> DRM node -> vaInitialize -> H.264 entrypoints
+        try:
+            from .. import ops
+
+            ext = ops.hip_ext()
+            if ext is not None and hasattr(ext, "vcn_probe"):
+                r = ext.vcn_probe()
+                return r if r.get("available") else None
+        except Exception:
+            pass
         for name in ("libva.so.2", "libva.so", "libva-drm.so.2"):
             try:
                 return ctypes.CDLL(name)

@@ -181,6 +181,17 @@ torch::Tensor postprocess_u8(torch::Tensor img) {
   return out;
 }
 
+pybind11::dict vcn_probe() {
+  char buf[512];
+  int rc = airtc_vcn_probe(buf, sizeof(buf));
+  pybind11::dict d;
+  d["available"] = rc >= 0;
+  d["h264_decode"] = rc >= 0 && (rc & 1);
+  d["h264_encode"] = rc >= 0 && (rc & 2);
+  d["detail"] = std::string(buf);
+  return d;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -194,4 +205,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("upsample2x", &upsample2x);
   m.def("preprocess_u8", &preprocess_u8);
   m.def("postprocess_u8", &postprocess_u8);
+  m.def("vcn_probe", &vcn_probe, "probe the VCN VA-API stack");
 }

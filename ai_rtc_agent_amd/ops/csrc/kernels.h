@@ -59,3 +59,9 @@ void airtc_attention(const uint16_t* q, const uint16_t* k, const uint16_t* v,
                      hipStream_t s);
 
 }  // extern "C"
+
+// vcn ------------------------------------------------------------------------
+// Probe the VCN video block's VA-API userspace (runtime dlopen). Returns
+// -1 = unavailable, else bit0 = H.264 decode, bit1 = H.264 encode; buf gets
+// a stage/detail summary. Implemented in vcn.cpp.
+extern "C" int airtc_vcn_probe(char* buf, int buflen);

@@ -116,3 +116,22 @@ def test_stun_binding_flow():
     assert resp.msg_type == stun.BINDING_RESPONSE
     assert resp.transaction_id == req.transaction_id
     assert stun.ATTR_XOR_MAPPED_ADDRESS in resp.attributes
+
+
+def test_vcn_native_probe_reports():
+    """The native VA-API probe (ops/csrc/vcn.cpp) must load and report a
+    precise stage/detail even on boxes without the VCN userspace."""
+    from ai_rtc_agent_amd.ops import _load_ext
+
+    try:
+        ext = _load_ext.load()
+    except ImportError:
+        pytest.skip("extension not built")
+    r = ext.vcn_probe()
+    assert set(r) == {"available", "h264_decode", "h264_encode", "detail"}
+    assert "stage=" in r["detail"]
+    # selection layer must fall back to software when unavailable
+    from ai_rtc_agent_amd.media.codec import SoftwareCodec, select_codec
+
+    if not r["available"]:
+        assert isinstance(select_codec(), SoftwareCodec)
