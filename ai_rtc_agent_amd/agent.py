@@ -145,9 +145,11 @@ async def whep(request: web.Request) -> web.Response:
 
     pc = PeerConnection()
     st["pcs"].add(pc)
-    # direct track attach (the reference comments out MediaRelay.subscribe,
-    # agent.py:248-252); our relay is available for multi-viewer fan-out
-    pc.add_track(st["source_track"])
+    # relay fan-out so N viewers share one pipeline pull (the reference
+    # attaches the track directly and leaves its MediaRelay unused,
+    # agent.py:248-252 — with >1 viewer they would steal frames from each
+    # other; the relay fixes that)
+    pc.add_track(st["relay"].subscribe(st["source_track"]))
 
     @pc.on("connectionstatechange")
     def on_state() -> None:
@@ -212,6 +214,20 @@ async def on_shutdown(app: web.Application) -> None:
     st["pcs"].clear()
 
 
+@web.middleware
+async def cors_middleware(request: web.Request, handler):
+    """CORS for browser clients (the reference pulls in aiohttp_middlewares'
+    cors_middleware, agent.py:459; ours is self-contained)."""
+    if request.method == "OPTIONS":
+        resp = web.Response(status=204)
+    else:
+        resp = await handler(request)
+    resp.headers["Access-Control-Allow-Origin"] = "*"
+    resp.headers["Access-Control-Allow-Methods"] = "GET, POST, DELETE, OPTIONS"
+    resp.headers["Access-Control-Allow-Headers"] = "Content-Type, Authorization"
+    return resp
+
+
 def create_app(
     model_id: str = "lykon/dreamshaper-8",
     udp_ports: Optional[list] = None,
@@ -220,7 +236,7 @@ def create_app(
     n_gpus: int = 1,
     use_turn: bool = True,
 ) -> web.Application:
-    app = web.Application()
+    app = web.Application(middlewares=[cors_middleware])
     app["model_id"] = model_id
     app["udp_ports"] = udp_ports
     app["host"] = host

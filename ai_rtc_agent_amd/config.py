@@ -125,6 +125,9 @@ class EngineConfig:
     lora_dict: Optional[dict] = None
     lcm_lora_id: Optional[str] = None
     vae_id: Optional[str] = None
+    use_controlnet: bool = False
+    controlnet_scale: float = 1.0
+    use_safety_checker: bool = False
     similarity_filter: SimilarityFilterConfig = field(default_factory=SimilarityFilterConfig)
     encoder: EncoderConfig = field(default_factory=EncoderConfig)
     device: str = "cuda"

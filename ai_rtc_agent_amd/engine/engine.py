@@ -79,6 +79,17 @@ class StreamDiffusionEngine:
         self.ctx_dim = ucfg.cross_attention_dim
         self.addition_embed_dim = ucfg.addition_embed_dim
 
+        self.controlnet = None
+        if cfg.use_controlnet:
+            from ..models.controlnet import ControlNet
+
+            self.controlnet = ControlNet(ucfg)
+        self.safety_checker = None
+        if cfg.use_safety_checker:
+            from ..models.safety import SafetyChecker
+
+            self.safety_checker = SafetyChecker()
+
         # real weights when available: model_id as a local diffusers-style
         # dir, or an HF-cache snapshot (random init otherwise — offline)
         self._load_weights_if_present(cfg.model_id)
@@ -99,6 +110,10 @@ class StreamDiffusionEngine:
         self.unet = self.unet.to(self.device, self.dtype).eval()
         self.vae = self.vae.to(self.device, self.dtype).eval()
         self.text_encoder = self.text_encoder.to(self.device).eval()
+        if self.controlnet is not None:
+            self.controlnet = self.controlnet.to(self.device, self.dtype).eval()
+        if self.safety_checker is not None:
+            self.safety_checker = self.safety_checker.to(self.device, self.dtype).eval()
 
         self.timers = StageTimers(use_cuda=self.device.type == "cuda")
         self._graph: Optional[torch.cuda.CUDAGraph] = None
@@ -290,12 +305,19 @@ class StreamDiffusionEngine:
         if added is not None and (cfg.cfg_type in ("full", "initialize") and self.rcfg.active):
             extra = added.shape[0] if cfg.cfg_type == "full" else fbs
             added = torch.cat([added[:extra], added], dim=0)
-        eps = self.unet(
-            self._unet_batch_input(x_t),
-            self._unet_batch_timesteps(),
-            self._unet_batch_embeds(),
-            added_cond=added,
-        )
+        unet_in = self._unet_batch_input(x_t)
+        unet_ts = self._unet_batch_timesteps()
+        unet_emb = self._unet_batch_embeds()
+        control = None
+        if self.controlnet is not None and cfg.mode == "img2img":
+            # hint = the current input frame (per in-flight stage we reuse
+            # the newest frame's hint; per-stage hints would need a hint
+            # FIFO mirroring the latent buffer)
+            hint = self._img_in.expand(unet_in.shape[0], -1, -1, -1).contiguous()
+            control = self.controlnet(
+                unet_in, unet_ts, unet_emb, hint, scale=cfg.controlnet_scale
+            )
+        eps = self.unet(unet_in, unet_ts, unet_emb, added_cond=added, control=control)
         eps = self.rcfg.apply(eps, fbs)
         denoised = self.scheduler.step_batch(eps, x_t, co)
 
@@ -309,7 +331,10 @@ class StreamDiffusionEngine:
             ) if cfg.do_add_noise else denoised[:-fbs]
             self._x_t_buffer.copy_(nxt)
 
-        return self.vae.decode(denoised[-fbs:])
+        decoded = self.vae.decode(denoised[-fbs:])
+        if self.safety_checker is not None:
+            decoded = self.safety_checker.filter(decoded)
+        return decoded
 
     def _maybe_capture(self) -> None:
         if (

@@ -378,6 +378,7 @@ class UNet2DCondition(nn.Module):
         timesteps: torch.Tensor,
         encoder_hidden_states: torch.Tensor,
         added_cond: torch.Tensor | None = None,
+        control: tuple | None = None,
     ) -> torch.Tensor:
         """sample: (B,H,W,C_in) NHWC latent; timesteps: (B,);
         encoder_hidden_states: (B,77,ctx)."""
@@ -401,9 +402,15 @@ class UNet2DCondition(nn.Module):
                 x = self.downsamplers[bi](x)
                 skips.append(x)
 
+        if control is not None:
+            # ControlNet residuals: one per skip entry + one for mid
+            skip_res, mid_res = control
+            skips = [s + c for s, c in zip(skips, skip_res)]
         x = self.mid_res1(x, temb)
         x = self.mid_attn(x, encoder_hidden_states)
         x = self.mid_res2(x, temb)
+        if control is not None:
+            x = x + mid_res
 
         ri = 0
         for bi in range(len(cfg.block_out_channels)):
