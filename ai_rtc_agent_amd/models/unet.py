@@ -165,18 +165,45 @@ class LayerNorm(nn.Module):
 
 
 class CrossAttention(nn.Module):
+    """Attention with fused projections and WEIGHT-level head-dim padding.
+
+    Head dims that the MFMA kernel doesn't serve natively (sd15's 40/80)
+    are padded to the next supported size by inserting ZERO ROWS into the
+    q/k/v projection weights (and zero input-columns into the out
+    projection) ONCE at cache build — the per-frame path then runs the
+    native kernel with no rearrange copies. Zero rows leave every dot
+    product and the softmax unchanged, so numerics are identical.
+    """
+
+    _DIMS = (32, 64, 96, 128, 160)
+
     def __init__(self, dim: int, ctx_dim: int, heads: int):
         super().__init__()
         self.heads = heads
+        self.head_dim = dim // heads
+        self.dpad = (
+            self.head_dim
+            if self.head_dim in self._DIMS
+            else min(x for x in self._DIMS if x >= self.head_dim)
+        )
         self.to_q = Linear(dim, dim, bias=False)
         self.to_k = Linear(ctx_dim, dim, bias=False)
         self.to_v = Linear(ctx_dim, dim, bias=False)
         self.to_out = Linear(dim, dim)
 
-    def _fused_w(self, name: str, parts, ref: torch.Tensor) -> torch.Tensor:
+    def _pad_heads(self, w: torch.Tensor) -> torch.Tensor:
+        """(heads*d, src) -> (heads*dpad, src) with zero rows per head."""
+        if self.dpad == self.head_dim:
+            return w
+        o, src = w.shape
+        wp = w.view(self.heads, self.head_dim, src)
+        wp = torch.nn.functional.pad(wp, (0, 0, 0, self.dpad - self.head_dim))
+        return wp.reshape(self.heads * self.dpad, src)
+
+    def _cached_w(self, name: str, ref: torch.Tensor, build) -> torch.Tensor:
         w = getattr(self, name, None)
         if w is None or w.device != ref.device or w.dtype != ref.dtype:
-            w = torch.cat([p.weight.detach().to(ref.device, ref.dtype) for p in parts], dim=0)
+            w = build().detach().to(ref.device, ref.dtype)
             setattr(self, name, w)
         return w
 
@@ -186,18 +213,31 @@ class CrossAttention(nn.Module):
         ctx: torch.Tensor | None = None,
         residual: torch.Tensor | None = None,
     ) -> torch.Tensor:
-        # fused projections: one GEMM for QKV (self) / KV (cross); the
-        # attention kernel consumes the chunk views zero-copy (strided q/k/v)
         if ctx is None:
-            qkv = ops.linear(x, self._fused_w("_wqkv", (self.to_q, self.to_k, self.to_v), x))
+            wqkv = self._cached_w(
+                "_wqkv", x,
+                lambda: torch.cat([self._pad_heads(p.weight) for p in (self.to_q, self.to_k, self.to_v)]),
+            )
+            qkv = ops.linear(x, wqkv)
             q, k, v = qkv.chunk(3, dim=-1)
         else:
-            q = self.to_q(x)
-            kv = ops.linear(ctx, self._fused_w("_wkv", (self.to_k, self.to_v), x))
+            wq = self._cached_w("_wq", x, lambda: self._pad_heads(self.to_q.weight))
+            wkv = self._cached_w(
+                "_wkv", x,
+                lambda: torch.cat([self._pad_heads(p.weight) for p in (self.to_k, self.to_v)]),
+            )
+            q = ops.linear(x, wq)
+            kv = ops.linear(ctx, wkv)
             k, v = kv.chunk(2, dim=-1)
-        o = ops.attention(q, k, v, self.heads)
-        # residual fused into the out-projection epilogue
-        return self.to_out(o, residual=residual)
+        o = ops.attention(q, k, v, self.heads, scale=1.0 / math.sqrt(self.head_dim))
+        # out projection consumes the padded layout (zero input columns)
+        wout = self._cached_w(
+            "_wout", x,
+            lambda: self._pad_heads(
+                self.to_out.weight.t().contiguous()
+            ).reshape(self.heads * self.dpad, -1).t().contiguous(),
+        )
+        return ops.linear(o, wout, self.to_out.bias, residual=residual)
 
 
 class FeedForwardGEGLU(nn.Module):

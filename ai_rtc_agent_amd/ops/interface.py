@@ -194,21 +194,24 @@ _ATTN_DIMS = (32, 64, 96, 128, 160)
 
 
 def attention(
-    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, num_heads: int
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, num_heads: int,
+    scale: float | None = None,
 ) -> torch.Tensor:
     """q: (B, Lq, C); k,v: (B, Lk, C). Returns (B, Lq, C).
 
     GPU: flash-style fused kernel (online softmax, MFMA QK^T and PV,
     LDS-tiled K/V) — ops/csrc/attention.hip. head_dim % 32 == 0 runs
-    zero-copy on (b,h)-strided views; other head dims (sd15's 40/80) are
-    zero-padded to the next supported size (softmax-invariant).
+    zero-copy on (b,h)-strided views; other head dims are zero-padded to
+    the next supported size (softmax-invariant — but pass the TRUE-dim
+    scale when the inputs were padded upstream).
     """
     if _use_hip(q):
         ext = _require_ext()
         b, lq, c = q.shape
         lk = k.shape[1]
         d = c // num_heads
-        scale = 1.0 / math.sqrt(d)
+        if scale is None:
+            scale = 1.0 / math.sqrt(d)
         if d in _ATTN_DIMS:
             return ext.attention_bhlc(q, k, v, num_heads, scale)
         dpad = min(x for x in _ATTN_DIMS if x >= d)
@@ -227,7 +230,7 @@ def attention(
     qh = q.reshape(b, lq, num_heads, d).permute(0, 2, 1, 3).float()
     kh = k.reshape(b, -1, num_heads, d).permute(0, 2, 1, 3).float()
     vh = v.reshape(b, -1, num_heads, d).permute(0, 2, 1, 3).float()
-    o = F.scaled_dot_product_attention(qh, kh, vh)
+    o = F.scaled_dot_product_attention(qh, kh, vh, scale=scale)
     return o.permute(0, 2, 1, 3).reshape(b, lq, c).to(q.dtype)
 
 
