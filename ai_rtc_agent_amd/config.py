@@ -134,6 +134,10 @@ class EngineConfig:
     # engine acceleration: "hip" (hand-written kernels + hipGraph) or "eager"
     acceleration: str = "hip"
     use_hip_graph: bool = True
+    # overlap VAE-decode+postprocess of frame i with frame i+1's denoise on a
+    # second HIP stream (pipelined graphs; disabled automatically when the
+    # similarity filter needs a per-frame decision)
+    pipeline_overlap: bool = True
 
     @property
     def denoising_steps(self) -> int:

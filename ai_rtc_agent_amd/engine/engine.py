@@ -195,6 +195,8 @@ class StreamDiffusionEngine:
         self._ts_batch = self._coeff["sub_timesteps_tensor"].to(dev)
         self._prev_out: Optional[torch.Tensor] = None
         self._graph = None
+        self._pipelined = False
+        self._last_done = None
         self._prepared = True
 
     @torch.no_grad()
@@ -279,11 +281,10 @@ class StreamDiffusionEngine:
         return self._ts_batch
 
     @torch.no_grad()
-    def _step_core(self) -> torch.Tensor:
-        """One stream-batch denoise round over static buffers.
-
-        _img_in (fbs,H,W,3 dtype) -> output image (fbs,H,W,3 dtype).
-        """
+    def _denoise_core(self) -> torch.Tensor:
+        """Stream-batch round WITHOUT the VAE decode: _img_in -> denoised
+        latent (fbs, lh, lw, 4). Split out so the decode half can run on a
+        second HIP stream overlapped with the next frame's denoise."""
         cfg = self.cfg
         fbs = cfg.frame_buffer_size
         co = self._coeff
@@ -331,29 +332,96 @@ class StreamDiffusionEngine:
             ) if cfg.do_add_noise else denoised[:-fbs]
             self._x_t_buffer.copy_(nxt)
 
-        decoded = self.vae.decode(denoised[-fbs:])
+        return denoised[-fbs:]
+
+    @torch.no_grad()
+    def _decode_core(self, latent: torch.Tensor) -> torch.Tensor:
+        decoded = self.vae.decode(latent)
         if self.safety_checker is not None:
             decoded = self.safety_checker.filter(decoded)
-        return decoded
+        return ops.postprocess_to_u8(decoded)
+
+    @torch.no_grad()
+    def _step_core(self) -> torch.Tensor:
+        """Sequential full step (CPU / eager GPU path)."""
+        return self._decode_core(self._denoise_core())
 
     def _maybe_capture(self) -> None:
+        """Capture the per-frame step into hipGraphs (torch.cuda.CUDAGraph is
+        hipGraph on ROCm).
+
+        PIPELINED mode (default): the step is split into two graphs —
+        g1 = preprocess-side denoise (VAE encode -> stream-batch UNet ->
+        scheduler -> latent handoff buffer) on stream A, and
+        g2 = VAE decode + postprocess on stream B. g2(frame i) runs
+        CONCURRENTLY with g1(frame i+1): steady-state throughput becomes
+        max(g1, g2) instead of g1+g2. Latent handoff buffers are ping-pong
+        (two g1/g2 captures sharing pools per stream); event chain:
+          sA: wait done[pp] -> g1 -> record lat[pp]
+          sB: wait lat[pp]  -> g2 -> record done[pp]
+        The CALLER's stream is never made to wait (that would transitively
+        serialise the next frame's g1 behind this frame's decode) —
+        consumers that read the output tensor call sync_output() first.
+        """
         if (
             self._graph is not None
             or self.device.type != "cuda"
             or not self.cfg.use_hip_graph
         ):
             return
-        # warmup on a side stream, then capture (hipGraph on ROCm)
-        s = torch.cuda.Stream()
-        s.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(s):
+        fbs = self.cfg.frame_buffer_size
+        lh, lw = self.cfg.latent_height, self.cfg.latent_width
+        self._pipelined = self.sim_filter is None and self.cfg.pipeline_overlap
+        sA = torch.cuda.Stream()
+        sB = torch.cuda.Stream()
+        self._sA, self._sB = sA, sB
+        self._lat_out = [
+            torch.zeros((fbs, lh, lw, 4), device=self.device, dtype=self.dtype)
+            for _ in range(2)
+        ]
+        # warmup (weight-transform caches etc.) on a side stream
+        sA.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(sA):
             for _ in range(2):
+                lat = self._denoise_core()
+                self._lat_out[0].copy_(lat)
+                _ = self._decode_core(self._lat_out[0])
+        torch.cuda.current_stream().wait_stream(sA)
+
+        if not self._pipelined:
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g, stream=sA):
                 self._graph_out = self._step_core()
-        torch.cuda.current_stream().wait_stream(s)
-        g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
-            self._graph_out = self._step_core()
-        self._graph = g
+            self._graph = g
+            return
+
+        self._g1, self._g2, self._out_u8 = [], [], []
+        for pp in range(2):
+            g1 = torch.cuda.CUDAGraph()
+            kw = {"pool": self._g1[0].pool()} if pp else {}
+            with torch.cuda.graph(g1, stream=sA, **kw):
+                self._lat_out[pp].copy_(self._denoise_core())
+            self._g1.append(g1)
+            g2 = torch.cuda.CUDAGraph()
+            kw = {"pool": self._g2[0].pool()} if pp else {}
+            with torch.cuda.graph(g2, stream=sB, **kw):
+                self._out_u8.append(self._decode_core(self._lat_out[pp]))
+            self._g2.append(g2)
+        self._ev_in = torch.cuda.Event()
+        self._ev_lat = [torch.cuda.Event() for _ in range(2)]
+        self._ev_done = [torch.cuda.Event() for _ in range(2)]
+        for e in self._ev_done:
+            e.record(sB)
+        self._pp = 0
+        self._graph = True  # sentinel: pipelined graphs ready
+
+    def sync_output(self) -> None:
+        """Make the caller's stream wait until the last returned output is
+        fully produced (pipelined mode defers this so the next frame's
+        denoise can overlap the decode)."""
+        ev = getattr(self, "_last_done", None)
+        if ev is not None:
+            torch.cuda.current_stream().wait_event(ev)
 
     # ------------------------------------------------------------------
     # public frame API
@@ -379,16 +447,33 @@ class StreamDiffusionEngine:
                 return out[0] if squeeze else out
 
         with self.timers.stage("diffusion"):
-            self._img_in.copy_(img)
             if self.device.type == "cuda" and self.cfg.use_hip_graph:
                 self._maybe_capture()
-                self._graph.replay()
-                decoded = self._graph_out
+                if self._pipelined:
+                    cur = torch.cuda.current_stream()
+                    pp = self._pp
+                    self._pp ^= 1
+                    self._ev_in.record(cur)
+                    with torch.cuda.stream(self._sA):
+                        self._sA.wait_event(self._ev_in)
+                        self._sA.wait_event(self._ev_done[pp])
+                        self._img_in.copy_(img)
+                        img.record_stream(self._sA)
+                        self._g1[pp].replay()
+                        self._ev_lat[pp].record()
+                    with torch.cuda.stream(self._sB):
+                        self._sB.wait_event(self._ev_lat[pp])
+                        self._g2[pp].replay()
+                        self._ev_done[pp].record()
+                    self._last_done = self._ev_done[pp]
+                    out = self._out_u8[pp]
+                else:
+                    self._img_in.copy_(img)
+                    self._graph.replay()
+                    out = self._graph_out
             else:
-                decoded = self._step_core()
-
-        with self.timers.stage("postprocess"):
-            out = ops.postprocess_to_u8(decoded)
+                self._img_in.copy_(img)
+                out = self._step_core()
 
         self._prev_out = out
         self.timers.frame_done()

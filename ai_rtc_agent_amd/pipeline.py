@@ -36,7 +36,11 @@ class StreamDiffusionPipeline:
         )
 
     def __call__(self, frame: torch.Tensor) -> torch.Tensor:
-        return self.engine(frame)
+        out = self.engine(frame)
+        # media-plane consumers read the tensor on their own stream; fence it
+        # against the engine's pipelined decode stream
+        self.engine.sync_output()
+        return out
 
     def update_prompt(self, prompt: str) -> None:
         self.engine.update_prompt(prompt)

@@ -28,16 +28,50 @@ def test_sd_turbo_end_to_end():
 
 
 def test_graph_replay_matches_eager():
-    """hipGraph capture must be bit-identical to the same kernels eager."""
+    """hipGraph capture (pipelined two-stream mode) must be bit-identical to
+    the same kernels eager. Outputs are cloned after sync_output: pipelined
+    outputs live in ping-pong buffers valid until the same-parity frame two
+    calls later."""
     outs = {}
     for use_graph in (False, True):
         cfg = sd_turbo_config(device="cuda", use_hip_graph=use_graph)
         e = StreamDiffusionEngine(cfg)
         e.prepare()
-        res = [e(frame(seed=10 + i)) for i in range(4)]
+        res = []
+        for i in range(4):
+            out = e(frame(seed=10 + i))
+            e.sync_output()
+            res.append(out.clone())
         outs[use_graph] = res
-    for a, b in zip(outs[False], outs[True]):
-        assert torch.equal(a, b), "graph replay diverged from eager kernels"
+    for i, (a, b) in enumerate(zip(outs[False], outs[True])):
+        assert torch.equal(a, b), f"graph replay diverged from eager at frame {i}"
+
+
+def test_pipelined_vs_sequential_graph():
+    """Two-stream pipelined replay must produce the same frames as the
+    single-graph sequential replay."""
+    outs = {}
+    for overlap in (False, True):
+        cfg = sd_turbo_config(device="cuda", use_hip_graph=True)
+        cfg.pipeline_overlap = overlap
+        e = StreamDiffusionEngine(cfg)
+        e.prepare()
+        res = []
+        for i in range(6):
+            out = e(frame(seed=40 + i))
+            e.sync_output()
+            res.append(out.clone())
+        outs[overlap] = res
+    for i, (a, b) in enumerate(zip(outs[False], outs[True])):
+        assert torch.equal(a, b), f"pipelined diverged at frame {i}"
+
+
+def _run_clone(e, f, n=3):
+    out = None
+    for _ in range(n):
+        out = e(f)
+    e.sync_output()
+    return out.clone()
 
 
 def test_prompt_update_through_graph():
@@ -45,9 +79,9 @@ def test_prompt_update_through_graph():
     e = StreamDiffusionEngine(cfg)
     e.prepare()
     f = frame(seed=3)
-    base = [e(f) for _ in range(3)][-1]
+    base = _run_clone(e, f)
     e.update_prompt("an entirely different style prompt")
-    after = [e(f) for _ in range(3)][-1]
+    after = _run_clone(e, f)
     assert not torch.equal(base, after), "graph-external embed update must take effect"
 
 
