@@ -290,7 +290,10 @@ class StreamDiffusionEngine:
         co = self._coeff
 
         if cfg.mode == "img2img":
-            x0 = self.vae.encode(self._img_in)
+            # preprocess lives INSIDE the graph: _frame_in (u8) is the only
+            # per-frame input buffer
+            self._img_proc = ops.preprocess_from_u8(self._frame_in, self.dtype)
+            x0 = self.vae.encode(self._img_proc)
             x_t0 = self.scheduler.add_noise(
                 x0,
                 self._init_noise[:fbs],
@@ -314,7 +317,7 @@ class StreamDiffusionEngine:
             # hint = the current input frame (per in-flight stage we reuse
             # the newest frame's hint; per-stage hints would need a hint
             # FIFO mirroring the latent buffer)
-            hint = self._img_in.expand(unet_in.shape[0], -1, -1, -1).contiguous()
+            hint = self._img_proc.expand(unet_in.shape[0], -1, -1, -1).contiguous()
             control = self.controlnet(
                 unet_in, unet_ts, unet_emb, hint, scale=cfg.controlnet_scale
             )
@@ -438,10 +441,11 @@ class StreamDiffusionEngine:
 
         with self.timers.stage("preprocess"):
             frame_u8 = frame_u8.to(self.device, non_blocking=True)
-            img = ops.preprocess_from_u8(frame_u8, self.dtype)
 
         if self.sim_filter is not None and self._prev_out is not None:
-            if self.sim_filter.should_skip(img):
+            # centre u8 before cosine similarity (raw u8 is all-positive,
+            # which would inflate similarity between unrelated frames)
+            if self.sim_filter.should_skip(frame_u8.float() - 127.5):
                 out = self._prev_out
                 self.timers.frame_done()
                 return out[0] if squeeze else out
@@ -457,8 +461,8 @@ class StreamDiffusionEngine:
                     with torch.cuda.stream(self._sA):
                         self._sA.wait_event(self._ev_in)
                         self._sA.wait_event(self._ev_done[pp])
-                        self._img_in.copy_(img)
-                        img.record_stream(self._sA)
+                        self._frame_in.copy_(frame_u8)
+                        frame_u8.record_stream(self._sA)
                         self._g1[pp].replay()
                         self._ev_lat[pp].record()
                     with torch.cuda.stream(self._sB):
@@ -468,11 +472,11 @@ class StreamDiffusionEngine:
                     self._last_done = self._ev_done[pp]
                     out = self._out_u8[pp]
                 else:
-                    self._img_in.copy_(img)
+                    self._frame_in.copy_(frame_u8)
                     self._graph.replay()
                     out = self._graph_out
             else:
-                self._img_in.copy_(img)
+                self._frame_in.copy_(frame_u8)
                 out = self._step_core()
 
         self._prev_out = out
