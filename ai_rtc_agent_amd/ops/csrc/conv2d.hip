@@ -65,10 +65,30 @@ __device__ __forceinline__ f16 epilogue(float acc, const float* bias,
 }
 
 // ---------------------------------------------------------------------------
+// GLDS variant: K-tiles stream HBM -> LDS via buffer_load...lds (async DMA,
+// no VGPR round-trip, no ds_write pass — guide §5 "glds, 2 LDS buffers,
+// BK=64, vmcnt(0) + plain __syncthreads()"). Zero-padding comes FREE from
+// the buffer descriptor's bounds check (OOB voffset -> 0 written to LDS).
+// The LDS image is lane-linear (glds writes base + lane*16), so the
+// bank-conflict fix moves to the SOURCE address (guide rule 21): 16B unit
+// u of row r holds global unit u ^ ((r>>1)&7), and fragment reads apply
+// the same XOR. Schedule: double-buffered, one barrier per K-step; hipcc's
+// syncthreads drains the in-flight glds (vmcnt(0)) exactly where the
+// recipe wants it.
+// ---------------------------------------------------------------------------
+typedef __attribute__((address_space(3))) f16 lds_f16;
+
+__device__ __forceinline__ unsigned row_swz(int row) {
+  return (unsigned)((row >> 1) & 7);
+}
+
+// ---------------------------------------------------------------------------
 // templated MFMA conv: MFRAG = M-fragments per wave (4 -> BM=128, 2 -> BM=64)
 // BK = K-tile depth: 64 when IC%64==0 (all SD/TAESD layers — halves the
 // barrier count per K element, 2 MFMA K-chunks per stage), else 32.
 // SPLITK > 1: partials go to ws (f32), finalize pass reduces.
+// DBUF: double-buffered single-barrier register-staged schedule (A/B'd
+// slower at SD shapes; kept for experiments). GLDS: the DMA schedule above.
 // ---------------------------------------------------------------------------
 template <int MFRAG, int BK, bool DBUF>
 __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
@@ -231,6 +251,174 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
     }
   } else {
     // plain f32 slab store; finalize kernel reduces + applies the epilogue
+    float* wsb = ws + ((long)b * splitk + split) * M * OC;
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int col = n0 + wn * 32 + ni * 16 + (lane & 15);
+      if (col >= OC) continue;
+#pragma unroll
+      for (int mi = 0; mi < MFRAG; ++mi)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int m = m0 + wm * (MFRAG * 16) + mi * 16 + (lane >> 4) * 4 + j;
+          if (m < M) wsb[(long)m * OC + col] = acc[mi][ni][j];
+        }
+    }
+  }
+}
+
+template <int MFRAG>
+__global__ __launch_bounds__(256) void conv2d_mfma_glds_kernel(
+    const f16* __restrict__ x, const f16* __restrict__ w,
+    const float* __restrict__ bias, const f16* __restrict__ cbias,
+    const f16* __restrict__ residual, f16* __restrict__ out,
+    float* __restrict__ ws, int H, int W, int IC, int HO, int WO, int OC,
+    int R, int S, int stride, int pad, int act, int K, int splitk) {
+  constexpr int BK = 64;
+  constexpr int BM = MFRAG * 32;
+  constexpr int ALOADS = MFRAG;  // 16B-per-lane glds issues per thread (A)
+  __shared__ f16 ldsA[2 * BM * BK];
+  __shared__ f16 ldsB[2 * BN * BK];
+
+  const int M = HO * WO;
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const int b = blockIdx.z / splitk;
+  const int split = blockIdx.z - b * splitk;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid & 1;
+  const int wn = wid >> 1;
+
+  // buffer descriptors: hardware bounds check gives OOB -> 0 (= the conv's
+  // zero padding, free)
+  const long img_bytes = (long)H * W * IC * 2;
+  auto rsrcA = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)(x + (long)b * H * W * IC), (short)0, (int)img_bytes, 0x00020000);
+  auto rsrcB = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)w, (short)0, (int)((long)OC * K * 2), 0x00020000);
+  constexpr unsigned OOB = 0x7ffffff0u;
+
+  // staging map: lane's LDS slot is (row, u_log); it LOADS global unit
+  // u_log ^ row_swz(row) (source-side swizzle, guide rule 21)
+  int a_ho[ALOADS], a_wo[ALOADS], a_src8[ALOADS];
+  bool a_mok[ALOADS];
+#pragma unroll
+  for (int i = 0; i < ALOADS; ++i) {
+    const int flat = tid + i * 256;
+    const int row = flat >> 3, u = flat & 7;
+    const int m = m0 + row;
+    a_mok[i] = m < M;
+    const int mm = min(m, M - 1);
+    a_ho[i] = (mm / WO) * stride;
+    a_wo[i] = (mm % WO) * stride;
+    a_src8[i] = (int)(u ^ row_swz(row)) * 8;
+  }
+  int b_off[2];  // BLOADS = 2 at BK=64
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int flat = tid + i * 256;
+    const int row = flat >> 3, u = flat & 7;
+    b_off[i] = (min(n0 + row, OC - 1) * K + (int)(u ^ row_swz(row)) * 8) * 2;
+  }
+
+  f32x4 acc[MFRAG][2];
+#pragma unroll
+  for (int mi = 0; mi < MFRAG; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nk = K / BK;
+  const int per = (nk + splitk - 1) / splitk;
+  const int k_lo = split * per;
+  const int k_hi = min(nk, k_lo + per);
+
+  auto issue = [&](int buf, int kt) {
+    KPos p = kpos_at(kt * BK, IC, S);
+#pragma unroll
+    for (int i = 0; i < ALOADS; ++i) {
+      const int hi = a_ho[i] + p.r - pad;
+      const int wi = a_wo[i] + p.s - pad;
+      const bool ok = a_mok[i] && (unsigned)hi < (unsigned)H &&
+                      (unsigned)wi < (unsigned)W;
+      const unsigned voff =
+          ok ? (unsigned)((((long)hi * W + wi) * IC + p.ic0 + a_src8[i]) * 2)
+             : OOB;
+      const int base = __builtin_amdgcn_readfirstlane(
+          buf * BM * BK + (i * 256 + wid * 64) * 8);
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(
+          rsrcA, (__attribute__((address_space(3))) void*)&ldsA[base], 16,
+          (int)voff, 0, 0, 0);
+    }
+    const int kbyte = kt * BK * 2;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int base = __builtin_amdgcn_readfirstlane(
+          buf * BN * BK + (i * 256 + wid * 64) * 8);
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(
+          rsrcB, (__attribute__((address_space(3))) void*)&ldsB[base], 16,
+          b_off[i] + kbyte, 0, 0, 0);
+    }
+  };
+
+  const int arow_base = wm * (MFRAG * 16) + (lane & 15);
+  auto compute = [&](int buf) {
+    const f16* la = &ldsA[buf * BM * BK];
+    const f16* lb = &ldsB[buf * BN * BK];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int u = kk * 4 + (lane >> 4);
+      f16x8 bfrag[2];
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        const int row = wn * 32 + ni * 16 + (lane & 15);
+        bfrag[ni] = *reinterpret_cast<const f16x8*>(
+            &lb[row * BK + (u ^ row_swz(row)) * 8]);
+      }
+#pragma unroll
+      for (int mi = 0; mi < MFRAG; ++mi) {
+        const int row = arow_base + mi * 16;
+        f16x8 afrag = *reinterpret_cast<const f16x8*>(
+            &la[row * BK + (u ^ row_swz(row)) * 8]);
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = mfma16x16x32(afrag, bfrag[ni], acc[mi][ni]);
+      }
+    }
+  };
+
+  if (k_lo < k_hi) issue(0, k_lo);
+  __syncthreads();  // hipcc drains the in-flight glds here (vmcnt(0))
+  for (int kt = k_lo; kt < k_hi; ++kt) {
+    const int cur = (kt - k_lo) & 1;
+    if (kt + 1 < k_hi) issue(cur ^ 1, kt + 1);
+    compute(cur);
+    __syncthreads();
+  }
+
+  // epilogue (same as the register-staged kernel)
+  if (splitk == 1) {
+    f16* ob = out + (long)b * M * OC;
+    const long cb_off = (long)b * OC;
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int col = n0 + wn * 32 + ni * 16 + (lane & 15);
+      if (col >= OC) continue;
+#pragma unroll
+      for (int mi = 0; mi < MFRAG; ++mi)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int m = m0 + wm * (MFRAG * 16) + mi * 16 + (lane >> 4) * 4 + j;
+          if (m < M) {
+            const long idx = (long)b * M * OC + (long)m * OC + col;
+            ob[(long)m * OC + col] = epilogue(acc[mi][ni][j], bias, cbias,
+                                              cb_off, residual, idx, col, act);
+          }
+        }
+    }
+  } else {
     float* wsb = ws + ((long)b * splitk + split) * M * OC;
 #pragma unroll
     for (int ni = 0; ni < 2; ++ni) {
@@ -410,27 +598,36 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
   const f16* cb1 = splitk == 1 ? cb : nullptr;
   const f16* res1 = splitk == 1 ? res : nullptr;
   const bool bk64 = (IC % 64 == 0);
-  static int dbuf = -1;
+  static int dbuf = -1, glds = -1;
   if (dbuf < 0) {
-    // Measured A/B on MI355X: the double-buffer single-barrier schedule is
-    // SLOWER here (114.8 vs 118.6 fps end-to-end) — the 2x LDS footprint
-    // costs more occupancy than the barrier removal buys at these tiles
-    // (guide §5.5: the lever is regime-gated). Default stays 2-barrier.
+    // Measured A/B on MI355X: the double-buffer single-barrier REGISTER
+    // schedule is SLOWER here (114.8 vs 118.6 fps end-to-end) — the 2x LDS
+    // footprint costs more occupancy than the barrier removal buys at these
+    // tiles (guide §5.5: regime-gated). Default stays 2-barrier.
     const char* e = getenv("AIRTC_CONV_DBUF");
     dbuf = e ? atoi(e) : 0;
+    const char* g = getenv("AIRTC_CONV_GLDS");
+    glds = g ? atoi(g) : 0;
   }
 #define CONV_LAUNCH(MF, BKV, DB)                                              \
   hipLaunchKernelGGL((conv2d_mfma_kernel<MF, BKV, DB>), grid, dim3(256), 0,   \
                      s, xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC,  \
                      R, S, stride, pad, act, K, splitk)
+#define CONV_LAUNCH_GLDS(MF)                                                  \
+  hipLaunchKernelGGL((conv2d_mfma_glds_kernel<MF>), grid, dim3(256), 0, s,    \
+                     xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC, R,  \
+                     S, stride, pad, act, K, splitk)
   if (path > 0) {
-    if (bk64) { if (dbuf) CONV_LAUNCH(4, 64, true); else CONV_LAUNCH(4, 64, false); }
+    if (bk64 && glds) CONV_LAUNCH_GLDS(4);
+    else if (bk64) { if (dbuf) CONV_LAUNCH(4, 64, true); else CONV_LAUNCH(4, 64, false); }
     else CONV_LAUNCH(4, 32, false);
   } else {
-    if (bk64) { if (dbuf) CONV_LAUNCH(2, 64, true); else CONV_LAUNCH(2, 64, false); }
+    if (bk64 && glds) CONV_LAUNCH_GLDS(2);
+    else if (bk64) { if (dbuf) CONV_LAUNCH(2, 64, true); else CONV_LAUNCH(2, 64, false); }
     else CONV_LAUNCH(2, 32, false);
   }
 #undef CONV_LAUNCH
+#undef CONV_LAUNCH_GLDS
   if (splitk > 1) {
     long total = (long)B * M * OC;
     int blocks = (int)min((long)2048, (total + 255) / 256);
