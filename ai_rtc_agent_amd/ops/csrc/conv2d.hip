@@ -277,8 +277,10 @@ __global__ __launch_bounds__(256) void conv2d_mfma_glds_kernel(
   constexpr int BK = 64;
   constexpr int BM = MFRAG * 32;
   constexpr int ALOADS = MFRAG;  // 16B-per-lane glds issues per thread (A)
-  __shared__ f16 ldsA[2 * BM * BK];
-  __shared__ f16 ldsB[2 * BN * BK];
+  constexpr int NBUF = 3;        // 3-deep ring: 2 tiles in flight across barriers
+  constexpr int PER_TILE = ALOADS + 2;  // glds instructions per wave per tile
+  __shared__ f16 ldsA[NBUF * BM * BK];
+  __shared__ f16 ldsB[NBUF * BN * BK];
 
   const int M = HO * WO;
   const int m0 = blockIdx.x * BM;
@@ -389,13 +391,32 @@ __global__ __launch_bounds__(256) void conv2d_mfma_glds_kernel(
     }
   };
 
+  // 3-buffer pipeline with COUNTED vmcnt + raw barriers (guide §5
+  // 'Pipelining across barriers'): one tile stays in flight across every
+  // barrier; __syncthreads() would drain it (hipcc emits vmcnt(0) inside),
+  // so the barrier is the raw s_barrier and the waits are hand-counted.
+  // Safety: each wave's counted vmcnt covers its OWN tile-t glds before the
+  // barrier; after the barrier every wave's tile t is complete. A buffer is
+  // re-issued 3 tiles later — one full barrier after its last ds_read
+  // (lgkmcnt(0) drains reads before the barrier).
   if (k_lo < k_hi) issue(0, k_lo);
-  __syncthreads();  // hipcc drains the in-flight glds here (vmcnt(0))
+  if (k_lo + 1 < k_hi) issue(1, k_lo + 1);
   for (int kt = k_lo; kt < k_hi; ++kt) {
-    const int cur = (kt - k_lo) & 1;
-    if (kt + 1 < k_hi) issue(cur ^ 1, kt + 1);
-    compute(cur);
-    __syncthreads();
+    if (kt + 1 < k_hi) {
+      // wait tile kt; leave tile kt+1 in flight
+      if constexpr (PER_TILE == 6)
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    // the barrier also separates iter t-1's reads (each wave drained its
+    // lgkm below before arriving) from this issue's overwrite of buf(t-1+3)
+    if (kt + 2 < k_hi) issue((kt + 2 - k_lo) % NBUF, kt + 2);
+    compute((kt - k_lo) % NBUF);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   }
 
   // epilogue (same as the register-staged kernel)
