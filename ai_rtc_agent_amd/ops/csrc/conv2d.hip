@@ -82,6 +82,22 @@ __device__ __forceinline__ unsigned row_swz(int row) {
   return (unsigned)((row >> 1) & 7);
 }
 
+// XCD-aware tile mapping (guide T1, bijective form): the dispatcher places
+// workgroup b on XCD b%8 with a PRIVATE L2 per XCD. The flat (mtile,ntile)
+// grid is remapped so each XCD walks a CONTIGUOUS id range, with the
+// OC-tile index fastest: the n-tiles sharing one activation panel run on
+// one XCD back-to-back, so the panel (e.g. 737 KB per 128-pixel m-tile)
+// stays L2-resident across its OC re-reads instead of re-fetching from
+// LLC/HBM (PMC showed the conv at 4-5% MfmaUtil, memory-stalled).
+__device__ __forceinline__ void xcd_tile_map(int wg, int nwg, int n_tiles,
+                                             int* mt, int* nt) {
+  const int x = wg & 7, pos = wg >> 3;
+  const int q = nwg >> 3, r = nwg & 7;
+  const int wgid = (x < r ? x * (q + 1) : r * (q + 1) + (x - r) * q) + pos;
+  *nt = wgid % n_tiles;
+  *mt = wgid / n_tiles;
+}
+
 // ---------------------------------------------------------------------------
 // templated MFMA conv: MFRAG = M-fragments per wave (4 -> BM=128, 2 -> BM=64)
 // BK = K-tile depth: 64 when IC%64==0 (all SD/TAESD layers — halves the
@@ -108,10 +124,19 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
   __shared__ f16 ldsB[NBUF * BN * KPITCH];
 
   const int M = HO * WO;
-  const int m0 = blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
-  const int b = blockIdx.z / splitk;
-  const int split = blockIdx.z - b * splitk;
+  const int n_tiles = ceil_div_dev(OC, BN);
+  int mt, nt;
+  if (splitk > 0)
+    xcd_tile_map(blockIdx.x, gridDim.x, n_tiles, &mt, &nt);
+  else {  // splitk<0 encodes "plain mapping" for A/B (|splitk| used below)
+    mt = blockIdx.x / n_tiles;
+    nt = blockIdx.x - mt * n_tiles;
+  }
+  const int spk = splitk > 0 ? splitk : -splitk;
+  const int m0 = mt * BM;
+  const int n0 = nt * BN;
+  const int b = blockIdx.z / spk;
+  const int split = blockIdx.z - b * spk;
   const f16* xb = x + (long)b * H * W * IC;
 
   const int tid = threadIdx.x;
@@ -149,7 +174,7 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
 
   // K-step range for this split
   const int nk = K / BK;
-  const int per = (nk + splitk - 1) / splitk;
+  const int per = (nk + spk - 1) / spk;
   const int k_lo = split * per;
   const int k_hi = min(nk, k_lo + per);
   // NOTE: an empty split (uneven tail) still stores its zero slab below —
@@ -230,7 +255,7 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
     }
   }
 
-  if (splitk == 1) {
+  if (spk == 1) {
     f16* ob = out + (long)b * M * OC;
     const long cb_off = (long)b * OC;
 #pragma unroll
@@ -251,7 +276,7 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
     }
   } else {
     // plain f32 slab store; finalize kernel reduces + applies the epilogue
-    float* wsb = ws + ((long)b * splitk + split) * M * OC;
+    float* wsb = ws + ((long)b * spk + split) * M * OC;
 #pragma unroll
     for (int ni = 0; ni < 2; ++ni) {
       const int col = n0 + wn * 32 + ni * 16 + (lane & 15);
@@ -283,10 +308,19 @@ __global__ __launch_bounds__(256) void conv2d_mfma_glds_kernel(
   __shared__ f16 ldsB[NBUF * BN * BK];
 
   const int M = HO * WO;
-  const int m0 = blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
-  const int b = blockIdx.z / splitk;
-  const int split = blockIdx.z - b * splitk;
+  const int n_tiles = ceil_div_dev(OC, BN);
+  int mt, nt;
+  if (splitk > 0)
+    xcd_tile_map(blockIdx.x, gridDim.x, n_tiles, &mt, &nt);
+  else {
+    mt = blockIdx.x / n_tiles;
+    nt = blockIdx.x - mt * n_tiles;
+  }
+  const int spk = splitk > 0 ? splitk : -splitk;
+  const int m0 = mt * BM;
+  const int n0 = nt * BN;
+  const int b = blockIdx.z / spk;
+  const int split = blockIdx.z - b * spk;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -333,7 +367,7 @@ __global__ __launch_bounds__(256) void conv2d_mfma_glds_kernel(
     for (int ni = 0; ni < 2; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
 
   const int nk = K / BK;
-  const int per = (nk + splitk - 1) / splitk;
+  const int per = (nk + spk - 1) / spk;
   const int k_lo = split * per;
   const int k_hi = min(nk, k_lo + per);
 
@@ -420,7 +454,7 @@ __global__ __launch_bounds__(256) void conv2d_mfma_glds_kernel(
   }
 
   // epilogue (same as the register-staged kernel)
-  if (splitk == 1) {
+  if (spk == 1) {
     f16* ob = out + (long)b * M * OC;
     const long cb_off = (long)b * OC;
 #pragma unroll
@@ -440,7 +474,7 @@ __global__ __launch_bounds__(256) void conv2d_mfma_glds_kernel(
         }
     }
   } else {
-    float* wsb = ws + ((long)b * splitk + split) * M * OC;
+    float* wsb = ws + ((long)b * spk + split) * M * OC;
 #pragma unroll
     for (int ni = 0; ni < 2; ++ni) {
       const int col = n0 + wn * 32 + ni * 16 + (lane & 15);
@@ -614,12 +648,12 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
 
   const int splitk = path > 0 ? path : -path;
   const int bm = path > 0 ? 128 : 64;
-  dim3 grid(ceil_div(M, bm), ceil_div(OC, BN), B * splitk);
+  dim3 grid(ceil_div(M, bm) * ceil_div(OC, BN), 1, B * splitk);
   const float* b1 = splitk == 1 ? bias : nullptr;
   const f16* cb1 = splitk == 1 ? cb : nullptr;
   const f16* res1 = splitk == 1 ? res : nullptr;
   const bool bk64 = (IC % 64 == 0);
-  static int dbuf = -1, glds = -1;
+  static int dbuf = -1, glds = -1, xcdmap = -1;
   if (dbuf < 0) {
     // Measured A/B on MI355X: the double-buffer single-barrier REGISTER
     // schedule is SLOWER here (114.8 vs 118.6 fps end-to-end) — the 2x LDS
@@ -629,15 +663,20 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
     dbuf = e ? atoi(e) : 0;
     const char* g = getenv("AIRTC_CONV_GLDS");
     glds = g ? atoi(g) : 0;
+    const char* xm = getenv("AIRTC_CONV_XCD");
+    xcdmap = xm ? atoi(xm) : 1;
   }
+  // splitk sign selects the tile mapping (positive = XCD-chunked, negative
+  // = plain) — kernels take |splitk| as the split factor
+  const int spk_arg = xcdmap ? splitk : -splitk;
 #define CONV_LAUNCH(MF, BKV, DB)                                              \
   hipLaunchKernelGGL((conv2d_mfma_kernel<MF, BKV, DB>), grid, dim3(256), 0,   \
                      s, xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC,  \
-                     R, S, stride, pad, act, K, splitk)
+                     R, S, stride, pad, act, K, spk_arg)
 #define CONV_LAUNCH_GLDS(MF)                                                  \
   hipLaunchKernelGGL((conv2d_mfma_glds_kernel<MF>), grid, dim3(256), 0, s,    \
                      xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC, R,  \
-                     S, stride, pad, act, K, splitk)
+                     S, stride, pad, act, K, spk_arg)
   if (path > 0) {
     if (bk64 && glds) CONV_LAUNCH_GLDS(4);
     else if (bk64) { if (dbuf) CONV_LAUNCH(4, 64, true); else CONV_LAUNCH(4, 64, false); }
