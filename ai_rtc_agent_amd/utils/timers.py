@@ -36,6 +36,13 @@ class _Percentile:
 
 
 class StageTimers:
+    """Per-stage timing + rocTX ranges.
+
+    On ROCm, torch.cuda.nvtx maps to rocTX: every stage() context also
+    emits a range so rocprofv3 --marker-trace groups kernels by pipeline
+    stage (SURVEY.md §5.1 — the reference has no tracing hooks at all).
+    """
+
     def __init__(self, use_cuda: bool = False):
         self.use_cuda = use_cuda and torch.cuda.is_available()
         self.stages: Dict[str, _Percentile] = {}
@@ -52,6 +59,7 @@ class StageTimers:
         def __enter__(self):
             p = self.parent
             if p.use_cuda:
+                torch.cuda.nvtx.range_push(f"airtc/{self.name}")  # rocTX
                 self.e0 = torch.cuda.Event(enable_timing=True)
                 self.e0.record()
             else:
@@ -63,6 +71,7 @@ class StageTimers:
             if p.use_cuda:
                 e1 = torch.cuda.Event(enable_timing=True)
                 e1.record()
+                torch.cuda.nvtx.range_pop()
                 p._pending.append((self.name, self.e0, e1))
             else:
                 p.stages.setdefault(self.name, _Percentile()).add(
