@@ -68,7 +68,7 @@ __device__ __forceinline__ f16x8 tr_bfrag(const f16* tile, int pitch,
   return cvt.f8;
 }
 
-template <int D>
+template <int D, int QF>  // QF = 16-row q-fragments per wave (1 or 2)
 __global__ __launch_bounds__(256) void attention_kernel(
     const f16* __restrict__ q, const f16* __restrict__ k,
     const f16* __restrict__ v, f16* __restrict__ out, int H, int Lq, int Lk,
@@ -76,13 +76,14 @@ __global__ __launch_bounds__(256) void attention_kernel(
     long o_sb, long o_sh, long o_row, float scale) {
   constexpr int KPITCH = D + 8;
   constexpr int D8 = D / 8, D32 = D / 32, D16 = D / 16;
+  constexpr int QTILE = 64 * QF;  // q rows per workgroup
   __shared__ f16 ldsK[KVT * KPITCH];
   __shared__ f16 ldsV[KVT * KPITCH];  // row-major like K; PV reads via tr_b16
-  __shared__ f16 ldsP[4 * 16 * PPITCH];
+  __shared__ f16 ldsP[4 * QF * 16 * PPITCH];
 
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
-  const int q0 = blockIdx.x * QT;
+  const int q0 = blockIdx.x * QTILE;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
@@ -93,23 +94,29 @@ __global__ __launch_bounds__(256) void attention_kernel(
   f16* ob = out + b * o_sb + h * o_sh;
 
   // ---- Q fragments in registers (one load, reused every kv tile) ----
-  f16x8 aq[D32];
-  {
-    int qrow = q0 + wid * 16 + (lane & 15);
+  f16x8 aq[QF][D32];
+#pragma unroll
+  for (int qi = 0; qi < QF; ++qi) {
+    int qrow = q0 + (wid * QF + qi) * 16 + (lane & 15);
     if (qrow >= Lq) qrow = Lq - 1;
     const f16* qr = qb + (long)qrow * q_row + (lane >> 4) * 8;
 #pragma unroll
     for (int i = 0; i < D32; ++i)
-      aq[i] = *reinterpret_cast<const f16x8*>(qr + i * 32);
+      aq[qi][i] = *reinterpret_cast<const f16x8*>(qr + i * 32);
   }
 
-  f32x4 o_acc[D16];
+  f32x4 o_acc[QF][D16];
 #pragma unroll
-  for (int i = 0; i < D16; ++i) o_acc[i] = {0.f, 0.f, 0.f, 0.f};
-  float m_i[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
-  float l_i[4] = {0.f, 0.f, 0.f, 0.f};
+  for (int qi = 0; qi < QF; ++qi)
+#pragma unroll
+    for (int i = 0; i < D16; ++i) o_acc[qi][i] = {0.f, 0.f, 0.f, 0.f};
+  float m_i[QF][4], l_i[QF][4];
+#pragma unroll
+  for (int qi = 0; qi < QF; ++qi)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) { m_i[qi][j] = -1e30f; l_i[qi][j] = 0.f; }
 
-  f16* myP = &ldsP[wid * 16 * PPITCH];
+  f16* myP = &ldsP[wid * QF * 16 * PPITCH];
   const int fcol = (lane >> 4) * 8;
 
   // T14 async-stage split: each thread owns LOADS_PT row-chunks of the K and
@@ -148,90 +155,101 @@ __global__ __launch_bounds__(256) void attention_kernel(
     __syncthreads();
     if (t0 + KVT < Lk) stage_load(t0 + KVT);
 
-    // ---- S = scale * Q K^T  (4 col fragments of 16) ----
-    f32x4 sfrag[4];
+    // ---- S = scale * Q K^T  (QF x 4 col fragments of 16) ----
+    f32x4 sfrag[QF][4];
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) {
-      sfrag[nf] = {0.f, 0.f, 0.f, 0.f};
       const f16* kr = &ldsK[(nf * 16 + (lane & 15)) * KPITCH + fcol];
 #pragma unroll
       for (int ds = 0; ds < D32; ++ds) {
         f16x8 bfrag = *reinterpret_cast<const f16x8*>(kr + ds * 32);
-        sfrag[nf] = mfma16x16x32(aq[ds], bfrag, sfrag[nf]);
+#pragma unroll
+        for (int qi = 0; qi < QF; ++qi) {
+          if (ds == 0) sfrag[qi][nf] = {0.f, 0.f, 0.f, 0.f};
+          sfrag[qi][nf] = mfma16x16x32(aq[qi][ds], bfrag, sfrag[qi][nf]);
+        }
       }
     }
 
-    // ---- online softmax update (rows lane-local) ----
-    float p[4][4];  // [nf][reg j]
-    float mnew[4];
+    // ---- online softmax update (rows lane-local), per q-fragment ----
 #pragma unroll
-    for (int j = 0; j < 4; ++j) mnew[j] = -1e30f;
+    for (int qi = 0; qi < QF; ++qi) {
+      float p[4][4];  // [nf][reg j]
+      float mnew[4];
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf) {
-      const int kcol = t0 + nf * 16 + (lane & 15);
-      const bool valid = kcol < Lk;
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        float val = valid ? sfrag[nf][j] * scale : -1e30f;
-        p[nf][j] = val;
-        mnew[j] = fmaxf(mnew[j], val);
-      }
-    }
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1)
-        mnew[j] = fmaxf(mnew[j], __shfl_xor(mnew[j], off, 64));
-      const float mn = fmaxf(m_i[j], mnew[j]);
-      const float alpha = __expf(m_i[j] - mn);
-      float rs = 0.f;
+      for (int j = 0; j < 4; ++j) mnew[j] = -1e30f;
 #pragma unroll
       for (int nf = 0; nf < 4; ++nf) {
-        p[nf][j] = __expf(p[nf][j] - mn);
-        rs += p[nf][j];
+        const int kcol = t0 + nf * 16 + (lane & 15);
+        const bool valid = kcol < Lk;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float val = valid ? sfrag[qi][nf][j] * scale : -1e30f;
+          p[nf][j] = val;
+          mnew[j] = fmaxf(mnew[j], val);
+        }
       }
 #pragma unroll
-      for (int off = 1; off < 16; off <<= 1) rs += __shfl_xor(rs, off, 64);
-      l_i[j] = l_i[j] * alpha + rs;
-      m_i[j] = mn;
+      for (int j = 0; j < 4; ++j) {
 #pragma unroll
-      for (int f = 0; f < D16; ++f) o_acc[f][j] *= alpha;
-    }
+        for (int off = 1; off < 16; off <<= 1)
+          mnew[j] = fmaxf(mnew[j], __shfl_xor(mnew[j], off, 64));
+        const float mn = fmaxf(m_i[qi][j], mnew[j]);
+        const float alpha = __expf(m_i[qi][j] - mn);
+        float rs = 0.f;
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf) {
+          p[nf][j] = __expf(p[nf][j] - mn);
+          rs += p[nf][j];
+        }
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) rs += __shfl_xor(rs, off, 64);
+        l_i[qi][j] = l_i[qi][j] * alpha + rs;
+        m_i[qi][j] = mn;
+#pragma unroll
+        for (int f = 0; f < D16; ++f) o_acc[qi][f][j] *= alpha;
+      }
 
-    // ---- P -> LDS (C-frag layout -> A-frag layout bridge) ----
+      // ---- P -> LDS (C-frag layout -> A-frag layout bridge) ----
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf)
+      for (int nf = 0; nf < 4; ++nf)
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
-        myP[((lane >> 4) * 4 + j) * PPITCH + nf * 16 + (lane & 15)] =
-            (f16)p[nf][j];
+        for (int j = 0; j < 4; ++j)
+          myP[(qi * 16 + (lane >> 4) * 4 + j) * PPITCH + nf * 16 + (lane & 15)] =
+              (f16)p[nf][j];
+    }
     // same-wave LDS write->read: wait for the writes, keep reads below
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
     // ---- O += P V  (V B-fragments via hardware transpose read) ----
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      f16x8 afrag = *reinterpret_cast<const f16x8*>(
-          &myP[(lane & 15) * PPITCH + ks * 32 + fcol]);
 #pragma unroll
       for (int f = 0; f < D16; ++f) {
         f16x8 bfrag = tr_bfrag(ldsV, KPITCH, ks * 32, f * 16, lane);
-        o_acc[f] = mfma16x16x32(afrag, bfrag, o_acc[f]);
+#pragma unroll
+        for (int qi = 0; qi < QF; ++qi) {
+          f16x8 afrag = *reinterpret_cast<const f16x8*>(
+              &myP[(qi * 16 + (lane & 15)) * PPITCH + ks * 32 + fcol]);
+          o_acc[qi][f] = mfma16x16x32(afrag, bfrag, o_acc[qi][f]);
+        }
       }
     }
   }
 
   // ---- epilogue: O /= l, masked stores ----
 #pragma unroll
-  for (int j = 0; j < 4; ++j) {
-    const int qrow = q0 + wid * 16 + (lane >> 4) * 4 + j;
-    if (qrow >= Lq) continue;
-    const float inv_l = 1.0f / l_i[j];
+  for (int qi = 0; qi < QF; ++qi)
 #pragma unroll
-    for (int f = 0; f < D16; ++f)
-      ob[(long)qrow * o_row + f * 16 + (lane & 15)] =
-          (f16)(o_acc[f][j] * inv_l);
-  }
+    for (int j = 0; j < 4; ++j) {
+      const int qrow = q0 + (wid * QF + qi) * 16 + (lane >> 4) * 4 + j;
+      if (qrow >= Lq) continue;
+      const float inv_l = 1.0f / l_i[qi][j];
+#pragma unroll
+      for (int f = 0; f < D16; ++f)
+        ob[(long)qrow * o_row + f * 16 + (lane & 15)] =
+            (f16)(o_acc[qi][f][j] * inv_l);
+    }
 }
 
 extern "C" void airtc_attention(const uint16_t* q, const uint16_t* k,
@@ -240,21 +258,30 @@ extern "C" void airtc_attention(const uint16_t* q, const uint16_t* k,
                                 long q_row, long k_sb, long k_sh, long k_row,
                                 long o_sb, long o_sh, long o_row, float scale,
                                 hipStream_t s) {
-  dim3 grid(ceil_div(Lq, QT), B * H);
+  // QF=2 (128-row q-tiles) halves the K/V re-read traffic — the dominant
+  // cost for long self-attention (each q-tile block streams the whole K/V);
+  // keep QF=1 for short Lq so the grid still fills the chip.
+  const int qf = (Lq >= 2048 && Lk >= 1024) ? 2 : 1;
+  dim3 grid(ceil_div(Lq, 64 * qf), B * H);
   const f16* qp = reinterpret_cast<const f16*>(q);
   const f16* kp = reinterpret_cast<const f16*>(k);
   const f16* vp = reinterpret_cast<const f16*>(v);
   f16* op = reinterpret_cast<f16*>(out);
-#define LAUNCH(D)                                                           \
-  hipLaunchKernelGGL(attention_kernel<D>, grid, dim3(256), 0, s, qp, kp,    \
-                     vp, op, H, Lq, Lk, q_sb, q_sh, q_row, k_sb, k_sh,      \
+#define LAUNCH(D, QF)                                                       \
+  hipLaunchKernelGGL((attention_kernel<D, QF>), grid, dim3(256), 0, s, qp,  \
+                     kp, vp, op, H, Lq, Lk, q_sb, q_sh, q_row, k_sb, k_sh,  \
                      k_row, o_sb, o_sh, o_row, scale)
-  switch (d) {
-    case 32: LAUNCH(32); break;
-    case 64: LAUNCH(64); break;
-    case 96: LAUNCH(96); break;
-    case 128: LAUNCH(128); break;
-    case 160: LAUNCH(160); break;
+  switch (d * 10 + qf) {
+    case 321: LAUNCH(32, 1); break;
+    case 322: LAUNCH(32, 2); break;
+    case 641: LAUNCH(64, 1); break;
+    case 642: LAUNCH(64, 2); break;
+    case 961: LAUNCH(96, 1); break;
+    case 962: LAUNCH(96, 2); break;
+    case 1281: LAUNCH(128, 1); break;
+    case 1282: LAUNCH(128, 2); break;
+    case 1601: LAUNCH(160, 1); break;
+    case 1602: LAUNCH(160, 2); break;
     default: break;  // host wrapper guarantees one of the above
   }
 #undef LAUNCH
