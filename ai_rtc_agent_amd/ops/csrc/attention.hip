@@ -24,6 +24,8 @@
 // Addressing: q/k/v/out are (b, h)-strided views (no host-side copies for
 // d%32==0 models; ops/interface.py pads other head sizes).
 
+#include <stdlib.h>
+
 #include "common.h"
 
 #define KVT 64  // kv tile
@@ -261,7 +263,13 @@ extern "C" void airtc_attention(const uint16_t* q, const uint16_t* k,
   // QF=2 (128-row q-tiles) halves the K/V re-read traffic — the dominant
   // cost for long self-attention (each q-tile block streams the whole K/V);
   // keep QF=1 for short Lq so the grid still fills the chip.
-  const int qf = (Lq >= 2048 && Lk >= 1024) ? 2 : 1;
+  static int qf_force = -2;
+  if (qf_force == -2) {
+    const char* e = getenv("AIRTC_ATTN_QF");
+    qf_force = e ? atoi(e) : 0;
+  }
+  int qf = (Lq >= 2048 && Lk >= 1024) ? 2 : 1;
+  if (qf_force > 0) qf = qf_force;
   dim3 grid(ceil_div(Lq, 64 * qf), B * H);
   const f16* qp = reinterpret_cast<const f16*>(q);
   const f16* kp = reinterpret_cast<const f16*>(k);
