@@ -268,7 +268,11 @@ extern "C" void airtc_attention(const uint16_t* q, const uint16_t* k,
     const char* e = getenv("AIRTC_ATTN_QF");
     qf_force = e ? atoi(e) : 0;
   }
-  int qf = (Lq >= 2048 && Lk >= 1024) ? 2 : 1;
+  // QF=2 measured SLOWER end-to-end (112.4 vs 118.7 fps same-box A/B):
+  // the doubled accumulator/S-fragment footprint costs more occupancy than
+  // the halved K/V traffic saves (K/V is LLC-resident at SD sizes). QF=1
+  // stays the default; the template is kept for larger-context models.
+  int qf = 1;
   if (qf_force > 0) qf = qf_force;
   dim3 grid(ceil_div(Lq, 64 * qf), B * H);
   const f16* qp = reinterpret_cast<const f16*>(q);
