@@ -121,3 +121,21 @@ def test_txt2img_gpu():
     e.prepare()
     out = e.txt2img()
     assert out.shape == (1, 512, 512, 3)
+
+
+def test_agent_serving_path_on_gpu():
+    """The serving stack (PipelinePool -> StreamDiffusionPipeline -> engine)
+    on a real GPU: what the agent runs per frame."""
+    from ai_rtc_agent_amd.config import sd_turbo_config
+    from ai_rtc_agent_amd.parallel.dispatch import PipelinePool
+
+    cfg = sd_turbo_config(device="cuda")
+    pool = PipelinePool.create(model_id="stabilityai/sd-turbo", n_gpus=1, cfg=cfg)
+    p = pool.assign("stream-1")
+    f = frame(seed=77)
+    out = p(f)
+    assert out.shape == (512, 512, 3) and out.is_cuda
+    stats = pool.stats()
+    assert stats["replicas"] == 1
+    assert stats["per_replica"][0]["frames"] >= 1
+    pool.release("stream-1")
