@@ -112,3 +112,15 @@ def test_stats_surface(tiny_cfg):
     s = e.stats()
     assert s["frames"] == 1
     assert "diffusion" in s["stages_ms"]
+
+
+def test_non_square_resolution():
+    from ai_rtc_agent_amd.config import EngineConfig
+    cfg = EngineConfig(
+        model_family="tiny", width=96, height=64, device="cpu",
+        use_hip_graph=False, use_lcm_lora=False, t_index_list=[0, 25],
+    )
+    e = StreamDiffusionEngine(cfg)
+    e.prepare()
+    out = e(torch.randint(0, 256, (64, 96, 3), dtype=torch.uint8))
+    assert out.shape == (64, 96, 3)
