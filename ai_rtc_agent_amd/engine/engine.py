@@ -189,9 +189,9 @@ class StreamDiffusionEngine:
         self._x_t_buffer = torch.zeros((max(0, B - fbs), lh, lw, 4), device=dev, dtype=dt)
         self.rcfg.reset(self._init_noise)
 
-        # static I/O buffers (graph-stable addresses)
+        # static I/O buffers (graph-stable addresses); _frame_in (u8) is the
+        # only per-frame input — preprocess lives inside the graph
         self._frame_in = torch.zeros((fbs, cfg.height, cfg.width, 3), device=dev, dtype=torch.uint8)
-        self._img_in = torch.zeros((fbs, cfg.height, cfg.width, 3), device=dev, dtype=dt)
         self._ts_batch = self._coeff["sub_timesteps_tensor"].to(dev)
         self._prev_out: Optional[torch.Tensor] = None
         self._graph = None

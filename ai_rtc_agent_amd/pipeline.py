@@ -28,7 +28,17 @@ class StreamDiffusionPipeline:
                 cfg.device = "cpu"
                 cfg.use_hip_graph = False
         self.cfg = cfg
-        self.engine = StreamDiffusionEngine(cfg)
+        # load-else-build ladder (reference lib/wrapper.py:611-615): a
+        # pre-built engine plan (python build.py) skips model init + LoRA
+        # fusion; otherwise build fresh.
+        from .engine.plan import load_plan
+
+        eng = None
+        try:
+            eng = load_plan(cfg)
+        except Exception:
+            eng = None
+        self.engine = eng if eng is not None else StreamDiffusionEngine(cfg)
         self.engine.prepare(
             prompt=cfg.prompt,
             num_inference_steps=cfg.num_inference_steps,

@@ -72,3 +72,21 @@ def test_download_offline(tmp_path, monkeypatch):
     from ai_rtc_agent_amd.utils.paths import civitai_model_path
 
     assert os.path.exists(civitai_model_path(dl.CIVITAI_MODEL_ID, dl.CIVITAI_VERSION_ID))
+
+
+def test_pipeline_uses_plan_cache(tmp_path, monkeypatch):
+    """Serving path picks up a pre-built plan (load-else-build ladder)."""
+    monkeypatch.setenv("ENGINES_CACHE", str(tmp_path))
+    import build as build_mod
+
+    build_mod.build(model_id="test/pipe-tiny", family="tiny", width=64)
+    from ai_rtc_agent_amd.config import EngineConfig
+    from ai_rtc_agent_amd.pipeline import StreamDiffusionPipeline
+
+    cfg = EngineConfig(
+        model_id="test/pipe-tiny", model_family="tiny", width=64, height=64,
+        device="cpu", use_hip_graph=False, use_lcm_lora=False,
+    )
+    p = StreamDiffusionPipeline(cfg=cfg)
+    out = p(torch.randint(0, 256, (64, 64, 3), dtype=torch.uint8))
+    assert out.shape == (64, 64, 3)
