@@ -490,6 +490,112 @@ __global__ __launch_bounds__(256) void conv2d_mfma_glds_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Halo-tiled 3x3 stride-1 conv: the im2col kernels above re-read every
+// activation element ~9x through L2 (the 3x3 gather) — measured as the
+// remaining conv bottleneck (TAESD 512²x64ch conv: 302 TF-equivalent but
+// ~5x the minimal activation traffic). Here each block owns an 8x8 output
+// tile and stages the 10x10 input PATCH in LDS once per 64-deep IC chunk;
+// the nine (r,s) taps re-read it from LDS. Weights stream straight from
+// global into register B-fragments (every block reads the same lines ->
+// L2-hot), so the only barriers are the IC/64 patch swaps.
+// Block: 64 px x 64 oc, 4 waves (2 px x 2 oc), wave = 32px x 32oc.
+// ---------------------------------------------------------------------------
+#define PPITCH_C 72  // patch channel pitch (f16): 64 + 8 pad (bank spread)
+
+__global__ __launch_bounds__(256) void conv3x3_tiled_kernel(
+    const f16* __restrict__ x, const f16* __restrict__ w,
+    const float* __restrict__ bias, const f16* __restrict__ cbias,
+    const f16* __restrict__ residual, f16* __restrict__ out, int H, int W,
+    int IC, int OC, int act, int K) {
+  __shared__ f16 patch[10 * 10 * PPITCH_C];
+
+  const int px_tiles_w = W >> 3;
+  const int pt = blockIdx.x;
+  const int tr0 = (pt / px_tiles_w) << 3;
+  const int tc0 = (pt - (pt / px_tiles_w) * px_tiles_w) << 3;
+  const int n0 = blockIdx.y * BN;
+  const long b = blockIdx.z;
+  const f16* xb = x + b * (long)H * W * IC;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid & 1;   // px half (32 px)
+  const int wn = wid >> 1;  // oc half (32 oc)
+
+  // patch staging map: 10*10 cells x 8 ic-octets = 800 16B loads
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nic = IC >> 6;  // 64-deep ic chunks
+  for (int icc = 0; icc < nic; ++icc) {
+    if (icc) __syncthreads();  // previous chunk's reads complete
+    const int ic0 = icc << 6;
+    for (int i = tid; i < 800; i += 256) {
+      const int cell = i >> 3, icq = i & 7;
+      const int pr = cell / 10, pc = cell - pr * 10;
+      const int hi = tr0 + pr - 1, wi = tc0 + pc - 1;
+      const bool ok = (unsigned)hi < (unsigned)H && (unsigned)wi < (unsigned)W;
+      f16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (ok)
+        v = *reinterpret_cast<const f16x8*>(
+            &xb[((long)hi * W + wi) * IC + ic0 + icq * 8]);
+      *reinterpret_cast<f16x8*>(&patch[cell * PPITCH_C + icq * 8]) = v;
+    }
+    __syncthreads();
+
+    // 9 taps x 2 K-chunks of 32 over this ic chunk
+#pragma unroll
+    for (int r = 0; r < 3; ++r)
+#pragma unroll
+      for (int sc = 0; sc < 3; ++sc) {
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          const int kglob = (r * 3 + sc) * IC + ic0 + kk * 32 + (lane >> 4) * 8;
+          f16x8 bfrag[2];
+#pragma unroll
+          for (int ni = 0; ni < 2; ++ni) {
+            const int oc = min(n0 + wn * 32 + ni * 16 + (lane & 15), OC - 1);
+            bfrag[ni] = *reinterpret_cast<const f16x8*>(&w[(long)oc * K + kglob]);
+          }
+#pragma unroll
+          for (int mi = 0; mi < 2; ++mi) {
+            const int m = wm * 32 + mi * 16 + (lane & 15);
+            const int pr = (m >> 3) + r, pc = (m & 7) + sc;
+            f16x8 afrag = *reinterpret_cast<const f16x8*>(
+                &patch[(pr * 10 + pc) * PPITCH_C + kk * 32 + (lane >> 4) * 8]);
+#pragma unroll
+            for (int ni = 0; ni < 2; ++ni)
+              acc[mi][ni] = mfma16x16x32(afrag, bfrag[ni], acc[mi][ni]);
+          }
+        }
+      }
+  }
+
+  // fused epilogue; output rows = this tile's pixels
+  f16* ob = out + b * (long)H * W * OC;
+  const long cb_off = b * OC;
+#pragma unroll
+  for (int ni = 0; ni < 2; ++ni) {
+    const int col = n0 + wn * 32 + ni * 16 + (lane & 15);
+    if (col >= OC) continue;
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int m = wm * 32 + mi * 16 + (lane >> 4) * 4 + j;
+        const int ho = tr0 + (m >> 3), wo = tc0 + (m & 7);
+        const long idx = (b * (long)H * W + (long)ho * W + wo) * OC + col;
+        ob[((long)ho * W + wo) * OC + col] =
+            epilogue(acc[mi][ni][j], bias, cbias, cb_off, residual, idx, col, act);
+      }
+  }
+}
+
 __global__ void conv_splitk_finalize(const float* __restrict__ ws,
                                      const float* __restrict__ bias,
                                      const f16* __restrict__ cbias,
@@ -646,6 +752,18 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
   f16* op = reinterpret_cast<f16*>(out);
   const int M = HO * WO;
 
+  static int tiled = -1;
+  if (tiled < 0) {
+    const char* t = getenv("AIRTC_CONV_TILED");
+    tiled = t ? atoi(t) : 1;
+  }
+  if (tiled && R == 3 && S == 3 && stride == 1 && pad == 1 && IC % 64 == 0 &&
+      (H & 7) == 0 && (W & 7) == 0 && M >= 4096) {
+    dim3 tgrid((H >> 3) * (W >> 3), ceil_div(OC, BN), B);
+    hipLaunchKernelGGL(conv3x3_tiled_kernel, tgrid, dim3(256), 0, s, xp, wp,
+                       bias, cb, res, op, H, W, IC, OC, act, K);
+    return;
+  }
   const int splitk = path > 0 ? path : -path;
   const int bm = path > 0 ? 128 : 64;
   dim3 grid(ceil_div(M, bm) * ceil_div(OC, BN), 1, B * splitk);
