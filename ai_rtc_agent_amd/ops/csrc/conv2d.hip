@@ -548,32 +548,44 @@ __global__ __launch_bounds__(256) void conv3x3_tiled_kernel(
     }
     __syncthreads();
 
-    // 9 taps x 2 K-chunks of 32 over this ic chunk
+    // 18 (tap, K-chunk) steps over this ic chunk; B-fragments are
+    // software-pipelined one step ahead so the L2 load latency hides under
+    // the current step's MFMAs
+    const long ocrow0 = (long)min(n0 + wn * 32 + (lane & 15), OC - 1) * K;
+    const long ocrow1 = (long)min(n0 + wn * 32 + 16 + (lane & 15), OC - 1) * K;
+    const int klane = (lane >> 4) * 8;
+    auto kglob_at = [&](int step) {
+      // step = (r*3+sc)*2 + kk
+      return (step >> 1) * IC + ic0 + (step & 1) * 32 + klane;
+    };
+    f16x8 bcur[2], bnext[2];
+    {
+      const int kg = kglob_at(0);
+      bcur[0] = *reinterpret_cast<const f16x8*>(&w[ocrow0 + kg]);
+      bcur[1] = *reinterpret_cast<const f16x8*>(&w[ocrow1 + kg]);
+    }
 #pragma unroll
-    for (int r = 0; r < 3; ++r)
-#pragma unroll
-      for (int sc = 0; sc < 3; ++sc) {
-#pragma unroll
-        for (int kk = 0; kk < 2; ++kk) {
-          const int kglob = (r * 3 + sc) * IC + ic0 + kk * 32 + (lane >> 4) * 8;
-          f16x8 bfrag[2];
-#pragma unroll
-          for (int ni = 0; ni < 2; ++ni) {
-            const int oc = min(n0 + wn * 32 + ni * 16 + (lane & 15), OC - 1);
-            bfrag[ni] = *reinterpret_cast<const f16x8*>(&w[(long)oc * K + kglob]);
-          }
-#pragma unroll
-          for (int mi = 0; mi < 2; ++mi) {
-            const int m = wm * 32 + mi * 16 + (lane & 15);
-            const int pr = (m >> 3) + r, pc = (m & 7) + sc;
-            f16x8 afrag = *reinterpret_cast<const f16x8*>(
-                &patch[(pr * 10 + pc) * PPITCH_C + kk * 32 + (lane >> 4) * 8]);
-#pragma unroll
-            for (int ni = 0; ni < 2; ++ni)
-              acc[mi][ni] = mfma16x16x32(afrag, bfrag[ni], acc[mi][ni]);
-          }
-        }
+    for (int step = 0; step < 18; ++step) {
+      if (step + 1 < 18) {
+        const int kg = kglob_at(step + 1);
+        bnext[0] = *reinterpret_cast<const f16x8*>(&w[ocrow0 + kg]);
+        bnext[1] = *reinterpret_cast<const f16x8*>(&w[ocrow1 + kg]);
       }
+      const int rs = step >> 1, kk = step & 1;
+      const int r = rs / 3, sc = rs - (rs / 3) * 3;
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        const int m = wm * 32 + mi * 16 + (lane & 15);
+        const int pr = (m >> 3) + r, pc = (m & 7) + sc;
+        f16x8 afrag = *reinterpret_cast<const f16x8*>(
+            &patch[(pr * 10 + pc) * PPITCH_C + kk * 32 + klane]);
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = mfma16x16x32(afrag, bcur[ni], acc[mi][ni]);
+      }
+      bcur[0] = bnext[0];
+      bcur[1] = bnext[1];
+    }
   }
 
   // fused epilogue; output rows = this tile's pixels
