@@ -766,8 +766,12 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
 
   static int tiled = -1;
   if (tiled < 0) {
+    // Measured SLOWER than im2col (109.5-110.4 vs 119.2-119.6 fps, two
+    // same-box A/Bs, with and without B-prefetch): the 3x3 activation
+    // re-gather is already L2-absorbed, while per-wave B-fragment loads
+    // cost 4x the issue of cooperative LDS staging. Kept for experiments.
     const char* t = getenv("AIRTC_CONV_TILED");
-    tiled = t ? atoi(t) : 1;
+    tiled = t ? atoi(t) : 0;
   }
   if (tiled && R == 3 && S == 3 && stride == 1 && pad == 1 && IC % 64 == 0 &&
       (H & 7) == 0 && (W & 7) == 0 && M >= 4096) {
