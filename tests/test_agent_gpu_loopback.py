@@ -45,9 +45,19 @@ def test_agent_gpu_loopback(monkeypatch):
 
         codec = SoftwareCodec()
         pkz = RtpPacketizer(ssrc=42)
-        g = torch.Generator().manual_seed(0)
-        frames = [torch.randint(0, 256, (512, 512, 3), generator=g, dtype=torch.uint8)
-                  for _ in range(3)]
+        # compressible frames (gradients): random 512² noise is ~700 KB of
+        # zlib output = a 600-packet UDP burst that overflows loopback
+        # socket buffers; real camera frames compress, synthetic ones must too
+        base = torch.arange(512, dtype=torch.uint8).view(1, 512, 1).expand(512, 512, 3)
+        frames = [(base.int() + 13 * i).clamp(0, 255).to(torch.uint8).contiguous()
+                  for i in range(3)]
+
+        async def send_frame(i):
+            pkts = pkz.packetize(codec.encode(frames[i % 3]), timestamp=i * 3000)
+            for j, pkt in enumerate(pkts):
+                pub_t.sendto(pkt.serialize(), ("127.0.0.1", srv_port))
+                if j % 40 == 39:
+                    await asyncio.sleep(0.005)  # pace the burst
 
         sub_t, sub_p = await loop.create_datagram_endpoint(
             _ClientProto, local_addr=("127.0.0.1", 0))
@@ -55,8 +65,7 @@ def test_agent_gpu_loopback(monkeypatch):
 
         # publish a few frames first so the source track registers
         for i in range(3):
-            for pkt in pkz.packetize(codec.encode(frames[i % 3]), timestamp=i * 3000):
-                pub_t.sendto(pkt.serialize(), ("127.0.0.1", srv_port))
+            await send_frame(i)
             await asyncio.sleep(0.05)
 
         r2 = await http.post("/whep", data=_offer_sdp(sub_port),
@@ -65,8 +74,7 @@ def test_agent_gpu_loopback(monkeypatch):
 
         got = None
         for i in range(3, 120):
-            for pkt in pkz.packetize(codec.encode(frames[i % 3]), timestamp=i * 3000):
-                pub_t.sendto(pkt.serialize(), ("127.0.0.1", srv_port))
+            await send_frame(i)
             try:
                 got = await asyncio.wait_for(sub_p.frames.get(), timeout=0.5)
                 break
