@@ -157,8 +157,15 @@ class PeerConnection:
                     continue
                 data = self._encoder.encode(frame.tensor, keyframe=counter == 0)
                 ts = frame.pts if frame.pts else counter * 3000
-                for pkt in self._packetizer.packetize(data, ts):
+                # paced send: large frames fragment into hundreds of
+                # datagrams; an unpaced burst overflows receiver socket
+                # buffers (and starves the event loop's read side)
+                for j, pkt in enumerate(self._packetizer.packetize(data, ts)):
                     self._transport.sendto(pkt.serialize(), self._remote_addr)
+                    if j % 32 == 31:
+                        await asyncio.sleep(0)
+                    if j % 256 == 255:
+                        await asyncio.sleep(0.002)
                 counter += 1
         except asyncio.CancelledError:
             pass
