@@ -255,5 +255,7 @@ class MediaRelay:
                 frame = await track.recv()
                 for sub in self._subscribers.get(key, []):
                     sub.push(frame)
-        except (asyncio.CancelledError, Exception):
+        except asyncio.CancelledError:
             pass
+        except Exception:
+            logger.exception("media relay pump failed")
