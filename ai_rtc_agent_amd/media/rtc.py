@@ -38,6 +38,28 @@ logger = logging.getLogger(__name__)
 
 CONFIG_MAGIC = b"AIRC"  # JSON config datagrams (datachannel-lite)
 
+MEDIA_SOCKET_BUF = 8 << 20  # 8 MiB: a 512p frame fragments to ~700 KB of RTP
+
+
+def tune_socket_buffers(transport) -> None:
+    """Large SO_RCVBUF/SO_SNDBUF on a datagram transport: one video frame
+    bursts hundreds of datagrams, far beyond the 212 KB Linux default (the
+    kernel silently clamps plain setsockopt to rmem_max, so try the FORCE
+    variants first — we run as root in the container)."""
+    sock = transport.get_extra_info("socket")
+    if sock is None:
+        return
+    for opt_force, opt in ((33, socket.SO_RCVBUF), (32, socket.SO_SNDBUF)):
+        # SO_RCVBUFFORCE=33, SO_SNDBUFFORCE=32 (linux)
+        try:
+            sock.setsockopt(socket.SOL_SOCKET, opt_force, MEDIA_SOCKET_BUF)
+        except (OSError, PermissionError):
+            try:
+                sock.setsockopt(socket.SOL_SOCKET, opt, MEDIA_SOCKET_BUF)
+            except OSError:
+                pass
+
+
 _port_pool: Optional[List[int]] = None
 
 
@@ -134,6 +156,7 @@ class PeerConnection:
                 self._transport, _ = await loop.create_datagram_endpoint(
                     lambda: _Proto(self), local_addr=("0.0.0.0", p)
                 )
+                tune_socket_buffers(self._transport)
                 self.port = self._transport.get_extra_info("sockname")[1]
                 return
             except OSError as e:  # port in use: walk the pool

@@ -99,7 +99,10 @@ class _ClientProto(asyncio.DatagramProtocol):
         self.codec = SoftwareCodec()
 
     def connection_made(self, transport):
+        from ai_rtc_agent_amd.media.rtc import tune_socket_buffers
+
         self.transport = transport
+        tune_socket_buffers(transport)
 
     def datagram_received(self, data, addr):
         if stun.is_stun(data):
