@@ -201,3 +201,124 @@ def test_loopback_whip_whep(monkeypatch):
         await http.close()
 
     run(body(), timeout=50)
+
+
+@pytest.mark.timeout(60)
+def test_multi_viewer_whep_fanout(monkeypatch):
+    """Two WHEP subscribers share one pipeline pull via the relay."""
+    monkeypatch.setenv("WARMUP_FRAMES", "0")
+    monkeypatch.setenv("DROP_FRAMES", "0")
+
+    async def body():
+        from aiohttp.test_utils import TestClient, TestServer
+
+        from ai_rtc_agent_amd.agent import create_app
+        from ai_rtc_agent_amd.parallel.dispatch import PipelinePool
+
+        calls = {"n": 0}
+
+        def stylize(t):
+            calls["n"] += 1
+            return (t.int() + 10).clamp(0, 255).to(torch.uint8)
+
+        app = create_app(pool=PipelinePool.single(stylize), use_turn=False)
+        http = TestClient(TestServer(app))
+        await http.start_server()
+        loop = asyncio.get_event_loop()
+
+        pub_t, _ = await loop.create_datagram_endpoint(
+            _ClientProto, local_addr=("127.0.0.1", 0))
+        pub_port = pub_t.get_extra_info("sockname")[1]
+        r = await http.post("/whip", data=_offer_sdp(pub_port),
+                            headers={"Content-Type": "application/sdp"})
+        srv_port = SessionDescription.parse(await r.text()).media[0].port
+        pub_t.sendto(stun.make_binding_request("u:p", b"k"), ("127.0.0.1", srv_port))
+        await asyncio.sleep(0.05)
+
+        codec = SoftwareCodec()
+        pkz = RtpPacketizer(ssrc=7)
+        g = torch.Generator().manual_seed(1)
+        frames = [torch.randint(0, 200, (16, 16, 3), generator=g, dtype=torch.uint8)
+                  for _ in range(4)]
+        for i in range(4):
+            for pkt in pkz.packetize(codec.encode(frames[i]), timestamp=i * 3000):
+                pub_t.sendto(pkt.serialize(), ("127.0.0.1", srv_port))
+            await asyncio.sleep(0.02)
+
+        subs = []
+        for _ in range(2):
+            t, p = await loop.create_datagram_endpoint(
+                _ClientProto, local_addr=("127.0.0.1", 0))
+            port = t.get_extra_info("sockname")[1]
+            r2 = await http.post("/whep", data=_offer_sdp(port),
+                                 headers={"Content-Type": "application/sdp"})
+            assert r2.status == 201
+            subs.append((t, p))
+
+        got = [None, None]
+        for i in range(4, 60):
+            for pkt in pkz.packetize(codec.encode(frames[i % 4]), timestamp=i * 3000):
+                pub_t.sendto(pkt.serialize(), ("127.0.0.1", srv_port))
+            for si, (_, p) in enumerate(subs):
+                if got[si] is None:
+                    try:
+                        got[si] = await asyncio.wait_for(p.frames.get(), timeout=0.1)
+                    except asyncio.TimeoutError:
+                        pass
+            if all(g is not None for g in got):
+                break
+        assert all(g is not None for g in got), "both viewers must receive frames"
+        # fan-out means ONE pipeline invocation per source frame, not one per viewer
+        assert calls["n"] <= 70
+
+        pub_t.close()
+        for t, _ in subs:
+            t.close()
+        await http.close()
+
+    run(body(), timeout=50)
+
+
+@pytest.mark.timeout(60)
+def test_datachannel_config_over_media_socket(monkeypatch):
+    """In-band config updates: AIRC-magic JSON datagrams on the media port
+    (parity with the reference's datachannel config, agent.py:154-168)."""
+    monkeypatch.setenv("WARMUP_FRAMES", "0")
+
+    async def body():
+        import json
+
+        from aiohttp.test_utils import TestClient, TestServer
+
+        from ai_rtc_agent_amd.agent import create_app
+        from ai_rtc_agent_amd.media.rtc import CONFIG_MAGIC
+        from ai_rtc_agent_amd.parallel.dispatch import PipelinePool
+        from tests.test_agent_http import StubPipeline
+
+        stub = StubPipeline()
+        app = create_app(pool=PipelinePool.single(stub), use_turn=False)
+        http = TestClient(TestServer(app))
+        await http.start_server()
+        loop = asyncio.get_event_loop()
+
+        pub_t, _ = await loop.create_datagram_endpoint(
+            _ClientProto, local_addr=("127.0.0.1", 0))
+        pub_port = pub_t.get_extra_info("sockname")[1]
+        r = await http.post("/whip", data=_offer_sdp(pub_port),
+                            headers={"Content-Type": "application/sdp"})
+        srv_port = SessionDescription.parse(await r.text()).media[0].port
+
+        msg = CONFIG_MAGIC + json.dumps(
+            {"prompt": "via datachannel", "t_index_list": [2, 4]}
+        ).encode()
+        pub_t.sendto(msg, ("127.0.0.1", srv_port))
+        for _ in range(50):
+            await asyncio.sleep(0.02)
+            if stub.prompt == "via datachannel":
+                break
+        assert stub.prompt == "via datachannel"
+        assert stub.t_index == [2, 4]
+        pub_t.close()
+        await http.close()
+
+    run(body(), timeout=50)
