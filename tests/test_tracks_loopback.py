@@ -256,13 +256,13 @@ def test_multi_viewer_whep_fanout(monkeypatch):
             subs.append((t, p))
 
         got = [None, None]
-        for i in range(4, 60):
+        for i in range(4, 120):
             for pkt in pkz.packetize(codec.encode(frames[i % 4]), timestamp=i * 3000):
                 pub_t.sendto(pkt.serialize(), ("127.0.0.1", srv_port))
             for si, (_, p) in enumerate(subs):
                 if got[si] is None:
                     try:
-                        got[si] = await asyncio.wait_for(p.frames.get(), timeout=0.1)
+                        got[si] = await asyncio.wait_for(p.frames.get(), timeout=0.15)
                     except asyncio.TimeoutError:
                         pass
             if all(g is not None for g in got):
