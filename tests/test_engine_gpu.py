@@ -139,3 +139,24 @@ def test_agent_serving_path_on_gpu():
     assert stats["replicas"] == 1
     assert stats["per_replica"][0]["frames"] >= 1
     pool.release("stream-1")
+
+
+def test_t_index_update_and_recapture_on_gpu():
+    """Runtime t_index updates under graph mode: same-length updates write
+    coefficient buffers in place (no re-capture); a length change triggers
+    a clean re-prepare + re-capture (reference lib/wrapper.py:389-407)."""
+    cfg = EngineConfig(
+        device="cuda", model_family="sd21", model_id="stabilityai/sd-turbo",
+        t_index_list=[10, 30], num_inference_steps=50, cfg_type="none",
+        use_lcm_lora=False,
+    )
+    e = StreamDiffusionEngine(cfg)
+    e.prepare()
+    f = frame(seed=9)
+    base = _run_clone(e, f)
+    e.update_t_index_list([5, 45])  # same length: in-place, graph kept
+    after = _run_clone(e, f)
+    assert not torch.equal(base, after)
+    e.update_t_index_list([0])      # length change: full re-prepare
+    out = _run_clone(e, f)
+    assert out.shape == (512, 512, 3)
