@@ -21,7 +21,6 @@ graph-external buffers in place; the graph is never re-captured
 """
 from __future__ import annotations
 
-import time
 from typing import List, Optional, Sequence
 
 import torch
@@ -437,7 +436,6 @@ class StreamDiffusionEngine:
         squeeze = frame_u8.dim() == 3
         if squeeze:
             frame_u8 = frame_u8.unsqueeze(0)
-        t_in = time.perf_counter()
 
         with self.timers.stage("preprocess"):
             frame_u8 = frame_u8.to(self.device, non_blocking=True)
