@@ -214,6 +214,39 @@ class StreamDiffusionEngine:
         self._added_cond.copy_(vec.expand_as(self._added_cond))
 
     # ------------------------------------------------------------------
+    # weight hot-swap (LoRA) — SURVEY.md §5.8 / N12: fuse, invalidate the
+    # kernel-side weight caches, re-broadcast over RCCL when distributed,
+    # and drop the hipGraphs (re-captured on the next frame)
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def load_lora(self, path_or_sd, scale: float = 1.0) -> int:
+        sd = load_lora_file(path_or_sd) if isinstance(path_or_sd, str) else path_or_sd
+        n = fuse_lora_state_dict(self.unet, sd, scale=scale)
+        self.refresh_weights()
+        return n
+
+    @torch.no_grad()
+    def refresh_weights(self) -> None:
+        """After any in-place weight change: drop the lazily-built GPU weight
+        transforms (conv GEMM layouts, fused QKV, f32 bias copies), re-sync
+        replicas, and force a graph re-capture."""
+        for module in (self.unet, self.vae, self.controlnet):
+            if module is None:
+                continue
+            for m in module.modules():
+                for attr in ("_wqkv", "_wkv", "_wq", "_wout"):
+                    if hasattr(m, attr):
+                        delattr(m, attr)
+                for t in list(m.parameters(recurse=False)):
+                    for attr in ("_airtc_wperm", "_airtc_b32", "_airtc_g32", "_airtc_w16"):
+                        if hasattr(t, attr):
+                            delattr(t, attr)
+        from ..parallel.collectives import broadcast_engine_weights
+
+        broadcast_engine_weights(self)  # no-op at world_size 1
+        self._graph = None  # re-capture lazily with the new weights
+
+    # ------------------------------------------------------------------
     # runtime config updates (POST /config + datachannel; SURVEY.md §3.5)
     # ------------------------------------------------------------------
     @torch.no_grad()

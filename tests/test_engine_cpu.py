@@ -124,3 +124,28 @@ def test_non_square_resolution():
     e.prepare()
     out = e(torch.randint(0, 256, (64, 96, 3), dtype=torch.uint8))
     assert out.shape == (64, 96, 3)
+
+
+def test_lora_hot_swap():
+    """Identical engines diverge exactly when one hot-swaps a LoRA (the
+    stream-batch FIFO advances every call, so the control engine isolates
+    the LoRA effect)."""
+    from ai_rtc_agent_amd.models.lora import make_random_lora
+
+    def fresh():
+        cfg = EngineConfig(
+            model_family="tiny", width=64, height=64, device="cpu",
+            use_hip_graph=False, use_lcm_lora=False,
+        )
+        e = StreamDiffusionEngine(cfg)
+        e.prepare()
+        return e
+
+    e1, e2 = fresh(), fresh()
+    for i in range(3):
+        a, b = e1(frame(seed=i)), e2(frame(seed=i))
+        assert torch.equal(a, b), "engines must match before the swap"
+    sd = make_random_lora(e2.unet, rank=2, seed=99, limit=20)
+    assert e2.load_lora(sd, scale=2.0) > 0
+    a, b = e1(frame(seed=7)), e2(frame(seed=7))
+    assert not torch.equal(a, b), "hot-swapped LoRA must change output"
