@@ -64,8 +64,9 @@ def fuse_lora_state_dict(
                 break
         if target is None:
             continue
-        down = slots["down"].float()
-        up = slots["up"].float()
+        target_dev = target.weight.device
+        down = slots["down"].float().to(target_dev)
+        up = slots["up"].float().to(target_dev)
         rank = down.shape[0]
         alpha = float(slots.get("alpha", torch.tensor(float(rank))))
         w = target.weight.data
