@@ -149,3 +149,14 @@ def test_lora_hot_swap():
     assert e2.load_lora(sd, scale=2.0) > 0
     a, b = e1(frame(seed=7)), e2(frame(seed=7))
     assert not torch.equal(a, b), "hot-swapped LoRA must change output"
+
+
+def test_frame_buffer_batching(tiny_cfg):
+    """frame_buffer_size > 1: batched frames in, batched frames out
+    (reference frame_bff_size, lib/wrapper.py:159-163)."""
+    tiny_cfg.frame_buffer_size = 2
+    e = make_engine(tiny_cfg)
+    assert e.cfg.unet_batch == 8
+    frames = torch.randint(0, 256, (2, 64, 64, 3), dtype=torch.uint8)
+    out = e(frames)
+    assert out.shape == (2, 64, 64, 3)
