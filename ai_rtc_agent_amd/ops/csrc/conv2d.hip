@@ -106,7 +106,7 @@ __device__ __forceinline__ void xcd_tile_map(int wg, int nwg, int n_tiles,
 // DBUF: double-buffered single-barrier register-staged schedule (A/B'd
 // slower at SD shapes; kept for experiments). GLDS: the DMA schedule above.
 // ---------------------------------------------------------------------------
-template <int MFRAG, int BK, bool DBUF>
+template <int MFRAG, int NFRAG, int BK, bool DBUF>
 __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
     const f16* __restrict__ x, const f16* __restrict__ w,
     const float* __restrict__ bias, const f16* __restrict__ cbias,
@@ -114,17 +114,18 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
     float* __restrict__ ws, int H, int W, int IC, int HO, int WO, int OC,
     int R, int S, int stride, int pad, int act, int K, int splitk) {
   constexpr int BM = MFRAG * 32;              // 128 or 64
+  constexpr int BNK = NFRAG * 32;             // 64 or 128 out-channels
   constexpr int KPITCH = BK + 8;              // +16B row pad (guide G4)
   constexpr int ALOADS = MFRAG * BK / 64;     // staged 16B A-loads per thread
-  constexpr int BLOADS = BK / 32;             // staged 16B B-loads per thread
+  constexpr int BLOADS = NFRAG * BK / 64;     // staged 16B B-loads per thread
   constexpr int KSH = BK == 64 ? 3 : 2;       // flat -> (row, k8) shifts
   constexpr int KMSK = BK / 8 - 1;
   constexpr int NBUF = DBUF ? 2 : 1;
   __shared__ f16 ldsA[NBUF * BM * KPITCH];
-  __shared__ f16 ldsB[NBUF * BN * KPITCH];
+  __shared__ f16 ldsB[NBUF * BNK * KPITCH];
 
   const int M = HO * WO;
-  const int n_tiles = ceil_div_dev(OC, BN);
+  const int n_tiles = ceil_div_dev(OC, BNK);
   int mt, nt;
   if (splitk > 0)
     xcd_tile_map(blockIdx.x, gridDim.x, n_tiles, &mt, &nt);
@@ -134,7 +135,7 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
   }
   const int spk = splitk > 0 ? splitk : -splitk;
   const int m0 = mt * BM;
-  const int n0 = nt * BN;
+  const int n0 = nt * BNK;
   const int b = blockIdx.z / spk;
   const int split = blockIdx.z - b * spk;
   const f16* xb = x + (long)b * H * W * IC;
@@ -166,11 +167,11 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
     wrow[i] = w + (long)min(n0 + b_row[i], OC - 1) * K + b_k8[i];
   }
 
-  f32x4 acc[MFRAG][2];
+  f32x4 acc[MFRAG][NFRAG];
 #pragma unroll
   for (int mi = 0; mi < MFRAG; ++mi)
 #pragma unroll
-    for (int ni = 0; ni < 2; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+    for (int ni = 0; ni < NFRAG; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
 
   // K-step range for this split
   const int nk = K / BK;
@@ -205,21 +206,21 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
   const int arow_base = wm * (MFRAG * 16) + (lane & 15);
   auto compute_tile = [&](int buf) {
     const f16* la = &ldsA[buf * BM * KPITCH];
-    const f16* lb = &ldsB[buf * BN * KPITCH];
+    const f16* lb = &ldsB[buf * BNK * KPITCH];
 #pragma unroll
     for (int kk = 0; kk < BK / 32; ++kk) {
       const int fcol = kk * 32 + (lane >> 4) * 8;
-      f16x8 bfrag[2];
+      f16x8 bfrag[NFRAG];
 #pragma unroll
-      for (int ni = 0; ni < 2; ++ni)
+      for (int ni = 0; ni < NFRAG; ++ni)
         bfrag[ni] = *reinterpret_cast<const f16x8*>(
-            &lb[(wn * 32 + ni * 16 + (lane & 15)) * KPITCH + fcol]);
+            &lb[(wn * (NFRAG * 16) + ni * 16 + (lane & 15)) * KPITCH + fcol]);
 #pragma unroll
       for (int mi = 0; mi < MFRAG; ++mi) {
         f16x8 afrag = *reinterpret_cast<const f16x8*>(
             &la[(arow_base + mi * 16) * KPITCH + fcol]);
 #pragma unroll
-        for (int ni = 0; ni < 2; ++ni)
+        for (int ni = 0; ni < NFRAG; ++ni)
           acc[mi][ni] = mfma16x16x32(afrag, bfrag[ni], acc[mi][ni]);
       }
     }
@@ -259,8 +260,8 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
     f16* ob = out + (long)b * M * OC;
     const long cb_off = (long)b * OC;
 #pragma unroll
-    for (int ni = 0; ni < 2; ++ni) {
-      const int col = n0 + wn * 32 + ni * 16 + (lane & 15);
+    for (int ni = 0; ni < NFRAG; ++ni) {
+      const int col = n0 + wn * (NFRAG * 16) + ni * 16 + (lane & 15);
       if (col >= OC) continue;
 #pragma unroll
       for (int mi = 0; mi < MFRAG; ++mi)
@@ -278,8 +279,8 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
     // plain f32 slab store; finalize kernel reduces + applies the epilogue
     float* wsb = ws + ((long)b * spk + split) * M * OC;
 #pragma unroll
-    for (int ni = 0; ni < 2; ++ni) {
-      const int col = n0 + wn * 32 + ni * 16 + (lane & 15);
+    for (int ni = 0; ni < NFRAG; ++ni) {
+      const int col = n0 + wn * (NFRAG * 16) + ni * 16 + (lane & 15);
       if (col >= OC) continue;
 #pragma unroll
       for (int mi = 0; mi < MFRAG; ++mi)
@@ -723,6 +724,15 @@ __global__ void conv2d_direct_kernel(
 // ---------------------------------------------------------------------------
 // launchers
 // ---------------------------------------------------------------------------
+static int conv_bn() {
+  static int bn = -1;
+  if (bn < 0) {
+    const char* e = getenv("AIRTC_CONV_BN128");
+    bn = (e && atoi(e)) ? 128 : 64;
+  }
+  return bn;
+}
+
 extern "C" int airtc_conv2d_splitk_for(int B, int HO, int WO, int OC, int IC) {
   // path/geometry decision, exported so the host can size the workspace:
   // returns 0 = direct, +k = BM128 split-K k, -k = BM64 split-K k.
@@ -730,19 +740,20 @@ extern "C" int airtc_conv2d_splitk_for(int B, int HO, int WO, int OC, int IC) {
   // hide latency with its own ILP (measured: 64x64x320 conv at 160 blocks
   // ran 87us = 86 TF; split-K over the K loop fills the chip).
   if (IC % 32 != 0) return 0;
+  const int BNSEL = conv_bn();
   const int M = HO * WO;
   // NOTE: a BM256 (MFRAG8) geometry measured SLOWER on the TAESD hi-res
   // layers (24.5 -> 37.1us @256²x64ch: fewer blocks + bigger staging
   // footprint cost more than the barrier amortisation bought) — path 100
   // exists but is never selected.
   if (M >= 2048) {
-    const long blocks = (long)ceil_div(M, 128) * ceil_div(OC, BN) * B;
+    const long blocks = (long)ceil_div(M, 128) * ceil_div(OC, BNSEL) * B;
     long k = (480 + blocks - 1) / blocks;
     if (k < 1) k = 1;
     if (k > 8) k = 8;
     return (int)k;
   }
-  const long blocks64 = (long)ceil_div(M, 64) * ceil_div(OC, BN) * B;
+  const long blocks64 = (long)ceil_div(M, 64) * ceil_div(OC, BNSEL) * B;
   long k = (512 + blocks64 - 1) / blocks64;
   if (k < 1) k = 1;
   if (k > 32) k = 32;
@@ -782,7 +793,8 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
   }
   const int splitk = path > 0 ? path : -path;
   const int bm = path > 0 ? 128 : 64;
-  dim3 grid(ceil_div(M, bm) * ceil_div(OC, BN), 1, B * splitk);
+  const int bn = (IC % 64 == 0) ? conv_bn() : 64;  // BN128 needs BK64 staging
+  dim3 grid(ceil_div(M, bm) * ceil_div(OC, bn), 1, B * splitk);
   const float* b1 = splitk == 1 ? bias : nullptr;
   const f16* cb1 = splitk == 1 ? cb : nullptr;
   const f16* res1 = splitk == 1 ? res : nullptr;
@@ -805,22 +817,25 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
   // splitk sign selects the tile mapping (positive = XCD-chunked, negative
   // = plain) — kernels take |splitk| as the split factor
   const int spk_arg = xcdmap ? splitk : -splitk;
-#define CONV_LAUNCH(MF, BKV, DB)                                              \
-  hipLaunchKernelGGL((conv2d_mfma_kernel<MF, BKV, DB>), grid, dim3(256), 0,   \
-                     s, xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC,  \
-                     R, S, stride, pad, act, K, spk_arg)
+#define CONV_LAUNCH(MF, NF, BKV, DB)                                          \
+  hipLaunchKernelGGL((conv2d_mfma_kernel<MF, NF, BKV, DB>), grid, dim3(256),  \
+                     0, s, xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO,   \
+                     OC, R, S, stride, pad, act, K, spk_arg)
 #define CONV_LAUNCH_GLDS(MF)                                                  \
   hipLaunchKernelGGL((conv2d_mfma_glds_kernel<MF>), grid, dim3(256), 0, s,    \
                      xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC, R,  \
                      S, stride, pad, act, K, spk_arg)
+  const bool bn128 = bn == 128;
   if (path > 0) {
     if (bk64 && glds) CONV_LAUNCH_GLDS(4);
-    else if (bk64) { if (dbuf) CONV_LAUNCH(4, 64, true); else CONV_LAUNCH(4, 64, false); }
-    else CONV_LAUNCH(4, 32, false);
+    else if (bk64 && bn128) CONV_LAUNCH(4, 4, 64, false);
+    else if (bk64) { if (dbuf) CONV_LAUNCH(4, 2, 64, true); else CONV_LAUNCH(4, 2, 64, false); }
+    else CONV_LAUNCH(4, 2, 32, false);
   } else {
     if (bk64 && glds) CONV_LAUNCH_GLDS(2);
-    else if (bk64) { if (dbuf) CONV_LAUNCH(2, 64, true); else CONV_LAUNCH(2, 64, false); }
-    else CONV_LAUNCH(2, 32, false);
+    else if (bk64 && bn128) CONV_LAUNCH(2, 4, 64, false);
+    else if (bk64) { if (dbuf) CONV_LAUNCH(2, 2, 64, true); else CONV_LAUNCH(2, 2, 64, false); }
+    else CONV_LAUNCH(2, 2, 32, false);
   }
 #undef CONV_LAUNCH
 #undef CONV_LAUNCH_GLDS
