@@ -26,7 +26,7 @@ def main():
     eng = StreamDiffusionEngine(sd_turbo_config(device="cuda"))
     eng.prepare()
     g = torch.Generator().manual_seed(0)
-    frames = [torch.randint(0, 256, (512, 512, 3), generator=g, dtype=torch.uint8, device="cuda")
+    frames = [torch.randint(0, 256, (512, 512, 3), generator=g, dtype=torch.uint8).cuda()
               for _ in range(8)]
     # warmup + capture
     for i in range(20):
