@@ -20,7 +20,6 @@ reference's single shared pipeline (agent.py:423).
 from __future__ import annotations
 
 import argparse
-import asyncio
 import json
 import logging
 import uuid

@@ -21,7 +21,7 @@ graph-external buffers in place; the graph is never re-captured
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence
+from typing import Optional, Sequence
 
 import torch
 

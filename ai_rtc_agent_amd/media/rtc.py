@@ -26,8 +26,6 @@ import socket
 import time
 from typing import Callable, Dict, List, Optional, Tuple
 
-import torch
-
 from .codec import select_codec
 from .rtp import RtpDefragmenter, RtpPacket, RtpPacketizer
 from .sdp import SessionDescription, build_answer

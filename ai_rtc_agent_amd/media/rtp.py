@@ -8,7 +8,7 @@ marker bit flags the last packet of a frame (standard for video RTP).
 from __future__ import annotations
 
 import struct
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional
 
 RTP_VERSION = 2

@@ -13,7 +13,7 @@ the UNet so the HIP kernels cover it with no new ops.
 """
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 import torch.nn as nn
@@ -22,7 +22,6 @@ from .. import ops
 from .unet import (
     Conv2d,
     Downsample,
-    GroupNormSiLU,
     Linear,
     ResnetBlock,
     SpatialTransformer,

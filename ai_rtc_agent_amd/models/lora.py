@@ -18,7 +18,7 @@ from __future__ import annotations
 
 import math
 import re
-from typing import Dict, Tuple
+from typing import Dict
 
 import torch
 import torch.nn as nn

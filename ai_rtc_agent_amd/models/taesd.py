@@ -14,8 +14,6 @@ decode() accepts the same — matching how the engine feeds UNet latents.
 """
 from __future__ import annotations
 
-import math
-
 import torch
 import torch.nn as nn
 import torch.nn.functional as F

@@ -19,8 +19,6 @@ per-link bound (MI355X design notes) better than per-tensor calls.
 from __future__ import annotations
 
 import os
-from typing import Iterable
-
 import torch
 import torch.distributed as dist
 

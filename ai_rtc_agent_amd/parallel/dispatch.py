@@ -19,7 +19,7 @@ For CPU tests the pool degrades to one CPU pipeline replica.
 from __future__ import annotations
 
 import logging
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import torch
 
