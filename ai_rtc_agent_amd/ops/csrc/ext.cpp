@@ -181,6 +181,13 @@ torch::Tensor postprocess_u8(torch::Tensor img) {
   return out;
 }
 
+pybind11::bytes h264_sps_pps(int width, int height) {
+  uint8_t buf[256];
+  int n = airtc_h264_sps_pps(width, height, buf, sizeof(buf));
+  TORCH_CHECK(n > 0, "sps/pps generation failed");
+  return pybind11::bytes(reinterpret_cast<const char*>(buf), n);
+}
+
 pybind11::dict vcn_probe() {
   char buf[512];
   int rc = airtc_vcn_probe(buf, sizeof(buf));
@@ -206,4 +213,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("preprocess_u8", &preprocess_u8);
   m.def("postprocess_u8", &postprocess_u8);
   m.def("vcn_probe", &vcn_probe, "probe the VCN VA-API stack");
+  m.def("h264_sps_pps", &h264_sps_pps, "Annex-B SPS+PPS for (w, h)");
 }

@@ -65,3 +65,7 @@ void airtc_attention(const uint16_t* q, const uint16_t* k, const uint16_t* v,
 // -1 = unavailable, else bit0 = H.264 decode, bit1 = H.264 encode; buf gets
 // a stage/detail summary. Implemented in vcn.cpp.
 extern "C" int airtc_vcn_probe(char* buf, int buflen);
+
+// H.264 parameter-set generation (vcn.cpp): Annex-B SPS+PPS for (w, h).
+extern "C" int airtc_h264_sps_pps(int width, int height, uint8_t* buf,
+                                  int buflen);

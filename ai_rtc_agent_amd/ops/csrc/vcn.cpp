@@ -170,3 +170,123 @@ extern "C" int airtc_vcn_probe(char* buf, int buflen) {
   if (!r.available) return -1;
   return (r.h264_decode ? 1 : 0) | (r.h264_encode ? 2 : 0);
 }
+
+// ---------------------------------------------------------------------------
+// H.264 bitstream layer (encoder side): SPS/PPS generation with exp-Golomb
+// coding and emulation prevention. This is the hardware-independent half of
+// the VCN encode session — VA-API encodes slices, but the parameter sets
+// (and their RTP/SDP advertisement) are the application's job. Baseline
+// profile, progressive, CAVLC: the low-latency WebRTC configuration.
+// ---------------------------------------------------------------------------
+namespace h264 {
+
+struct BitWriter {
+  std::vector<uint8_t> bytes;
+  uint32_t cur = 0;
+  int nbits = 0;
+
+  void put(uint32_t value, int width) {
+    for (int i = width - 1; i >= 0; --i) {
+      cur = (cur << 1) | ((value >> i) & 1);
+      if (++nbits == 8) {
+        bytes.push_back((uint8_t)cur);
+        cur = 0;
+        nbits = 0;
+      }
+    }
+  }
+  void ue(uint32_t v) {  // unsigned exp-Golomb
+    uint32_t vp1 = v + 1;
+    int lead = 0;
+    for (uint32_t t = vp1; t > 1; t >>= 1) ++lead;
+    put(0, lead);
+    put(vp1, lead + 1);
+  }
+  void se(int32_t v) {  // signed exp-Golomb
+    ue(v <= 0 ? (uint32_t)(-2 * v) : (uint32_t)(2 * v - 1));
+  }
+  void rbsp_trailing() {
+    put(1, 1);
+    if (nbits) put(0, 8 - nbits);
+  }
+};
+
+// RBSP -> NAL with emulation prevention (00 00 {00,01,02,03} -> 00 00 03 xx)
+inline void emit_nal(std::vector<uint8_t>* out, uint8_t nal_header,
+                     const std::vector<uint8_t>& rbsp) {
+  out->insert(out->end(), {0, 0, 0, 1, nal_header});
+  int zeros = 0;
+  for (uint8_t b : rbsp) {
+    if (zeros >= 2 && b <= 3) {
+      out->push_back(3);
+      zeros = 0;
+    }
+    out->push_back(b);
+    zeros = (b == 0) ? zeros + 1 : 0;
+  }
+}
+
+inline std::vector<uint8_t> build_sps_pps(int width, int height) {
+  std::vector<uint8_t> out;
+  // --- SPS (nal 7, baseline/constrained) ---
+  {
+    BitWriter w;
+    w.put(66, 8);   // profile_idc: baseline
+    w.put(0xC0, 8); // constraint_set0+1, reserved
+    w.put(31, 8);   // level_idc 3.1
+    w.ue(0);        // sps_id
+    w.ue(0);        // log2_max_frame_num_minus4
+    w.ue(2);        // pic_order_cnt_type: 2 (low-latency, no reordering)
+    w.ue(1);        // max_num_ref_frames
+    w.put(0, 1);    // gaps_in_frame_num_value_allowed
+    const int mbs_w = (width + 15) / 16, mbs_h = (height + 15) / 16;
+    w.ue(mbs_w - 1);
+    w.ue(mbs_h - 1);
+    w.put(1, 1);    // frame_mbs_only
+    w.put(1, 1);    // direct_8x8_inference
+    const int crop_r = mbs_w * 16 - width, crop_b = mbs_h * 16 - height;
+    if (crop_r || crop_b) {
+      w.put(1, 1);  // frame_cropping
+      w.ue(0); w.ue(crop_r / 2); w.ue(0); w.ue(crop_b / 2);
+    } else {
+      w.put(0, 1);
+    }
+    w.put(0, 1);    // vui_parameters_present
+    w.rbsp_trailing();
+    emit_nal(&out, 0x67, w.bytes);
+  }
+  // --- PPS (nal 8) ---
+  {
+    BitWriter w;
+    w.ue(0);        // pps_id
+    w.ue(0);        // sps_id
+    w.put(0, 1);    // entropy_coding_mode: CAVLC
+    w.put(0, 1);    // bottom_field_pic_order_in_frame_present
+    w.ue(0);        // num_slice_groups_minus1
+    w.ue(0);        // num_ref_idx_l0_default_active_minus1
+    w.ue(0);        // num_ref_idx_l1_default_active_minus1
+    w.put(0, 1);    // weighted_pred
+    w.put(0, 2);    // weighted_bipred_idc
+    w.se(0);        // pic_init_qp_minus26
+    w.se(0);        // pic_init_qs_minus26
+    w.se(0);        // chroma_qp_index_offset
+    w.put(1, 1);    // deblocking_filter_control_present
+    w.put(0, 1);    // constrained_intra_pred
+    w.put(0, 1);    // redundant_pic_cnt_present
+    w.rbsp_trailing();
+    emit_nal(&out, 0x68, w.bytes);
+  }
+  return out;
+}
+
+}  // namespace h264
+
+// C ABI: fills buf with Annex-B SPS+PPS for (width, height); returns length
+// or -1 if the buffer is too small.
+extern "C" int airtc_h264_sps_pps(int width, int height, uint8_t* buf,
+                                  int buflen) {
+  std::vector<uint8_t> v = h264::build_sps_pps(width, height);
+  if ((int)v.size() > buflen) return -1;
+  memcpy(buf, v.data(), v.size());
+  return (int)v.size();
+}
