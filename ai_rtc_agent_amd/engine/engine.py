@@ -197,6 +197,34 @@ class StreamDiffusionEngine:
         self._pipelined = False
         self._last_done = None
         self._prepared = True
+        self._refresh_static_kv()
+
+    def _cross_attn_modules(self):
+        from ..models.unet import CrossAttention
+
+        mods = []
+        for root in (self.unet, self.controlnet):
+            if root is None:
+                continue
+            for name, m in root.named_modules():
+                if isinstance(m, CrossAttention) and name.endswith("attn2"):
+                    mods.append(m)
+        return mods
+
+    @torch.no_grad()
+    def _refresh_static_kv(self) -> None:
+        """Precompute every cross-attention layer's K|V from the (static)
+        text embeddings. K/V depend only on the prompt, so these GEMMs run
+        at prepare()/update_prompt() instead of once per frame; the hot
+        graph reads the static buffers (stable addresses, copy_ on update)."""
+        ctx = self._unet_batch_embeds()
+        for m in self._cross_attn_modules():
+            kv = m.compute_kv(ctx)
+            cur = getattr(m, "static_kv", None)
+            if cur is None or cur.shape != kv.shape or cur.dtype != kv.dtype:
+                m.static_kv = kv.contiguous()
+            else:
+                cur.copy_(kv)
 
     @torch.no_grad()
     def _refresh_added_cond(self) -> None:
@@ -244,6 +272,8 @@ class StreamDiffusionEngine:
         from ..parallel.collectives import broadcast_engine_weights
 
         broadcast_engine_weights(self)  # no-op at world_size 1
+        if self._prepared:
+            self._refresh_static_kv()
         self._graph = None  # re-capture lazily with the new weights
 
     # ------------------------------------------------------------------
@@ -262,6 +292,7 @@ class StreamDiffusionEngine:
             self._embeds_full[B:].copy_(self._embeds_batch)
         if self._added_cond is not None:
             self._refresh_added_cond()
+        self._refresh_static_kv()
 
     @torch.no_grad()
     def update_t_index_list(self, t_index_list: Sequence[int]) -> None:

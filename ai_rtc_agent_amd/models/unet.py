@@ -207,6 +207,19 @@ class CrossAttention(nn.Module):
             setattr(self, name, w)
         return w
 
+    def compute_kv(self, ctx: torch.Tensor) -> torch.Tensor:
+        """Fused K|V projection of a context. The engine calls this at
+        prepare()/update_prompt() to PRECOMPUTE text K/V into a static
+        buffer (`static_kv`): cross-attention K/V depend only on the prompt,
+        so the per-frame graph skips these GEMMs entirely (the reference
+        keeps prompt encoding out of its TRT hot path the same way,
+        lib/pipeline.py:44-45)."""
+        wkv = self._cached_w(
+            "_wkv", ctx,
+            lambda: torch.cat([self._pad_heads(p.weight) for p in (self.to_k, self.to_v)]),
+        )
+        return ops.linear(ctx, wkv)
+
     def forward(
         self,
         x: torch.Tensor,
@@ -222,12 +235,10 @@ class CrossAttention(nn.Module):
             q, k, v = qkv.chunk(3, dim=-1)
         else:
             wq = self._cached_w("_wq", x, lambda: self._pad_heads(self.to_q.weight))
-            wkv = self._cached_w(
-                "_wkv", x,
-                lambda: torch.cat([self._pad_heads(p.weight) for p in (self.to_k, self.to_v)]),
-            )
             q = ops.linear(x, wq)
-            kv = ops.linear(ctx, wkv)
+            kv = getattr(self, "static_kv", None)
+            if kv is None:
+                kv = self.compute_kv(ctx)
             k, v = kv.chunk(2, dim=-1)
         o = ops.attention(q, k, v, self.heads, scale=1.0 / math.sqrt(self.head_dim))
         # out projection consumes the padded layout (zero input columns)
