@@ -50,6 +50,8 @@ class VcnH264Codec:
     NVENC/NVDEC on/off envs (Dockerfile:54-56).
     """
 
+    rtp_mode = "rfc6184"  # Annex-B NALs on the wire (media/h264.py)
+
     def __init__(self, cfg: EncoderConfig | None = None):
         self.cfg = cfg or EncoderConfig()
         self._lib = self._probe()
@@ -98,6 +100,11 @@ class SoftwareCodec:
     encode(): u8 RGB (H, W, 3) tensor -> bytes
     decode(): bytes -> u8 RGB tensor (or None until a keyframe arrives)
     """
+
+    # transport hint: frames are opaque blobs -> generic RTP fragmentation.
+    # H.264 codecs report "rfc6184" and the transport switches to
+    # single-NAL/FU-A payloads (media/h264.py).
+    rtp_mode = "raw"
 
     def __init__(self, keyframe_interval: int = 30, level: int = 1):
         self.keyframe_interval = keyframe_interval

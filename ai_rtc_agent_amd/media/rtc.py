@@ -27,6 +27,7 @@ import time
 from typing import Callable, Dict, List, Optional, Tuple
 
 from .codec import select_codec
+from .h264 import H264Depacketizer, join_annexb, packetize_h264, split_annexb
 from .rtp import RtpDefragmenter, RtpPacket, RtpPacketizer
 from .sdp import SessionDescription, build_answer
 from .stun import StunMessage, BINDING_REQUEST, is_stun, make_binding_response
@@ -94,6 +95,8 @@ class PeerConnection:
         self._decoder = select_codec(role="decode")
         self._packetizer = RtpPacketizer(ssrc=random.randint(1, 2**31))
         self._defrag = RtpDefragmenter()
+        self._h264_depack = H264Depacketizer()
+        self._rx_nals: list = []
         self.port: Optional[int] = None
         self._ice_pwd = ""
 
@@ -178,10 +181,27 @@ class PeerConnection:
                     continue
                 data = self._encoder.encode(frame.tensor, keyframe=counter == 0)
                 ts = frame.pts if frame.pts else counter * 3000
+                if getattr(self._encoder, "rtp_mode", "raw") == "rfc6184":
+                    # H.264: standard single-NAL / FU-A payloads per NAL
+                    pkts = []
+                    nals = split_annexb(data)
+                    payloads = packetize_h264(nals)
+                    for pi, pl in enumerate(payloads):
+                        pkts.append(RtpPacket(
+                            payload_type=self._packetizer.payload_type,
+                            sequence_number=self._packetizer._seq,
+                            timestamp=ts,
+                            ssrc=self._packetizer.ssrc,
+                            marker=1 if pi == len(payloads) - 1 else 0,
+                            payload=pl,
+                        ))
+                        self._packetizer._seq = (self._packetizer._seq + 1) & 0xFFFF
+                else:
+                    pkts = self._packetizer.packetize(data, ts)
                 # paced send: large frames fragment into hundreds of
                 # datagrams; an unpaced burst overflows receiver socket
                 # buffers (and starves the event loop's read side)
-                for j, pkt in enumerate(self._packetizer.packetize(data, ts)):
+                for j, pkt in enumerate(pkts):
                     self._transport.sendto(pkt.serialize(), self._remote_addr)
                     if j % 32 == 31:
                         await asyncio.sleep(0)
@@ -221,7 +241,16 @@ class PeerConnection:
         self._remote_addr = addr
         if self.connection_state == "connecting":
             self._set_state("connected")
-        frame_bytes = self._defrag.push(pkt)
+        if getattr(self._decoder, "rtp_mode", "raw") == "rfc6184":
+            nal = self._h264_depack.push(pkt.payload)
+            if nal is not None:
+                self._rx_nals.append(nal)
+            if not pkt.marker or not self._rx_nals:
+                return
+            frame_bytes = join_annexb(self._rx_nals)
+            self._rx_nals = []
+        else:
+            frame_bytes = self._defrag.push(pkt)
         if frame_bytes is None:
             return
         try:
