@@ -72,9 +72,10 @@ __device__ __forceinline__ f16 epilogue(float acc, const float* bias,
 // The LDS image is lane-linear (glds writes base + lane*16), so the
 // bank-conflict fix moves to the SOURCE address (guide rule 21): 16B unit
 // u of row r holds global unit u ^ ((r>>1)&7), and fragment reads apply
-// the same XOR. Schedule: double-buffered, one barrier per K-step; hipcc's
-// syncthreads drains the in-flight glds (vmcnt(0)) exactly where the
-// recipe wants it.
+// the same XOR. Schedule: 3-deep buffer ring with COUNTED vmcnt + raw
+// barriers (one tile stays in flight across every barrier). Measured: ties
+// the register-staged default within noise at SD shapes (ladder) — kept
+// env-gated (AIRTC_CONV_GLDS=1) as the starting point for deeper pipelines.
 // ---------------------------------------------------------------------------
 typedef __attribute__((address_space(3))) f16 lds_f16;
 
