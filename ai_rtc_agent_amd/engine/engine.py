@@ -274,6 +274,9 @@ class StreamDiffusionEngine:
         broadcast_engine_weights(self)  # no-op at world_size 1
         if self._prepared:
             self._refresh_static_kv()
+        if self.device.type == "cuda":
+            # quiesce in-flight replays before their graphs are dropped
+            torch.cuda.synchronize()
         self._graph = None  # re-capture lazily with the new weights
 
     # ------------------------------------------------------------------
