@@ -245,6 +245,8 @@ class PeerConnection:
             nal = self._h264_depack.push(pkt.payload)
             if nal is not None:
                 self._rx_nals.append(nal)
+                if len(self._rx_nals) > 64:  # lost marker: drop the stale AU
+                    self._rx_nals = self._rx_nals[-1:]
             if not pkt.marker or not self._rx_nals:
                 return
             frame_bytes = join_annexb(self._rx_nals)
