@@ -20,14 +20,12 @@ reference's single shared pipeline (agent.py:423).
 from __future__ import annotations
 
 import argparse
-import json
 import logging
 import uuid
 from typing import Optional
 
 from aiohttp import web
 
-from . import config as cfgmod
 from .media.ice import get_ice_servers, get_link_headers
 from .media.rtc import MediaRelay, PeerConnection, set_udp_port_pool
 from .media.tracks import VideoStreamTrack
@@ -200,8 +198,16 @@ async def on_startup(app: web.Application) -> None:
     from .parallel.dispatch import PipelinePool
 
     if st.get("pool") is None:
+        from .config import EngineConfig
+
+        cfg = EngineConfig(
+            model_id=app["model_id"],
+            model_family=app.get("family", "sd15"),
+            width=app.get("resolution", 512),
+            height=app.get("resolution", 512),
+        )
         st["pool"] = PipelinePool.create(
-            model_id=app["model_id"], n_gpus=app["n_gpus"]
+            model_id=app["model_id"], n_gpus=app["n_gpus"], cfg=cfg
         )
     st["ice_servers"] = get_ice_servers() if app["use_turn"] else []
 
@@ -234,6 +240,8 @@ def create_app(
     host: str = "127.0.0.1",
     n_gpus: int = 1,
     use_turn: bool = True,
+    family: str = "sd15",
+    resolution: int = 512,
 ) -> web.Application:
     app = web.Application(middlewares=[cors_middleware])
     app["model_id"] = model_id
@@ -241,6 +249,8 @@ def create_app(
     app["host"] = host
     app["n_gpus"] = n_gpus
     app["use_turn"] = use_turn
+    app["family"] = family
+    app["resolution"] = resolution
     app["state"] = {
         "pcs": set(),
         "source_track": None,
@@ -271,6 +281,9 @@ def main() -> None:
     parser.add_argument("--log-level", default="INFO")
     parser.add_argument("--gpus", type=int, default=1, help="pipeline replicas (one per GPU)")
     parser.add_argument("--host", default="0.0.0.0")
+    parser.add_argument("--family", default="sd15", choices=["sd15", "sd21", "sdxl"],
+                        help="UNet family served by the pipeline")
+    parser.add_argument("--resolution", type=int, default=512)
     args = parser.parse_args()
 
     logging.basicConfig(level=getattr(logging, args.log_level.upper(), logging.INFO))
@@ -278,7 +291,8 @@ def main() -> None:
     if args.udp_ports:
         lo, _, hi = args.udp_ports.partition("-")
         ports = list(range(int(lo), int(hi or lo) + 1))
-    app = create_app(model_id=args.model_id, udp_ports=ports, n_gpus=args.gpus)
+    app = create_app(model_id=args.model_id, udp_ports=ports, n_gpus=args.gpus,
+                     family=args.family, resolution=args.resolution)
     web.run_app(app, host=args.host, port=args.port)
 
 
