@@ -38,15 +38,19 @@ class PipelinePool:
         from ..config import EngineConfig
         from ..pipeline import StreamDiffusionPipeline
 
+        import copy
+
         pipelines = []
         if torch.cuda.is_available():
             n = min(n_gpus, torch.cuda.device_count())
             for i in range(n):
-                c = cfg or EngineConfig(model_id=model_id)
+                # per-replica config copy: replicas must not share (and
+                # last-write) one mutable config object
+                c = copy.deepcopy(cfg) if cfg is not None else EngineConfig(model_id=model_id)
                 c.device = f"cuda:{i}"
                 pipelines.append(StreamDiffusionPipeline(model_id, cfg=c))
         else:
-            c = cfg or EngineConfig(model_id=model_id)
+            c = copy.deepcopy(cfg) if cfg is not None else EngineConfig(model_id=model_id)
             c.device = "cpu"
             c.use_hip_graph = False
             pipelines.append(StreamDiffusionPipeline(model_id, cfg=c))
