@@ -42,13 +42,21 @@ def test_bench_single_process():
     assert d["value"] > 0 and d["ms_per_step"] > 0
 
 
+def _free_port():
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 @pytest.mark.timeout(600)
 def test_bench_torchrun_2rank_gloo():
     env = dict(os.environ, AIRTC_BENCH_FAMILY="tiny")
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29611", BENCH,
+         "--master-port", str(_free_port()), BENCH,
          "--gpus", "2", "--steps", "2", "--warmup", "1",
          "--width", "64", "--latency-frames", "2"],
         capture_output=True, text=True, timeout=580, cwd=ROOT, env=env,
