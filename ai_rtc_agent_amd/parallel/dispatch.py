@@ -80,7 +80,14 @@ class PipelinePool:
         return self._pipelines
 
     def stats(self) -> dict:
+        device = None
+        if torch.cuda.is_available():
+            device = {
+                "name": torch.cuda.get_device_name(0),
+                "memory_allocated_mb": round(torch.cuda.memory_allocated() / 1e6, 1),
+            }
         return {
+            "device": device,
             "replicas": len(self._pipelines),
             "sessions": {k: v for k, v in self._sessions.items()},
             "per_replica": [
