@@ -28,6 +28,7 @@ from typing import Callable, Dict, List, Optional, Tuple
 
 from .codec import select_codec
 from .h264 import H264Depacketizer, join_annexb, packetize_h264, split_annexb
+from .rtcp import is_rtcp, make_pli, parse_pli
 from .rtp import RtpDefragmenter, RtpPacket, RtpPacketizer
 from .sdp import SessionDescription, build_answer
 from .stun import StunMessage, BINDING_REQUEST, is_stun, make_binding_response
@@ -97,6 +98,9 @@ class PeerConnection:
         self._defrag = RtpDefragmenter()
         self._h264_depack = H264Depacketizer()
         self._rx_nals: list = []
+        self._force_keyframe = False
+        self._decode_misses = 0
+        self._last_pli = 0.0
         self.port: Optional[int] = None
         self._ice_pwd = ""
 
@@ -179,7 +183,9 @@ class PeerConnection:
                 frame = await self._send_track.recv()
                 if self._remote_addr is None or self._transport is None:
                     continue
-                data = self._encoder.encode(frame.tensor, keyframe=counter == 0)
+                want_key = counter == 0 or self._force_keyframe
+                self._force_keyframe = False
+                data = self._encoder.encode(frame.tensor, keyframe=want_key)
                 ts = frame.pts if frame.pts else counter * 3000
                 if getattr(self._encoder, "rtp_mode", "raw") == "rfc6184":
                     # H.264: standard single-NAL / FU-A payloads per NAL
@@ -227,6 +233,11 @@ class PeerConnection:
                 self._remote_addr = addr  # peer-reflexive
                 self._set_state("connected")
             return
+        if is_rtcp(data):
+            if parse_pli(data) is not None:
+                # peer lost decodability: force a keyframe on the next frame
+                self._force_keyframe = True
+            return
         if data[:4] == CONFIG_MAGIC:
             try:
                 payload = json.loads(data[4:].decode())
@@ -261,7 +272,16 @@ class PeerConnection:
             logger.exception("decode failed")
             return
         if tensor is None:
+            # undecodable (lost reference frame): ask for a keyframe
+            self._decode_misses += 1
+            now = time.monotonic()
+            if self._decode_misses >= 2 and now - self._last_pli > 0.5 \
+                    and self._transport is not None and self._remote_addr:
+                self._transport.sendto(
+                    make_pli(self._packetizer.ssrc, pkt.ssrc), self._remote_addr)
+                self._last_pli = now
             return
+        self._decode_misses = 0
         if self._recv_track is None:
             self._recv_track = QueueTrack()
         self._recv_track.push(VideoFrame(tensor=tensor, pts=pkt.timestamp))

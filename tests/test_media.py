@@ -135,3 +135,59 @@ def test_vcn_native_probe_reports():
 
     if not r["available"]:
         assert isinstance(select_codec(), SoftwareCodec)
+
+
+def test_rtcp_pli_roundtrip():
+    from ai_rtc_agent_amd.media import rtcp
+
+    pkt = rtcp.make_pli(111, 222)
+    assert rtcp.is_rtcp(pkt)
+    assert rtcp.parse_pli(pkt) == (111, 222)
+    assert rtcp.parse_pli(b"\x80\x00\x00\x00") is None
+    # RTP packets must not be mistaken for RTCP
+    from ai_rtc_agent_amd.media.rtp import RtpPacket
+
+    rtp = RtpPacket(payload_type=97, payload=b"x").serialize()
+    assert not rtcp.is_rtcp(rtp)
+
+
+def test_pli_keyframe_recovery():
+    """Receiver desyncs (lost keyframe) -> sends PLI; sender receives PLI ->
+    forces a keyframe on the next frame."""
+    import asyncio
+
+    from ai_rtc_agent_amd.media.rtc import PeerConnection
+    from ai_rtc_agent_amd.media.rtp import RtpPacketizer
+    from ai_rtc_agent_amd.media import rtcp
+
+    async def body():
+        rx = PeerConnection()
+        sent = []
+
+        class FakeTransport:
+            def sendto(self, data, addr):
+                sent.append(data)
+
+        rx._transport = FakeTransport()
+        rx._remote_addr = ("127.0.0.1", 5000)
+
+        enc = SoftwareCodec(keyframe_interval=1000)
+        pkz = RtpPacketizer(ssrc=9)
+        f0 = torch.zeros(8, 8, 3, dtype=torch.uint8)
+        _ = enc.encode(f0)                    # I-frame (never delivered)
+        for i in range(1, 4):                 # deliver only P-frames
+            for pkt in pkz.packetize(enc.encode(f0), timestamp=i * 3000):
+                rx._on_datagram(pkt.serialize(), ("127.0.0.1", 5000))
+        plis = [d for d in sent if rtcp.is_rtcp(d)]
+        assert plis, "desynced receiver must emit a PLI"
+
+        # sender side: receiving that PLI forces a keyframe
+        tx = PeerConnection()
+        tx._on_datagram(plis[0], ("127.0.0.1", 6000))
+        assert tx._force_keyframe
+
+    loop = asyncio.new_event_loop()
+    try:
+        loop.run_until_complete(body())
+    finally:
+        loop.close()
