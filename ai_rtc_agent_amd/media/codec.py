@@ -106,33 +106,66 @@ class SoftwareCodec:
     # single-NAL/FU-A payloads (media/h264.py).
     rtp_mode = "raw"
 
-    def __init__(self, keyframe_interval: int = 30, level: int = 1):
+    FPS_ASSUMED = 30  # per-frame byte budget = bitrate / 8 / FPS
+
+    def __init__(self, keyframe_interval: int = 30, level: int = 1,
+                 cfg: EncoderConfig | None = None):
         self.keyframe_interval = keyframe_interval
         self.level = level
+        self.cfg = cfg
         self._enc_prev: Optional[np.ndarray] = None
         self._enc_count = 0
         self._dec_prev: Optional[np.ndarray] = None
 
+    def _budget(self) -> Optional[int]:
+        if self.cfg is None:
+            return None
+        bps = max(self.cfg.min_bitrate, min(self.cfg.max_bitrate, self.cfg.default_bitrate))
+        return max(1024, bps // 8 // self.FPS_ASSUMED)
+
     def encode(self, frame_u8: torch.Tensor, keyframe: bool = False) -> bytes:
         arr = frame_u8.detach().cpu().numpy().astype(np.uint8)
         h, w, _ = arr.shape
-        hdr = struct.pack("!HH", h, w)
         force_key = keyframe or self._enc_prev is None or self._enc_count % self.keyframe_interval == 0
         self._enc_count += 1
-        if force_key:
-            self._enc_prev = arr
-            return _MAGIC_I + hdr + zlib.compress(arr.tobytes(), self.level)
-        delta = (arr.astype(np.int16) - self._enc_prev.astype(np.int16)).astype(np.int8)
+        budget = self._budget()
+
+        if not force_key:
+            hdr = struct.pack("!HHB", h, w, 0)
+            delta = (arr.astype(np.int16) - self._enc_prev.astype(np.int16)).astype(np.int8)
+            body = zlib.compress(delta.tobytes(), self.level)
+            if budget is None or len(body) <= budget:
+                self._enc_prev = arr
+                return _MAGIC_P + hdr + body
+            # over budget: fall through to a (possibly downscaled) keyframe
+
+        # RATE CONTROL (honours the EncoderConfig bitrate knobs, parity with
+        # the reference's NVENC_* envs): escalate zlib level, then spatially
+        # downscale by powers of two until the frame fits its byte budget.
+        scale = 0
+        src = arr
+        body = zlib.compress(src.tobytes(), self.level)
+        if budget is not None and len(body) > budget:
+            body = zlib.compress(src.tobytes(), 9)
+        while budget is not None and len(body) > budget and scale < 3 \
+                and src.shape[0] > 16 and src.shape[1] > 16:
+            src = src[::2, ::2]
+            scale += 1
+            body = zlib.compress(src.tobytes(), 9)
         self._enc_prev = arr
-        return _MAGIC_P + hdr + zlib.compress(delta.tobytes(), self.level)
+        hdr = struct.pack("!HHB", h, w, scale)
+        return _MAGIC_I + hdr + body
 
     def decode(self, data: bytes) -> Optional[torch.Tensor]:
-        magic, hdr, body = data[:4], data[4:8], data[8:]
-        h, w = struct.unpack("!HH", hdr)
+        magic, hdr, body = data[:4], data[4:9], data[9:]
+        h, w, scale = struct.unpack("!HHB", hdr)
         if magic == _MAGIC_I:
-            arr = np.frombuffer(zlib.decompress(body), dtype=np.uint8).reshape(h, w, 3)
-            self._dec_prev = arr
-            return torch.from_numpy(arr.copy())
+            hs, ws = (h + (1 << scale) - 1) >> scale, (w + (1 << scale) - 1) >> scale
+            arr = np.frombuffer(zlib.decompress(body), dtype=np.uint8).reshape(hs, ws, 3)
+            if scale:
+                arr = arr.repeat(1 << scale, axis=0).repeat(1 << scale, axis=1)[:h, :w]
+            self._dec_prev = np.ascontiguousarray(arr)
+            return torch.from_numpy(self._dec_prev.copy())
         if magic == _MAGIC_P:
             if self._dec_prev is None:
                 return None  # wait for a keyframe
@@ -152,4 +185,4 @@ def select_codec(cfg: EncoderConfig | None = None, role: str = "encode"):
             return VcnH264Codec(cfg)
         except CodecUnavailable:
             pass
-    return SoftwareCodec()
+    return SoftwareCodec(cfg=cfg or EncoderConfig())

@@ -191,3 +191,42 @@ def test_pli_keyframe_recovery():
         loop.run_until_complete(body())
     finally:
         loop.close()
+
+
+def test_software_codec_rate_control():
+    """EncoderConfig bitrate knobs bound the encoded frame size (escalating
+    compression, then spatial downscale); decode recovers full resolution."""
+    from ai_rtc_agent_amd.config import EncoderConfig
+
+    g = torch.Generator().manual_seed(3)
+    noise = torch.randint(0, 256, (256, 256, 3), generator=g, dtype=torch.uint8)
+
+    # generous budget: lossless, full size
+    rich = SoftwareCodec(cfg=EncoderConfig(default_bitrate=200_000_000,
+                                           min_bitrate=1_000, max_bitrate=300_000_000))
+    d = SoftwareCodec()
+    out = d.decode(rich.encode(noise, keyframe=True))
+    assert torch.equal(out, noise)
+
+    # tight budget: frame must shrink to fit, decode keeps the shape
+    tight_cfg = EncoderConfig(default_bitrate=2_000_000, min_bitrate=1_000,
+                              max_bitrate=2_000_000)
+    tight = SoftwareCodec(cfg=tight_cfg)
+    data = tight.encode(noise, keyframe=True)
+    assert len(data) <= 2_000_000 // 8 // SoftwareCodec.FPS_ASSUMED + 16
+    out2 = SoftwareCodec().decode(data)
+    assert out2.shape == (256, 256, 3), "decoder restores full resolution"
+
+
+def test_p_frame_over_budget_becomes_keyframe():
+    from ai_rtc_agent_amd.config import EncoderConfig
+
+    cfg = EncoderConfig(default_bitrate=1_500_000, min_bitrate=1_000,
+                        max_bitrate=1_500_000)
+    c = SoftwareCodec(keyframe_interval=1000, cfg=cfg)
+    g = torch.Generator().manual_seed(4)
+    f0 = torch.randint(0, 256, (128, 128, 3), generator=g, dtype=torch.uint8)
+    f1 = torch.randint(0, 256, (128, 128, 3), generator=g, dtype=torch.uint8)
+    c.encode(f0, keyframe=True)
+    data = c.encode(f1)  # delta of two noise frames blows the budget
+    assert data[:4] == b"RZI1", "over-budget P-frame must fall back to I"
