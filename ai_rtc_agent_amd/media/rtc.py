@@ -28,7 +28,7 @@ from typing import Callable, Dict, List, Optional, Tuple
 
 from .codec import select_codec
 from .h264 import H264Depacketizer, join_annexb, packetize_h264, split_annexb
-from .rtcp import is_rtcp, make_pli, parse_pli
+from .rtcp import is_rtcp, make_pli, make_rr, parse_pli, parse_rr
 from .rtp import RtpDefragmenter, RtpPacket, RtpPacketizer
 from .sdp import SessionDescription, build_answer
 from .stun import StunMessage, BINDING_REQUEST, is_stun, make_binding_response
@@ -101,6 +101,11 @@ class PeerConnection:
         self._force_keyframe = False
         self._decode_misses = 0
         self._last_pli = 0.0
+        # loss accounting (receiver) + adaptive bitrate (sender)
+        self._rx_count = 0
+        self._rx_base_seq: Optional[int] = None
+        self._rx_high_seq = 0
+        self._rr_last_sent = (0, 0)  # (expected, received) at last RR
         self.port: Optional[int] = None
         self._ice_pwd = ""
 
@@ -237,6 +242,9 @@ class PeerConnection:
             if parse_pli(data) is not None:
                 # peer lost decodability: force a keyframe on the next frame
                 self._force_keyframe = True
+            rr = parse_rr(data)
+            if rr is not None:
+                self._adapt_bitrate(rr[1])
             return
         if data[:4] == CONFIG_MAGIC:
             try:
@@ -252,6 +260,7 @@ class PeerConnection:
         self._remote_addr = addr
         if self.connection_state == "connecting":
             self._set_state("connected")
+        self._account_rx(pkt)
         if getattr(self._decoder, "rtp_mode", "raw") == "rfc6184":
             nal = self._h264_depack.push(pkt.payload)
             if nal is not None:
@@ -288,6 +297,44 @@ class PeerConnection:
         if not self._track_fired:
             self._track_fired = True
             self._emit("track", self._recv_track)
+
+    def _account_rx(self, pkt: RtpPacket) -> None:
+        """Track loss from RTP sequence numbers; report every ~128 packets
+        so the sender's rate control sees fresh loss fractions."""
+        seq = pkt.sequence_number
+        if self._rx_base_seq is None:
+            self._rx_base_seq = seq
+            self._rx_high_seq = seq
+        # unwrap 16-bit sequence space
+        high = self._rx_high_seq
+        diff = (seq - (high & 0xFFFF)) & 0xFFFF
+        if 0 < diff < 0x8000:  # in-order advance (16-bit space unwrapped)
+            self._rx_high_seq = high + diff
+        self._rx_count += 1
+        expected = self._rx_high_seq - self._rx_base_seq + 1
+        exp0, rcv0 = self._rr_last_sent
+        if expected - exp0 >= 128 and self._transport is not None and self._remote_addr:
+            interval_exp = expected - exp0
+            interval_rcv = self._rx_count - rcv0
+            lost_frac = max(0.0, 1.0 - interval_rcv / max(1, interval_exp))
+            self._transport.sendto(
+                make_rr(self._packetizer.ssrc, pkt.ssrc, lost_frac,
+                        max(0, expected - self._rx_count), self._rx_high_seq),
+                self._remote_addr,
+            )
+            self._rr_last_sent = (expected, self._rx_count)
+
+    def _adapt_bitrate(self, fraction_lost: float) -> None:
+        """AIMD-style sender rate control on RR loss feedback, within the
+        EncoderConfig min/max bitrate bounds (the knobs NVENC would own in
+        the reference)."""
+        cfg = getattr(self._encoder, "cfg", None)
+        if cfg is None:
+            return
+        if fraction_lost > 0.05:
+            cfg.default_bitrate = max(cfg.min_bitrate, int(cfg.default_bitrate * 0.7))
+        elif fraction_lost < 0.01:
+            cfg.default_bitrate = min(cfg.max_bitrate, int(cfg.default_bitrate * 1.08))
 
     # -- teardown --------------------------------------------------------
     async def close(self) -> None:

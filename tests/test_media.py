@@ -230,3 +230,60 @@ def test_p_frame_over_budget_becomes_keyframe():
     c.encode(f0, keyframe=True)
     data = c.encode(f1)  # delta of two noise frames blows the budget
     assert data[:4] == b"RZI1", "over-budget P-frame must fall back to I"
+
+
+def test_rtcp_rr_roundtrip():
+    from ai_rtc_agent_amd.media import rtcp
+
+    rr = rtcp.make_rr(1, 2, fraction_lost=0.25, cumulative_lost=100, highest_seq=5000)
+    assert rtcp.is_rtcp(rr)
+    ssrc, frac, cum = rtcp.parse_rr(rr)
+    assert ssrc == 2 and abs(frac - 0.25) < 0.01 and cum == 100
+    assert rtcp.parse_pli(rr) is None  # type discrimination
+
+
+def test_loss_feedback_adapts_bitrate():
+    """Receiver counts RTP loss -> emits RR; sender shrinks its codec's
+    bitrate under loss and grows it back when clean."""
+    import asyncio
+
+    from ai_rtc_agent_amd.config import EncoderConfig
+    from ai_rtc_agent_amd.media.rtc import PeerConnection
+    from ai_rtc_agent_amd.media.rtp import RtpPacket
+    from ai_rtc_agent_amd.media import rtcp
+
+    async def body():
+        rx = PeerConnection()
+        sent = []
+
+        class T:
+            def sendto(self, data, addr):
+                sent.append(data)
+
+        rx._transport = T()
+        rx._remote_addr = ("127.0.0.1", 1)
+        # feed 300 packets with 50% loss (every other seq missing)
+        for seq in range(0, 600, 2):
+            pkt = RtpPacket(payload_type=97, sequence_number=seq, timestamp=0,
+                            ssrc=9, payload=b"\x00\x00x")
+            rx._on_datagram(pkt.serialize(), ("127.0.0.1", 1))
+        rrs = [d for d in sent if rtcp.parse_rr(d) is not None]
+        assert rrs, "receiver must emit RRs"
+        _, frac, _ = rtcp.parse_rr(rrs[-1])
+        assert frac > 0.3, f"loss fraction should reflect ~50% loss, got {frac}"
+
+        tx = PeerConnection()
+        tx._encoder.cfg = EncoderConfig(default_bitrate=4_000_000,
+                                        min_bitrate=500_000, max_bitrate=8_000_000)
+        tx._on_datagram(rrs[-1], ("127.0.0.1", 2))
+        assert tx._encoder.cfg.default_bitrate < 4_000_000, "lossy RR must shrink bitrate"
+        clean = rtcp.make_rr(1, 9, 0.0, 0, 1000)
+        for _ in range(3):
+            tx._on_datagram(clean, ("127.0.0.1", 2))
+        assert tx._encoder.cfg.default_bitrate > 0.7 * 4_000_000 * 1.05
+
+    loop = asyncio.new_event_loop()
+    try:
+        loop.run_until_complete(body())
+    finally:
+        loop.close()
