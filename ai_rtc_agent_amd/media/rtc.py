@@ -97,7 +97,7 @@ class PeerConnection:
         self._packetizer = RtpPacketizer(ssrc=random.randint(1, 2**31))
         self._defrag = RtpDefragmenter()
         self._h264_depack = H264Depacketizer()
-        self._rx_nals: list = []
+        self._rx_au: Dict[int, dict] = {}
         self._force_keyframe = False
         self._decode_misses = 0
         self._last_pli = 0.0
@@ -262,15 +262,25 @@ class PeerConnection:
             self._set_state("connected")
         self._account_rx(pkt)
         if getattr(self._decoder, "rtp_mode", "raw") == "rfc6184":
-            nal = self._h264_depack.push(pkt.payload)
-            if nal is not None:
-                self._rx_nals.append(nal)
-                if len(self._rx_nals) > 64:  # lost marker: drop the stale AU
-                    self._rx_nals = self._rx_nals[-1:]
-            if not pkt.marker or not self._rx_nals:
+            # buffer the access unit's packets per timestamp and reassemble
+            # in SEQUENCE order on the marker (packets may arrive reordered)
+            au = self._rx_au.setdefault(pkt.timestamp, {})
+            au[pkt.sequence_number] = (pkt.payload, pkt.marker)
+            if len(self._rx_au) > 8:  # stale AUs (lost markers): drop oldest
+                for old in sorted(self._rx_au)[:-4]:
+                    self._rx_au.pop(old, None)
+            if not pkt.marker:
                 return
-            frame_bytes = join_annexb(self._rx_nals)
-            self._rx_nals = []
+            self._h264_depack = H264Depacketizer()
+            nals = []
+            for seq in sorted(au):
+                nal = self._h264_depack.push(au[seq][0])
+                if nal is not None:
+                    nals.append(nal)
+            self._rx_au.pop(pkt.timestamp, None)
+            if not nals:
+                return
+            frame_bytes = join_annexb(nals)
         else:
             frame_bytes = self._defrag.push(pkt)
         if frame_bytes is None:

@@ -175,3 +175,46 @@ def test_rfc6184_transport_path():
         loop.run_until_complete(body())
     finally:
         loop.close()
+
+
+def test_rfc6184_reordered_packets():
+    """The receive path must reassemble an access unit whose RTP packets
+    arrive out of order (sequence-sorted per timestamp)."""
+    import asyncio
+    import random as rnd
+
+    import torch
+
+    from ai_rtc_agent_amd.media.h264 import packetize_h264, split_annexb
+    from ai_rtc_agent_amd.media.rtc import PeerConnection
+    from ai_rtc_agent_amd.media.rtp import RtpPacket
+
+    class MockDec:
+        rtp_mode = "rfc6184"
+
+        def decode(self, data):
+            self.decoded = data
+            return torch.zeros(8, 8, 3, dtype=torch.uint8)
+
+    async def body():
+        rx = PeerConnection()
+        rx._decoder = MockDec()
+        stream = b"\x00\x00\x00\x01\x67\x42" + b"\x00\x00\x00\x01" + b"\x65" + bytes(range(250)) * 20
+        payloads = packetize_h264(split_annexb(stream))
+        pkts = [RtpPacket(payload_type=97, sequence_number=i, timestamp=3000,
+                          ssrc=5, marker=1 if i == len(payloads) - 1 else 0,
+                          payload=pl).serialize()
+                for i, pl in enumerate(payloads)]
+        order = list(range(len(pkts)))
+        # shuffle everything except keep the marker last (it triggers flush)
+        body_idx = order[:-1]
+        rnd.Random(1).shuffle(body_idx)
+        for i in body_idx + [order[-1]]:
+            rx._on_datagram(pkts[i], ("127.0.0.1", 1))
+        assert getattr(rx._decoder, "decoded", None) == stream
+
+    loop = asyncio.new_event_loop()
+    try:
+        loop.run_until_complete(body())
+    finally:
+        loop.close()
