@@ -142,6 +142,7 @@ async def whep(request: web.Request) -> web.Response:
 
     pc = PeerConnection()
     st["pcs"].add(pc)
+    st.setdefault("whep_pcs", set()).add(pc)
     # relay fan-out so N viewers share one pipeline pull (the reference
     # attaches the track directly and leaves its MediaRelay unused,
     # agent.py:248-252 — with >1 viewer they would steal frames from each
@@ -152,6 +153,7 @@ async def whep(request: web.Request) -> web.Response:
     def on_state() -> None:
         if pc.connection_state in ("failed", "closed"):
             st["pcs"].discard(pc)
+            st.get("whep_pcs", set()).discard(pc)
 
     await pc.set_remote_description(offer_sdp)
     sdp = await pc.create_answer(host=request.app["host"], direction="sendonly")
@@ -162,6 +164,13 @@ async def whep(request: web.Request) -> web.Response:
 
 
 async def whep_delete(request: web.Request) -> web.Response:
+    """Close every WHEP subscriber PC (the WHEP resource URL carries no
+    per-session id in this minimal server, so DELETE tears down all)."""
+    st = _state(request.app)
+    for pc in list(st.get("whep_pcs", ())):
+        await pc.close()
+        st["pcs"].discard(pc)
+        st["whep_pcs"].discard(pc)
     return web.Response(status=200)
 
 

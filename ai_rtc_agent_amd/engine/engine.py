@@ -305,6 +305,12 @@ class StreamDiffusionEngine:
         if t_index_list == self.cfg.t_index_list:
             return
         if len(t_index_list) != len(self.cfg.t_index_list):
+            # prepare() reallocates graph-captured buffers (_coeff, _x_t_buffer,
+            # _frame_in) and drops the graphs; quiesce first so in-flight
+            # pipelined replays on streams A/B are not reading freed memory
+            # (same rule refresh_weights follows).
+            if self.device.type == "cuda":
+                torch.cuda.synchronize(self.device)
             self.cfg.t_index_list = t_index_list
             self.prepare()
             return

@@ -152,7 +152,15 @@ class SoftwareCodec:
             src = src[::2, ::2]
             scale += 1
             body = zlib.compress(src.tobytes(), 9)
-        self._enc_prev = arr
+        # reconstructed-reference rule: the encoder's prediction reference
+        # must be what the DECODER will hold — for a downscaled keyframe
+        # that is the nearest-upsampled reconstruction, not the original
+        # (otherwise every following P-frame carries a persistent error).
+        if scale:
+            recon = src.repeat(1 << scale, axis=0).repeat(1 << scale, axis=1)[:h, :w]
+            self._enc_prev = np.ascontiguousarray(recon)
+        else:
+            self._enc_prev = arr
         hdr = struct.pack("!HHB", h, w, scale)
         return _MAGIC_I + hdr + body
 

@@ -267,13 +267,35 @@ class PeerConnection:
             au = self._rx_au.setdefault(pkt.timestamp, {})
             au[pkt.sequence_number] = (pkt.payload, pkt.marker)
             if len(self._rx_au) > 8:  # stale AUs (lost markers): drop oldest
-                for old in sorted(self._rx_au)[:-4]:
+                # order by SIGNED wrapped distance from the first-seen AU
+                # timestamp so the drop survives the 32-bit timestamp wrap
+                ts_base = next(iter(self._rx_au))
+                for old in sorted(
+                    self._rx_au,
+                    key=lambda t: ((t - ts_base + 0x80000000) & 0xFFFFFFFF)
+                    - 0x80000000,
+                )[:-4]:
                     self._rx_au.pop(old, None)
-            if not pkt.marker:
+            # reassemble only once the AU is COMPLETE: its (unwrapped)
+            # sequence numbers form a contiguous run whose last packet
+            # carries the marker. A marker that arrives before reordered
+            # fragments keeps the AU buffered instead of dropping them
+            # (the stale-AU GC above bounds how long we wait).
+            if not any(m for (_, m) in au.values()):
                 return
+            # SIGNED wrapped distance from the first-seen packet: correct
+            # ordering across the 0xFFFF->0 wrap and for late low-seq arrivals
+            seq_base = next(iter(au))
+            ordered = sorted(
+                au, key=lambda s: ((s - seq_base + 0x8000) & 0xFFFF) - 0x8000
+            )
+            offs = [((s - seq_base + 0x8000) & 0xFFFF) - 0x8000 for s in ordered]
+            if offs != list(range(offs[0], offs[0] + len(offs))) \
+                    or not au[ordered[-1]][1]:
+                return  # gap mid-AU or marker not last yet: wait
             self._h264_depack = H264Depacketizer()
             nals = []
-            for seq in sorted(au):
+            for seq in ordered:
                 nal = self._h264_depack.push(au[seq][0])
                 if nal is not None:
                     nals.append(nal)

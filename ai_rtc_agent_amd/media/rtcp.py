@@ -36,19 +36,21 @@ PT_RR = 201
 
 def make_rr(sender_ssrc: int, media_ssrc: int, fraction_lost: float,
             cumulative_lost: int, highest_seq: int) -> bytes:
+    # length=7 -> 32 bytes total: header word + sender SSRC + a full RFC 3550
+    # report block (SSRC, lost, highest seq, jitter, LSR, DLSR)
     header = bytes([(2 << 6) | 1, PT_RR]) + struct.pack("!H", 7)
     fl = min(255, max(0, int(fraction_lost * 256)))
     cl = min(0xFFFFFF, max(0, cumulative_lost))
-    block = struct.pack("!IIBBHIII", sender_ssrc & 0xFFFFFFFF,
+    block = struct.pack("!IIBBHIIII", sender_ssrc & 0xFFFFFFFF,
                         media_ssrc & 0xFFFFFFFF, fl,
                         (cl >> 16) & 0xFF, cl & 0xFFFF,
-                        highest_seq & 0xFFFFFFFF, 0, 0)
+                        highest_seq & 0xFFFFFFFF, 0, 0, 0)
     return header + block
 
 
 def parse_rr(data: bytes):
     """Returns (media_ssrc, fraction_lost [0..1], cumulative_lost) or None."""
-    if not is_rtcp(data) or data[1] != PT_RR or len(data) < 28:
+    if not is_rtcp(data) or data[1] != PT_RR or len(data) < 32:
         return None
-    _, media_ssrc, fl, cl_hi, cl_lo, _, _, _ = struct.unpack("!IIBBHIII", data[4:28])
+    _, media_ssrc, fl, cl_hi, cl_lo, _, _, _, _ = struct.unpack("!IIBBHIIII", data[4:32])
     return media_ssrc, fl / 256.0, (cl_hi << 16) | cl_lo

@@ -190,37 +190,52 @@ def load_diffusers_unet(
     return loaded
 
 
+# TAESD sequential layouts. madebyollin/taesd ships two namings for the SAME
+# flat-Sequential architecture:
+#   raw   ("taesd_encoder/decoder.safetensors"):  encoder.N... / decoder.N...
+#   diffusers AutoencoderTiny ("diffusion_pytorch_model.safetensors" at the
+#   snapshot TOP LEVEL): encoder.layers.N... / decoder.layers.N...
+# Encoder indices coincide (no parameter-free modules before conv_out);
+# decoder indices differ by 1 (raw has a Clamp at 0 and a ReLU at 2, diffusers
+# folds both into forward()). Each entry: (seq index, our module, is_block).
 TAESD_SEQ_ENCODER = [
-    ("0", "conv_in"),
-    ("1", "stage1"),
-    ("2", "down1"), ("3", "stage2.0"), ("4", "stage2.1"), ("5", "stage2.2"),
-    ("6", "down2"), ("7", "stage3.0"), ("8", "stage3.1"), ("9", "stage3.2"),
-    ("10", "down3"), ("11", "stage4.0"), ("12", "stage4.1"), ("13", "stage4.2"),
-    ("14", "conv_out"),
+    (0, "conv_in", False),
+    (1, "stage1", True),
+    (2, "down1", False), (3, "stage2.0", True), (4, "stage2.1", True), (5, "stage2.2", True),
+    (6, "down2", False), (7, "stage3.0", True), (8, "stage3.1", True), (9, "stage3.2", True),
+    (10, "down3", False), (11, "stage4.0", True), (12, "stage4.1", True), (13, "stage4.2", True),
+    (14, "conv_out", False),
 ]
+TAESD_SEQ_DECODER_RAW = [
+    (1, "conv_in", False),
+    (3, "stage1.0", True), (4, "stage1.1", True), (5, "stage1.2", True),
+    (7, "up1", False),
+    (8, "stage2.0", True), (9, "stage2.1", True), (10, "stage2.2", True),
+    (12, "up2", False),
+    (13, "stage3.0", True), (14, "stage3.1", True), (15, "stage3.2", True),
+    (17, "up3", False),
+    (18, "stage4", True),
+    (19, "conv_out", False),
+]
+TAESD_SEQ_DECODER_DIFFUSERS = [(i - 1, dst, b) for i, dst, b in TAESD_SEQ_DECODER_RAW]
 
 
-def load_taesd_encoder(model, sd: Dict[str, torch.Tensor], prefix: str = "encoder.") -> int:
-    """TAESD encoder weights (madebyollin/taesd layout: a flat nn.Sequential
-    with _Block sub-Sequentials conv.0/conv.2/conv.4)."""
+def _load_taesd_seq(model, sd: Dict[str, torch.Tensor], prefix: str, table) -> int:
+    """Copy a flat-Sequential TAESD state dict into one of our TAESD halves.
+
+    Blocks are madebyollin _Block / diffusers AutoencoderTinyBlock: a `conv`
+    Sequential whose parameterized entries are conv.0/conv.2/conv.4 → our
+    c1/c2/c3."""
     own = dict(model.state_dict())
     n = 0
-    for src_i, dst in TAESD_SEQ_ENCODER:
-        if "stage" in dst and dst.count(".") == 1:
-            base = f"{prefix}{src_i}.conv"
+    for src_i, dst, is_block in table:
+        if is_block:
             for ci, our in (("0", "c1"), ("2", "c2"), ("4", "c3")):
                 for p in ("weight", "bias"):
-                    k = f"{base}.{ci}.{p}"
-                    if k in sd:
-                        own[f"{dst}.{our}.{p}"].copy_(sd[k].to(own[f"{dst}.{our}.{p}"].dtype))
-                        n += 1
-        elif dst == "stage1":
-            base = f"{prefix}{src_i}.conv"
-            for ci, our in (("0", "c1"), ("2", "c2"), ("4", "c3")):
-                for p in ("weight", "bias"):
-                    k = f"{base}.{ci}.{p}"
-                    if k in sd:
-                        own[f"stage1.{our}.{p}"].copy_(sd[k].to(own[f"stage1.{our}.{p}"].dtype))
+                    k = f"{prefix}{src_i}.conv.{ci}.{p}"
+                    ours = f"{dst}.{our}.{p}"
+                    if k in sd and ours in own:
+                        own[ours].copy_(sd[k].to(own[ours].dtype))
                         n += 1
         else:
             for p in ("weight", "bias"):
@@ -232,23 +247,85 @@ def load_taesd_encoder(model, sd: Dict[str, torch.Tensor], prefix: str = "encode
     return n
 
 
+def load_taesd_encoder(model, sd: Dict[str, torch.Tensor], prefix: str = "encoder.") -> int:
+    if any(k.startswith(prefix + "layers.") for k in sd):
+        prefix = prefix + "layers."
+    return _load_taesd_seq(model, sd, prefix, TAESD_SEQ_ENCODER)
+
+
+def load_taesd_decoder(model, sd: Dict[str, torch.Tensor], prefix: str = "decoder.") -> int:
+    if any(k.startswith(prefix + "layers.") for k in sd):
+        return _load_taesd_seq(model, sd, prefix + "layers.", TAESD_SEQ_DECODER_DIFFUSERS)
+    return _load_taesd_seq(model, sd, prefix, TAESD_SEQ_DECODER_RAW)
+
+
+def load_taesd(vae, sd: Dict[str, torch.Tensor]) -> Tuple[int, int]:
+    """Load both halves of a TinyVAE from a combined TAESD state dict
+    (either naming scheme). Returns (n_encoder, n_decoder) tensors loaded."""
+    return (
+        load_taesd_encoder(vae.encoder, sd, prefix="encoder."),
+        load_taesd_decoder(vae.decoder, sd, prefix="decoder."),
+    )
+
+
+def _is_autoencoder_kl(sd: Dict[str, torch.Tensor]) -> bool:
+    return any(k.startswith("encoder.down_blocks.") for k in sd)
+
+
 def load_model_dir(engine, model_dir: str) -> bool:
-    """Load UNet (+ optional TAESD) safetensors from a local diffusers-style
-    directory; returns False when nothing was found (random init stays)."""
+    """Load UNet (+ TAESD enc+dec) safetensors from a local diffusers-style
+    directory; returns False when nothing was found (random init stays).
+
+    TAESD probe order matches what download.py actually fetches for
+    madebyollin/taesd: the diffusers AutoencoderTiny file sits at the
+    snapshot TOP LEVEL; the raw split files are taesd_encoder/decoder;
+    dreamshaper-8's vae/ subdir is a full AutoencoderKL that cannot populate
+    a TinyVAE — it is detected and skipped with a warning."""
+    import logging
+
     from safetensors.torch import load_file
 
+    log = logging.getLogger(__name__)
     found = False
     for sub in ("unet/diffusion_pytorch_model.safetensors", "unet.safetensors"):
         p = os.path.join(model_dir, sub)
         if os.path.exists(p):
-            load_diffusers_unet(engine.unet, load_file(p), strict=False)
-            found = True
+            n = load_diffusers_unet(engine.unet, load_file(p), strict=False)
+            if n == 0:
+                log.warning("UNet checkpoint %s matched 0 tensors", p)
+            else:
+                found = True
             break
-    for sub in ("taesd.safetensors", "vae/diffusion_pytorch_model.safetensors"):
+    n_enc = n_dec = 0
+    for sub in (
+        "taesd.safetensors",
+        "diffusion_pytorch_model.safetensors",  # AutoencoderTiny, top level
+        "vae/diffusion_pytorch_model.safetensors",
+    ):
         p = os.path.join(model_dir, sub)
-        if os.path.exists(p):
-            sd = load_file(p)
-            load_taesd_encoder(engine.vae.encoder, sd, prefix="encoder.")
+        if not os.path.exists(p):
+            continue
+        sd = load_file(p)
+        if _is_autoencoder_kl(sd):
+            log.warning("%s is a full AutoencoderKL, not TAESD; skipping "
+                        "(fetch madebyollin/taesd for the TinyVAE weights)", p)
+            continue
+        n_enc, n_dec = load_taesd(engine.vae, sd)
+        if n_enc or n_dec:
             found = True
             break
+    # raw split files (madebyollin/taesd also ships these)
+    if n_enc == 0:
+        p = os.path.join(model_dir, "taesd_encoder.safetensors")
+        if os.path.exists(p):
+            n_enc = load_taesd_encoder(engine.vae.encoder, load_file(p), prefix="")
+            found = found or n_enc > 0
+    if n_dec == 0:
+        p = os.path.join(model_dir, "taesd_decoder.safetensors")
+        if os.path.exists(p):
+            n_dec = load_taesd_decoder(engine.vae.decoder, load_file(p), prefix="")
+            found = found or n_dec > 0
+    if found and n_enc == 0 and n_dec == 0:
+        log.warning("model dir %s: UNet loaded but no TAESD weights matched — "
+                    "VAE stays random-init", model_dir)
     return found

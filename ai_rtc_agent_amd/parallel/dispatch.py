@@ -79,7 +79,28 @@ class PipelinePool:
     def active(self) -> List:
         return self._pipelines
 
+    @staticmethod
+    def _device_info(p) -> dict | None:
+        dev = getattr(getattr(p, "cfg", None), "device", None)
+        if dev is None or not str(dev).startswith("cuda") or not torch.cuda.is_available():
+            return None
+        d = torch.device(dev)
+        return {
+            "device": str(d),
+            "name": torch.cuda.get_device_name(d),
+            "memory_allocated_mb": round(torch.cuda.memory_allocated(d) / 1e6, 1),
+        }
+
     def stats(self) -> dict:
+        per_replica = []
+        for i, p in enumerate(self._pipelines):
+            entry = {"load": self._load[i]}
+            info = self._device_info(p)
+            if info:
+                entry.update(info)
+            if hasattr(p, "stats"):
+                entry.update(p.stats())
+            per_replica.append(entry)
         device = None
         if torch.cuda.is_available():
             device = {
@@ -90,8 +111,5 @@ class PipelinePool:
             "device": device,
             "replicas": len(self._pipelines),
             "sessions": {k: v for k, v in self._sessions.items()},
-            "per_replica": [
-                {"load": self._load[i], **(p.stats() if hasattr(p, "stats") else {})}
-                for i, p in enumerate(self._pipelines)
-            ],
+            "per_replica": per_replica,
         }
