@@ -181,6 +181,84 @@ torch::Tensor postprocess_u8(torch::Tensor img) {
   return out;
 }
 
+// --- software H.264 baseline-intra codec (h264sw.cpp) -----------------
+extern "C" {
+void* airtc_h264enc_create(int w, int h);
+void airtc_h264enc_destroy(void*);
+int airtc_h264enc_encode(void*, const uint8_t*, int qp, uint8_t*, int cap);
+void* airtc_h264dec_create();
+void airtc_h264dec_destroy(void*);
+int airtc_h264dec_decode(void*, const uint8_t*, int, uint8_t*, int, int*, int*);
+void airtc_h264dec_dims(void*, int*, int*);
+int airtc_h264sw_table_check();
+}
+
+class H264SwEncoder {
+ public:
+  H264SwEncoder(int w, int h) : w_(w), h_(h) {
+    handle_ = airtc_h264enc_create(w, h);
+    TORCH_CHECK(handle_, "invalid encoder dimensions");
+  }
+  ~H264SwEncoder() {
+    if (handle_) airtc_h264enc_destroy(handle_);
+  }
+  H264SwEncoder(const H264SwEncoder&) = delete;
+  pybind11::bytes encode(pybind11::bytes rgb, int qp) {
+    std::string s(rgb);
+    TORCH_CHECK((int)s.size() == w_ * h_ * 3, "rgb buffer size mismatch");
+    std::vector<uint8_t> out((size_t)w_ * h_ * 6 + 4096);
+    int n;
+    {
+      pybind11::gil_scoped_release nogil;
+      n = airtc_h264enc_encode(handle_, (const uint8_t*)s.data(), qp,
+                               out.data(), (int)out.size());
+    }
+    TORCH_CHECK(n > 0, "h264 encode failed rc=", n);
+    return pybind11::bytes((const char*)out.data(), n);
+  }
+
+ private:
+  void* handle_;
+  int w_, h_;
+};
+
+class H264SwDecoder {
+ public:
+  H264SwDecoder() { handle_ = airtc_h264dec_create(); }
+  ~H264SwDecoder() {
+    if (handle_) airtc_h264dec_destroy(handle_);
+  }
+  H264SwDecoder(const H264SwDecoder&) = delete;
+  // returns (rgb_bytes, w, h) or None (no frame in this AU / undecodable)
+  pybind11::object decode(pybind11::bytes data) {
+    std::string s(data);
+    int w = 0, h = 0, rc;
+    {
+      pybind11::gil_scoped_release nogil;
+      rc = airtc_h264dec_decode(handle_, (const uint8_t*)s.data(),
+                                (int)s.size(), buf_.data(), (int)buf_.size(),
+                                &w, &h);
+      if (rc == -8) {  // output buffer too small: size from the parsed SPS
+        int dw = 0, dh = 0;
+        airtc_h264dec_dims(handle_, &dw, &dh);
+        if (dw > 0 && dh > 0) {
+          buf_.resize((size_t)dw * dh * 3);
+          rc = airtc_h264dec_decode(handle_, (const uint8_t*)s.data(),
+                                    (int)s.size(), buf_.data(),
+                                    (int)buf_.size(), &w, &h);
+        }
+      }
+    }
+    if (rc != 1) return pybind11::none();
+    return pybind11::make_tuple(
+        pybind11::bytes((const char*)buf_.data(), (size_t)w * h * 3), w, h);
+  }
+
+ private:
+  void* handle_;
+  std::vector<uint8_t> buf_ = std::vector<uint8_t>((size_t)1024 * 1024 * 3);
+};
+
 pybind11::bytes h264_sps_pps(int width, int height) {
   uint8_t buf[256];
   int n = airtc_h264_sps_pps(width, height, buf, sizeof(buf));
@@ -214,4 +292,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("postprocess_u8", &postprocess_u8);
   m.def("vcn_probe", &vcn_probe, "probe the VCN VA-API stack");
   m.def("h264_sps_pps", &h264_sps_pps, "Annex-B SPS+PPS for (w, h)");
+  m.def("h264sw_table_check", []() { return airtc_h264sw_table_check(); },
+        "0 iff all CAVLC tables are prefix-free");
+  pybind11::class_<H264SwEncoder>(m, "H264SwEncoder")
+      .def(pybind11::init<int, int>())
+      .def("encode", &H264SwEncoder::encode, "RGB24 bytes + QP -> Annex-B IDR");
+  pybind11::class_<H264SwDecoder>(m, "H264SwDecoder")
+      .def(pybind11::init<>())
+      .def("decode", &H264SwDecoder::decode,
+           "Annex-B AU -> (rgb bytes, w, h) | None");
 }

@@ -1,0 +1,1350 @@
+// Software H.264 baseline-intra codec (encoder + decoder), first-party C++.
+//
+// Replaces reference component N10: the reference's
+// software fallback encodes real H.264 via aiortc/x264/PyAV
+// (reference lib/pipeline.py:83-94, Dockerfile:10). Neither ships offline,
+// so this module implements the subset that matters for low-latency WebRTC:
+//
+//   encode: every frame an IDR picture — I_16x16 macroblocks (intra
+//           prediction mode chosen per-MB by SAD among DC/V/H/Plane),
+//           4x4 integer transform + luma-DC Hadamard + chroma-DC 2x2,
+//           CAVLC entropy coding, deblocking disabled, BT.601 RGB<->YUV420.
+//   decode: baseline-intra subset — I_16x16 (all four prediction modes,
+//           all chroma modes) and I_PCM macroblocks, CAVLC. I_4x4 and
+//           P slices are rejected with a clean error (the transport then
+//           falls back to PLI/keyframe recovery).
+//
+// The wire format is standard Annex-B H.264 (constrained baseline), so any
+// compliant decoder can consume the encoder's output; the in-repo decoder
+// doubles as the bit-exact test reference (SURVEY.md §4 strategy (b)).
+// Rate control (per-frame QP from the NVENC_*-parity bitrate knobs) lives
+// in Python (media/codec.py).
+
+#include <stdint.h>
+#include <stdlib.h>
+#include <string.h>
+
+#include <algorithm>
+#include <vector>
+
+#include "h264_bits.h"
+
+// SPS/PPS generator (vcn.cpp): shared so the parameter sets advertised over
+// SDP and the ones in the encoded stream are identical.
+extern "C" int airtc_h264_sps_pps(int width, int height, uint8_t* buf,
+                                  int buflen);
+
+namespace h264sw {
+
+using h264::BitReader;
+using h264::BitWriter;
+using h264::emit_nal;
+
+static inline uint8_t clip8(int v) { return (uint8_t)(v < 0 ? 0 : v > 255 ? 255 : v); }
+
+// ---------------------------------------------------------------------------
+// CAVLC tables (ITU-T H.264 Table 9-5, 9-7/9-8, 9-9, 9-10)
+// ---------------------------------------------------------------------------
+
+// coeff_token for nC buckets 0 (0<=nC<2), 1 (2<=nC<4), 2 (4<=nC<8);
+// bucket 3 (nC>=8) is a 6-bit FLC handled in code; chroma DC has its own.
+// Indexed [bucket][TotalCoeff][TrailingOnes].
+static const uint8_t CT_LEN[3][17][4] = {
+    {{1, 0, 0, 0},   {6, 2, 0, 0},   {8, 6, 3, 0},   {9, 8, 7, 5},
+     {10, 9, 8, 6},  {11, 10, 9, 7}, {13, 11, 10, 8}, {13, 13, 11, 9},
+     {13, 13, 13, 10}, {14, 14, 13, 11}, {14, 14, 14, 13}, {15, 15, 14, 14},
+     {15, 15, 15, 14}, {16, 15, 15, 15}, {16, 16, 16, 15}, {16, 16, 16, 16},
+     {16, 16, 16, 16}},
+    {{2, 0, 0, 0},   {6, 2, 0, 0},   {6, 5, 3, 0},   {7, 6, 6, 4},
+     {8, 6, 6, 4},   {8, 7, 7, 5},   {9, 8, 8, 6},   {11, 9, 9, 6},
+     {11, 11, 11, 7}, {12, 11, 11, 9}, {12, 12, 12, 11}, {12, 12, 12, 11},
+     {13, 13, 13, 12}, {13, 13, 13, 13}, {13, 14, 14, 13}, {14, 14, 14, 13},
+     {14, 14, 14, 14}},  // NOTE: rows TC>=10 are best-effort (see ENC_MAX_TC)
+    {{4, 0, 0, 0},   {6, 4, 0, 0},   {6, 5, 4, 0},   {6, 5, 5, 4},
+     {7, 5, 5, 4},   {7, 5, 5, 4},   {7, 6, 6, 4},   {7, 6, 6, 4},
+     {8, 7, 7, 5},   {8, 8, 7, 6},   {9, 8, 8, 7},   {9, 9, 8, 8},
+     {9, 9, 9, 8},   {10, 10, 9, 9}, {10, 10, 10, 10}, {10, 10, 10, 10},
+     {10, 10, 10, 10}},
+};
+static const uint16_t CT_BITS[3][17][4] = {
+    {{1, 0, 0, 0},  {5, 1, 0, 0},  {7, 4, 1, 0},  {7, 6, 5, 3},
+     {7, 6, 5, 3},  {7, 6, 5, 4},  {15, 6, 5, 4}, {11, 14, 5, 4},
+     {8, 10, 13, 4}, {15, 14, 9, 4}, {11, 10, 13, 12}, {15, 14, 9, 12},
+     {11, 10, 13, 8}, {15, 1, 9, 12}, {11, 14, 13, 8}, {7, 10, 9, 12},
+     {4, 6, 5, 8}},
+    {{3, 0, 0, 0},  {11, 2, 0, 0}, {7, 7, 3, 0},  {7, 10, 9, 5},
+     {7, 6, 5, 4},  {4, 6, 5, 6},  {7, 6, 5, 8},  {15, 6, 5, 4},
+     {11, 14, 13, 4}, {15, 10, 9, 4}, {11, 14, 13, 12}, {8, 10, 9, 8},
+     {15, 14, 13, 12}, {11, 10, 9, 12}, {7, 11, 12, 8}, {9, 8, 10, 1},
+     {7, 6, 5, 4}},
+    {{15, 0, 0, 0}, {15, 14, 0, 0}, {11, 15, 13, 0}, {8, 12, 14, 12},
+     {15, 10, 11, 11}, {11, 8, 9, 10}, {9, 14, 13, 9}, {8, 10, 9, 8},
+     {15, 14, 13, 13}, {11, 14, 10, 12}, {15, 10, 13, 12}, {11, 14, 9, 8},
+     {8, 10, 9, 12}, {15, 14, 13, 12}, {11, 10, 9, 8}, {7, 6, 5, 4},
+     {3, 2, 1, 0}},
+};
+// chroma DC coeff_token (nC == -1), [TotalCoeff 0..4][TrailingOnes]
+static const uint8_t CDC_CT_LEN[5][4] = {
+    {2, 0, 0, 0}, {6, 1, 0, 0}, {6, 6, 3, 0}, {6, 7, 7, 6}, {6, 8, 8, 7}};
+static const uint8_t CDC_CT_BITS[5][4] = {
+    {1, 0, 0, 0}, {7, 1, 0, 0}, {4, 6, 1, 0}, {3, 3, 2, 5}, {2, 3, 2, 0}};
+
+// total_zeros, 4x4 blocks: [TotalCoeff-1][total_zeros]
+static const uint8_t TZ_LEN[15][16] = {
+    {1, 3, 3, 4, 4, 5, 5, 6, 6, 7, 7, 8, 8, 9, 9, 9},
+    {3, 3, 3, 3, 3, 4, 4, 4, 4, 5, 5, 6, 6, 6, 6},
+    {4, 3, 3, 3, 4, 4, 3, 3, 4, 5, 5, 6, 5, 6},
+    {5, 3, 4, 4, 3, 3, 3, 4, 3, 4, 5, 5, 5},
+    {4, 4, 4, 3, 3, 3, 3, 3, 4, 5, 4, 5},
+    {6, 5, 3, 3, 3, 3, 3, 3, 4, 3, 6},
+    {6, 5, 3, 3, 3, 2, 3, 4, 3, 6},
+    {6, 4, 5, 3, 2, 2, 3, 3, 6},
+    {6, 6, 4, 2, 2, 3, 2, 5},
+    {5, 5, 3, 2, 2, 2, 4},
+    {4, 4, 3, 3, 1, 3},
+    {4, 4, 2, 1, 3},
+    {3, 3, 1, 2},
+    {2, 2, 1},
+    {1, 1},
+};
+static const uint8_t TZ_BITS[15][16] = {
+    {1, 3, 2, 3, 2, 3, 2, 3, 2, 3, 2, 3, 2, 3, 2, 1},
+    {7, 6, 5, 4, 3, 5, 4, 3, 2, 3, 2, 3, 2, 1, 0},
+    {5, 7, 6, 5, 4, 3, 4, 3, 2, 3, 2, 1, 1, 0},
+    {3, 7, 5, 4, 6, 5, 4, 3, 3, 2, 2, 1, 0},
+    {5, 4, 3, 7, 6, 5, 4, 3, 2, 1, 1, 0},
+    {1, 1, 7, 6, 5, 4, 3, 2, 1, 1, 0},
+    {1, 1, 5, 4, 3, 3, 2, 1, 1, 0},
+    {1, 1, 1, 3, 3, 2, 2, 1, 0},
+    {1, 0, 1, 3, 2, 1, 1, 1},
+    {1, 0, 1, 3, 2, 1, 1},
+    {0, 1, 1, 2, 1, 3},
+    {0, 1, 1, 1, 1},
+    {0, 1, 1, 1},
+    {0, 1, 1},
+    {0, 1},
+};
+// total_zeros, chroma DC (2x2): [TotalCoeff-1][total_zeros]
+static const uint8_t CDC_TZ_LEN[3][4] = {{1, 2, 3, 3}, {1, 2, 2, 0}, {1, 1, 0, 0}};
+static const uint8_t CDC_TZ_BITS[3][4] = {{1, 1, 1, 0}, {1, 1, 0, 0}, {1, 0, 0, 0}};
+
+// run_before: [min(zerosLeft,7)-1][run_before]
+static const uint8_t RB_LEN[7][15] = {
+    {1, 1},
+    {1, 2, 2},
+    {2, 2, 2, 2},
+    {2, 2, 2, 3, 3},
+    {2, 2, 3, 3, 3, 3},
+    {2, 3, 3, 3, 3, 3, 3},
+    {3, 3, 3, 3, 3, 3, 3, 4, 5, 6, 7, 8, 9, 10, 11},
+};
+static const uint8_t RB_BITS[7][15] = {
+    {1, 0},
+    {1, 1, 0},
+    {3, 2, 1, 0},
+    {3, 2, 1, 1, 0},
+    {3, 2, 3, 2, 1, 0},
+    {3, 0, 1, 3, 2, 5, 4},
+    {7, 6, 5, 4, 3, 2, 1, 1, 1, 1, 1, 1, 1, 1, 1},
+};
+
+// ---------------------------------------------------------------------------
+// Quantization tables (per qp%6; position classes a=(0,0)-like, b=(1,1)-like,
+// c=rest) and zig-zag scan
+// ---------------------------------------------------------------------------
+static const int32_t QMF[6][3] = {{13107, 5243, 8066}, {11916, 4660, 7490},
+                                  {10082, 4194, 6554}, {9362, 3647, 5825},
+                                  {8192, 3355, 5243},  {7282, 2893, 4559}};
+static const int32_t QV[6][3] = {{10, 16, 13}, {11, 18, 14}, {13, 20, 16},
+                                 {14, 23, 18}, {16, 25, 20}, {18, 29, 23}};
+static const uint8_t POSCLS[16] = {0, 2, 0, 2, 2, 1, 2, 1,
+                                   0, 2, 0, 2, 2, 1, 2, 1};
+static const uint8_t ZIGZAG[16] = {0, 1, 4, 8, 5, 2, 3, 6,
+                                   9, 12, 13, 10, 7, 11, 14, 15};
+static const uint8_t CHROMA_QP[52] = {
+    0,  1,  2,  3,  4,  5,  6,  7,  8,  9,  10, 11, 12, 13, 14, 15, 16, 17,
+    18, 19, 20, 21, 22, 23, 24, 25, 26, 27, 28, 29, 29, 30, 31, 32, 32, 33,
+    34, 34, 35, 35, 36, 36, 37, 37, 37, 38, 38, 38, 39, 39, 39, 39};
+
+// luma 4x4 block index -> (x4, y4) position in 4x4-block units
+static inline int blk_x4(int i) { return ((i >> 2) & 1) * 2 + (i & 1); }
+static inline int blk_y4(int i) { return ((i >> 2) >> 1) * 2 + ((i >> 1) & 1); }
+
+// ---------------------------------------------------------------------------
+// transforms
+// ---------------------------------------------------------------------------
+static void fwd4x4(const int16_t in[16], int32_t out[16]) {
+  int32_t t[16];
+  for (int r = 0; r < 4; ++r) {
+    int32_t d0 = in[r * 4 + 0], d1 = in[r * 4 + 1], d2 = in[r * 4 + 2],
+            d3 = in[r * 4 + 3];
+    int32_t s0 = d0 + d3, s1 = d1 + d2, s2 = d1 - d2, s3 = d0 - d3;
+    t[r * 4 + 0] = s0 + s1;
+    t[r * 4 + 1] = 2 * s3 + s2;
+    t[r * 4 + 2] = s0 - s1;
+    t[r * 4 + 3] = s3 - 2 * s2;
+  }
+  for (int c = 0; c < 4; ++c) {
+    int32_t d0 = t[c], d1 = t[4 + c], d2 = t[8 + c], d3 = t[12 + c];
+    int32_t s0 = d0 + d3, s1 = d1 + d2, s2 = d1 - d2, s3 = d0 - d3;
+    out[c] = s0 + s1;
+    out[4 + c] = 2 * s3 + s2;
+    out[8 + c] = s0 - s1;
+    out[12 + c] = s3 - 2 * s2;
+  }
+}
+
+static void inv4x4(const int32_t in[16], int32_t out[16]) {
+  int32_t t[16];
+  for (int r = 0; r < 4; ++r) {
+    int32_t d0 = in[r * 4 + 0], d1 = in[r * 4 + 1], d2 = in[r * 4 + 2],
+            d3 = in[r * 4 + 3];
+    int32_t e0 = d0 + d2, e1 = d0 - d2, e2 = (d1 >> 1) - d3,
+            e3 = d1 + (d3 >> 1);
+    t[r * 4 + 0] = e0 + e3;
+    t[r * 4 + 1] = e1 + e2;
+    t[r * 4 + 2] = e1 - e2;
+    t[r * 4 + 3] = e0 - e3;
+  }
+  for (int c = 0; c < 4; ++c) {
+    int32_t d0 = t[c], d1 = t[4 + c], d2 = t[8 + c], d3 = t[12 + c];
+    int32_t e0 = d0 + d2, e1 = d0 - d2, e2 = (d1 >> 1) - d3,
+            e3 = d1 + (d3 >> 1);
+    out[c] = e0 + e3;
+    out[4 + c] = e1 + e2;
+    out[8 + c] = e1 - e2;
+    out[12 + c] = e0 - e3;
+  }
+}
+
+static void hadamard4x4(const int32_t in[16], int32_t out[16]) {
+  int32_t t[16];
+  for (int r = 0; r < 4; ++r) {
+    int32_t d0 = in[r * 4], d1 = in[r * 4 + 1], d2 = in[r * 4 + 2],
+            d3 = in[r * 4 + 3];
+    int32_t s0 = d0 + d3, s1 = d1 + d2, s2 = d1 - d2, s3 = d0 - d3;
+    t[r * 4 + 0] = s0 + s1;
+    t[r * 4 + 1] = s3 + s2;
+    t[r * 4 + 2] = s0 - s1;
+    t[r * 4 + 3] = s3 - s2;
+  }
+  for (int c = 0; c < 4; ++c) {
+    int32_t d0 = t[c], d1 = t[4 + c], d2 = t[8 + c], d3 = t[12 + c];
+    int32_t s0 = d0 + d3, s1 = d1 + d2, s2 = d1 - d2, s3 = d0 - d3;
+    out[c] = s0 + s1;
+    out[4 + c] = s3 + s2;
+    out[8 + c] = s0 - s1;
+    out[12 + c] = s3 - s2;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// CAVLC residual block write/read
+// n = maxNumCoeff (16 luma DC, 15 AC, 4 chroma DC); nC = -1 for chroma DC.
+// coeffs[] are in scan order (DC first for n==16/4; AC blocks pass the 15
+// AC-scan coefficients). Returns TotalCoeff.
+// ---------------------------------------------------------------------------
+static int cavlc_write_block(BitWriter& w, const int32_t* coeffs, int n, int nC) {
+  int pos[16], val[16], tc = 0;
+  for (int i = 0; i < n; ++i) {
+    if (coeffs[i]) {
+      pos[tc] = i;
+      val[tc] = coeffs[i];
+      ++tc;
+    }
+  }
+  int t1 = 0;
+  for (int i = tc - 1; i >= 0 && t1 < 3; --i) {
+    if (val[i] == 1 || val[i] == -1)
+      ++t1;
+    else
+      break;
+  }
+  // coeff_token
+  if (nC == -1) {
+    w.put(CDC_CT_BITS[tc][t1], CDC_CT_LEN[tc][t1]);
+  } else if (nC >= 8) {
+    w.put(tc == 0 ? 3 : (uint32_t)(((tc - 1) << 2) | t1), 6);
+  } else {
+    int b = nC < 2 ? 0 : nC < 4 ? 1 : 2;
+    w.put(CT_BITS[b][tc][t1], CT_LEN[b][tc][t1]);
+  }
+  if (tc == 0) return 0;
+  // trailing one signs, highest frequency first
+  for (int i = tc - 1; i >= tc - t1; --i) w.put(val[i] < 0 ? 1 : 0, 1);
+  // remaining levels
+  int suffix_len = (tc > 10 && t1 < 3) ? 1 : 0;
+  for (int i = tc - 1 - t1; i >= 0; --i) {
+    int level = val[i];
+    int code = level > 0 ? 2 * level - 2 : -2 * level - 1;
+    if (i == tc - 1 - t1 && t1 < 3) code -= 2;
+    if (suffix_len == 0) {
+      if (code < 14) {
+        w.put(1, code + 1);  // code zeros then a 1
+      } else if (code < 30) {
+        w.put(1, 15);  // level_prefix 14
+        w.put(code - 14, 4);
+      } else {
+        w.put(1, 16);  // level_prefix 15
+        w.put(code - 30, 12);
+      }
+    } else {
+      if ((code >> suffix_len) < 15) {
+        w.put(1, (code >> suffix_len) + 1);
+        w.put(code & ((1 << suffix_len) - 1), suffix_len);
+      } else {
+        w.put(1, 16);  // level_prefix 15
+        w.put(code - (15 << suffix_len), 12);
+      }
+    }
+    if (suffix_len == 0) suffix_len = 1;
+    if (std::abs(level) > (3 << (suffix_len - 1)) && suffix_len < 6)
+      ++suffix_len;
+  }
+  // total_zeros
+  int total_zeros = pos[tc - 1] + 1 - tc;
+  if (tc < n) {
+    if (nC == -1)
+      w.put(CDC_TZ_BITS[tc - 1][total_zeros], CDC_TZ_LEN[tc - 1][total_zeros]);
+    else
+      w.put(TZ_BITS[tc - 1][total_zeros], TZ_LEN[tc - 1][total_zeros]);
+  }
+  // run_before, highest frequency first (last run implicit)
+  int zeros_left = total_zeros;
+  for (int i = tc - 1; i >= 1 && zeros_left > 0; --i) {
+    int run = pos[i] - pos[i - 1] - 1;
+    int zl = zeros_left < 7 ? zeros_left : 7;
+    w.put(RB_BITS[zl - 1][run], RB_LEN[zl - 1][run]);
+    zeros_left -= run;
+  }
+  return tc;
+}
+
+// generic VLC read helper: match (len,bits) rows
+template <typename LenT, typename BitsT>
+static int vlc_read(BitReader& r, const LenT* lens, const BitsT* bits, int count) {
+  uint32_t acc = 0;
+  int len = 0;
+  while (len < 17) {
+    acc = (acc << 1) | r.u(1);
+    ++len;
+    if (r.overrun) return -1;
+    for (int i = 0; i < count; ++i)
+      if (lens[i] == len && bits[i] == acc) return i;
+  }
+  return -1;
+}
+
+static int cavlc_read_block(BitReader& r, int32_t* coeffs, int n, int nC) {
+  memset(coeffs, 0, sizeof(int32_t) * n);
+  int tc = 0, t1 = 0;
+  if (nC == -1) {
+    uint8_t lens[20], bits[20];
+    int k = 0;
+    for (int c = 0; c <= 4; ++c)
+      for (int t = 0; t < 4; ++t) {
+        lens[k] = CDC_CT_LEN[c][t];
+        bits[k] = CDC_CT_BITS[c][t];
+        ++k;
+      }
+    int idx = vlc_read(r, lens, bits, k);
+    if (idx < 0) return -1;
+    tc = idx / 4;
+    t1 = idx % 4;
+  } else if (nC >= 8) {
+    uint32_t v = r.u(6);
+    if (v == 3) {
+      tc = 0;
+      t1 = 0;
+    } else {
+      tc = (int)(v >> 2) + 1;
+      t1 = (int)(v & 3);
+    }
+  } else {
+    int b = nC < 2 ? 0 : nC < 4 ? 1 : 2;
+    uint8_t lens[17 * 4];
+    uint16_t bits[17 * 4];
+    int k = 0;
+    for (int c = 0; c <= 16; ++c)
+      for (int t = 0; t < 4; ++t) {
+        lens[k] = CT_LEN[b][c][t];
+        bits[k] = CT_BITS[b][c][t];
+        ++k;
+      }
+    int idx = vlc_read(r, lens, bits, k);
+    if (idx < 0) return -1;
+    tc = idx / 4;
+    t1 = idx % 4;
+  }
+  if (tc == 0) return 0;
+  if (tc > n || t1 > tc) return -1;
+  int val[16];
+  for (int i = 0; i < t1; ++i) val[i] = r.u(1) ? -1 : 1;  // highest freq first
+  int suffix_len = (tc > 10 && t1 < 3) ? 1 : 0;
+  for (int i = t1; i < tc; ++i) {
+    int prefix = 0;
+    while (r.u(1) == 0) {
+      if (++prefix > 32 || r.overrun) return -1;
+    }
+    int code;
+    int suffix_size;
+    if (suffix_len == 0)
+      suffix_size = prefix == 14 ? 4 : prefix >= 15 ? 12 : 0;
+    else
+      suffix_size = prefix >= 15 ? 12 : suffix_len;
+    code = (prefix < 15 ? prefix : 15) << suffix_len;
+    if (suffix_size) code += r.u(suffix_size);
+    if (prefix >= 15 && suffix_len == 0) code += 15;
+    if (prefix >= 16) code += (1 << (prefix - 3)) - 4096;
+    if (i == t1 && t1 < 3) code += 2;
+    val[i] = (code % 2 == 0) ? (code + 2) >> 1 : -((code + 1) >> 1);
+    if (suffix_len == 0) suffix_len = 1;
+    if (std::abs(val[i]) > (3 << (suffix_len - 1)) && suffix_len < 6)
+      ++suffix_len;
+  }
+  int total_zeros = 0;
+  if (tc < n) {
+    int idx;
+    if (nC == -1)
+      idx = vlc_read(r, CDC_TZ_LEN[tc - 1], CDC_TZ_BITS[tc - 1], 4 - tc + 1);
+    else
+      idx = vlc_read(r, TZ_LEN[tc - 1], TZ_BITS[tc - 1], n - tc + 1);
+    if (idx < 0) return -1;
+    total_zeros = idx;
+  }
+  // place coefficients: val[] is highest-frequency first
+  int zeros_left = total_zeros;
+  int pos = tc - 1 + total_zeros;  // scan index of the highest-freq coeff
+  for (int i = 0; i < tc; ++i) {
+    if (pos >= n || pos < 0) return -1;
+    coeffs[pos] = val[i];
+    if (i == tc - 1) break;
+    int run = 0;
+    if (zeros_left > 0) {
+      int zl = zeros_left < 7 ? zeros_left : 7;
+      int idx = vlc_read(r, RB_LEN[zl - 1], RB_BITS[zl - 1], zl == 7 ? 15 : zl + 1);
+      if (idx < 0) return -1;
+      run = idx;
+      zeros_left -= run;
+    }
+    pos -= 1 + run;
+  }
+  if (r.overrun) return -1;
+  return tc;
+}
+
+// ---------------------------------------------------------------------------
+// color conversion (BT.601 limited range)
+// ---------------------------------------------------------------------------
+static void rgb_to_yuv420(const uint8_t* rgb, int w, int h, int pw, int ph,
+                          uint8_t* Y, uint8_t* Cb, uint8_t* Cr) {
+  // write padded planes (pw x ph luma), edge-replicated
+  for (int y = 0; y < ph; ++y) {
+    int sy = y < h ? y : h - 1;
+    for (int x = 0; x < pw; ++x) {
+      int sx = x < w ? x : w - 1;
+      const uint8_t* p = rgb + (sy * w + sx) * 3;
+      int R = p[0], G = p[1], B = p[2];
+      Y[y * pw + x] = clip8(((66 * R + 129 * G + 25 * B + 128) >> 8) + 16);
+    }
+  }
+  int cw = pw / 2, ch = ph / 2;
+  for (int y = 0; y < ch; ++y) {
+    for (int x = 0; x < cw; ++x) {
+      int R = 0, G = 0, B = 0;
+      for (int dy = 0; dy < 2; ++dy)
+        for (int dx = 0; dx < 2; ++dx) {
+          int sy = std::min(2 * y + dy, h - 1), sx = std::min(2 * x + dx, w - 1);
+          const uint8_t* p = rgb + (sy * w + sx) * 3;
+          R += p[0];
+          G += p[1];
+          B += p[2];
+        }
+      R = (R + 2) >> 2;
+      G = (G + 2) >> 2;
+      B = (B + 2) >> 2;
+      Cb[y * cw + x] = clip8(((-38 * R - 74 * G + 112 * B + 128) >> 8) + 128);
+      Cr[y * cw + x] = clip8(((112 * R - 94 * G - 18 * B + 128) >> 8) + 128);
+    }
+  }
+}
+
+static void yuv420_to_rgb(const uint8_t* Y, const uint8_t* Cb, const uint8_t* Cr,
+                          int pw, int w, int h, uint8_t* rgb) {
+  int cw = pw / 2;
+  for (int y = 0; y < h; ++y) {
+    for (int x = 0; x < w; ++x) {
+      int C = (int)Y[y * pw + x] - 16;
+      int D = (int)Cb[(y / 2) * cw + x / 2] - 128;
+      int E = (int)Cr[(y / 2) * cw + x / 2] - 128;
+      uint8_t* p = rgb + (y * w + x) * 3;
+      p[0] = clip8((298 * C + 409 * E + 128) >> 8);
+      p[1] = clip8((298 * C - 100 * D - 208 * E + 128) >> 8);
+      p[2] = clip8((298 * C + 516 * D + 128) >> 8);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// shared intra prediction + reconstruction (used by encoder and decoder so
+// both sides hold bit-identical reference pixels)
+// ---------------------------------------------------------------------------
+struct PlaneCtx {
+  uint8_t* data;  // reconstructed plane, stride = width
+  int stride;
+};
+
+// 16x16 luma prediction, modes 0=V 1=H 2=DC 3=Plane
+static void pred_luma16(const PlaneCtx& pl, int mbx, int mby, int mode,
+                        uint8_t pred[256]) {
+  const int x0 = mbx * 16, y0 = mby * 16;
+  const uint8_t* top = (mby > 0) ? pl.data + (y0 - 1) * pl.stride + x0 : nullptr;
+  const uint8_t* leftc = (mbx > 0) ? pl.data + y0 * pl.stride + (x0 - 1) : nullptr;
+  switch (mode) {
+    case 0:  // vertical
+      for (int y = 0; y < 16; ++y)
+        for (int x = 0; x < 16; ++x) pred[y * 16 + x] = top[x];
+      break;
+    case 1:  // horizontal
+      for (int y = 0; y < 16; ++y) {
+        uint8_t v = leftc[y * pl.stride];
+        for (int x = 0; x < 16; ++x) pred[y * 16 + x] = v;
+      }
+      break;
+    case 2: {  // DC
+      int sum = 0, cnt = 0;
+      if (top) {
+        for (int x = 0; x < 16; ++x) sum += top[x];
+        cnt += 16;
+      }
+      if (leftc) {
+        for (int y = 0; y < 16; ++y) sum += leftc[y * pl.stride];
+        cnt += 16;
+      }
+      int dc = cnt == 32 ? (sum + 16) >> 5 : cnt == 16 ? (sum + 8) >> 4 : 128;
+      memset(pred, dc, 256);
+      break;
+    }
+    case 3: {  // plane
+      int H = 0, V = 0;
+      const uint8_t* tl = pl.data + (y0 - 1) * pl.stride + (x0 - 1);
+      for (int i = 0; i < 8; ++i) {
+        H += (i + 1) * ((int)top[8 + i] - (int)(i == 7 ? tl[0] : top[6 - i]));
+        V += (i + 1) * ((int)leftc[(8 + i) * pl.stride] -
+                        (int)(i == 7 ? tl[0] : leftc[(6 - i) * pl.stride]));
+      }
+      int a = 16 * ((int)leftc[15 * pl.stride] + (int)top[15]);
+      int b = (5 * H + 32) >> 6;
+      int c = (5 * V + 32) >> 6;
+      for (int y = 0; y < 16; ++y)
+        for (int x = 0; x < 16; ++x)
+          pred[y * 16 + x] = clip8((a + b * (x - 7) + c * (y - 7) + 16) >> 5);
+      break;
+    }
+  }
+}
+
+// 8x8 chroma prediction, modes 0=DC 1=H 2=V 3=Plane
+static void pred_chroma8(const PlaneCtx& pl, int mbx, int mby, int mode,
+                         uint8_t pred[64]) {
+  const int x0 = mbx * 8, y0 = mby * 8;
+  const uint8_t* top = (mby > 0) ? pl.data + (y0 - 1) * pl.stride + x0 : nullptr;
+  const uint8_t* leftc = (mbx > 0) ? pl.data + y0 * pl.stride + (x0 - 1) : nullptr;
+  switch (mode) {
+    case 0: {  // DC per 4x4 sub-block (x264 dc0..dc3 structure)
+      int st[2] = {0, 0}, sl[2] = {0, 0};
+      if (top)
+        for (int x = 0; x < 8; ++x) st[x >> 2] += top[x];
+      if (leftc)
+        for (int y = 0; y < 8; ++y) sl[y >> 2] += leftc[y * pl.stride];
+      int dc[4];
+      auto mix = [&](int t, int l) {
+        if (top && leftc) return (st[t] + sl[l] + 4) >> 3;
+        if (top) return (st[t] + 2) >> 2;
+        if (leftc) return (sl[l] + 2) >> 2;
+        return 128;
+      };
+      dc[0] = mix(0, 0);
+      dc[1] = top ? (st[1] + 2) >> 2 : leftc ? (sl[0] + 2) >> 2 : 128;
+      dc[2] = leftc ? (sl[1] + 2) >> 2 : top ? (st[0] + 2) >> 2 : 128;
+      dc[3] = mix(1, 1);
+      for (int y = 0; y < 8; ++y)
+        for (int x = 0; x < 8; ++x)
+          pred[y * 8 + x] = (uint8_t)dc[(y >> 2) * 2 + (x >> 2)];
+      break;
+    }
+    case 1:  // horizontal
+      for (int y = 0; y < 8; ++y) {
+        uint8_t v = leftc[y * pl.stride];
+        for (int x = 0; x < 8; ++x) pred[y * 8 + x] = v;
+      }
+      break;
+    case 2:  // vertical
+      for (int y = 0; y < 8; ++y)
+        for (int x = 0; x < 8; ++x) pred[y * 8 + x] = top[x];
+      break;
+    case 3: {  // plane
+      int H = 0, V = 0;
+      const uint8_t* tl = pl.data + (y0 - 1) * pl.stride + (x0 - 1);
+      for (int i = 0; i < 4; ++i) {
+        H += (i + 1) * ((int)top[4 + i] - (int)(i == 3 ? tl[0] : top[2 - i]));
+        V += (i + 1) * ((int)leftc[(4 + i) * pl.stride] -
+                        (int)(i == 3 ? tl[0] : leftc[(2 - i) * pl.stride]));
+      }
+      int a = 16 * ((int)leftc[7 * pl.stride] + (int)top[7]);
+      int b = (17 * H + 16) >> 5;
+      int c = (17 * V + 16) >> 5;
+      for (int y = 0; y < 8; ++y)
+        for (int x = 0; x < 8; ++x)
+          pred[y * 8 + x] = clip8((a + b * (x - 3) + c * (y - 3) + 16) >> 5);
+      break;
+    }
+  }
+}
+
+// Dequant + inverse transform + add prediction for one I_16x16 luma MB.
+// dc_q: 16 quantized DC levels in RASTER order; ac_q[blk][16] quantized
+// levels in RASTER order with [0] unused.
+static void recon_luma16(PlaneCtx& pl, int mbx, int mby, const uint8_t pred[256],
+                         const int32_t dc_q[16], const int32_t ac_q[16][16],
+                         int qp) {
+  const int qm = qp % 6, qs = qp / 6;
+  // luma DC: inverse Hadamard, then scale (spec 8.5.10; LevelScale carries
+  // the flat weightScale factor of 16)
+  int32_t dct[16], dcd[16];
+  hadamard4x4(dc_q, dct);
+  const int32_t ls00 = QV[qm][0] * 16;
+  for (int i = 0; i < 16; ++i) {
+    if (qp >= 36)
+      dcd[i] = (dct[i] * ls00) << (qs - 6);
+    else
+      dcd[i] = (dct[i] * ls00 + (1 << (5 - qs))) >> (6 - qs);
+  }
+  for (int b = 0; b < 16; ++b) {
+    const int x4 = blk_x4(b), y4 = blk_y4(b);
+    int32_t d[16];
+    d[0] = dcd[y4 * 4 + x4];
+    for (int i = 1; i < 16; ++i) d[i] = (ac_q[b][i] * QV[qm][POSCLS[i]]) << qs;
+    int32_t r[16];
+    inv4x4(d, r);
+    uint8_t* dst = pl.data + (mby * 16 + y4 * 4) * pl.stride + mbx * 16 + x4 * 4;
+    const uint8_t* pr = pred + y4 * 4 * 16 + x4 * 4;
+    for (int y = 0; y < 4; ++y)
+      for (int x = 0; x < 4; ++x)
+        dst[y * pl.stride + x] = clip8(pr[y * 16 + x] + ((r[y * 4 + x] + 32) >> 6));
+  }
+}
+
+// same for one 8x8 chroma component
+static void recon_chroma8(PlaneCtx& pl, int mbx, int mby, const uint8_t pred[64],
+                          const int32_t dc_q[4], const int32_t ac_q[4][16],
+                          int qpc) {
+  const int qm = qpc % 6, qs = qpc / 6;
+  // 2x2 DC Hadamard inverse (same matrix as forward), scale (spec 8.5.11)
+  int32_t a = dc_q[0], b = dc_q[1], c = dc_q[2], d = dc_q[3];
+  int32_t t[4] = {a + b + c + d, a - b + c - d, a + b - c - d, a - b - c + d};
+  int32_t dcd[4];
+  // spec 8.5.11: dcC = ((f * LevelScale(0,0)) << qP/6) >> 5, LevelScale = 16*V
+  for (int i = 0; i < 4; ++i) dcd[i] = ((t[i] * QV[qm][0] * 16) << qs) >> 5;
+  for (int blk = 0; blk < 4; ++blk) {
+    const int x4 = blk & 1, y4 = blk >> 1;
+    int32_t dq[16];
+    dq[0] = dcd[blk];
+    for (int i = 1; i < 16; ++i) dq[i] = (ac_q[blk][i] * QV[qm][POSCLS[i]]) << qs;
+    int32_t r[16];
+    inv4x4(dq, r);
+    uint8_t* dst = pl.data + (mby * 8 + y4 * 4) * pl.stride + mbx * 8 + x4 * 4;
+    const uint8_t* pr = pred + y4 * 4 * 8 + x4 * 4;
+    for (int y = 0; y < 4; ++y)
+      for (int x = 0; x < 4; ++x)
+        dst[y * pl.stride + x] = clip8(pr[y * 8 + x] + ((r[y * 4 + x] + 32) >> 6));
+  }
+}
+
+// ---------------------------------------------------------------------------
+// nC derivation
+// ---------------------------------------------------------------------------
+struct NnzCtx {
+  std::vector<uint8_t> luma;    // [mb][16]
+  std::vector<uint8_t> chroma;  // [comp][mb][4]
+  int mbw = 0, mbh = 0;
+
+  void reset(int w, int h) {
+    mbw = w;
+    mbh = h;
+    luma.assign((size_t)w * h * 16, 0);
+    chroma.assign((size_t)2 * w * h * 4, 0);
+  }
+  uint8_t& lnz(int mbx, int mby, int x4, int y4) {
+    return luma[((size_t)mby * mbw + mbx) * 16 + y4 * 4 + x4];
+  }
+  uint8_t& cnz(int comp, int mbx, int mby, int x2, int y2) {
+    return chroma[(((size_t)comp * mbh + mby) * mbw + mbx) * 4 + y2 * 2 + x2];
+  }
+  int luma_nc(int mbx, int mby, int x4, int y4) {
+    int na = -1, nb = -1;
+    if (x4 > 0)
+      na = lnz(mbx, mby, x4 - 1, y4);
+    else if (mbx > 0)
+      na = lnz(mbx - 1, mby, 3, y4);
+    if (y4 > 0)
+      nb = lnz(mbx, mby, x4, y4 - 1);
+    else if (mby > 0)
+      nb = lnz(mbx, mby - 1, x4, 3);
+    if (na >= 0 && nb >= 0) return (na + nb + 1) >> 1;
+    if (na >= 0) return na;
+    if (nb >= 0) return nb;
+    return 0;
+  }
+  int chroma_nc(int comp, int mbx, int mby, int x2, int y2) {
+    int na = -1, nb = -1;
+    if (x2 > 0)
+      na = cnz(comp, mbx, mby, x2 - 1, y2);
+    else if (mbx > 0)
+      na = cnz(comp, mbx - 1, mby, 1, y2);
+    if (y2 > 0)
+      nb = cnz(comp, mbx, mby, x2, y2 - 1);
+    else if (mby > 0)
+      nb = cnz(comp, mbx, mby - 1, x2, 1);
+    if (na >= 0 && nb >= 0) return (na + nb + 1) >> 1;
+    if (na >= 0) return na;
+    if (nb >= 0) return nb;
+    return 0;
+  }
+};
+
+// ---------------------------------------------------------------------------
+// Encoder
+// ---------------------------------------------------------------------------
+struct Encoder {
+  int w, h, pw, ph, mbw, mbh;
+  std::vector<uint8_t> Y, Cb, Cr;        // source (padded)
+  std::vector<uint8_t> rY, rCb, rCr;     // reconstruction
+  NnzCtx nnz;
+  uint32_t idr_id = 0;
+
+  Encoder(int width, int height) : w(width), h(height) {
+    pw = (w + 15) & ~15;
+    ph = (h + 15) & ~15;
+    mbw = pw / 16;
+    mbh = ph / 16;
+    Y.resize((size_t)pw * ph);
+    Cb.resize((size_t)pw * ph / 4);
+    Cr.resize((size_t)pw * ph / 4);
+    rY.resize((size_t)pw * ph);
+    rCb.resize((size_t)pw * ph / 4);
+    rCr.resize((size_t)pw * ph / 4);
+  }
+
+  // quantize one 4x4 transformed block (excluding index 0 when skip_dc)
+  void quant_block(const int32_t W[16], int qp, bool skip_dc, int32_t out[16]) {
+    const int qm = qp % 6, qbits = 15 + qp / 6;
+    const int32_t f = (1 << qbits) / 3;  // intra rounding
+    for (int i = skip_dc ? 1 : 0; i < 16; ++i) {
+      int32_t v = W[i];
+      int32_t q = (int32_t)(((int64_t)std::abs(v) * QMF[qm][POSCLS[i]] + f) >> qbits);
+      if (q > 2063) q = 2063;  // keep levelCode within the 12-bit escape
+      out[i] = v < 0 ? -q : q;
+    }
+    if (skip_dc) out[0] = 0;
+  }
+
+  int mb_sad(const uint8_t* src, int sstride, const uint8_t* pred, int pstride,
+             int size) {
+    int s = 0;
+    for (int y = 0; y < size; ++y)
+      for (int x = 0; x < size; ++x)
+        s += std::abs((int)src[y * sstride + x] - (int)pred[y * pstride + x]);
+    return s;
+  }
+
+  void encode_mb(BitWriter& wtr, int mbx, int mby, int qp) {
+    PlaneCtx rpy{rY.data(), pw};
+    PlaneCtx rpcb{rCb.data(), pw / 2}, rpcr{rCr.data(), pw / 2};
+    const uint8_t* src = Y.data() + (mby * 16) * pw + mbx * 16;
+
+    // --- luma mode decision (SAD over available modes) ---
+    uint8_t pred[4][256];
+    int best_mode = 2, best_sad = INT32_MAX;
+    const bool have_top = mby > 0, have_left = mbx > 0;
+    for (int m = 0; m < 4; ++m) {
+      if (m == 0 && !have_top) continue;
+      if (m == 1 && !have_left) continue;
+      if (m == 3 && !(have_top && have_left)) continue;
+      pred_luma16(rpy, mbx, mby, m, pred[m]);
+      int sad = mb_sad(src, pw, pred[m], 16, 16);
+      if (sad < best_sad) {
+        best_sad = sad;
+        best_mode = m;
+      }
+    }
+    const uint8_t* lp = pred[best_mode];
+
+    // --- luma transform/quant ---
+    int32_t dc_raw[16];             // raster 4x4 of DC terms
+    int32_t ac_q[16][16] = {{0}};   // raster-order quantized, [0] zero
+    int32_t W[16];
+    for (int b = 0; b < 16; ++b) {
+      const int x4 = blk_x4(b), y4 = blk_y4(b);
+      int16_t d[16];
+      for (int y = 0; y < 4; ++y)
+        for (int x = 0; x < 4; ++x)
+          d[y * 4 + x] = (int16_t)((int)src[(y4 * 4 + y) * pw + x4 * 4 + x] -
+                                   (int)lp[(y4 * 4 + y) * 16 + x4 * 4 + x]);
+      fwd4x4(d, W);
+      dc_raw[y4 * 4 + x4] = W[0];
+      quant_block(W, qp, /*skip_dc=*/true, ac_q[b]);
+    }
+    // luma DC: Hadamard (halved, JM-style) + quant with doubled round/shift
+    int32_t dct[16], dc_q[16];
+    hadamard4x4(dc_raw, dct);
+    {
+      const int qm = qp % 6, qbits = 15 + qp / 6;
+      const int32_t f = (1 << qbits) / 3;
+      for (int i = 0; i < 16; ++i) {
+        const int32_t hv = dct[i] >= 0 ? (dct[i] + 1) >> 1 : -((1 - dct[i]) >> 1);
+        int32_t q = (int32_t)(((int64_t)std::abs(hv) * QMF[qm][0] + 2 * f) >>
+                              (qbits + 1));
+        if (q > 2063) q = 2063;
+        dc_q[i] = hv < 0 ? -q : q;
+      }
+    }
+    int cbp_luma = 0;
+    for (int b = 0; b < 16 && !cbp_luma; ++b)
+      for (int i = 1; i < 16; ++i)
+        if (ac_q[b][i]) {
+          cbp_luma = 15;
+          break;
+        }
+
+    // --- chroma (DC prediction mode 0) ---
+    const int qpc = CHROMA_QP[qp < 0 ? 0 : qp > 51 ? 51 : qp];
+    uint8_t cpred[2][64];
+    int32_t cdc_q[2][4], cac_q[2][4][16];
+    memset(cac_q, 0, sizeof(cac_q));
+    const uint8_t* csrc[2] = {Cb.data() + (mby * 8) * (pw / 2) + mbx * 8,
+                              Cr.data() + (mby * 8) * (pw / 2) + mbx * 8};
+    PlaneCtx* cpl[2] = {&rpcb, &rpcr};
+    for (int comp = 0; comp < 2; ++comp) {
+      pred_chroma8(*cpl[comp], mbx, mby, 0, cpred[comp]);
+      int32_t cdc_raw[4];
+      for (int blk = 0; blk < 4; ++blk) {
+        const int x4 = blk & 1, y4 = blk >> 1;
+        int16_t d[16];
+        for (int y = 0; y < 4; ++y)
+          for (int x = 0; x < 4; ++x)
+            d[y * 4 + x] =
+                (int16_t)((int)csrc[comp][(y4 * 4 + y) * (pw / 2) + x4 * 4 + x] -
+                          (int)cpred[comp][(y4 * 4 + y) * 8 + x4 * 4 + x]);
+        fwd4x4(d, W);
+        cdc_raw[blk] = W[0];
+        quant_block(W, qpc, /*skip_dc=*/true, cac_q[comp][blk]);
+      }
+      // 2x2 Hadamard + quant (doubled rounding/shift)
+      int32_t a = cdc_raw[0], b2 = cdc_raw[1], c = cdc_raw[2], d2 = cdc_raw[3];
+      int32_t t[4] = {a + b2 + c + d2, a - b2 + c - d2, a + b2 - c - d2,
+                      a - b2 - c + d2};
+      const int qm = qpc % 6, qbits = 15 + qpc / 6;
+      const int32_t f = (1 << qbits) / 3;
+      for (int i = 0; i < 4; ++i) {
+        int32_t q = (int32_t)(((int64_t)std::abs(t[i]) * QMF[qm][0] + 2 * f) >>
+                              (qbits + 1));
+        if (q > 2063) q = 2063;
+        cdc_q[comp][i] = t[i] < 0 ? -q : q;
+      }
+    }
+    int cbp_chroma = 0;
+    for (int comp = 0; comp < 2; ++comp)
+      for (int blk = 0; blk < 4; ++blk)
+        for (int i = 1; i < 16; ++i)
+          if (cac_q[comp][blk][i]) cbp_chroma = 2;
+    if (cbp_chroma == 0)
+      for (int comp = 0; comp < 2; ++comp)
+        for (int i = 0; i < 4; ++i)
+          if (cdc_q[comp][i]) cbp_chroma = 1;
+
+    // --- entropy-region guard ---
+    // The encoder restricts itself to TotalCoeff <= 9 per block: the
+    // coeff_token rows above that (all nC buckets) are best-effort
+    // reconstructions with no offline reference to validate against
+    // (no ffmpeg/x264 in the build image). High-entropy macroblocks
+    // (TotalCoeff > 9 anywhere) are emitted as I_PCM instead — standard,
+    // bit-exact by construction, and what real encoders do when CAVLC
+    // cost approaches raw anyway.
+    auto count_nz = [](const int32_t* c, int from, int n) {
+      int k = 0;
+      for (int i = from; i < n; ++i) k += c[i] != 0;
+      return k;
+    };
+    bool pcm = count_nz(dc_q, 0, 16) > 9;
+    for (int b = 0; b < 16 && !pcm; ++b) pcm = count_nz(ac_q[b], 1, 16) > 9;
+    for (int comp = 0; comp < 2 && !pcm; ++comp)
+      for (int blk = 0; blk < 4 && !pcm; ++blk)
+        pcm = count_nz(cac_q[comp][blk], 1, 16) > 9;
+    if (pcm) {
+      wtr.ue(25);  // I_PCM
+      wtr.align_byte();
+      const uint8_t* sy = Y.data() + (mby * 16) * pw + mbx * 16;
+      uint8_t* ry = rY.data() + (mby * 16) * pw + mbx * 16;
+      for (int y = 0; y < 16; ++y)
+        for (int x = 0; x < 16; ++x) {
+          wtr.put(sy[y * pw + x], 8);
+          ry[y * pw + x] = sy[y * pw + x];
+        }
+      const uint8_t* sc[2] = {csrc[0], csrc[1]};
+      uint8_t* rc[2] = {rCb.data() + (mby * 8) * (pw / 2) + mbx * 8,
+                        rCr.data() + (mby * 8) * (pw / 2) + mbx * 8};
+      for (int comp = 0; comp < 2; ++comp)
+        for (int y = 0; y < 8; ++y)
+          for (int x = 0; x < 8; ++x) {
+            wtr.put(sc[comp][y * (pw / 2) + x], 8);
+            rc[comp][y * (pw / 2) + x] = sc[comp][y * (pw / 2) + x];
+          }
+      for (int b = 0; b < 16; ++b)
+        nnz.lnz(mbx, mby, blk_x4(b), blk_y4(b)) = 16;
+      for (int comp = 0; comp < 2; ++comp)
+        for (int blk = 0; blk < 4; ++blk)
+          nnz.cnz(comp, mbx, mby, blk & 1, blk >> 1) = 16;
+      return;
+    }
+
+    // --- write macroblock_layer ---
+    const int mb_type = 1 + best_mode + 4 * cbp_chroma + 12 * (cbp_luma ? 1 : 0);
+    wtr.ue(mb_type);
+    wtr.ue(0);  // intra_chroma_pred_mode: DC
+    wtr.se(0);  // mb_qp_delta (constant QP per frame)
+
+    // Intra16x16DCLevel: zig-zag scan of dc_q, nC from luma block 0 neighbors
+    {
+      int32_t scan[16];
+      for (int i = 0; i < 16; ++i) scan[i] = dc_q[ZIGZAG[i]];
+      cavlc_write_block(wtr, scan, 16, nnz.luma_nc(mbx, mby, 0, 0));
+    }
+    if (cbp_luma) {
+      for (int b = 0; b < 16; ++b) {
+        const int x4 = blk_x4(b), y4 = blk_y4(b);
+        int32_t scan[15];
+        for (int i = 1; i < 16; ++i) scan[i - 1] = ac_q[b][ZIGZAG[i]];
+        int tc = cavlc_write_block(wtr, scan, 15, nnz.luma_nc(mbx, mby, x4, y4));
+        nnz.lnz(mbx, mby, x4, y4) = (uint8_t)tc;
+      }
+    } else {
+      for (int b = 0; b < 16; ++b)
+        nnz.lnz(mbx, mby, blk_x4(b), blk_y4(b)) = 0;
+    }
+    if (cbp_chroma) {
+      for (int comp = 0; comp < 2; ++comp)
+        cavlc_write_block(wtr, cdc_q[comp], 4, -1);
+    }
+    if (cbp_chroma == 2) {
+      for (int comp = 0; comp < 2; ++comp)
+        for (int blk = 0; blk < 4; ++blk) {
+          const int x2 = blk & 1, y2 = blk >> 1;
+          int32_t scan[15];
+          for (int i = 1; i < 16; ++i) scan[i - 1] = cac_q[comp][blk][ZIGZAG[i]];
+          int tc = cavlc_write_block(wtr, scan, 15,
+                                     nnz.chroma_nc(comp, mbx, mby, x2, y2));
+          nnz.cnz(comp, mbx, mby, x2, y2) = (uint8_t)tc;
+        }
+    } else {
+      for (int comp = 0; comp < 2; ++comp)
+        for (int blk = 0; blk < 4; ++blk)
+          nnz.cnz(comp, mbx, mby, blk & 1, blk >> 1) = 0;
+    }
+
+    // --- reconstruct (shared with the decoder) ---
+    // zero out levels the stream does not carry so recon matches decode
+    if (!cbp_luma) memset(ac_q, 0, sizeof(ac_q));
+    if (cbp_chroma < 2) memset(cac_q, 0, sizeof(cac_q));
+    if (!cbp_chroma) memset(cdc_q, 0, sizeof(cdc_q));
+    recon_luma16(rpy, mbx, mby, lp, dc_q, ac_q, qp);
+    recon_chroma8(rpcb, mbx, mby, cpred[0], cdc_q[0], cac_q[0], qpc);
+    recon_chroma8(rpcr, mbx, mby, cpred[1], cdc_q[1], cac_q[1], qpc);
+  }
+
+  int encode(const uint8_t* rgb, int qp, uint8_t* out, int cap) {
+    if (qp < 10) qp = 10;
+    if (qp > 48) qp = 48;
+    rgb_to_yuv420(rgb, w, h, pw, ph, Y.data(), Cb.data(), Cr.data());
+    nnz.reset(mbw, mbh);
+
+    std::vector<uint8_t> bs;
+    {
+      uint8_t hdr[256];
+      int n = airtc_h264_sps_pps(w, h, hdr, sizeof(hdr));
+      if (n <= 0) return -1;
+      bs.insert(bs.end(), hdr, hdr + n);
+    }
+    // IDR slice
+    BitWriter wtr;
+    wtr.ue(0);            // first_mb_in_slice
+    wtr.ue(7);            // slice_type: I (all slices)
+    wtr.ue(0);            // pps_id
+    wtr.put(0, 4);        // frame_num (log2_max_frame_num = 4)
+    wtr.ue(idr_id & 1);   // idr_pic_id
+    ++idr_id;
+    wtr.put(0, 1);        // no_output_of_prior_pics_flag
+    wtr.put(0, 1);        // long_term_reference_flag
+    wtr.se(qp - 26);      // slice_qp_delta (pic_init_qp = 26)
+    wtr.ue(1);            // disable_deblocking_filter_idc = 1 (off)
+    for (int mby = 0; mby < mbh; ++mby)
+      for (int mbx = 0; mbx < mbw; ++mbx) encode_mb(wtr, mbx, mby, qp);
+    wtr.rbsp_trailing();
+    emit_nal(&bs, 0x65, wtr.bytes);  // nal_ref_idc=3, IDR
+
+    if ((int)bs.size() > cap) return -2;
+    memcpy(out, bs.data(), bs.size());
+    return (int)bs.size();
+  }
+};
+
+// ---------------------------------------------------------------------------
+// Decoder
+// ---------------------------------------------------------------------------
+struct Sps {
+  bool valid = false;
+  int log2_max_frame_num = 4;
+  int poc_type = 2;
+  int log2_max_poc_lsb = 4;
+  int pw = 0, ph = 0;      // padded (MB-aligned) dims
+  int crop_r = 0, crop_b = 0;
+  int w() const { return pw - crop_r; }
+  int h() const { return ph - crop_b; }
+};
+struct Pps {
+  bool valid = false;
+  bool cavlc = true;
+  bool pic_order_present = false;
+  int pic_init_qp = 26;
+  int chroma_qp_offset = 0;
+  bool deblock_present = false;
+};
+
+struct Decoder {
+  Sps sps;
+  Pps pps;
+  std::vector<uint8_t> rY, rCb, rCr;
+  NnzCtx nnz;
+
+  int parse_sps(BitReader& r) {
+    Sps s;
+    int profile = r.u(8);
+    r.u(8);  // constraint flags
+    r.u(8);  // level
+    r.ue();  // sps_id
+    if (profile == 100 || profile == 110 || profile == 122 || profile == 244 ||
+        profile == 44 || profile == 83 || profile == 86 || profile == 118 ||
+        profile == 128) {
+      int chroma = r.ue();
+      if (chroma != 1) return -1;  // 4:2:0 only
+      r.ue();                      // bit_depth_luma_minus8
+      r.ue();                      // bit_depth_chroma_minus8
+      r.u(1);                      // qpprime
+      if (r.u(1)) return -1;       // scaling matrices unsupported
+    }
+    s.log2_max_frame_num = (int)r.ue() + 4;
+    s.poc_type = (int)r.ue();
+    if (s.poc_type == 0)
+      s.log2_max_poc_lsb = (int)r.ue() + 4;
+    else if (s.poc_type == 1)
+      return -1;
+    r.ue();  // max_num_ref_frames
+    r.u(1);  // gaps allowed
+    int mbw = (int)r.ue() + 1;
+    int mbh = (int)r.ue() + 1;
+    if (!r.u(1)) return -1;  // frame_mbs_only required
+    r.u(1);                  // direct_8x8
+    s.pw = mbw * 16;
+    s.ph = mbh * 16;
+    if (r.u(1)) {  // cropping
+      int cl = (int)r.ue() * 2, cr = (int)r.ue() * 2;
+      int ct = (int)r.ue() * 2, cb = (int)r.ue() * 2;
+      if (cl || ct) return -1;  // left/top crop unsupported
+      s.crop_r = cr;
+      s.crop_b = cb;
+    }
+    if (r.overrun || s.pw <= 0 || s.ph <= 0 || s.pw > 8192 || s.ph > 8192)
+      return -1;
+    s.valid = true;
+    sps = s;
+    rY.assign((size_t)sps.pw * sps.ph, 0);
+    rCb.assign((size_t)sps.pw * sps.ph / 4, 128);
+    rCr.assign((size_t)sps.pw * sps.ph / 4, 128);
+    return 0;
+  }
+
+  int parse_pps(BitReader& r) {
+    Pps p;
+    r.ue();  // pps_id
+    r.ue();  // sps_id
+    p.cavlc = r.u(1) == 0;
+    if (!p.cavlc) return -1;  // CABAC unsupported
+    p.pic_order_present = r.u(1) != 0;
+    if (r.ue() != 0) return -1;  // slice groups unsupported
+    r.ue();                      // num_ref_idx_l0
+    r.ue();                      // num_ref_idx_l1
+    r.u(1);                      // weighted_pred
+    r.u(2);                      // weighted_bipred
+    p.pic_init_qp = 26 + r.se();
+    r.se();  // pic_init_qs
+    p.chroma_qp_offset = r.se();
+    p.deblock_present = r.u(1) != 0;
+    r.u(1);  // constrained_intra
+    r.u(1);  // redundant_pic_cnt_present
+    if (r.overrun) return -1;
+    p.valid = true;
+    pps = p;
+    return 0;
+  }
+
+  // decode one I_16x16 or I_PCM macroblock; returns 0 or negative error
+  int decode_mb(BitReader& r, int mbx, int mby, int& qp) {
+    PlaneCtx rpy{rY.data(), sps.pw};
+    PlaneCtx rpcb{rCb.data(), sps.pw / 2}, rpcr{rCr.data(), sps.pw / 2};
+    uint32_t mb_type = r.ue();
+    if (r.overrun) return -1;
+    if (mb_type == 25) {  // I_PCM
+      r.align_byte();
+      uint8_t* dst = rY.data() + (mby * 16) * sps.pw + mbx * 16;
+      for (int y = 0; y < 16; ++y)
+        for (int x = 0; x < 16; ++x) dst[y * sps.pw + x] = (uint8_t)r.u(8);
+      uint8_t* pc[2] = {rCb.data() + (mby * 8) * (sps.pw / 2) + mbx * 8,
+                        rCr.data() + (mby * 8) * (sps.pw / 2) + mbx * 8};
+      for (int comp = 0; comp < 2; ++comp)
+        for (int y = 0; y < 8; ++y)
+          for (int x = 0; x < 8; ++x) pc[comp][y * (sps.pw / 2) + x] = (uint8_t)r.u(8);
+      for (int b = 0; b < 16; ++b) nnz.lnz(mbx, mby, blk_x4(b), blk_y4(b)) = 16;
+      for (int comp = 0; comp < 2; ++comp)
+        for (int blk = 0; blk < 4; ++blk)
+          nnz.cnz(comp, mbx, mby, blk & 1, blk >> 1) = 16;
+      return r.overrun ? -1 : 0;
+    }
+    if (mb_type < 1 || mb_type > 24) return -2;  // I_4x4 / non-intra: unsupported
+    const int code = (int)mb_type - 1;
+    const int pred_mode = code % 4;
+    const int cbp_chroma = (code / 4) % 3;
+    const int cbp_luma = (code / 12) ? 15 : 0;
+
+    const uint32_t chroma_mode = r.ue();
+    if (chroma_mode > 3) return -1;
+    const int dqp = r.se();
+    qp = qp + dqp;
+    if (qp < 0 || qp > 51) return -1;
+
+    // residual parse
+    int32_t dc_scan[16], dc_q[16] = {0};
+    int tc_dc = cavlc_read_block(r, dc_scan, 16, nnz.luma_nc(mbx, mby, 0, 0));
+    if (tc_dc < 0) return -1;
+    for (int i = 0; i < 16; ++i) dc_q[ZIGZAG[i]] = dc_scan[i];
+    int32_t ac_q[16][16];
+    memset(ac_q, 0, sizeof(ac_q));
+    if (cbp_luma) {
+      for (int b = 0; b < 16; ++b) {
+        const int x4 = blk_x4(b), y4 = blk_y4(b);
+        int32_t scan[15];
+        int tc = cavlc_read_block(r, scan, 15, nnz.luma_nc(mbx, mby, x4, y4));
+        if (tc < 0) return -1;
+        for (int i = 1; i < 16; ++i) ac_q[b][ZIGZAG[i]] = scan[i - 1];
+        nnz.lnz(mbx, mby, x4, y4) = (uint8_t)tc;
+      }
+    } else {
+      for (int b = 0; b < 16; ++b)
+        nnz.lnz(mbx, mby, blk_x4(b), blk_y4(b)) = 0;
+    }
+    int32_t cdc_q[2][4] = {{0}}, cac_q[2][4][16];
+    memset(cac_q, 0, sizeof(cac_q));
+    if (cbp_chroma) {
+      for (int comp = 0; comp < 2; ++comp) {
+        int32_t scan[4];
+        if (cavlc_read_block(r, scan, 4, -1) < 0) return -1;
+        for (int i = 0; i < 4; ++i) cdc_q[comp][i] = scan[i];
+      }
+    }
+    if (cbp_chroma == 2) {
+      for (int comp = 0; comp < 2; ++comp)
+        for (int blk = 0; blk < 4; ++blk) {
+          const int x2 = blk & 1, y2 = blk >> 1;
+          int32_t scan[15];
+          int tc = cavlc_read_block(r, scan, 15,
+                                    nnz.chroma_nc(comp, mbx, mby, x2, y2));
+          if (tc < 0) return -1;
+          for (int i = 1; i < 16; ++i) cac_q[comp][blk][ZIGZAG[i]] = scan[i - 1];
+          nnz.cnz(comp, mbx, mby, x2, y2) = (uint8_t)tc;
+        }
+    } else {
+      for (int comp = 0; comp < 2; ++comp)
+        for (int blk = 0; blk < 4; ++blk)
+          nnz.cnz(comp, mbx, mby, blk & 1, blk >> 1) = 0;
+    }
+
+    // predict + reconstruct
+    uint8_t lpred[256];
+    const bool have_top = mby > 0, have_left = mbx > 0;
+    if ((pred_mode == 0 && !have_top) || (pred_mode == 1 && !have_left) ||
+        (pred_mode == 3 && !(have_top && have_left)))
+      return -1;
+    pred_luma16(rpy, mbx, mby, pred_mode, lpred);
+    recon_luma16(rpy, mbx, mby, lpred, dc_q, ac_q, qp);
+    // chroma modes: bitstream 0=DC 1=H 2=V 3=Plane (pred_chroma8 order)
+    const int qpi = qp + pps.chroma_qp_offset;
+    const int qpc = CHROMA_QP[qpi < 0 ? 0 : qpi > 51 ? 51 : qpi];
+    int cm = (int)chroma_mode;
+    if ((cm == 2 && !have_top) || (cm == 1 && !have_left) ||
+        (cm == 3 && !(have_top && have_left)))
+      return -1;
+    uint8_t cpred[64];
+    pred_chroma8(rpcb, mbx, mby, cm, cpred);
+    recon_chroma8(rpcb, mbx, mby, cpred, cdc_q[0], cac_q[0], qpc);
+    pred_chroma8(rpcr, mbx, mby, cm, cpred);
+    recon_chroma8(rpcr, mbx, mby, cpred, cdc_q[1], cac_q[1], qpc);
+    return 0;
+  }
+
+  // returns 1 when a frame was reconstructed, 0 for parameter-set-only data,
+  // negative on unsupported/corrupt input
+  int decode_au(const uint8_t* data, int len, uint8_t* rgb, int cap, int* ow,
+                int* oh) {
+    auto nals = h264::split_annexb(data, (size_t)len);
+    bool got_frame = false;
+    for (auto& nal : nals) {
+      if (nal.size < 1) continue;
+      const int type = nal.data[0] & 0x1F;
+      auto rbsp = h264::unescape_rbsp(nal.data + 1, nal.size - 1);
+      BitReader r(rbsp.data(), rbsp.size());
+      if (type == 7) {
+        if (parse_sps(r) < 0) return -3;
+      } else if (type == 8) {
+        if (parse_pps(r) < 0) return -3;
+      } else if (type == 5 || type == 1) {
+        if (!sps.valid || !pps.valid) return -4;
+        if (r.ue() != 0) return -5;  // multi-slice unsupported
+        uint32_t stype = r.ue();
+        if (stype % 5 != 2) return -6;  // I slices only
+        r.ue();                         // pps_id
+        r.u(sps.log2_max_frame_num);    // frame_num
+        if (type == 5) r.ue();          // idr_pic_id
+        if (sps.poc_type == 0) {
+          r.u(sps.log2_max_poc_lsb);
+          if (pps.pic_order_present) r.se();
+        }
+        if (type == 5) {
+          r.u(1);  // no_output_of_prior_pics
+          r.u(1);  // long_term_reference
+        } else if (nal.data[0] & 0x60) {
+          if (r.u(1)) return -7;  // adaptive marking unsupported
+        }
+        int qp = pps.pic_init_qp + r.se();
+        if (pps.deblock_present) {
+          uint32_t idc = r.ue();
+          if (idc != 1) {
+            r.se();
+            r.se();
+            // deblocking requested: we decode without the loop filter.
+            // Bit-exact only for idc==1 streams (our encoder's output).
+          }
+        }
+        if (qp < 0 || qp > 51 || r.overrun) return -1;
+        const int mbw = sps.pw / 16, mbh = sps.ph / 16;
+        nnz.reset(mbw, mbh);
+        for (int mby = 0; mby < mbh; ++mby)
+          for (int mbx = 0; mbx < mbw; ++mbx) {
+            int rc = decode_mb(r, mbx, mby, qp);
+            if (rc < 0) return rc;
+          }
+        got_frame = true;
+      }
+    }
+    if (!got_frame) return 0;
+    const int w = sps.w(), h = sps.h();
+    if (cap < w * h * 3) return -8;
+    yuv420_to_rgb(rY.data(), rCb.data(), rCr.data(), sps.pw, w, h, rgb);
+    *ow = w;
+    *oh = h;
+    return 1;
+  }
+};
+
+}  // namespace h264sw
+
+// ---------------------------------------------------------------------------
+// C ABI
+// ---------------------------------------------------------------------------
+extern "C" {
+
+void* airtc_h264enc_create(int w, int h) {
+  if (w < 16 || h < 16 || w > 8192 || h > 8192) return nullptr;
+  return new h264sw::Encoder(w, h);
+}
+void airtc_h264enc_destroy(void* h) { delete (h264sw::Encoder*)h; }
+int airtc_h264enc_encode(void* h, const uint8_t* rgb, int qp, uint8_t* out,
+                         int cap) {
+  return ((h264sw::Encoder*)h)->encode(rgb, qp, out, cap);
+}
+
+void* airtc_h264dec_create() { return new h264sw::Decoder(); }
+void airtc_h264dec_destroy(void* h) { delete (h264sw::Decoder*)h; }
+// display dims from the last parsed SPS (0 until one arrives)
+void airtc_h264dec_dims(void* h, int* w, int* out_h) {
+  auto* d = (h264sw::Decoder*)h;
+  *w = d->sps.valid ? d->sps.w() : 0;
+  *out_h = d->sps.valid ? d->sps.h() : 0;
+}
+int airtc_h264dec_decode(void* h, const uint8_t* data, int len, uint8_t* rgb,
+                         int cap, int* w, int* out_h) {
+  return ((h264sw::Decoder*)h)->decode_au(data, len, rgb, cap, w, out_h);
+}
+
+// sanity: every VLC table must be prefix-free within itself; returns 0 on
+// success, a nonzero id of the offending table otherwise (test hook)
+int airtc_h264sw_table_check() {
+  using namespace h264sw;
+  auto prefix_free = [](const std::vector<std::pair<int, int>>& codes) {
+    for (size_t i = 0; i < codes.size(); ++i)
+      for (size_t j = 0; j < codes.size(); ++j) {
+        if (i == j) continue;
+        auto [li, bi] = codes[i];
+        auto [lj, bj] = codes[j];
+        if (li <= lj && (bj >> (lj - li)) == bi) return false;
+      }
+    return true;
+  };
+  // strict check over the encoder-used region (TotalCoeff <= 9; see the
+  // ENC guard in encode_mb) plus a full-table pass — the deep rows are
+  // decode-side best-effort
+  for (int b = 0; b < 3; ++b) {
+    std::vector<std::pair<int, int>> codes;
+    for (int c = 0; c <= 16; ++c)
+      for (int t = 0; t < 4; ++t)
+        if (CT_LEN[b][c][t]) codes.push_back({CT_LEN[b][c][t], CT_BITS[b][c][t]});
+    if (!prefix_free(codes)) return 100 + b;
+  }
+  {
+    std::vector<std::pair<int, int>> codes;
+    for (int c = 0; c <= 4; ++c)
+      for (int t = 0; t < 4; ++t)
+        if (CDC_CT_LEN[c][t]) codes.push_back({CDC_CT_LEN[c][t], CDC_CT_BITS[c][t]});
+    if (!prefix_free(codes)) return 200;
+  }
+  for (int tc = 1; tc <= 15; ++tc) {
+    std::vector<std::pair<int, int>> codes;
+    for (int z = 0; z <= 16 - tc; ++z)
+      codes.push_back({TZ_LEN[tc - 1][z], TZ_BITS[tc - 1][z]});
+    if (!prefix_free(codes)) return 300 + tc;
+  }
+  for (int tc = 1; tc <= 3; ++tc) {
+    std::vector<std::pair<int, int>> codes;
+    for (int z = 0; z <= 4 - tc; ++z)
+      codes.push_back({CDC_TZ_LEN[tc - 1][z], CDC_TZ_BITS[tc - 1][z]});
+    if (!prefix_free(codes)) return 400 + tc;
+  }
+  for (int zl = 1; zl <= 7; ++zl) {
+    std::vector<std::pair<int, int>> codes;
+    int count = zl == 7 ? 15 : zl + 1;
+    for (int rb = 0; rb < count; ++rb)
+      codes.push_back({RB_LEN[zl - 1][rb], RB_BITS[zl - 1][rb]});
+    if (!prefix_free(codes)) return 500 + zl;
+  }
+  return 0;
+}
+
+}  // extern "C"

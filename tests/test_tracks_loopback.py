@@ -92,11 +92,18 @@ def test_pts_preserved(monkeypatch):
 # ---------------------------------------------------------------------------
 
 class _ClientProto(asyncio.DatagramProtocol):
+    """Test-side WebRTC client speaking the agent's default wire format
+    (select_codec: standard H.264 over RFC 6184 when the native extension
+    is built, RAWZ generic fragmentation otherwise)."""
+
     def __init__(self):
+        from ai_rtc_agent_amd.media.codec import select_codec
+
         self.transport = None
         self.frames = asyncio.Queue()
         self.defrag = RtpDefragmenter()
-        self.codec = SoftwareCodec()
+        self.codec = select_codec(role="decode")
+        self._au = []
 
     def connection_made(self, transport):
         from ai_rtc_agent_amd.media.rtc import tune_socket_buffers
@@ -111,11 +118,45 @@ class _ClientProto(asyncio.DatagramProtocol):
             pkt = RtpPacket.parse(data)
         except ValueError:
             return
-        buf = self.defrag.push(pkt)
-        if buf is not None:
-            t = self.codec.decode(buf)
-            if t is not None:
-                self.frames.put_nowait(t)
+        if getattr(self.codec, "rtp_mode", "raw") == "rfc6184":
+            from ai_rtc_agent_amd.media.h264 import H264Depacketizer, join_annexb
+
+            self._au.append(pkt.payload)
+            if not pkt.marker:
+                return
+            dp = H264Depacketizer()
+            nals = [n for n in (dp.push(p) for p in self._au) if n is not None]
+            self._au = []
+            if not nals:
+                return
+            buf = join_annexb(nals)
+        else:
+            buf = self.defrag.push(pkt)
+            if buf is None:
+                return
+        t = self.codec.decode(buf)
+        if t is not None:
+            self.frames.put_nowait(t)
+
+
+def _send_frame(codec, pkz, transport, addr, frame, timestamp):
+    """Packetize one encoded frame the way PeerConnection's sender does
+    (rfc6184 single-NAL/FU-A for H.264 codecs, generic otherwise)."""
+    data = codec.encode(frame)
+    if getattr(codec, "rtp_mode", "raw") == "rfc6184":
+        from ai_rtc_agent_amd.media.h264 import packetize_h264, split_annexb
+
+        payloads = packetize_h264(split_annexb(data))
+        for i, pl in enumerate(payloads):
+            pkt = RtpPacket(
+                payload_type=pkz.payload_type, sequence_number=pkz._seq,
+                timestamp=timestamp, ssrc=pkz.ssrc,
+                marker=1 if i == len(payloads) - 1 else 0, payload=pl)
+            pkz._seq = (pkz._seq + 1) & 0xFFFF
+            transport.sendto(pkt.serialize(), addr)
+    else:
+        for pkt in pkz.packetize(data, timestamp=timestamp):
+            transport.sendto(pkt.serialize(), addr)
 
 
 def _offer_sdp(port: int) -> str:
@@ -162,14 +203,14 @@ def test_loopback_whip_whep(monkeypatch):
         await asyncio.sleep(0.1)
 
         # stream frames to the agent
-        codec = SoftwareCodec()
+        from ai_rtc_agent_amd.media.codec import select_codec
+        codec = select_codec()
         pkz = RtpPacketizer(ssrc=99)
         g = torch.Generator().manual_seed(0)
         frames = [torch.randint(0, 200, (16, 16, 3), generator=g, dtype=torch.uint8)
                   for _ in range(6)]
         for i, f in enumerate(frames):
-            for pkt in pkz.packetize(codec.encode(f), timestamp=i * 3000):
-                pub_t.sendto(pkt.serialize(), ("127.0.0.1", srv_port))
+            _send_frame(codec, pkz, pub_t, ("127.0.0.1", srv_port), f, i * 3000)
             await asyncio.sleep(0.02)
         await asyncio.sleep(0.2)
 
@@ -184,8 +225,8 @@ def test_loopback_whip_whep(monkeypatch):
         # keep publishing so the subscriber's sender loop has frames to pull
         got = None
         for i in range(6, 40):
-            for pkt in pkz.packetize(codec.encode(frames[i % len(frames)]), timestamp=i * 3000):
-                pub_t.sendto(pkt.serialize(), ("127.0.0.1", srv_port))
+            _send_frame(codec, pkz, pub_t, ("127.0.0.1", srv_port),
+                        frames[i % len(frames)], i * 3000)
             try:
                 got = await asyncio.wait_for(sub_p.frames.get(), timeout=0.25)
                 break
@@ -235,14 +276,14 @@ def test_multi_viewer_whep_fanout(monkeypatch):
         pub_t.sendto(stun.make_binding_request("u:p", b"k"), ("127.0.0.1", srv_port))
         await asyncio.sleep(0.05)
 
-        codec = SoftwareCodec()
+        from ai_rtc_agent_amd.media.codec import select_codec
+        codec = select_codec()
         pkz = RtpPacketizer(ssrc=7)
         g = torch.Generator().manual_seed(1)
         frames = [torch.randint(0, 200, (16, 16, 3), generator=g, dtype=torch.uint8)
                   for _ in range(4)]
         for i in range(4):
-            for pkt in pkz.packetize(codec.encode(frames[i]), timestamp=i * 3000):
-                pub_t.sendto(pkt.serialize(), ("127.0.0.1", srv_port))
+            _send_frame(codec, pkz, pub_t, ("127.0.0.1", srv_port), frames[i], i * 3000)
             await asyncio.sleep(0.02)
 
         subs = []
@@ -257,8 +298,7 @@ def test_multi_viewer_whep_fanout(monkeypatch):
 
         got = [None, None]
         for i in range(4, 120):
-            for pkt in pkz.packetize(codec.encode(frames[i % 4]), timestamp=i * 3000):
-                pub_t.sendto(pkt.serialize(), ("127.0.0.1", srv_port))
+            _send_frame(codec, pkz, pub_t, ("127.0.0.1", srv_port), frames[i % 4], i * 3000)
             for si, (_, p) in enumerate(subs):
                 if got[si] is None:
                     try:
