@@ -108,6 +108,11 @@ class PeerConnection:
         self._rr_last_sent = (0, 0)  # (expected, received) at last RR
         self.port: Optional[int] = None
         self._ice_pwd = ""
+        # DTLS-SRTP (created when the remote SDP carries a fingerprint)
+        self._dtls = None
+        self._dtls_expected_fp: Optional[str] = None
+        self._dtls_verified = False
+        self._dtls_task: Optional[asyncio.Task] = None
 
     # -- event API (aiortc-style) --------------------------------------
     def on(self, event: str, handler: Optional[Callable] = None):
@@ -138,10 +143,28 @@ class PeerConnection:
                 if len(parts) >= 6 and parts[2].lower() == "udp":
                     self._remote_addr = (parts[4], int(parts[5]))
                     break
+        # a remote fingerprint means the peer (browser/OBS) requires
+        # DTLS-SRTP; we always answer setup:passive, so we are the server
+        fp = self.remote_description.fingerprint
+        for m in self.remote_description.media:
+            fp = m.fingerprint or fp
+        if fp is not None:
+            from . import dtls as dtls_mod
+
+            if dtls_mod.dtls_available():
+                self._dtls = dtls_mod.create_endpoint(server=True)
+                self._dtls_expected_fp = fp
+            else:
+                logger.error(
+                    "peer requires DTLS-SRTP but the native endpoint is "
+                    "unavailable; answering plain RTP (will not connect)"
+                )
 
     async def create_answer(self, host: str = "127.0.0.1", direction: str = "sendrecv") -> str:
         await self._bind(host)
         assert self.remote_description is not None, "set_remote_description first"
+        from . import dtls as dtls_mod
+
         ans = build_answer(
             self.remote_description,
             host,
@@ -149,11 +172,29 @@ class PeerConnection:
             codec_name="H264",
             ssrc=self._packetizer.ssrc,
             direction=direction,
+            fingerprint=dtls_mod.local_fingerprint(),
         )
         self.local_description = ans
         self._ice_pwd = ans.media[0].ice_pwd if ans.media else ""
+        if self._dtls is not None and self._dtls_task is None:
+            self._dtls_task = asyncio.ensure_future(self._dtls_timer())
         self._set_state("connecting")
         return ans.serialize()
+
+    async def _dtls_timer(self) -> None:
+        """Drive DTLS retransmissions until the handshake completes (the
+        transport is datagram-lossy; OpenSSL's timer needs a pump)."""
+        t0 = time.monotonic()
+        try:
+            while (self._dtls is not None and not self._dtls.established()
+                   and self.connection_state not in ("closed", "failed")
+                   and time.monotonic() - t0 < 30.0):
+                await asyncio.sleep(0.4)
+                if self._transport is not None and self._remote_addr:
+                    for out in self._dtls.handle_timeout():
+                        self._transport.sendto(out, self._remote_addr)
+        except asyncio.CancelledError:
+            pass
 
     async def _bind(self, host: str) -> None:
         if self._transport is not None:
@@ -213,7 +254,7 @@ class PeerConnection:
                 # datagrams; an unpaced burst overflows receiver socket
                 # buffers (and starves the event loop's read side)
                 for j, pkt in enumerate(pkts):
-                    self._transport.sendto(pkt.serialize(), self._remote_addr)
+                    self._send_media(pkt.serialize())
                     if j % 32 == 31:
                         await asyncio.sleep(0)
                     if j % 256 == 255:
@@ -224,6 +265,33 @@ class PeerConnection:
         except Exception:
             logger.exception("sender loop failed")
             self._set_state("failed")
+
+    # -- outbound media (SRTP protection when DTLS is active) ------------
+    def _send_media(self, data: bytes) -> None:
+        """Send one RTP/RTCP datagram, SRTP/SRTCP-protected when the
+        session negotiated DTLS; plain otherwise. Media is gated until the
+        handshake completes (RFC 5764: no SRTP keys before then)."""
+        if self._transport is None or self._remote_addr is None:
+            return
+        if self._dtls is not None:
+            if not self._dtls.established():
+                return
+            data = (self._dtls.protect_rtcp(data) if is_rtcp(data)
+                    else self._dtls.protect_rtp(data))
+        self._transport.sendto(data, self._remote_addr)
+
+    def _dtls_on_established(self) -> None:
+        from . import dtls as dtls_mod
+
+        self._dtls_verified = True
+        actual = self._dtls.peer_fingerprint()
+        if not dtls_mod.fingerprints_match(self._dtls_expected_fp or "", actual):
+            logger.error("DTLS peer fingerprint mismatch (%s != sdp %s)",
+                         actual, self._dtls_expected_fp)
+            self._set_state("failed")
+            return
+        logger.info("DTLS-SRTP established (peer fingerprint verified)")
+        self._set_state("connected")
 
     # -- inbound --------------------------------------------------------
     def _on_datagram(self, data: bytes, addr) -> None:
@@ -236,8 +304,25 @@ class PeerConnection:
                 resp = make_binding_response(msg, addr, self._ice_pwd.encode())
                 self._transport.sendto(resp, addr)
                 self._remote_addr = addr  # peer-reflexive
-                self._set_state("connected")
+                if self._dtls is None:
+                    self._set_state("connected")
             return
+        # RFC 5764 5.1.2 demux: DTLS record types occupy [20, 63]
+        if self._dtls is not None and data and 20 <= data[0] <= 63:
+            self._remote_addr = addr
+            for out in self._dtls.feed(data):
+                if self._transport is not None:
+                    self._transport.sendto(out, addr)
+            if self._dtls.established() and not self._dtls_verified:
+                self._dtls_on_established()
+            return
+        if self._dtls is not None and data and 128 <= data[0] <= 191:
+            if not self._dtls.established():
+                return
+            data = (self._dtls.unprotect_rtcp(data) if is_rtcp(data)
+                    else self._dtls.unprotect_rtp(data))
+            if data is None:
+                return  # auth failed / out of window: drop
         if is_rtcp(data):
             if parse_pli(data) is not None:
                 # peer lost decodability: force a keyframe on the next frame
@@ -318,8 +403,7 @@ class PeerConnection:
             now = time.monotonic()
             if self._decode_misses >= 2 and now - self._last_pli > 0.5 \
                     and self._transport is not None and self._remote_addr:
-                self._transport.sendto(
-                    make_pli(self._packetizer.ssrc, pkt.ssrc), self._remote_addr)
+                self._send_media(make_pli(self._packetizer.ssrc, pkt.ssrc))
                 self._last_pli = now
             return
         self._decode_misses = 0
@@ -349,11 +433,9 @@ class PeerConnection:
             interval_exp = expected - exp0
             interval_rcv = self._rx_count - rcv0
             lost_frac = max(0.0, 1.0 - interval_rcv / max(1, interval_exp))
-            self._transport.sendto(
+            self._send_media(
                 make_rr(self._packetizer.ssrc, pkt.ssrc, lost_frac,
-                        max(0, expected - self._rx_count), self._rx_high_seq),
-                self._remote_addr,
-            )
+                        max(0, expected - self._rx_count), self._rx_high_seq))
             self._rr_last_sent = (expected, self._rx_count)
 
     def _adapt_bitrate(self, fraction_lost: float) -> None:
@@ -373,6 +455,9 @@ class PeerConnection:
         if self.connection_state == "closed":
             return
         self._set_state("closed")
+        if self._dtls_task is not None:
+            self._dtls_task.cancel()
+            self._dtls_task = None
         if self._sender_task is not None:
             self._sender_task.cancel()
             try:

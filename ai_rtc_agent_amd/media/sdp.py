@@ -40,6 +40,7 @@ class MediaSection:
     mid: str = "0"
     ssrc: Optional[int] = None
     setup: str = "actpass"
+    fingerprint: Optional[str] = None  # "sha-256 AA:BB:..." (RFC 8122)
 
     def codec_by_name(self, name: str) -> Optional[RtpCodec]:
         for c in self.codecs:
@@ -52,6 +53,7 @@ class MediaSection:
 class SessionDescription:
     session_id: str = ""
     media: List[MediaSection] = field(default_factory=list)
+    fingerprint: Optional[str] = None  # session-level a=fingerprint
 
     @staticmethod
     def parse(sdp: str) -> "SessionDescription":
@@ -66,6 +68,8 @@ class SessionDescription:
                 parts = line[2:].split()
                 if len(parts) >= 2:
                     sd.session_id = parts[1]
+            elif line.startswith("a=fingerprint:") and cur is None:
+                sd.fingerprint = line.split(":", 1)[1].strip()
             elif line.startswith("m="):
                 parts = line[2:].split()
                 cur = MediaSection(kind=parts[0], port=int(parts[1]), protocol=parts[2])
@@ -103,6 +107,8 @@ class SessionDescription:
                     cur.mid = line.split(":", 1)[1]
                 elif line.startswith("a=setup:"):
                     cur.setup = line.split(":", 1)[1]
+                elif line.startswith("a=fingerprint:"):
+                    cur.fingerprint = line.split(":", 1)[1].strip()
                 elif line.startswith("a=ssrc:"):
                     try:
                         cur.ssrc = int(line[len("a=ssrc:"):].split()[0])
@@ -130,6 +136,8 @@ class SessionDescription:
             lines.append(f"a=mid:{m.mid}")
             lines.append(f"a={m.direction}")
             lines.append(f"a=setup:{m.setup}")
+            if m.fingerprint:
+                lines.append(f"a=fingerprint:{m.fingerprint}")
             for c in m.codecs:
                 lines.append(f"a=rtpmap:{c.payload_type} {c.name}/{c.clock_rate}")
                 if c.parameters:
@@ -156,9 +164,11 @@ def build_answer(
     codec_name: str,
     ssrc: int,
     direction: str = "sendrecv",
+    fingerprint: Optional[str] = None,
 ) -> "SessionDescription":
     """Answer an offer: echo media sections, pick our codec, attach our ICE
-    credentials + host candidate."""
+    credentials + host candidate (+ DTLS fingerprint, setup:passive — the
+    offerer is the DTLS client)."""
     ans = SessionDescription(session_id=str(random.randint(10**8, 10**9)))
     for i, m in enumerate(offer.media):
         sec = MediaSection(
@@ -171,6 +181,7 @@ def build_answer(
             ice_pwd=_rand(24),
             ssrc=ssrc,
             setup="passive",
+            fingerprint=fingerprint,
         )
         codec = m.codec_by_name(codec_name)
         if codec is None and m.codecs:

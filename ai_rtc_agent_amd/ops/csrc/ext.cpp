@@ -279,7 +279,10 @@ pybind11::dict vcn_probe() {
 
 }  // namespace
 
+void airtc_register_dtls(pybind11::module_& m);  // dtls.cpp
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  airtc_register_dtls(m);
   m.def("conv2d", &conv2d, "implicit-GEMM MFMA conv2d (NHWC)");
   m.def("group_norm_silu", &group_norm_silu);
   m.def("layer_norm", &layer_norm);

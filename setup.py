@@ -17,7 +17,7 @@ CSRC = os.path.join(ROOT, "ai_rtc_agent_amd", "ops", "csrc")
 
 sources = [
     os.path.join(CSRC, f)
-    for f in ["ext.cpp", "vcn.cpp", "h264sw.cpp", "elementwise.hip", "norms.hip", "conv2d.hip", "attention.hip"]
+    for f in ["ext.cpp", "vcn.cpp", "h264sw.cpp", "dtls.cpp", "elementwise.hip", "norms.hip", "conv2d.hip", "attention.hip"]
 ]
 
 setup(
@@ -26,6 +26,7 @@ setup(
         CUDAExtension(
             name="ai_rtc_agent_amd.ops._C",
             sources=sources,
+            libraries=["ssl", "crypto"],  # dtls.cpp (system OpenSSL 3)
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "--offload-arch=gfx950", "-std=c++17"],
