@@ -446,7 +446,10 @@ class StreamDiffusionEngine:
             return
         fbs = self.cfg.frame_buffer_size
         lh, lw = self.cfg.latent_height, self.cfg.latent_width
-        self._pipelined = self.sim_filter is None and self.cfg.pipeline_overlap
+        # pipeline overlap composes with the similarity filter: a skipped
+        # frame simply returns the persistent output buffer of the last
+        # replay without touching the stream A/B event chain
+        self._pipelined = self.cfg.pipeline_overlap
         sA = torch.cuda.Stream()
         sB = torch.cuda.Stream()
         self._sA, self._sB = sA, sB
@@ -514,9 +517,10 @@ class StreamDiffusionEngine:
             frame_u8 = frame_u8.to(self.device, non_blocking=True)
 
         if self.sim_filter is not None and self._prev_out is not None:
-            # centre u8 before cosine similarity (raw u8 is all-positive,
-            # which would inflate similarity between unrelated frames)
-            if self.sim_filter.should_skip(frame_u8.float() - 127.5):
+            # on-device pooled cosine (centering + 64x64 avg-pool inside
+            # the filter); the only CPU sync is one scalar read that waits
+            # on the frame upload, NOT on the pipelined streams
+            if self.sim_filter.should_skip(frame_u8):
                 out = self._prev_out
                 self.timers.frame_done()
                 return out[0] if squeeze else out
