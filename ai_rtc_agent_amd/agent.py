@@ -293,6 +293,11 @@ def main() -> None:
     parser.add_argument("--family", default="sd15", choices=["sd15", "sd21", "sdxl"],
                         help="UNet family served by the pipeline")
     parser.add_argument("--resolution", type=int, default=512)
+    parser.add_argument("--workers", type=int, default=0,
+                        help="process-per-GPU serving: spawn N worker agents "
+                             "(one per GPU, own media sockets) behind a "
+                             "signalling front-end (SURVEY.md §5.8); 0 = "
+                             "single-process mode")
     args = parser.parse_args()
 
     logging.basicConfig(level=getattr(logging, args.log_level.upper(), logging.INFO))
@@ -300,6 +305,24 @@ def main() -> None:
     if args.udp_ports:
         lo, _, hi = args.udp_ports.partition("-")
         ports = list(range(int(lo), int(hi or lo) + 1))
+    if args.workers > 0:
+        import asyncio
+
+        import torch
+
+        from .parallel.frontend import WorkerFrontend
+
+        fe = WorkerFrontend(args.workers, model_id=args.model_id,
+                            family=args.family, resolution=args.resolution,
+                            pin_gpu=torch.cuda.is_available(),
+                            udp_ports=ports)
+        fe.spawn()
+        asyncio.get_event_loop().run_until_complete(fe.wait_ready())
+        try:
+            web.run_app(fe.create_app(), host=args.host, port=args.port)
+        finally:
+            fe.shutdown()
+        return
     app = create_app(model_id=args.model_id, udp_ports=ports, n_gpus=args.gpus,
                      family=args.family, resolution=args.resolution)
     web.run_app(app, host=args.host, port=args.port)
