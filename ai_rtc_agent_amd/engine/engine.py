@@ -70,9 +70,21 @@ class StreamDiffusionEngine:
         ucfg = _unet_config_for(cfg.model_family)
         self.unet = unet if unet is not None else UNet2DCondition(ucfg)
         self.vae = vae if vae is not None else TinyVAE()
+        # family conventions: SD1.5 = CLIP ViT-L/14 (quick-gelu, last
+        # layer); SD2.x = OpenCLIP ViT-H (gelu, PENULTIMATE layer, 23
+        # transformer blocks); sdxl keeps a single encoder stand-in with
+        # the pooled text_projection the addition-embed path consumes
+        if cfg.model_family == "tiny":
+            te_layers, te_act, te_skip = 2, "quick_gelu", 0
+        elif cfg.model_family == "sd21":
+            te_layers, te_act, te_skip = 23, "gelu", 1
+        else:
+            te_layers, te_act, te_skip = 12, "quick_gelu", 0
         self.text_encoder = text_encoder if text_encoder is not None else TextEncoder(
             hidden=ucfg.cross_attention_dim,
-            layers=2 if cfg.model_family == "tiny" else 12,
+            layers=te_layers,
+            act=te_act,
+            clip_skip=te_skip,
             pooled_dim=1280 if ucfg.addition_embed_dim else None,
         )
         self.ctx_dim = ucfg.cross_attention_dim

@@ -272,6 +272,56 @@ def _is_autoencoder_kl(sd: Dict[str, torch.Tensor]) -> bool:
     return any(k.startswith("encoder.down_blocks.") for k in sd)
 
 
+# ---------------------------------------------------------------------------
+# CLIP text encoder (transformers CLIPTextModel naming, as shipped in a
+# diffusers snapshot's text_encoder/ directory — reference loads the same
+# artifact via CLIPTextModel.from_pretrained, lib/wrapper.py:468-473)
+# ---------------------------------------------------------------------------
+def clip_text_key_map(n_layers: int) -> List[Tuple[str, str]]:
+    pairs: List[Tuple[str, str]] = [
+        ("text_model.embeddings.token_embedding.weight", "token_emb.weight"),
+        ("text_model.embeddings.position_embedding.weight", "pos_emb.weight"),
+        ("text_model.final_layer_norm.weight", "final_ln.weight"),
+        ("text_model.final_layer_norm.bias", "final_ln.bias"),
+    ]
+    for i in range(n_layers):
+        src = f"text_model.encoder.layers.{i}"
+        dst = f"blocks.{i}"
+        for a, b in (
+            ("layer_norm1", "ln1"), ("layer_norm2", "ln2"),
+            ("self_attn.q_proj", "q_proj"), ("self_attn.k_proj", "k_proj"),
+            ("self_attn.v_proj", "v_proj"), ("self_attn.out_proj", "out_proj"),
+            ("mlp.fc1", "fc1"), ("mlp.fc2", "fc2"),
+        ):
+            for p in ("weight", "bias"):
+                pairs.append((f"{src}.{a}.{p}", f"{dst}.{b}.{p}"))
+    pairs.append(("text_projection.weight", "pooled_proj.weight"))
+    return pairs
+
+
+def load_clip_text_encoder(model, sd: Dict[str, torch.Tensor],
+                           strict: bool = False) -> int:
+    """Load a CLIPTextModel state dict into our TextEncoder. With strict,
+    every expected key (except the optional text_projection) must match."""
+    own = dict(model.state_dict())
+    n = 0
+    missing: List[str] = []
+    for src, dst in clip_text_key_map(len(model.blocks)):
+        optional = dst.startswith("pooled_proj")
+        if src not in sd or dst not in own:
+            if not optional:
+                missing.append(src)
+            continue
+        v = sd[src]
+        if v.shape != own[dst].shape:
+            raise ValueError(f"shape mismatch {src}: {v.shape} vs {own[dst].shape}")
+        own[dst].copy_(v.to(own[dst].dtype))
+        n += 1
+    if strict and missing:
+        raise KeyError(f"missing {len(missing)} CLIP keys, e.g. {missing[:5]}")
+    return n
+
+
 def load_model_dir(engine, model_dir: str) -> bool:
     """Load UNet (+ TAESD enc+dec) safetensors from a local diffusers-style
     directory; returns False when nothing was found (random init stays).
@@ -328,4 +378,22 @@ def load_model_dir(engine, model_dir: str) -> bool:
     if found and n_enc == 0 and n_dec == 0:
         log.warning("model dir %s: UNet loaded but no TAESD weights matched — "
                     "VAE stays random-init", model_dir)
+    # CLIP text encoder + its BPE tokenizer (reference lib/wrapper.py:468-473)
+    te = getattr(engine, "text_encoder", None)
+    if te is not None:
+        for sub in ("text_encoder/model.safetensors",
+                    "text_encoder/pytorch_model.safetensors"):
+            p = os.path.join(model_dir, sub)
+            if os.path.exists(p):
+                try:
+                    n = load_clip_text_encoder(te, load_file(p))
+                    if n:
+                        found = True
+                    else:
+                        log.warning("CLIP checkpoint %s matched 0 tensors", p)
+                except ValueError as e:
+                    log.warning("CLIP checkpoint %s skipped: %s", p, e)
+                break
+        if hasattr(te, "load_tokenizer_dir") and te.load_tokenizer_dir(model_dir):
+            log.info("BPE tokenizer loaded from %s", model_dir)
     return found

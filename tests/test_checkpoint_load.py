@@ -171,3 +171,132 @@ def test_sdxl_map_builds():
     srcs = [s for s, _ in pairs]
     assert "add_embedding.linear_1.weight" in srcs
     assert any("transformer_blocks.9" in s for s in srcs), "depth-10 mid stack"
+
+
+# ---------------------------------------------------------------------------
+# CLIP text encoder + BPE tokenizer (round-1 verdict, Missing #4)
+# ---------------------------------------------------------------------------
+
+def _synth_clip_sd(model):
+    from ai_rtc_agent_amd.models.load import clip_text_key_map
+
+    own = dict(model.state_dict())
+    sd = {}
+    for src, dst in clip_text_key_map(len(model.blocks)):
+        if dst not in own:
+            continue
+        g = torch.Generator().manual_seed(abs(hash(src)) % (2**31))
+        sd[src] = torch.randn(own[dst].shape, generator=g) * 0.02
+    return sd
+
+
+def test_clip_key_map_covers_every_parameter():
+    from ai_rtc_agent_amd.models.load import clip_text_key_map
+    from ai_rtc_agent_amd.models.text_encoder import TextEncoder
+
+    te = TextEncoder(hidden=64, layers=2, vocab_size=128, pooled_dim=96)
+    mapped = {dst for _, dst in clip_text_key_map(2)}
+    own = set(dict(te.state_dict()).keys())
+    assert not own - mapped, sorted(own - mapped)[:8]
+
+
+def test_clip_strict_load_and_checkpoint_determined_embeddings():
+    """The reference contract (lib/wrapper.py:468-473): prompt conditioning
+    comes from the checkpoint. Two encoders with DIFFERENT random seeds
+    must produce IDENTICAL embeddings after loading the same synthesized
+    CLIPTextModel state dict."""
+    from ai_rtc_agent_amd.models.load import load_clip_text_encoder
+    from ai_rtc_agent_amd.models.text_encoder import TextEncoder
+
+    a = TextEncoder(hidden=64, layers=2, vocab_size=128, seed=0)
+    b = TextEncoder(hidden=64, layers=2, vocab_size=128, seed=123)
+    with torch.no_grad():
+        assert not torch.allclose(a.encode("a cat"), b.encode("a cat"))
+    sd = _synth_clip_sd(a)
+    na = load_clip_text_encoder(a, sd, strict=True)
+    nb = load_clip_text_encoder(b, sd, strict=True)
+    assert na == nb == len(sd)
+    with torch.no_grad():
+        ea, eb = a.encode("a cat"), b.encode("a cat")
+    assert torch.allclose(ea, eb, atol=1e-6)
+    assert ea.shape == (1, 77, 64)
+
+
+def test_clip_strict_load_rejects_incomplete():
+    from ai_rtc_agent_amd.models.load import load_clip_text_encoder
+    from ai_rtc_agent_amd.models.text_encoder import TextEncoder
+
+    te = TextEncoder(hidden=64, layers=2, vocab_size=128)
+    sd = _synth_clip_sd(te)
+    sd.pop("text_model.encoder.layers.1.mlp.fc1.weight")
+    import pytest as _pytest
+
+    with _pytest.raises(KeyError):
+        load_clip_text_encoder(te, sd, strict=True)
+
+
+def test_clip_penultimate_layer_option():
+    from ai_rtc_agent_amd.models.text_encoder import TextEncoder
+
+    torch.manual_seed(0)
+    last = TextEncoder(hidden=32, layers=3, vocab_size=64, seed=7, clip_skip=0)
+    pen = TextEncoder(hidden=32, layers=3, vocab_size=64, seed=7, clip_skip=1)
+    with torch.no_grad():
+        assert not torch.allclose(last.encode("x"), pen.encode("x"))
+
+
+def test_bpe_tokenizer_algorithm():
+    """Real CLIP BPE over a synthetic vocab: merges apply by rank, the
+    end-of-word marker binds, unknowns fall back to EOS."""
+    from ai_rtc_agent_amd.models.text_encoder import ClipBpeTokenizer
+
+    vocab = {}
+    for tok in ["h", "e", "l", "o", "w", "r", "d",
+                "he", "ll", "hell", "o</w>", "hello</w>",
+                "w", "o", "r</w>", "d</w>",
+                "<|startoftext|>", "<|endoftext|>"]:
+        vocab.setdefault(tok, len(vocab))
+    merges = [("h", "e"), ("l", "l"), ("he", "ll"), ("hell", "o</w>")]
+    tk = ClipBpeTokenizer(vocab, merges, max_length=10)
+    ids = tk("Hello")[0].tolist()
+    assert ids[0] == vocab["<|startoftext|>"]
+    assert ids[1] == vocab["hello</w>"]          # full merge chain applied
+    assert ids[2] == vocab["<|endoftext|>"]
+    assert all(i == vocab["<|endoftext|>"] for i in ids[2:])  # EOS padding
+    # a word without merges splits to chars + </w> on the last
+    ids2 = tk("word")[0].tolist()
+    assert ids2[1] == vocab["w"] and ids2[4] == vocab["d</w>"]
+
+
+def test_load_model_dir_wires_clip_and_tokenizer(tmp_path):
+    import json as _json
+
+    from safetensors.torch import save_file
+
+    from ai_rtc_agent_amd.models.load import load_model_dir
+    from ai_rtc_agent_amd.models.taesd import TinyVAE
+    from ai_rtc_agent_amd.models.text_encoder import (
+        ClipBpeTokenizer,
+        TextEncoder,
+    )
+
+    class _Eng:
+        pass
+
+    eng = _Eng()
+    eng.unet = UNet2DCondition(UNetConfig.tiny())
+    eng.vae = TinyVAE(width=16)
+    eng.text_encoder = TextEncoder(hidden=64, layers=2, vocab_size=128)
+    (tmp_path / "text_encoder").mkdir()
+    save_file(_synth_clip_sd(eng.text_encoder),
+              str(tmp_path / "text_encoder" / "model.safetensors"))
+    (tmp_path / "tokenizer").mkdir()
+    vocab = {c: i for i, c in enumerate("abcdefgh")}
+    vocab["<|startoftext|>"] = 126
+    vocab["<|endoftext|>"] = 127
+    (tmp_path / "tokenizer" / "vocab.json").write_text(_json.dumps(vocab))
+    (tmp_path / "tokenizer" / "merges.txt").write_text("#version: 0.2\na b\n")
+    before = eng.text_encoder.token_emb.weight.clone()
+    assert load_model_dir(eng, str(tmp_path))
+    assert not torch.allclose(before, eng.text_encoder.token_emb.weight)
+    assert isinstance(eng.text_encoder.tokenizer, ClipBpeTokenizer)
