@@ -87,6 +87,10 @@ def main() -> None:
                    help="frames for the full wire-to-wire latency sample "
                         "(RTP depacketize -> H.264 decode -> pipeline -> "
                         "H.264 encode -> RTP packetize); 0 disables")
+    p.add_argument("--fbs", type=int, default=1,
+                   help="frame_buffer_size: frames batched per engine call "
+                        "(multi-stream batched serving; aggregate FPS = "
+                        "steps*fbs/elapsed). 1 = headline single-stream")
     p.add_argument(
         "--model", default="sd-turbo", choices=["sd-turbo", "sd15", "sdxl"],
         help="sd-turbo 1-step (headline) | sd15 4-step LCM+RCFG+filter "
@@ -102,6 +106,8 @@ def main() -> None:
         # no-GPU smoke only: the full SD-Turbo UNet on CPU is ~0.5 s/frame
         args.steps, args.warmup, args.latency_frames = 4, 1, 4
         args.e2e_frames = min(args.e2e_frames, 2)
+    if args.fbs > 1:
+        args.e2e_frames = 0  # wire path is per-stream; measured at fbs=1
     device = f"cuda:{local}" if use_cuda else "cpu"
 
     graph = not args.no_graph and use_cuda
@@ -123,14 +129,19 @@ def main() -> None:
     # the 2-rank gloo run finishes in seconds (never set on GPU benches)
     if os.environ.get("AIRTC_BENCH_FAMILY"):
         cfg.model_family = os.environ["AIRTC_BENCH_FAMILY"]
+    if args.fbs > 1:
+        cfg.frame_buffer_size = args.fbs
     eng = StreamDiffusionEngine(cfg)
     broadcast_engine_weights(eng)  # RCCL over xGMI; no-op at world=1
     eng.prepare()
 
     # synthetic stream: a small ring of random frames, resident on device
+    # (fbs>1 = one frame per concurrently-served stream per engine call)
     g = torch.Generator().manual_seed(1234 + rank)
+    fshape = ((args.width, args.width, 3) if args.fbs == 1
+              else (args.fbs, args.width, args.width, 3))
     frames = [
-        torch.randint(0, 256, (args.width, args.width, 3), generator=g, dtype=torch.uint8).to(device)
+        torch.randint(0, 256, fshape, generator=g, dtype=torch.uint8).to(device)
         for _ in range(4)
     ]
 
@@ -204,7 +215,7 @@ def main() -> None:
             p50_e2e = float(lt.item()) or None
 
     ms_per_step = elapsed / args.steps * 1000.0
-    fps_total = world * args.steps / elapsed
+    fps_total = world * args.steps * args.fbs / elapsed
 
     if rank == 0:
         print(json.dumps({
