@@ -19,12 +19,16 @@ def test_agent_gpu_loopback(monkeypatch):
 
         from ai_rtc_agent_amd.agent import create_app
         from ai_rtc_agent_amd.config import sd_turbo_config
-        from ai_rtc_agent_amd.media.codec import SoftwareCodec
+        from ai_rtc_agent_amd.media.codec import select_codec
         from ai_rtc_agent_amd.media.rtp import RtpPacketizer
         from ai_rtc_agent_amd.media.sdp import SessionDescription
         from ai_rtc_agent_amd.media import stun
         from ai_rtc_agent_amd.parallel.dispatch import PipelinePool
-        from tests.test_tracks_loopback import _ClientProto, _offer_sdp
+        from tests.test_tracks_loopback import (
+            _ClientProto,
+            _offer_sdp,
+            _send_frame,
+        )
 
         cfg = sd_turbo_config(device="cuda")
         pool = PipelinePool.create("stabilityai/sd-turbo", n_gpus=1, cfg=cfg)
@@ -43,21 +47,19 @@ def test_agent_gpu_loopback(monkeypatch):
         pub_t.sendto(stun.make_binding_request("u:p", b"k"), ("127.0.0.1", srv_port))
         await asyncio.sleep(0.2)
 
-        codec = SoftwareCodec()
+        codec = select_codec()  # the default wire format (H.264 when built)
         pkz = RtpPacketizer(ssrc=42)
-        # compressible frames (gradients): random 512² noise is ~700 KB of
-        # zlib output = a 600-packet UDP burst that overflows loopback
-        # socket buffers; real camera frames compress, synthetic ones must too
+        # compressible frames (gradients): random 512² noise would make a
+        # huge UDP burst that overflows loopback socket buffers; real
+        # camera frames compress, synthetic ones must too
         base = torch.arange(512, dtype=torch.uint8).view(1, 512, 1).expand(512, 512, 3)
         frames = [(base.int() + 13 * i).clamp(0, 255).to(torch.uint8).contiguous()
                   for i in range(3)]
 
         async def send_frame(i):
-            pkts = pkz.packetize(codec.encode(frames[i % 3]), timestamp=i * 3000)
-            for j, pkt in enumerate(pkts):
-                pub_t.sendto(pkt.serialize(), ("127.0.0.1", srv_port))
-                if j % 40 == 39:
-                    await asyncio.sleep(0.005)  # pace the burst
+            _send_frame(codec, pkz, pub_t, ("127.0.0.1", srv_port),
+                        frames[i % 3], (i + 1) * 3000)
+            await asyncio.sleep(0.005)
 
         sub_t, sub_p = await loop.create_datagram_endpoint(
             _ClientProto, local_addr=("127.0.0.1", 0))
