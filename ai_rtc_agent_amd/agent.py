@@ -216,7 +216,8 @@ async def on_startup(app: web.Application) -> None:
             height=app.get("resolution", 512),
         )
         st["pool"] = PipelinePool.create(
-            model_id=app["model_id"], n_gpus=app["n_gpus"], cfg=cfg
+            model_id=app["model_id"], n_gpus=app["n_gpus"], cfg=cfg,
+            streams_per_replica=app.get("streams_per_gpu", 1),
         )
     st["ice_servers"] = get_ice_servers() if app["use_turn"] else []
 
@@ -251,6 +252,7 @@ def create_app(
     use_turn: bool = True,
     family: str = "sd15",
     resolution: int = 512,
+    streams_per_gpu: int = 1,
 ) -> web.Application:
     app = web.Application(middlewares=[cors_middleware])
     app["model_id"] = model_id
@@ -260,6 +262,7 @@ def create_app(
     app["use_turn"] = use_turn
     app["family"] = family
     app["resolution"] = resolution
+    app["streams_per_gpu"] = streams_per_gpu
     app["state"] = {
         "pcs": set(),
         "source_track": None,
@@ -293,6 +296,11 @@ def main() -> None:
     parser.add_argument("--family", default="sd15", choices=["sd15", "sd21", "sdxl"],
                         help="UNet family served by the pipeline")
     parser.add_argument("--resolution", type=int, default=512)
+    parser.add_argument("--streams-per-gpu", type=int, default=1,
+                        help="multi-stream batched serving: sessions per "
+                             "GPU batched through one engine (fbs=K; "
+                             "profiles/batching_ab.md measured +124% "
+                             "aggregate at K=8)")
     parser.add_argument("--workers", type=int, default=0,
                         help="process-per-GPU serving: spawn N worker agents "
                              "(one per GPU, own media sockets) behind a "
@@ -324,7 +332,8 @@ def main() -> None:
             fe.shutdown()
         return
     app = create_app(model_id=args.model_id, udp_ports=ports, n_gpus=args.gpus,
-                     family=args.family, resolution=args.resolution)
+                     family=args.family, resolution=args.resolution,
+                     streams_per_gpu=args.streams_per_gpu)
     web.run_app(app, host=args.host, port=args.port)
 
 

@@ -53,19 +53,32 @@ class VideoStreamTrack(MediaStreamTrack):
         self._warmup_left = config.warmup_frames()
         self._drop = config.drop_frames()
 
+    async def _invoke(self, tensor: torch.Tensor) -> torch.Tensor:
+        """Run the pipeline; batched-serving proxies return awaitables
+        (parallel/batching.py), plain pipelines return tensors."""
+        import inspect
+
+        out = self.pipeline(tensor)
+        if inspect.isawaitable(out):
+            out = await out
+        return out
+
     async def recv(self) -> VideoFrame:
         if self._warmup_left > 0:
             # burn-in: run the pipeline, discard the output
             self._warmup_left -= 1
             frame = await self.track.recv()
-            _ = self.pipeline(frame.tensor)
+            try:
+                _ = await self._invoke(frame.tensor)
+            except asyncio.CancelledError:
+                pass  # superseded by a newer frame (batched mode)
             if self._warmup_left == 0:
                 self.warmed_up = True
             return frame
         for _ in range(self._drop):
             await self.track.recv()
         frame = await self.track.recv()
-        out = self.pipeline(frame.tensor)
+        out = await self._invoke(frame.tensor)
         return VideoFrame(tensor=out, pts=frame.pts, time_base=frame.time_base)
 
 

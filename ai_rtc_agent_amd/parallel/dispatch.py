@@ -34,11 +34,31 @@ class PipelinePool:
 
     # -- construction ----------------------------------------------------
     @staticmethod
-    def create(model_id: str, n_gpus: int = 1, cfg=None) -> "PipelinePool":
+    def create(model_id: str, n_gpus: int = 1, cfg=None,
+               streams_per_replica: int = 1) -> "PipelinePool":
+        """streams_per_replica > 1 enables multi-stream batched serving:
+        each replica's engine runs frame_buffer_size=K and up to K sessions
+        share it through the asyncio collation loop (parallel/batching.py;
+        measured +124% aggregate throughput at K=8 on MI355X,
+        profiles/batching_ab.md)."""
         from ..config import EngineConfig
         from ..pipeline import StreamDiffusionPipeline
 
         import copy
+
+        def build(device: str):
+            c = copy.deepcopy(cfg) if cfg is not None else EngineConfig(model_id=model_id)
+            c.device = device
+            if device == "cpu":
+                c.use_hip_graph = False
+            if streams_per_replica > 1:
+                c.frame_buffer_size = streams_per_replica
+            p = StreamDiffusionPipeline(model_id, cfg=c)
+            if streams_per_replica > 1:
+                from .batching import BatchedPipeline
+
+                return BatchedPipeline(p, streams_per_replica)
+            return p
 
         pipelines = []
         if torch.cuda.is_available():
@@ -46,15 +66,11 @@ class PipelinePool:
             for i in range(n):
                 # per-replica config copy: replicas must not share (and
                 # last-write) one mutable config object
-                c = copy.deepcopy(cfg) if cfg is not None else EngineConfig(model_id=model_id)
-                c.device = f"cuda:{i}"
-                pipelines.append(StreamDiffusionPipeline(model_id, cfg=c))
+                pipelines.append(build(f"cuda:{i}"))
         else:
-            c = copy.deepcopy(cfg) if cfg is not None else EngineConfig(model_id=model_id)
-            c.device = "cpu"
-            c.use_hip_graph = False
-            pipelines.append(StreamDiffusionPipeline(model_id, cfg=c))
-        logger.info("pipeline pool: %d replica(s)", len(pipelines))
+            pipelines.append(build("cpu"))
+        logger.info("pipeline pool: %d replica(s) x %d stream slot(s)",
+                    len(pipelines), streams_per_replica)
         return PipelinePool(pipelines)
 
     @staticmethod
@@ -63,18 +79,45 @@ class PipelinePool:
 
     # -- session affinity -------------------------------------------------
     def assign(self, stream_id: str):
+        from .batching import BatchedPipeline
+
         if stream_id in self._sessions:
-            return self._pipelines[self._sessions[stream_id]]
+            idx = self._sessions[stream_id]
+            p = self._pipelines[idx]
+            return p.acquire(stream_id) if isinstance(p, BatchedPipeline) else p
+        for idx in sorted(self._load, key=lambda i: self._load[i]):
+            p = self._pipelines[idx]
+            if isinstance(p, BatchedPipeline):
+                proxy = p.acquire(stream_id)
+                if proxy is None:
+                    continue  # replica's slots are full
+                self._load[idx] += 1
+                self._sessions[stream_id] = idx
+                logger.info("stream %s -> replica %d slot %d",
+                            stream_id, idx, proxy.slot)
+                return proxy
+            self._load[idx] += 1
+            self._sessions[stream_id] = idx
+            logger.info("stream %s -> replica %d", stream_id, idx)
+            return p
+        # everything full: overload the least-loaded batched replica's base
         idx = min(self._load, key=lambda i: self._load[i])
         self._load[idx] += 1
         self._sessions[stream_id] = idx
-        logger.info("stream %s -> replica %d", stream_id, idx)
-        return self._pipelines[idx]
+        logger.warning("all stream slots busy; stream %s overloads replica %d",
+                       stream_id, idx)
+        p = self._pipelines[idx]
+        return p.base if isinstance(p, BatchedPipeline) else p
 
     def release(self, stream_id: str) -> None:
+        from .batching import BatchedPipeline
+
         idx = self._sessions.pop(stream_id, None)
         if idx is not None:
             self._load[idx] = max(0, self._load[idx] - 1)
+            p = self._pipelines[idx]
+            if isinstance(p, BatchedPipeline):
+                p.release(stream_id)
 
     def active(self) -> List:
         return self._pipelines
