@@ -1,15 +1,16 @@
 """Peer connection + ICE-lite UDP media transport — from scratch.
 
 Stands in for the aiortc RTCPeerConnection the reference uses
-(agent.py:136,299). Scope honestly stated: SDP offer/answer signalling, ICE-
-lite (STUN binding answerer, peer-reflexive address learning), RTP media
-with the codec HAL, and a lightweight JSON config channel over the same
-socket (magic-prefixed datagrams) mirroring the reference's datachannel
-config updates (agent.py:154-168). DTLS-SRTP is NOT implemented — this
-image has no DTLS-capable library (no aiortc/cryptography; stdlib ssl has
-no DTLS), so browser interop requires adding a DTLS stack on a deployment
-box; LAN/loopback peers use this transport directly. docs/webrtc.md carries
-the full gap statement.
+(agent.py:136,299). Scope: SDP offer/answer signalling, ICE-lite (STUN
+binding answerer, peer-reflexive address learning), DTLS-SRTP (native
+OpenSSL endpoint, media/dtls.py — armed whenever the remote SDP carries a
+fingerprint, which is what every browser/OBS offer does), a TURN relay
+client (media/turn.py — allocated when a turn: ice server with credentials
+is configured, advertised as an extra relay candidate), RTP media with the
+codec HAL (standard H.264 by default), and a lightweight JSON config
+channel over the same socket (magic-prefixed datagrams) mirroring the
+reference's datachannel config updates (agent.py:154-168). Offers WITHOUT
+a fingerprint (this repo's own test peers) run plain RTP.
 
 UDP port pinning: the reference monkey-patches asyncio's datagram endpoint
 factory to force media onto operator ports (agent.py:32-69, for firewalls /
@@ -113,6 +114,9 @@ class PeerConnection:
         self._dtls_expected_fp: Optional[str] = None
         self._dtls_verified = False
         self._dtls_task: Optional[asyncio.Task] = None
+        # TURN relay (allocated when a turn: ice server is configured)
+        self._relay = None
+        self._relay_peers: set = set()
 
     # -- event API (aiortc-style) --------------------------------------
     def on(self, event: str, handler: Optional[Callable] = None):
@@ -165,6 +169,7 @@ class PeerConnection:
         assert self.remote_description is not None, "set_remote_description first"
         from . import dtls as dtls_mod
 
+        await self._maybe_allocate_relay()
         ans = build_answer(
             self.remote_description,
             host,
@@ -174,12 +179,65 @@ class PeerConnection:
             direction=direction,
             fingerprint=dtls_mod.local_fingerprint(),
         )
+        if self._relay is not None and self._relay.client.relayed_addr:
+            rip, rport = self._relay.client.relayed_addr
+            for m in ans.media:
+                m.candidates.append(
+                    f"candidate:2 1 udp 16777215 {rip} {rport} typ relay "
+                    f"raddr 0.0.0.0 rport 0")
         self.local_description = ans
         self._ice_pwd = ans.media[0].ice_pwd if ans.media else ""
         if self._dtls is not None and self._dtls_task is None:
             self._dtls_task = asyncio.ensure_future(self._dtls_timer())
         self._set_state("connecting")
         return ans.serialize()
+
+    async def _maybe_allocate_relay(self) -> None:
+        """Allocate a TURN relay when a turn: ice server with credentials
+        is configured (reference gets this from aioice + Twilio,
+        agent.py:80-109). Host candidates remain primary; the relayed
+        address is advertised as an extra candidate and any peer that
+        reaches us through it is answered through it."""
+        if self._relay is not None or not self.ice_servers:
+            return
+        from .turn import TurnClient, TurnTransport, parse_turn_url
+
+        for srv in self.ice_servers:
+            urls = getattr(srv, "urls", None) if not isinstance(srv, dict) \
+                else srv.get("urls")
+            username = getattr(srv, "username", None) if not isinstance(srv, dict) \
+                else srv.get("username")
+            credential = getattr(srv, "credential", None) if not isinstance(srv, dict) \
+                else srv.get("credential")
+            if not urls or not username:
+                continue
+            for url in urls if isinstance(urls, list) else [urls]:
+                addr = parse_turn_url(url)
+                if addr is None:
+                    continue
+                tt = TurnTransport(TurnClient(username, credential or ""),
+                                   addr, self._on_relay_data)
+                try:
+                    if await tt.allocate(timeout=3.0):
+                        self._relay = tt
+                        logger.info("TURN relay allocated at %s:%d",
+                                    *tt.client.relayed_addr)
+                        return
+                except OSError:
+                    pass
+                tt.close()
+
+    def _on_relay_data(self, data: bytes, peer) -> None:
+        self._relay_peers.add(peer)
+        self._on_datagram(data, peer)
+
+    def _sendto(self, data: bytes, addr) -> None:
+        """Route to the peer: via the TURN relay for peers that reached us
+        through it, else directly over our UDP socket."""
+        if self._relay is not None and addr in self._relay_peers:
+            self._relay.sendto(data, addr)
+        elif self._transport is not None:
+            self._transport.sendto(data, addr)
 
     async def _dtls_timer(self) -> None:
         """Drive DTLS retransmissions until the handshake completes (the
@@ -192,7 +250,7 @@ class PeerConnection:
                 await asyncio.sleep(0.4)
                 if self._transport is not None and self._remote_addr:
                     for out in self._dtls.handle_timeout():
-                        self._transport.sendto(out, self._remote_addr)
+                        self._sendto(out, self._remote_addr)
         except asyncio.CancelledError:
             pass
 
@@ -278,7 +336,7 @@ class PeerConnection:
                 return
             data = (self._dtls.protect_rtcp(data) if is_rtcp(data)
                     else self._dtls.protect_rtp(data))
-        self._transport.sendto(data, self._remote_addr)
+        self._sendto(data, self._remote_addr)
 
     def _dtls_on_established(self) -> None:
         from . import dtls as dtls_mod
@@ -302,7 +360,7 @@ class PeerConnection:
                 return
             if msg.msg_type == BINDING_REQUEST and self._transport is not None:
                 resp = make_binding_response(msg, addr, self._ice_pwd.encode())
-                self._transport.sendto(resp, addr)
+                self._sendto(resp, addr)
                 self._remote_addr = addr  # peer-reflexive
                 if self._dtls is None:
                     self._set_state("connected")
@@ -311,8 +369,7 @@ class PeerConnection:
         if self._dtls is not None and data and 20 <= data[0] <= 63:
             self._remote_addr = addr
             for out in self._dtls.feed(data):
-                if self._transport is not None:
-                    self._transport.sendto(out, addr)
+                self._sendto(out, addr)
             if self._dtls.established() and not self._dtls_verified:
                 self._dtls_on_established()
             return
@@ -464,6 +521,9 @@ class PeerConnection:
                 await self._sender_task
             except (asyncio.CancelledError, Exception):
                 pass
+        if self._relay is not None:
+            self._relay.close()
+            self._relay = None
         if self._transport is not None:
             self._transport.close()
             self._transport = None
