@@ -259,6 +259,44 @@ class H264SwDecoder {
   std::vector<uint8_t> buf_ = std::vector<uint8_t>((size_t)1024 * 1024 * 3);
 };
 
+// --- VCN hardware encode session (vcn.cpp; hardware-unvalidated, see the
+// scope note there — gated behind AIRTC_VCN_EXPERIMENTAL in media/codec.py)
+extern "C" {
+void* airtc_vcn_enc_create(int w, int h, char* errbuf, int errlen);
+void airtc_vcn_enc_destroy(void*);
+int airtc_vcn_enc_encode(void*, const uint8_t*, int qp, uint8_t*, int cap);
+}
+
+class VcnEncoder {
+ public:
+  VcnEncoder(int w, int h) : w_(w), h_(h) {
+    char err[256] = {0};
+    handle_ = airtc_vcn_enc_create(w, h, err, sizeof(err));
+    if (!handle_) throw std::runtime_error(std::string("VCN encode: ") + err);
+  }
+  ~VcnEncoder() {
+    if (handle_) airtc_vcn_enc_destroy(handle_);
+  }
+  VcnEncoder(const VcnEncoder&) = delete;
+  pybind11::bytes encode(pybind11::bytes rgb, int qp) {
+    std::string s(rgb);
+    TORCH_CHECK((int)s.size() == w_ * h_ * 3, "rgb buffer size mismatch");
+    std::vector<uint8_t> out((size_t)w_ * h_ * 2 + 65536);
+    int n;
+    {
+      pybind11::gil_scoped_release nogil;
+      n = airtc_vcn_enc_encode(handle_, (const uint8_t*)s.data(), qp,
+                               out.data(), (int)out.size());
+    }
+    TORCH_CHECK(n > 0, "VCN encode failed rc=", n);
+    return pybind11::bytes((const char*)out.data(), n);
+  }
+
+ private:
+  void* handle_;
+  int w_, h_;
+};
+
 pybind11::bytes h264_sps_pps(int width, int height) {
   uint8_t buf[256];
   int n = airtc_h264_sps_pps(width, height, buf, sizeof(buf));
@@ -304,4 +342,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def(pybind11::init<>())
       .def("decode", &H264SwDecoder::decode,
            "Annex-B AU -> (rgb bytes, w, h) | None");
+  pybind11::class_<VcnEncoder>(m, "VcnEncoder")
+      .def(pybind11::init<int, int>())
+      .def("encode", &VcnEncoder::encode,
+           "RGB24 bytes + QP -> Annex-B IDR via the VCN hardware encoder");
 }
