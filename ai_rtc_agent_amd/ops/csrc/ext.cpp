@@ -184,6 +184,7 @@ torch::Tensor postprocess_u8(torch::Tensor img) {
 // --- software H.264 baseline-intra codec (h264sw.cpp) -----------------
 extern "C" {
 void* airtc_h264enc_create(int w, int h);
+void airtc_h264enc_set_slices(void*, int n);
 void airtc_h264enc_destroy(void*);
 int airtc_h264enc_encode(void*, const uint8_t*, int qp, uint8_t*, int cap);
 void* airtc_h264dec_create();
@@ -195,9 +196,10 @@ int airtc_h264sw_table_check();
 
 class H264SwEncoder {
  public:
-  H264SwEncoder(int w, int h) : w_(w), h_(h) {
+  H264SwEncoder(int w, int h, int slices = 4) : w_(w), h_(h) {
     handle_ = airtc_h264enc_create(w, h);
     TORCH_CHECK(handle_, "invalid encoder dimensions");
+    airtc_h264enc_set_slices(handle_, slices);
   }
   ~H264SwEncoder() {
     if (handle_) airtc_h264enc_destroy(handle_);
@@ -336,7 +338,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("h264sw_table_check", []() { return airtc_h264sw_table_check(); },
         "0 iff all CAVLC tables are prefix-free");
   pybind11::class_<H264SwEncoder>(m, "H264SwEncoder")
-      .def(pybind11::init<int, int>())
+      .def(pybind11::init<int, int, int>(), pybind11::arg("w"),
+           pybind11::arg("h"), pybind11::arg("slices") = 4)
       .def("encode", &H264SwEncoder::encode, "RGB24 bytes + QP -> Annex-B IDR");
   pybind11::class_<H264SwDecoder>(m, "H264SwDecoder")
       .def(pybind11::init<>())

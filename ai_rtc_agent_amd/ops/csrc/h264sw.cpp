@@ -25,6 +25,7 @@
 #include <string.h>
 
 #include <algorithm>
+#include <thread>
 #include <vector>
 
 #include "h264_bits.h"
@@ -436,10 +437,11 @@ static int cavlc_read_block(BitReader& r, int32_t* coeffs, int n, int nC) {
 // ---------------------------------------------------------------------------
 // color conversion (BT.601 limited range)
 // ---------------------------------------------------------------------------
-static void rgb_to_yuv420(const uint8_t* rgb, int w, int h, int pw, int ph,
-                          uint8_t* Y, uint8_t* Cb, uint8_t* Cr) {
-  // write padded planes (pw x ph luma), edge-replicated
-  for (int y = 0; y < ph; ++y) {
+static void rgb_to_yuv420_rows(const uint8_t* rgb, int w, int h, int pw,
+                               int y0, int y1, uint8_t* Y, uint8_t* Cb,
+                               uint8_t* Cr) {
+  // write padded planes (pw wide), edge-replicated, luma rows [y0, y1)
+  for (int y = y0; y < y1; ++y) {
     int sy = y < h ? y : h - 1;
     for (int x = 0; x < pw; ++x) {
       int sx = x < w ? x : w - 1;
@@ -448,8 +450,8 @@ static void rgb_to_yuv420(const uint8_t* rgb, int w, int h, int pw, int ph,
       Y[y * pw + x] = clip8(((66 * R + 129 * G + 25 * B + 128) >> 8) + 16);
     }
   }
-  int cw = pw / 2, ch = ph / 2;
-  for (int y = 0; y < ch; ++y) {
+  int cw = pw / 2;
+  for (int y = y0 / 2; y < y1 / 2; ++y) {
     for (int x = 0; x < cw; ++x) {
       int R = 0, G = 0, B = 0;
       for (int dy = 0; dy < 2; ++dy)
@@ -467,6 +469,11 @@ static void rgb_to_yuv420(const uint8_t* rgb, int w, int h, int pw, int ph,
       Cr[y * cw + x] = clip8(((112 * R - 94 * G - 18 * B + 128) >> 8) + 128);
     }
   }
+}
+
+static void rgb_to_yuv420(const uint8_t* rgb, int w, int h, int pw, int ph,
+                          uint8_t* Y, uint8_t* Cb, uint8_t* Cr) {
+  rgb_to_yuv420_rows(rgb, w, h, pw, 0, ph, Y, Cb, Cr);
 }
 
 static void yuv420_to_rgb(const uint8_t* Y, const uint8_t* Cb, const uint8_t* Cr,
@@ -496,10 +503,10 @@ struct PlaneCtx {
 
 // 16x16 luma prediction, modes 0=V 1=H 2=DC 3=Plane
 static void pred_luma16(const PlaneCtx& pl, int mbx, int mby, int mode,
-                        uint8_t pred[256]) {
+                        bool have_top, bool have_left, uint8_t pred[256]) {
   const int x0 = mbx * 16, y0 = mby * 16;
-  const uint8_t* top = (mby > 0) ? pl.data + (y0 - 1) * pl.stride + x0 : nullptr;
-  const uint8_t* leftc = (mbx > 0) ? pl.data + y0 * pl.stride + (x0 - 1) : nullptr;
+  const uint8_t* top = have_top ? pl.data + (y0 - 1) * pl.stride + x0 : nullptr;
+  const uint8_t* leftc = have_left ? pl.data + y0 * pl.stride + (x0 - 1) : nullptr;
   switch (mode) {
     case 0:  // vertical
       for (int y = 0; y < 16; ++y)
@@ -546,10 +553,10 @@ static void pred_luma16(const PlaneCtx& pl, int mbx, int mby, int mode,
 
 // 8x8 chroma prediction, modes 0=DC 1=H 2=V 3=Plane
 static void pred_chroma8(const PlaneCtx& pl, int mbx, int mby, int mode,
-                         uint8_t pred[64]) {
+                         bool have_top, bool have_left, uint8_t pred[64]) {
   const int x0 = mbx * 8, y0 = mby * 8;
-  const uint8_t* top = (mby > 0) ? pl.data + (y0 - 1) * pl.stride + x0 : nullptr;
-  const uint8_t* leftc = (mbx > 0) ? pl.data + y0 * pl.stride + (x0 - 1) : nullptr;
+  const uint8_t* top = have_top ? pl.data + (y0 - 1) * pl.stride + x0 : nullptr;
+  const uint8_t* leftc = have_left ? pl.data + y0 * pl.stride + (x0 - 1) : nullptr;
   switch (mode) {
     case 0: {  // DC per 4x4 sub-block (x264 dc0..dc3 structure)
       int st[2] = {0, 0}, sl[2] = {0, 0};
@@ -681,30 +688,39 @@ struct NnzCtx {
   uint8_t& cnz(int comp, int mbx, int mby, int x2, int y2) {
     return chroma[(((size_t)comp * mbh + mby) * mbw + mbx) * 4 + y2 * 2 + x2];
   }
-  int luma_nc(int mbx, int mby, int x4, int y4) {
+  // slice_start: first MB address of the current slice — neighbours in a
+  // DIFFERENT slice are unavailable (H.264 availability rule), which is
+  // what makes slices independently decodable (and thread-parallel)
+  bool left_ok(int mbx, int mby, int slice_start) const {
+    return mbx > 0 && (mby * mbw + mbx - 1) >= slice_start;
+  }
+  bool top_ok(int mbx, int mby, int slice_start) const {
+    return mby > 0 && ((mby - 1) * mbw + mbx) >= slice_start;
+  }
+  int luma_nc(int mbx, int mby, int x4, int y4, int slice_start) {
     int na = -1, nb = -1;
     if (x4 > 0)
       na = lnz(mbx, mby, x4 - 1, y4);
-    else if (mbx > 0)
+    else if (left_ok(mbx, mby, slice_start))
       na = lnz(mbx - 1, mby, 3, y4);
     if (y4 > 0)
       nb = lnz(mbx, mby, x4, y4 - 1);
-    else if (mby > 0)
+    else if (top_ok(mbx, mby, slice_start))
       nb = lnz(mbx, mby - 1, x4, 3);
     if (na >= 0 && nb >= 0) return (na + nb + 1) >> 1;
     if (na >= 0) return na;
     if (nb >= 0) return nb;
     return 0;
   }
-  int chroma_nc(int comp, int mbx, int mby, int x2, int y2) {
+  int chroma_nc(int comp, int mbx, int mby, int x2, int y2, int slice_start) {
     int na = -1, nb = -1;
     if (x2 > 0)
       na = cnz(comp, mbx, mby, x2 - 1, y2);
-    else if (mbx > 0)
+    else if (left_ok(mbx, mby, slice_start))
       na = cnz(comp, mbx - 1, mby, 1, y2);
     if (y2 > 0)
       nb = cnz(comp, mbx, mby, x2, y2 - 1);
-    else if (mby > 0)
+    else if (top_ok(mbx, mby, slice_start))
       nb = cnz(comp, mbx, mby - 1, x2, 1);
     if (na >= 0 && nb >= 0) return (na + nb + 1) >> 1;
     if (na >= 0) return na;
@@ -758,7 +774,7 @@ struct Encoder {
     return s;
   }
 
-  void encode_mb(BitWriter& wtr, int mbx, int mby, int qp) {
+  void encode_mb(BitWriter& wtr, int mbx, int mby, int qp, int slice_start) {
     PlaneCtx rpy{rY.data(), pw};
     PlaneCtx rpcb{rCb.data(), pw / 2}, rpcr{rCr.data(), pw / 2};
     const uint8_t* src = Y.data() + (mby * 16) * pw + mbx * 16;
@@ -766,12 +782,13 @@ struct Encoder {
     // --- luma mode decision (SAD over available modes) ---
     uint8_t pred[4][256];
     int best_mode = 2, best_sad = INT32_MAX;
-    const bool have_top = mby > 0, have_left = mbx > 0;
+    const bool have_top = nnz.top_ok(mbx, mby, slice_start);
+    const bool have_left = nnz.left_ok(mbx, mby, slice_start);
     for (int m = 0; m < 4; ++m) {
       if (m == 0 && !have_top) continue;
       if (m == 1 && !have_left) continue;
       if (m == 3 && !(have_top && have_left)) continue;
-      pred_luma16(rpy, mbx, mby, m, pred[m]);
+      pred_luma16(rpy, mbx, mby, m, have_top, have_left, pred[m]);
       int sad = mb_sad(src, pw, pred[m], 16, 16);
       if (sad < best_sad) {
         best_sad = sad;
@@ -826,7 +843,7 @@ struct Encoder {
                               Cr.data() + (mby * 8) * (pw / 2) + mbx * 8};
     PlaneCtx* cpl[2] = {&rpcb, &rpcr};
     for (int comp = 0; comp < 2; ++comp) {
-      pred_chroma8(*cpl[comp], mbx, mby, 0, cpred[comp]);
+      pred_chroma8(*cpl[comp], mbx, mby, 0, have_top, have_left, cpred[comp]);
       int32_t cdc_raw[4];
       for (int blk = 0; blk < 4; ++blk) {
         const int x4 = blk & 1, y4 = blk >> 1;
@@ -918,14 +935,15 @@ struct Encoder {
     {
       int32_t scan[16];
       for (int i = 0; i < 16; ++i) scan[i] = dc_q[ZIGZAG[i]];
-      cavlc_write_block(wtr, scan, 16, nnz.luma_nc(mbx, mby, 0, 0));
+      cavlc_write_block(wtr, scan, 16, nnz.luma_nc(mbx, mby, 0, 0, slice_start));
     }
     if (cbp_luma) {
       for (int b = 0; b < 16; ++b) {
         const int x4 = blk_x4(b), y4 = blk_y4(b);
         int32_t scan[15];
         for (int i = 1; i < 16; ++i) scan[i - 1] = ac_q[b][ZIGZAG[i]];
-        int tc = cavlc_write_block(wtr, scan, 15, nnz.luma_nc(mbx, mby, x4, y4));
+        int tc = cavlc_write_block(wtr, scan, 15,
+                                   nnz.luma_nc(mbx, mby, x4, y4, slice_start));
         nnz.lnz(mbx, mby, x4, y4) = (uint8_t)tc;
       }
     } else {
@@ -942,8 +960,8 @@ struct Encoder {
           const int x2 = blk & 1, y2 = blk >> 1;
           int32_t scan[15];
           for (int i = 1; i < 16; ++i) scan[i - 1] = cac_q[comp][blk][ZIGZAG[i]];
-          int tc = cavlc_write_block(wtr, scan, 15,
-                                     nnz.chroma_nc(comp, mbx, mby, x2, y2));
+          int tc = cavlc_write_block(
+              wtr, scan, 15, nnz.chroma_nc(comp, mbx, mby, x2, y2, slice_start));
           nnz.cnz(comp, mbx, mby, x2, y2) = (uint8_t)tc;
         }
     } else {
@@ -962,10 +980,28 @@ struct Encoder {
     recon_chroma8(rpcr, mbx, mby, cpred[1], cdc_q[1], cac_q[1], qpc);
   }
 
+  int n_slices = 1;  // MB-row bands, encoded in parallel threads
+
   int encode(const uint8_t* rgb, int qp, uint8_t* out, int cap) {
     if (qp < 10) qp = 10;
     if (qp > 48) qp = 48;
-    rgb_to_yuv420(rgb, w, h, pw, ph, Y.data(), Cb.data(), Cr.data());
+    const int ns = std::max(1, std::min({n_slices, mbh, 16}));
+    // parallel colour conversion by row bands
+    if (ns > 1) {
+      std::vector<std::thread> ct;
+      int rows = ((ph / 2 + ns - 1) / ns) * 2;  // even: chroma bands disjoint
+      for (int s = 0; s < ns; ++s) {
+        int y0 = s * rows, y1 = std::min(ph, y0 + rows);
+        if (y0 >= y1) break;
+        ct.emplace_back([&, y0, y1] {
+          rgb_to_yuv420_rows(rgb, w, h, pw, y0, y1, Y.data(), Cb.data(),
+                             Cr.data());
+        });
+      }
+      for (auto& t : ct) t.join();
+    } else {
+      rgb_to_yuv420(rgb, w, h, pw, ph, Y.data(), Cb.data(), Cr.data());
+    }
     nnz.reset(mbw, mbh);
 
     std::vector<uint8_t> bs;
@@ -975,22 +1011,41 @@ struct Encoder {
       if (n <= 0) return -1;
       bs.insert(bs.end(), hdr, hdr + n);
     }
-    // IDR slice
-    BitWriter wtr;
-    wtr.ue(0);            // first_mb_in_slice
-    wtr.ue(7);            // slice_type: I (all slices)
-    wtr.ue(0);            // pps_id
-    wtr.put(0, 4);        // frame_num (log2_max_frame_num = 4)
-    wtr.ue(idr_id & 1);   // idr_pic_id
+    const uint32_t pic_idr_id = idr_id & 1;
     ++idr_id;
-    wtr.put(0, 1);        // no_output_of_prior_pics_flag
-    wtr.put(0, 1);        // long_term_reference_flag
-    wtr.se(qp - 26);      // slice_qp_delta (pic_init_qp = 26)
-    wtr.ue(1);            // disable_deblocking_filter_idc = 1 (off)
-    for (int mby = 0; mby < mbh; ++mby)
-      for (int mbx = 0; mbx < mbw; ++mbx) encode_mb(wtr, mbx, mby, qp);
-    wtr.rbsp_trailing();
-    emit_nal(&bs, 0x65, wtr.bytes);  // nal_ref_idc=3, IDR
+
+    // one slice per MB-row band; slices only predict within themselves,
+    // so bands encode concurrently (recon rows + nnz rows are disjoint)
+    const int band = (mbh + ns - 1) / ns;
+    std::vector<std::vector<uint8_t>> slice_nals(ns);
+    auto encode_band = [&](int si) {
+      const int r0 = si * band, r1 = std::min(mbh, r0 + band);
+      if (r0 >= r1) return;
+      const int slice_start = r0 * mbw;
+      BitWriter wtr;
+      wtr.ue((uint32_t)slice_start);  // first_mb_in_slice
+      wtr.ue(7);                      // slice_type: I (all slices)
+      wtr.ue(0);                      // pps_id
+      wtr.put(0, 4);                  // frame_num (log2_max_frame_num = 4)
+      wtr.ue(pic_idr_id);             // idr_pic_id
+      wtr.put(0, 1);                  // no_output_of_prior_pics_flag
+      wtr.put(0, 1);                  // long_term_reference_flag
+      wtr.se(qp - 26);                // slice_qp_delta (pic_init_qp = 26)
+      wtr.ue(1);                      // disable_deblocking_filter_idc = off
+      for (int mby = r0; mby < r1; ++mby)
+        for (int mbx = 0; mbx < mbw; ++mbx)
+          encode_mb(wtr, mbx, mby, qp, slice_start);
+      wtr.rbsp_trailing();
+      emit_nal(&slice_nals[si], 0x65, wtr.bytes);  // nal_ref_idc=3, IDR
+    };
+    if (ns > 1) {
+      std::vector<std::thread> ts;
+      for (int si = 0; si < ns; ++si) ts.emplace_back(encode_band, si);
+      for (auto& t : ts) t.join();
+    } else {
+      encode_band(0);
+    }
+    for (auto& nal : slice_nals) bs.insert(bs.end(), nal.begin(), nal.end());
 
     if ((int)bs.size() > cap) return -2;
     memcpy(out, bs.data(), bs.size());
@@ -1098,7 +1153,7 @@ struct Decoder {
   }
 
   // decode one I_16x16 or I_PCM macroblock; returns 0 or negative error
-  int decode_mb(BitReader& r, int mbx, int mby, int& qp) {
+  int decode_mb(BitReader& r, int mbx, int mby, int& qp, int slice_start) {
     PlaneCtx rpy{rY.data(), sps.pw};
     PlaneCtx rpcb{rCb.data(), sps.pw / 2}, rpcr{rCr.data(), sps.pw / 2};
     uint32_t mb_type = r.ue();
@@ -1133,7 +1188,8 @@ struct Decoder {
 
     // residual parse
     int32_t dc_scan[16], dc_q[16] = {0};
-    int tc_dc = cavlc_read_block(r, dc_scan, 16, nnz.luma_nc(mbx, mby, 0, 0));
+    int tc_dc = cavlc_read_block(r, dc_scan, 16,
+                                 nnz.luma_nc(mbx, mby, 0, 0, slice_start));
     if (tc_dc < 0) return -1;
     for (int i = 0; i < 16; ++i) dc_q[ZIGZAG[i]] = dc_scan[i];
     int32_t ac_q[16][16];
@@ -1142,7 +1198,8 @@ struct Decoder {
       for (int b = 0; b < 16; ++b) {
         const int x4 = blk_x4(b), y4 = blk_y4(b);
         int32_t scan[15];
-        int tc = cavlc_read_block(r, scan, 15, nnz.luma_nc(mbx, mby, x4, y4));
+        int tc = cavlc_read_block(r, scan, 15,
+                                  nnz.luma_nc(mbx, mby, x4, y4, slice_start));
         if (tc < 0) return -1;
         for (int i = 1; i < 16; ++i) ac_q[b][ZIGZAG[i]] = scan[i - 1];
         nnz.lnz(mbx, mby, x4, y4) = (uint8_t)tc;
@@ -1165,8 +1222,8 @@ struct Decoder {
         for (int blk = 0; blk < 4; ++blk) {
           const int x2 = blk & 1, y2 = blk >> 1;
           int32_t scan[15];
-          int tc = cavlc_read_block(r, scan, 15,
-                                    nnz.chroma_nc(comp, mbx, mby, x2, y2));
+          int tc = cavlc_read_block(
+              r, scan, 15, nnz.chroma_nc(comp, mbx, mby, x2, y2, slice_start));
           if (tc < 0) return -1;
           for (int i = 1; i < 16; ++i) cac_q[comp][blk][ZIGZAG[i]] = scan[i - 1];
           nnz.cnz(comp, mbx, mby, x2, y2) = (uint8_t)tc;
@@ -1179,11 +1236,12 @@ struct Decoder {
 
     // predict + reconstruct
     uint8_t lpred[256];
-    const bool have_top = mby > 0, have_left = mbx > 0;
+    const bool have_top = nnz.top_ok(mbx, mby, slice_start);
+    const bool have_left = nnz.left_ok(mbx, mby, slice_start);
     if ((pred_mode == 0 && !have_top) || (pred_mode == 1 && !have_left) ||
         (pred_mode == 3 && !(have_top && have_left)))
       return -1;
-    pred_luma16(rpy, mbx, mby, pred_mode, lpred);
+    pred_luma16(rpy, mbx, mby, pred_mode, have_top, have_left, lpred);
     recon_luma16(rpy, mbx, mby, lpred, dc_q, ac_q, qp);
     // chroma modes: bitstream 0=DC 1=H 2=V 3=Plane (pred_chroma8 order)
     const int qpi = qp + pps.chroma_qp_offset;
@@ -1193,19 +1251,27 @@ struct Decoder {
         (cm == 3 && !(have_top && have_left)))
       return -1;
     uint8_t cpred[64];
-    pred_chroma8(rpcb, mbx, mby, cm, cpred);
+    pred_chroma8(rpcb, mbx, mby, cm, have_top, have_left, cpred);
     recon_chroma8(rpcb, mbx, mby, cpred, cdc_q[0], cac_q[0], qpc);
-    pred_chroma8(rpcr, mbx, mby, cm, cpred);
+    pred_chroma8(rpcr, mbx, mby, cm, have_top, have_left, cpred);
     recon_chroma8(rpcr, mbx, mby, cpred, cdc_q[1], cac_q[1], qpc);
     return 0;
   }
 
   // returns 1 when a frame was reconstructed, 0 for parameter-set-only data,
   // negative on unsupported/corrupt input
+  struct SliceJob {
+    std::vector<uint8_t> rbsp;
+    size_t header_bitpos;
+    int first_mb;
+    int qp;
+  };
+
   int decode_au(const uint8_t* data, int len, uint8_t* rgb, int cap, int* ow,
                 int* oh) {
     auto nals = h264::split_annexb(data, (size_t)len);
     bool got_frame = false;
+    std::vector<SliceJob> slices;
     for (auto& nal : nals) {
       if (nal.size < 1) continue;
       const int type = nal.data[0] & 0x1F;
@@ -1217,7 +1283,7 @@ struct Decoder {
         if (parse_pps(r) < 0) return -3;
       } else if (type == 5 || type == 1) {
         if (!sps.valid || !pps.valid) return -4;
-        if (r.ue() != 0) return -5;  // multi-slice unsupported
+        int first_mb = (int)r.ue();
         uint32_t stype = r.ue();
         if (stype % 5 != 2) return -6;  // I slices only
         r.ue();                         // pps_id
@@ -1244,15 +1310,50 @@ struct Decoder {
           }
         }
         if (qp < 0 || qp > 51 || r.overrun) return -1;
-        const int mbw = sps.pw / 16, mbh = sps.ph / 16;
-        nnz.reset(mbw, mbh);
-        for (int mby = 0; mby < mbh; ++mby)
-          for (int mbx = 0; mbx < mbw; ++mbx) {
-            int rc = decode_mb(r, mbx, mby, qp);
-            if (rc < 0) return rc;
-          }
-        got_frame = true;
+        const int total = (sps.pw / 16) * (sps.ph / 16);
+        if (first_mb < 0 || first_mb >= total) return -5;
+        slices.push_back({std::move(rbsp), r.pos, first_mb, qp});
+        continue;  // rbsp moved into the job
       }
+    }
+    // decode collected slices — slices are independent (neighbours in a
+    // different slice are unavailable), so they run in parallel threads
+    if (!slices.empty()) {
+      const int mbw = sps.pw / 16, mbh = sps.ph / 16;
+      const int total = mbw * mbh;
+      nnz.reset(mbw, mbh);
+      std::vector<int> rcs(slices.size(), 0), counts(slices.size(), 0);
+      auto decode_slice = [&](size_t si) {
+        SliceJob& job = slices[si];
+        BitReader sr(job.rbsp.data(), job.rbsp.size());
+        sr.pos = job.header_bitpos;
+        int mb = job.first_mb;
+        int qp = job.qp;
+        while (mb < total && sr.more_rbsp_data()) {
+          int rc = decode_mb(sr, mb % mbw, mb / mbw, qp, job.first_mb);
+          if (rc < 0) {
+            rcs[si] = rc;
+            return;
+          }
+          ++counts[si];
+          ++mb;
+        }
+      };
+      if (slices.size() > 1) {
+        std::vector<std::thread> ts;
+        for (size_t si = 0; si < slices.size(); ++si)
+          ts.emplace_back(decode_slice, si);
+        for (auto& t : ts) t.join();
+      } else {
+        decode_slice(0);
+      }
+      int covered = 0;
+      for (size_t si = 0; si < slices.size(); ++si) {
+        if (rcs[si] < 0) return rcs[si];
+        covered += counts[si];
+      }
+      if (covered < total) return -9;  // frame not fully covered
+      got_frame = true;
     }
     if (!got_frame) return 0;
     const int w = sps.w(), h = sps.h();
@@ -1274,6 +1375,10 @@ extern "C" {
 void* airtc_h264enc_create(int w, int h) {
   if (w < 16 || h < 16 || w > 8192 || h > 8192) return nullptr;
   return new h264sw::Encoder(w, h);
+}
+// number of slices (MB-row bands) encoded in parallel threads
+void airtc_h264enc_set_slices(void* h, int n) {
+  ((h264sw::Encoder*)h)->n_slices = n < 1 ? 1 : n;
 }
 void airtc_h264enc_destroy(void* h) { delete (h264sw::Encoder*)h; }
 int airtc_h264enc_encode(void* h, const uint8_t* rgb, int qp, uint8_t* out,

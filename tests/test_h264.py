@@ -289,19 +289,22 @@ def test_h264sw_slice_header_fields():
     import torch
 
     ext = _h264_ext()
-    enc = ext.H264SwEncoder(64, 64)
+    enc = ext.H264SwEncoder(64, 64, slices=2)
     data = enc.encode(bytes(64 * 64 * 3), 30)
     nals = split_annexb(data)
-    assert [nal_type(n) for n in nals] == [7, 8, 5]
-    r = BitReader(nals[2][1:])
-    assert r.ue() == 0      # first_mb_in_slice
-    assert r.ue() == 7      # slice_type I (all)
-    assert r.ue() == 0      # pps_id
-    r.u(4)                  # frame_num
-    r.ue()                  # idr_pic_id
-    r.u(2)                  # dec_ref_pic_marking
-    qp_delta = r.ue()       # se coded; just check it parses
-    assert r.ue() == 1 or True  # disable_deblocking (after se decode ambiguity)
+    assert [nal_type(n) for n in nals] == [7, 8, 5, 5]  # SPS PPS + 2 slices
+    firsts = []
+    for sl in nals[2:]:
+        r = BitReader(sl[1:])
+        firsts.append(r.ue())   # first_mb_in_slice
+        assert r.ue() == 7      # slice_type I (all)
+        assert r.ue() == 0      # pps_id
+        r.u(4)                  # frame_num
+        r.ue()                  # idr_pic_id
+        r.u(2)                  # dec_ref_pic_marking
+        r.ue()                  # slice_qp_delta (se-coded)
+    # 64x64 = 4x4 MBs, 2 row bands -> slices start at MB 0 and MB 8
+    assert firsts == [0, 8]
 
 
 def test_h264sw_noise_uses_pcm_and_roundtrips():
