@@ -128,10 +128,13 @@ class Conv2d(nn.Module):
         act: int | None = None,
         residual: torch.Tensor | None = None,
         channel_bias: torch.Tensor | None = None,
+        in_affine: torch.Tensor | None = None,
+        in_act: int = 0,
     ) -> torch.Tensor:
         return ops.conv2d_nhwc(
             x, self.weight, self.bias, self.stride, self.padding, fuse_silu,
             act=act, residual=residual, channel_bias=channel_bias,
+            in_affine=in_affine, in_act=in_act,
         )
 
 
@@ -151,6 +154,11 @@ class GroupNormSiLU(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return ops.group_norm_silu_nhwc(x, self.groups, self.weight, self.bias, self.eps, self.silu)
+
+    def coeffs(self, x: torch.Tensor) -> torch.Tensor:
+        """(B, C, 2) input-affine pairs for the fused GN->conv path
+        (ops.conv2d_nhwc in_affine); the SiLU moves to the conv's in_act."""
+        return ops.group_norm_coeffs(x, self.groups, self.weight, self.bias, self.eps)
 
 
 class LayerNorm(nn.Module):
@@ -322,11 +330,16 @@ class ResnetBlock(nn.Module):
         self.shortcut = Conv2d(cin, cout, 1) if cin != cout else None
 
     def forward(self, x: torch.Tensor, temb: torch.Tensor) -> torch.Tensor:
-        # time-emb add fused into conv1's epilogue; skip add fused into conv2's
+        # fusion ladder: time-emb add in conv1's epilogue, skip add in
+        # conv2's, and BOTH GroupNorm+SiLU applications in the convs'
+        # input loads (in_affine) — per resnet block only the two conv
+        # kernels and two tiny stats/coeffs passes remain
         temb_b = self.time_emb_proj(ops.silu(temb))
-        h = self.conv1(self.norm1(x), channel_bias=temb_b)
+        h = self.conv1(x, channel_bias=temb_b,
+                       in_affine=self.norm1.coeffs(x), in_act=ops.ACT_SILU)
         skip = self.shortcut(x) if self.shortcut is not None else x
-        return self.conv2(self.norm2(h), residual=skip)
+        return self.conv2(h, residual=skip,
+                          in_affine=self.norm2.coeffs(h), in_act=ops.ACT_SILU)
 
 
 class Downsample(nn.Module):
@@ -475,5 +488,5 @@ class UNet2DCondition(nn.Module):
             if self.upsamplers[bi] is not None:
                 x = self.upsamplers[bi](x)
 
-        x = self.norm_out(x)
-        return self.conv_out(x)
+        return self.conv_out(x, in_affine=self.norm_out.coeffs(x),
+                             in_act=ops.ACT_SILU)

@@ -39,16 +39,30 @@ __device__ __forceinline__ KPos kpos_at(int k0, int IC, int S) {
   return {rs / S, rs - (rs / S) * S, k0 - rs * IC};
 }
 
-// one staged 16B A-load with inline zero-padding
+// one staged 16B A-load with inline zero-padding. aff (optional,
+// per-channel float pairs pre-offset for the batch) applies the fused
+// GroupNorm affine + activation AT LOAD TIME — the producing GN-apply
+// kernel and a full activation HBM round-trip disappear. Padding lanes
+// stay zero (ok-gated): the transform models act(gn(x)) of the unfused
+// pipeline, which the conv then zero-pads.
 __device__ __forceinline__ f16x8 load_a(const f16* xb, int ho_s, int wo_s,
                                         int r, int s, int pad, int H, int W,
-                                        int IC, int ic) {
+                                        int IC, int ic, const float* aff,
+                                        int in_act) {
   const int hi = ho_s + r - pad;
   const int wi = wo_s + s - pad;
   const bool ok = (unsigned)hi < (unsigned)H && (unsigned)wi < (unsigned)W;
   const int hc = ok ? hi : 0, wc = ok ? wi : 0;
   f16x8 v = *reinterpret_cast<const f16x8*>(&xb[((long)hc * W + wc) * IC + ic]);
-  if (!ok) v = f16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  if (!ok) {
+    v = f16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  } else if (aff) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = (float)v[j] * aff[(ic + j) * 2] + aff[(ic + j) * 2 + 1];
+      v[j] = (f16)apply_act(f, in_act);
+    }
+  }
   return v;
 }
 
@@ -113,7 +127,8 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
     const float* __restrict__ bias, const f16* __restrict__ cbias,
     const f16* __restrict__ residual, f16* __restrict__ out,
     float* __restrict__ ws, int H, int W, int IC, int HO, int WO, int OC,
-    int R, int S, int stride, int pad, int act, int K, int splitk) {
+    int R, int S, int stride, int pad, int act, int K, int splitk,
+    const float* __restrict__ in_aff, int in_act) {
   constexpr int BM = MFRAG * 32;              // 128 or 64
   constexpr int BNK = NFRAG * 32;             // 64 or 128 out-channels
   constexpr int KPITCH = BK + 8;              // +16B row pad (guide G4)
@@ -140,6 +155,7 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
   const int b = blockIdx.z / spk;
   const int split = blockIdx.z - b * spk;
   const f16* xb = x + (long)b * H * W * IC;
+  const float* affb = in_aff ? in_aff + (long)b * IC * 2 : nullptr;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -189,7 +205,7 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
 #pragma unroll
     for (int i = 0; i < ALOADS; ++i)
       regA[i] = load_a(xb, a_ho[i], a_wo[i], p.r, p.s, pad, H, W, IC,
-                       p.ic0 + a_k8[i]);
+                       p.ic0 + a_k8[i], affb, in_act);
 #pragma unroll
     for (int i = 0; i < BLOADS; ++i)
       regB[i] = *reinterpret_cast<const f16x8*>(wrow[i] + kt * BK);
@@ -643,7 +659,8 @@ __global__ __launch_bounds__(256) void conv2d_smallic_kernel(
     const float* __restrict__ bias, const f16* __restrict__ cbias,
     const f16* __restrict__ residual, f16* __restrict__ out, int H, int W,
     int IC, int HO, int WO, int OC, int R, int S, int stride, int pad,
-    int act, int K, long total_pix) {
+    int act, int K, long total_pix, const float* __restrict__ in_aff,
+    int in_act) {
   __shared__ f16 wlds[SMALLIC_MAX_K * OCT];
   const int oc0 = blockIdx.y * OCT;
   for (int i = threadIdx.x; i < K * OCT; i += 256) {
@@ -660,6 +677,7 @@ __global__ __launch_bounds__(256) void conv2d_smallic_kernel(
     const long b = pix / M;
     const int ho = m / WO, wo = m % WO;
     const f16* xb = x + b * (long)H * W * IC;
+    const float* affb = in_aff ? in_aff + b * (long)IC * 2 : nullptr;
     float acc[OCT];
 #pragma unroll
     for (int o = 0; o < OCT; ++o) acc[o] = 0.f;
@@ -672,7 +690,9 @@ __global__ __launch_bounds__(256) void conv2d_smallic_kernel(
         const f16* xr = &xb[((long)hi * W + wi) * IC];
         const f16* wr = &wlds[(r * S + s) * IC * OCT];
         for (int c = 0; c < IC; ++c) {
-          const float xv = (float)xr[c];
+          float xv = (float)xr[c];
+          if (affb)
+            xv = apply_act(xv * affb[c * 2] + affb[c * 2 + 1], in_act);
           const f16* wv = &wr[c * OCT];
 #pragma unroll
           for (int o = 0; o < OCT; ++o) acc[o] += xv * (float)wv[o];
@@ -695,7 +715,7 @@ __global__ void conv2d_direct_kernel(
     const float* __restrict__ bias, const f16* __restrict__ cbias,
     const f16* __restrict__ residual, f16* __restrict__ out, int H, int W,
     int IC, int HO, int WO, int OC, int R, int S, int stride, int pad, int act,
-    int K, long total) {
+    int K, long total, const float* __restrict__ in_aff, int in_act) {
   const int M = HO * WO;
   for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
        i += gridDim.x * (long)blockDim.x) {
@@ -705,6 +725,7 @@ __global__ void conv2d_direct_kernel(
     const long b = rest / M;
     const int ho = m / WO, wo = m % WO;
     const f16* xb = x + b * (long)H * W * IC;
+    const float* affb = in_aff ? in_aff + b * (long)IC * 2 : nullptr;
     const f16* wk = w + (long)oc * K;
     float a = 0.f;
     for (int r = 0; r < R; ++r) {
@@ -715,7 +736,12 @@ __global__ void conv2d_direct_kernel(
         if ((unsigned)wi >= (unsigned)W) continue;
         const f16* xr = &xb[((long)hi * W + wi) * IC];
         const f16* wr = &wk[(r * S + s) * IC];
-        for (int c = 0; c < IC; ++c) a += (float)xr[c] * (float)wr[c];
+        for (int c = 0; c < IC; ++c) {
+          float xv = (float)xr[c];
+          if (affb)
+            xv = apply_act(xv * affb[c * 2] + affb[c * 2 + 1], in_act);
+          a += xv * (float)wr[c];
+        }
       }
     }
     out[i] = epilogue(a, bias, cbias, b * OC, residual, i, oc, act);
@@ -767,6 +793,7 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
                                   float* ws, int B, int H, int W, int IC,
                                   int HO, int WO, int OC, int R, int S,
                                   int stride, int pad, int act, int path,
+                                  const float* in_aff, int in_act,
                                   hipStream_t s) {
   const int K = R * S * IC;
   const f16* xp = reinterpret_cast<const f16*>(x);
@@ -785,8 +812,11 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
     const char* t = getenv("AIRTC_CONV_TILED");
     tiled = t ? atoi(t) : 0;
   }
-  if (tiled && R == 3 && S == 3 && stride == 1 && pad == 1 && IC % 64 == 0 &&
-      (H & 7) == 0 && (W & 7) == 0 && M >= 4096) {
+  // fused input-affine runs on the register-staged schedule only (the
+  // glds DMA path writes HBM straight to LDS; the tiled path has its own
+  // gather) — both are env-gated experiments anyway
+  if (!in_aff && tiled && R == 3 && S == 3 && stride == 1 && pad == 1 &&
+      IC % 64 == 0 && (H & 7) == 0 && (W & 7) == 0 && M >= 4096) {
     dim3 tgrid((H >> 3) * (W >> 3), ceil_div(OC, BN), B);
     hipLaunchKernelGGL(conv3x3_tiled_kernel, tgrid, dim3(256), 0, s, xp, wp,
                        bias, cb, res, op, H, W, IC, OC, act, K);
@@ -821,19 +851,20 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
 #define CONV_LAUNCH(MF, NF, BKV, DB)                                          \
   hipLaunchKernelGGL((conv2d_mfma_kernel<MF, NF, BKV, DB>), grid, dim3(256),  \
                      0, s, xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO,   \
-                     OC, R, S, stride, pad, act, K, spk_arg)
+                     OC, R, S, stride, pad, act, K, spk_arg, in_aff, in_act)
 #define CONV_LAUNCH_GLDS(MF)                                                  \
   hipLaunchKernelGGL((conv2d_mfma_glds_kernel<MF>), grid, dim3(256), 0, s,    \
                      xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC, R,  \
                      S, stride, pad, act, K, spk_arg)
   const bool bn128 = bn == 128;
+  const int use_glds = glds && !in_aff;
   if (path > 0) {
-    if (bk64 && glds) CONV_LAUNCH_GLDS(4);
+    if (bk64 && use_glds) CONV_LAUNCH_GLDS(4);
     else if (bk64 && bn128) CONV_LAUNCH(4, 4, 64, false);
     else if (bk64) { if (dbuf) CONV_LAUNCH(4, 2, 64, true); else CONV_LAUNCH(4, 2, 64, false); }
     else CONV_LAUNCH(4, 2, 32, false);
   } else {
-    if (bk64 && glds) CONV_LAUNCH_GLDS(2);
+    if (bk64 && use_glds) CONV_LAUNCH_GLDS(2);
     else if (bk64 && bn128) CONV_LAUNCH(2, 4, 64, false);
     else if (bk64) { if (dbuf) CONV_LAUNCH(2, 2, 64, true); else CONV_LAUNCH(2, 2, 64, false); }
     else CONV_LAUNCH(2, 2, 32, false);
@@ -853,7 +884,8 @@ extern "C" void airtc_conv2d_direct(const uint16_t* x, const uint16_t* w,
                                     const uint16_t* residual, uint16_t* out,
                                     int B, int H, int W, int IC, int HO,
                                     int WO, int OC, int R, int S, int stride,
-                                    int pad, int act, hipStream_t s) {
+                                    int pad, int act, const float* in_aff,
+                                    int in_act, hipStream_t s) {
   const int K = R * S * IC;
   if (K <= SMALLIC_MAX_K) {
     long pix = (long)B * HO * WO;
@@ -865,7 +897,7 @@ extern "C" void airtc_conv2d_direct(const uint16_t* x, const uint16_t* w,
                        reinterpret_cast<const f16*>(cbias),
                        reinterpret_cast<const f16*>(residual),
                        reinterpret_cast<f16*>(out), H, W, IC, HO, WO, OC, R,
-                       S, stride, pad, act, K, pix);
+                       S, stride, pad, act, K, pix, in_aff, in_act);
     return;
   }
   long total = (long)B * HO * WO * OC;
@@ -876,5 +908,5 @@ extern "C" void airtc_conv2d_direct(const uint16_t* x, const uint16_t* w,
                      reinterpret_cast<const f16*>(cbias),
                      reinterpret_cast<const f16*>(residual),
                      reinterpret_cast<f16*>(out), H, W, IC, HO, WO, OC, R, S,
-                     stride, pad, act, K, total);
+                     stride, pad, act, K, total, in_aff, in_act);
 }

@@ -30,7 +30,8 @@ torch::Tensor conv2d(torch::Tensor x, torch::Tensor w_perm,
                      c10::optional<torch::Tensor> bias,
                      c10::optional<torch::Tensor> cbias,
                      c10::optional<torch::Tensor> residual, int64_t R,
-                     int64_t S, int64_t stride, int64_t pad, int64_t act) {
+                     int64_t S, int64_t stride, int64_t pad, int64_t act,
+                     c10::optional<torch::Tensor> in_affine, int64_t in_act) {
   CHECK_IN(x);
   CHECK_IN(w_perm);
   TORCH_CHECK(x.dtype() == torch::kHalf && w_perm.dtype() == torch::kHalf);
@@ -57,11 +58,19 @@ torch::Tensor conv2d(torch::Tensor x, torch::Tensor w_perm,
     TORCH_CHECK(residual->numel() == out.numel(), "residual shape mismatch");
     res = h_ptr(*residual);
   }
+  const float* aff = nullptr;
+  if (in_affine.has_value()) {
+    CHECK_IN((*in_affine));
+    TORCH_CHECK(in_affine->dtype() == torch::kFloat &&
+                    in_affine->numel() == (long)B * IC * 2,
+                "in_affine must be (B, IC, 2) f32");
+    aff = in_affine->data_ptr<float>();
+  }
   const int path = airtc_conv2d_splitk_for(B, HO, WO, OC, IC);
   if (path == 0) {
     airtc_conv2d_direct(h_ptr(x), h_ptr(w_perm), bp, cb, res, h_ptr_mut(out),
                         B, H, W, IC, HO, WO, OC, (int)R, (int)S, (int)stride,
-                        (int)pad, (int)act, cur_stream());
+                        (int)pad, (int)act, aff, (int)in_act, cur_stream());
     return out;
   }
   float* wsp = nullptr;
@@ -74,8 +83,26 @@ torch::Tensor conv2d(torch::Tensor x, torch::Tensor w_perm,
   }
   airtc_conv2d_mfma(h_ptr(x), h_ptr(w_perm), bp, cb, res, h_ptr_mut(out), wsp,
                     B, H, W, IC, HO, WO, OC, (int)R, (int)S, (int)stride,
-                    (int)pad, (int)act, path, cur_stream());
+                    (int)pad, (int)act, path, aff, (int)in_act, cur_stream());
   return out;
+}
+
+torch::Tensor group_norm_coeffs(torch::Tensor x, int64_t groups,
+                                torch::Tensor gamma, torch::Tensor beta,
+                                double eps) {
+  CHECK_IN(x);
+  const int B = x.size(0);
+  const int C = x.size(-1);
+  const long HW = x.numel() / ((long)B * C);
+  auto coeffs = torch::empty({B, C, 2}, x.options().dtype(torch::kFloat));
+  const int nchunk = airtc_group_norm_nchunk(B, (int)groups);
+  auto ws = torch::empty({(long)B * groups * nchunk * 2},
+                         x.options().dtype(torch::kFloat));
+  airtc_group_norm_coeffs(h_ptr(x), gamma.data_ptr<float>(),
+                          beta.data_ptr<float>(), coeffs.data_ptr<float>(),
+                          ws.data_ptr<float>(), B, (int)HW, C, (int)groups,
+                          (float)eps, cur_stream());
+  return coeffs;
 }
 
 torch::Tensor group_norm_silu(torch::Tensor x, int64_t groups,
@@ -323,8 +350,15 @@ void airtc_register_dtls(pybind11::module_& m);  // dtls.cpp
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   airtc_register_dtls(m);
-  m.def("conv2d", &conv2d, "implicit-GEMM MFMA conv2d (NHWC)");
+  m.def("conv2d", &conv2d, "implicit-GEMM MFMA conv2d (NHWC)",
+        pybind11::arg("x"), pybind11::arg("w_perm"), pybind11::arg("bias"),
+        pybind11::arg("cbias"), pybind11::arg("residual"), pybind11::arg("R"),
+        pybind11::arg("S"), pybind11::arg("stride"), pybind11::arg("pad"),
+        pybind11::arg("act"), pybind11::arg("in_affine") = pybind11::none(),
+        pybind11::arg("in_act") = 0);
   m.def("group_norm_silu", &group_norm_silu);
+  m.def("group_norm_coeffs", &group_norm_coeffs,
+        "(B,C,2) f32 affine pairs for the fused GN->conv input transform");
   m.def("layer_norm", &layer_norm);
   m.def("attention_bhlc", &attention_bhlc);
   m.def("silu", &silu);

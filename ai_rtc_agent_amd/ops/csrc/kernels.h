@@ -28,6 +28,11 @@ void airtc_group_norm_silu(const uint16_t* x, const float* gamma,
 void airtc_layer_norm(const uint16_t* x, const float* gamma, const float* beta,
                       uint16_t* out, long rows, int C, float eps,
                       hipStream_t s);
+// (B, C, 2) f32 affine pairs for the fused GN->conv input transform
+void airtc_group_norm_coeffs(const uint16_t* x, const float* gamma,
+                             const float* beta, float* coeffs, float* ws,
+                             int B, int HW, int C, int G, float eps,
+                             hipStream_t s);
 
 // conv ----------------------------------------------------------------------
 // x   : (B, H, W, IC) NHWC f16 (zero-padding handled inline)
@@ -38,16 +43,20 @@ void airtc_layer_norm(const uint16_t* x, const float* gamma, const float* beta,
 // path: from airtc_conv2d_splitk_for (0=direct, 1=BM128, -k=BM64 splitk k);
 // ws  : f32 workspace (B*k, HO*WO, OC) required when path < -1
 int airtc_conv2d_splitk_for(int B, int HO, int WO, int OC, int IC);
+// in_aff: optional (B, IC, 2) f32 per-channel input affine (fused GN) with
+// in_act applied after — transforms x AT LOAD TIME (padding stays zero)
 void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w, const float* bias,
                        const uint16_t* cbias, const uint16_t* residual,
                        uint16_t* out, float* ws, int B, int H, int W, int IC,
                        int HO, int WO, int OC, int R, int S, int stride,
-                       int pad, int act, int path, hipStream_t s);
+                       int pad, int act, int path, const float* in_aff,
+                       int in_act, hipStream_t s);
 void airtc_conv2d_direct(const uint16_t* x, const uint16_t* w,
                          const float* bias, const uint16_t* cbias,
                          const uint16_t* residual, uint16_t* out, int B, int H,
                          int W, int IC, int HO, int WO, int OC, int R, int S,
-                         int stride, int pad, int act, hipStream_t s);
+                         int stride, int pad, int act, const float* in_aff,
+                         int in_act, hipStream_t s);
 
 // attention -----------------------------------------------------------------
 // q: base+strides address (B,H) heads; row stride in elements.

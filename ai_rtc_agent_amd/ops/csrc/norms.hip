@@ -113,6 +113,54 @@ extern "C" int airtc_group_norm_nchunk(int B, int G) {
   return n;
 }
 
+// per-(b, channel) affine pairs for the fused GN->conv input transform
+// (conv2d.hip load_a): s = gamma_c * rstd_g, t = beta_c - mean_g * s.
+// Replaces the full-tensor apply pass when the ONLY consumer is a conv.
+__global__ void group_norm_coeffs_kernel(const float* __restrict__ ws,
+                                         const float* __restrict__ gamma,
+                                         const float* __restrict__ beta,
+                                         float* __restrict__ coeffs, int HW,
+                                         int C, int G, int nchunk,
+                                         float eps) {
+  const int b = blockIdx.x / G;
+  const int g = blockIdx.x % G;
+  const int Cg = C / G;
+  __shared__ float stats[2];
+  if (threadIdx.x == 0) {
+    const float* w = &ws[(long)blockIdx.x * nchunk * 2];
+    float s0 = 0.f, s1 = 0.f;
+    for (int i = 0; i < nchunk; ++i) {
+      s0 += w[2 * i];
+      s1 += w[2 * i + 1];
+    }
+    const float n = (float)HW * Cg;
+    const float mean = s0 / n;
+    stats[0] = mean;
+    stats[1] = rsqrtf(s1 / n - mean * mean + eps);
+  }
+  __syncthreads();
+  const float mean = stats[0], rstd = stats[1];
+  for (int c = threadIdx.x; c < Cg; c += blockDim.x) {
+    const int ch = g * Cg + c;
+    const float sc = gamma[ch] * rstd;
+    float* o = &coeffs[((long)b * C + ch) * 2];
+    o[0] = sc;
+    o[1] = beta[ch] - mean * sc;
+  }
+}
+
+extern "C" void airtc_group_norm_coeffs(const uint16_t* x, const float* gamma,
+                                        const float* beta, float* coeffs,
+                                        float* ws, int B, int HW, int C,
+                                        int G, float eps, hipStream_t s) {
+  const int nchunk = airtc_group_norm_nchunk(B, G);
+  dim3 grid(B * G, nchunk);
+  hipLaunchKernelGGL(group_norm_stats_kernel, grid, dim3(256), 0, s,
+                     reinterpret_cast<const f16*>(x), ws, HW, C, G, nchunk);
+  hipLaunchKernelGGL(group_norm_coeffs_kernel, dim3(B * G), dim3(256), 0, s,
+                     ws, gamma, beta, coeffs, HW, C, G, nchunk, eps);
+}
+
 extern "C" void airtc_group_norm_silu(const uint16_t* x, const float* gamma,
                                       const float* beta, uint16_t* out,
                                       float* ws, int B, int HW, int C, int G,

@@ -90,6 +90,8 @@ def conv2d_nhwc(
     act: int | None = None,
     residual: torch.Tensor | None = None,
     channel_bias: torch.Tensor | None = None,
+    in_affine: torch.Tensor | None = None,
+    in_act: int = ACT_NONE,
 ) -> torch.Tensor:
     """x: (B,H,W,C) contiguous; weight: (O,I,R,S) [torch layout]; out (B,H',W',O).
 
@@ -101,6 +103,10 @@ def conv2d_nhwc(
     Epilogue fusion (order): y = act(conv + bias + channel_bias + residual)
       channel_bias: (B, O) — the resnet time-embedding add
       residual:     broadcast-free tensor of the output shape — skip adds
+    Input fusion: in_affine (B, C, 2) f32 applies x*s+t (then in_act) to
+      every input element AT LOAD TIME — the fused-GroupNorm path
+      (group_norm_coeffs); the producing apply kernel and its activation
+      round-trip through HBM disappear.
     """
     if act is None:
         act = ACT_SILU if fuse_silu else ACT_NONE
@@ -126,8 +132,18 @@ def conv2d_nhwc(
             stride,
             padding,
             act,
+            None if in_affine is None else in_affine.contiguous(),
+            in_act,
         )
 
+    if in_affine is not None:
+        aff = in_affine.float()
+        xf = x.float() * aff[:, None, None, :, 0] + aff[:, None, None, :, 1]
+        if in_act == ACT_SILU:
+            xf = F.silu(xf)
+        elif in_act == ACT_RELU:
+            xf = F.relu(xf)
+        x = xf.to(x.dtype)
     xc = x.permute(0, 3, 1, 2)
     y = F.conv2d(xc.float(), weight.float(), None if bias is None else bias.float(),
                  stride=stride, padding=padding)
@@ -146,6 +162,35 @@ def conv2d_nhwc(
 # ---------------------------------------------------------------------------
 # normalisations
 # ---------------------------------------------------------------------------
+
+def group_norm_coeffs(
+    x: torch.Tensor,
+    num_groups: int,
+    gamma: torch.Tensor,
+    beta: torch.Tensor,
+    eps: float = 1e-5,
+) -> torch.Tensor:
+    """Per-(batch, channel) affine pairs (B, C, 2) f32 such that
+    gn(x)[..., c] == x[..., c] * s + t — the input-side half of the fused
+    GN->conv (conv2d_nhwc in_affine). GPU: stats kernel + a tiny coeffs
+    kernel; the full-tensor apply pass disappears."""
+    if _use_hip(x):
+        if (x.shape[-1] // num_groups) % 2 != 0:
+            raise ValueError("group_norm kernel needs even channels-per-group")
+        ext = _require_ext()
+        g32 = _cached(gamma, "_airtc_g32", lambda: gamma.detach().float().contiguous())
+        b32 = _cached(beta, "_airtc_b32", lambda: beta.detach().float().contiguous())
+        return ext.group_norm_coeffs(x, num_groups, g32, b32, eps)
+    b, c = x.shape[0], x.shape[-1]
+    xf = x.reshape(b, -1, num_groups, c // num_groups).permute(0, 2, 1, 3).float()
+    mean = xf.mean(dim=(2, 3))                             # (B, G)
+    rstd = (xf.var(dim=(2, 3), unbiased=False) + eps).rsqrt()
+    mean_c = mean.repeat_interleave(c // num_groups, dim=1)
+    rstd_c = rstd.repeat_interleave(c // num_groups, dim=1)
+    s = gamma.float()[None] * rstd_c
+    t = beta.float()[None] - mean_c * s
+    return torch.stack([s, t], dim=-1).contiguous()
+
 
 def group_norm_silu_nhwc(
     x: torch.Tensor,

@@ -78,3 +78,23 @@ def test_layer_norm():
     g, b = torch.randn(32), torch.randn(32)
     y = ops.layer_norm(x, g, b)
     assert torch.allclose(y, F.layer_norm(x, (32,), g, b), atol=1e-5)
+
+
+def test_fused_gn_conv_cpu_parity():
+    """CPU fallback: conv2d_nhwc(in_affine) == gn_silu -> conv."""
+    import math
+
+    import torch
+
+    from ai_rtc_agent_amd import ops
+
+    g = torch.Generator().manual_seed(0)
+    x = torch.randn(2, 8, 8, 16, generator=g)
+    w = torch.randn(24, 16, 3, 3, generator=g) / math.sqrt(16 * 9)
+    gamma = torch.randn(16, generator=g) * 0.3 + 1
+    beta = torch.randn(16, generator=g) * 0.2
+    ref = ops.conv2d_nhwc(ops.group_norm_silu_nhwc(x, 4, gamma, beta, 1e-5, True), w)
+    aff = ops.group_norm_coeffs(x, 4, gamma, beta, 1e-5)
+    got = ops.conv2d_nhwc(x, w, in_affine=aff, in_act=ops.ACT_SILU)
+    assert torch.allclose(got.float(), ref.float(), atol=1e-4), \
+        (got - ref).abs().max()

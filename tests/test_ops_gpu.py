@@ -147,3 +147,35 @@ def test_pre_post_process():
     assert_close(f, ref, 1e-2, "preprocess")
     back = ops.postprocess_to_u8(f)
     assert (back.int() - u8.int()).abs().max().item() <= 1
+
+
+@pytest.mark.parametrize("ic,oc,h,groups", [
+    (64, 64, 16, 32),      # BM64 register-staged path
+    (320, 320, 16, 32),    # SD resnet shape (split-K)
+    (64, 64, 64, 32),      # BM128 large-spatial path
+    (4, 32, 16, 2),        # small-IC direct path
+])
+def test_fused_gn_conv_vs_unfused(ic, oc, h, groups):
+    """in_affine fusion (GN apply + SiLU inside the conv's A-load) must
+    match gn_silu -> conv bit-for-bit-ish on every conv path, including
+    the zero-padding rule (padding applies AFTER the transform)."""
+    x = rnd(2, h, h, ic, seed=ic + h)
+    w = rnd(oc, ic, 3, 3, seed=2, scale=1.0 / math.sqrt(ic * 9))
+    gamma = (torch.randn(ic, generator=torch.Generator().manual_seed(3)) * 0.3 + 1).to(DEV)
+    beta = (torch.randn(ic, generator=torch.Generator().manual_seed(4)) * 0.2).to(DEV)
+
+    ref = ops.conv2d_nhwc(
+        ops.group_norm_silu_nhwc(x, groups, gamma, beta, 1e-5, True), w)
+    aff = ops.group_norm_coeffs(x, groups, gamma, beta, 1e-5)
+    assert aff.shape == (2, ic, 2)
+    got = ops.conv2d_nhwc(x, w, in_affine=aff, in_act=ops.ACT_SILU)
+    assert_close(got, ref, 2e-2, f"fused gn-conv ic{ic} h{h}")
+
+
+def test_group_norm_coeffs_vs_cpu():
+    x = rnd(2, 8, 8, 64, seed=9)
+    gamma = (torch.randn(64, generator=torch.Generator().manual_seed(5)) + 1).to(DEV)
+    beta = torch.randn(64, generator=torch.Generator().manual_seed(6)).to(DEV)
+    got = ops.group_norm_coeffs(x, 32, gamma, beta, 1e-5)
+    ref = ops.group_norm_coeffs(x.cpu(), 32, gamma.cpu(), beta.cpu(), 1e-5)
+    assert_close(got, ref.to(DEV), 2e-2, "gn coeffs")
