@@ -27,6 +27,16 @@ import torch.nn as nn
 from .. import ops
 
 
+def _fuse_gn() -> bool:
+    """AIRTC_FUSE_GN=1 routes GroupNorm+SiLU through the convs'
+    fused input-affine path — A/B'd slower at SD shapes (see ResnetBlock),
+    kept for experiments."""
+    import os
+
+    return os.environ.get("AIRTC_FUSE_GN", "0") == "1"
+
+
+
 @dataclass
 class UNetConfig:
     in_channels: int = 4
@@ -330,16 +340,24 @@ class ResnetBlock(nn.Module):
         self.shortcut = Conv2d(cin, cout, 1) if cin != cout else None
 
     def forward(self, x: torch.Tensor, temb: torch.Tensor) -> torch.Tensor:
-        # fusion ladder: time-emb add in conv1's epilogue, skip add in
-        # conv2's, and BOTH GroupNorm+SiLU applications in the convs'
-        # input loads (in_affine) — per resnet block only the two conv
-        # kernels and two tiny stats/coeffs passes remain
+        # time-emb add fused into conv1's epilogue; skip add fused into
+        # conv2's. The DEEPER fusion (GN apply + SiLU inside the conv's
+        # A-load, in_affine) measured SLOWER end-to-end on MI355X
+        # (123.1 -> 88.3 fps headline; sd15 64.4 -> 35.1): the per-element
+        # coefficient gather + transform sits in the register-staging
+        # critical path of kernels that are already issue/latency-bound —
+        # same regime as the rejected halo-tiled loads (ladder). Kept
+        # env-gated for experiments: AIRTC_FUSE_GN=1.
         temb_b = self.time_emb_proj(ops.silu(temb))
-        h = self.conv1(x, channel_bias=temb_b,
-                       in_affine=self.norm1.coeffs(x), in_act=ops.ACT_SILU)
+        if _fuse_gn():
+            h = self.conv1(x, channel_bias=temb_b,
+                           in_affine=self.norm1.coeffs(x), in_act=ops.ACT_SILU)
+            skip = self.shortcut(x) if self.shortcut is not None else x
+            return self.conv2(h, residual=skip,
+                              in_affine=self.norm2.coeffs(h), in_act=ops.ACT_SILU)
+        h = self.conv1(self.norm1(x), channel_bias=temb_b)
         skip = self.shortcut(x) if self.shortcut is not None else x
-        return self.conv2(h, residual=skip,
-                          in_affine=self.norm2.coeffs(h), in_act=ops.ACT_SILU)
+        return self.conv2(self.norm2(h), residual=skip)
 
 
 class Downsample(nn.Module):
@@ -488,5 +506,7 @@ class UNet2DCondition(nn.Module):
             if self.upsamplers[bi] is not None:
                 x = self.upsamplers[bi](x)
 
-        return self.conv_out(x, in_affine=self.norm_out.coeffs(x),
-                             in_act=ops.ACT_SILU)
+        if _fuse_gn():
+            return self.conv_out(x, in_affine=self.norm_out.coeffs(x),
+                                 in_act=ops.ACT_SILU)
+        return self.conv_out(self.norm_out(x))
