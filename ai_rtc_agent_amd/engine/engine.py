@@ -278,7 +278,7 @@ class StreamDiffusionEngine:
                     if hasattr(m, attr):
                         delattr(m, attr)
                 for t in list(m.parameters(recurse=False)):
-                    for attr in ("_airtc_wperm", "_airtc_b32", "_airtc_g32", "_airtc_w16"):
+                    for attr in ("_airtc_wperm", "_airtc_b32", "_airtc_g32", "_airtc_w16", "_airtc_wpad32"):
                         if hasattr(t, attr):
                             delattr(t, attr)
         from ..parallel.collectives import broadcast_engine_weights

@@ -113,6 +113,25 @@ def conv2d_nhwc(
     if _use_hip(x):
         ext = _require_ext()
         O, I, R, S = weight.shape
+        # Tiny-IC convs at large spatial (the 3/4-channel conv_in layers at
+        # 512²) measured 129us on the scalar small-IC kernel; zero-padding
+        # the input channels to 32 routes them through the MFMA path
+        # (numerics identical: zero channels contribute nothing). The pad
+        # copy is a ~1.5 MB -> 16 MB expansion, trivial against HBM3E.
+        if I < 32 and in_affine is None and x.shape[1] * x.shape[2] >= 16384:
+            xp = torch.zeros((*x.shape[:3], 32), dtype=x.dtype, device=x.device)
+            xp[..., :I].copy_(x)
+            w_pad = _cached(
+                weight,
+                "_airtc_wpad32",
+                lambda: torch.cat(
+                    [weight.detach(),
+                     torch.zeros((O, 32 - I, R, S), dtype=weight.dtype,
+                                 device=weight.device)], dim=1).contiguous(),
+            )
+            return conv2d_nhwc(xp, w_pad, bias, stride, padding, fuse_silu,
+                               act=act, residual=residual,
+                               channel_bias=channel_bias)
         w_perm = _cached(
             weight,
             "_airtc_wperm",

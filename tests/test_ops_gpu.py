@@ -44,6 +44,7 @@ def test_ext_loads():
     (640, 1280, 16, 2, 1, 3),   # stride-2 into split-K territory
     (64, 64, 64, 1, 1, 3),      # TAESD-shaped: BM128 large-spatial path
     (320, 64, 64, 1, 0, 1),     # 1x1 conv on the BM128 path
+    (3, 64, 160, 1, 1, 3),      # tiny-IC at large spatial: pad-to-32 MFMA route
 ])
 def test_conv2d_vs_torch(ic, oc, h, stride, pad, r):
     x = rnd(2, h, h, ic, seed=ic + oc)
