@@ -128,7 +128,8 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
     const f16* __restrict__ residual, f16* __restrict__ out,
     float* __restrict__ ws, int H, int W, int IC, int HO, int WO, int OC,
     int R, int S, int stride, int pad, int act, int K, int splitk,
-    const float* __restrict__ in_aff, int in_act) {
+    const float* __restrict__ in_aff, int in_act,
+    int* __restrict__ counters) {
   constexpr int BM = MFRAG * 32;              // 128 or 64
   constexpr int BNK = NFRAG * 32;             // 64 or 128 out-channels
   constexpr int KPITCH = BK + 8;              // +16B row pad (guide G4)
@@ -293,7 +294,12 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
         }
     }
   } else {
-    // plain f32 slab store; finalize kernel reduces + applies the epilogue
+    // f32 slab store; with `counters` the LAST split block for this
+    // (b, mt, nt) tile reduces every slab and applies the epilogue IN
+    // KERNEL (threadfence-reduction pattern) — the separate finalize
+    // launch disappears (it was 0.66 ms/frame across ~105 launches).
+    // Counters are self-cleaning: the last block resets its slot to 0,
+    // so one zeroed persistent buffer serves every launch/graph replay.
     float* wsb = ws + ((long)b * spk + split) * M * OC;
 #pragma unroll
     for (int ni = 0; ni < NFRAG; ++ni) {
@@ -306,6 +312,32 @@ __global__ __launch_bounds__(256) void conv2d_mfma_kernel(
           const int m = m0 + wm * (MFRAG * 16) + mi * 16 + (lane >> 4) * 4 + j;
           if (m < M) wsb[(long)m * OC + col] = acc[mi][ni][j];
         }
+    }
+    if (counters) {
+      __threadfence();
+      __shared__ int is_last;
+      if (tid == 0) {
+        const long slot = (long)b * gridDim.x + blockIdx.x;
+        const int old = atomicAdd(&counters[slot], 1);
+        is_last = (old == spk - 1);
+        if (is_last) counters[slot] = 0;
+      }
+      __syncthreads();
+      if (!is_last) return;
+      f16* ob = out + (long)b * M * OC;
+      const long cb_off = (long)b * OC;
+      const float* slab0 = ws + (long)b * spk * M * OC;
+      for (int i = tid; i < BM * BNK; i += 256) {
+        const int m = m0 + i / BNK;
+        const int col = n0 + i % BNK;
+        if (m >= M || col >= OC) continue;
+        const float* p = slab0 + (long)m * OC + col;
+        float a = 0.f;
+        for (int sp = 0; sp < spk; ++sp) a += p[(long)sp * M * OC];
+        const long oidx = (long)b * M * OC + (long)m * OC + col;
+        ob[(long)m * OC + col] =
+            epilogue(a, bias, cbias, cb_off, residual, oidx, col, act);
+      }
     }
   }
 }
@@ -794,7 +826,7 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
                                   int HO, int WO, int OC, int R, int S,
                                   int stride, int pad, int act, int path,
                                   const float* in_aff, int in_act,
-                                  hipStream_t s) {
+                                  int* counters, hipStream_t s) {
   const int K = R * S * IC;
   const f16* xp = reinterpret_cast<const f16*>(x);
   const f16* wp = reinterpret_cast<const f16*>(w);
@@ -826,9 +858,13 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
   const int bm = path > 0 ? 128 : 64;
   const int bn = (IC % 64 == 0) ? conv_bn() : 64;  // BN128 needs BK64 staging
   dim3 grid(ceil_div(M, bm) * ceil_div(OC, bn), 1, B * splitk);
-  const float* b1 = splitk == 1 ? bias : nullptr;
-  const f16* cb1 = splitk == 1 ? cb : nullptr;
-  const f16* res1 = splitk == 1 ? res : nullptr;
+  // with in-kernel fused finalize (counters) the kernel applies the real
+  // epilogue itself; the separate finalize launch happens only on the
+  // env-gated glds path (which has no counter support)
+  const bool fuse_fin = counters != nullptr && splitk > 1;
+  const float* b1 = (splitk == 1 || fuse_fin) ? bias : nullptr;
+  const f16* cb1 = (splitk == 1 || fuse_fin) ? cb : nullptr;
+  const f16* res1 = (splitk == 1 || fuse_fin) ? res : nullptr;
   const bool bk64 = (IC % 64 == 0);
   static int dbuf = -1, glds = -1, xcdmap = -1;
   if (dbuf < 0) {
@@ -851,13 +887,14 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
 #define CONV_LAUNCH(MF, NF, BKV, DB)                                          \
   hipLaunchKernelGGL((conv2d_mfma_kernel<MF, NF, BKV, DB>), grid, dim3(256),  \
                      0, s, xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO,   \
-                     OC, R, S, stride, pad, act, K, spk_arg, in_aff, in_act)
+                     OC, R, S, stride, pad, act, K, spk_arg, in_aff, in_act,  \
+                     fuse_fin ? counters : nullptr)
 #define CONV_LAUNCH_GLDS(MF)                                                  \
   hipLaunchKernelGGL((conv2d_mfma_glds_kernel<MF>), grid, dim3(256), 0, s,    \
                      xp, wp, b1, cb1, res1, op, ws, H, W, IC, HO, WO, OC, R,  \
                      S, stride, pad, act, K, spk_arg)
   const bool bn128 = bn == 128;
-  const int use_glds = glds && !in_aff;
+  const int use_glds = glds && !in_aff && !fuse_fin;
   if (path > 0) {
     if (bk64 && use_glds) CONV_LAUNCH_GLDS(4);
     else if (bk64 && bn128) CONV_LAUNCH(4, 4, 64, false);
@@ -871,7 +908,7 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
   }
 #undef CONV_LAUNCH
 #undef CONV_LAUNCH_GLDS
-  if (splitk > 1) {
+  if (splitk > 1 && !fuse_fin) {
     long total = (long)B * M * OC;
     int blocks = (int)min((long)2048, (total + 255) / 256);
     hipLaunchKernelGGL(conv_splitk_finalize, dim3(blocks), dim3(256), 0, s, ws,

@@ -31,7 +31,8 @@ torch::Tensor conv2d(torch::Tensor x, torch::Tensor w_perm,
                      c10::optional<torch::Tensor> cbias,
                      c10::optional<torch::Tensor> residual, int64_t R,
                      int64_t S, int64_t stride, int64_t pad, int64_t act,
-                     c10::optional<torch::Tensor> in_affine, int64_t in_act) {
+                     c10::optional<torch::Tensor> in_affine, int64_t in_act,
+                     c10::optional<torch::Tensor> counters) {
   CHECK_IN(x);
   CHECK_IN(w_perm);
   TORCH_CHECK(x.dtype() == torch::kHalf && w_perm.dtype() == torch::kHalf);
@@ -81,9 +82,19 @@ torch::Tensor conv2d(torch::Tensor x, torch::Tensor w_perm,
                       x.options().dtype(torch::kFloat));
     wsp = ws.data_ptr<float>();
   }
+  int* cnt = nullptr;
+  if (counters.has_value() && splitk > 1) {
+    CHECK_IN((*counters));
+    TORCH_CHECK(counters->dtype() == torch::kInt, "counters must be int32");
+    TORCH_CHECK(counters->numel() >= (long)B * ((HO * WO + 63) / 64) *
+                                         ((OC + 63) / 64),
+                "counters buffer too small");
+    cnt = counters->data_ptr<int>();
+  }
   airtc_conv2d_mfma(h_ptr(x), h_ptr(w_perm), bp, cb, res, h_ptr_mut(out), wsp,
                     B, H, W, IC, HO, WO, OC, (int)R, (int)S, (int)stride,
-                    (int)pad, (int)act, path, aff, (int)in_act, cur_stream());
+                    (int)pad, (int)act, path, aff, (int)in_act, cnt,
+                    cur_stream());
   return out;
 }
 
@@ -355,7 +366,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("cbias"), pybind11::arg("residual"), pybind11::arg("R"),
         pybind11::arg("S"), pybind11::arg("stride"), pybind11::arg("pad"),
         pybind11::arg("act"), pybind11::arg("in_affine") = pybind11::none(),
-        pybind11::arg("in_act") = 0);
+        pybind11::arg("in_act") = 0,
+        pybind11::arg("counters") = pybind11::none());
   m.def("group_norm_silu", &group_norm_silu);
   m.def("group_norm_coeffs", &group_norm_coeffs,
         "(B,C,2) f32 affine pairs for the fused GN->conv input transform");

@@ -50,7 +50,7 @@ void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w, const float* bias,
                        uint16_t* out, float* ws, int B, int H, int W, int IC,
                        int HO, int WO, int OC, int R, int S, int stride,
                        int pad, int act, int path, const float* in_aff,
-                       int in_act, hipStream_t s);
+                       int in_act, int* counters, hipStream_t s);
 void airtc_conv2d_direct(const uint16_t* x, const uint16_t* w,
                          const float* bias, const uint16_t* cbias,
                          const uint16_t* residual, uint16_t* out, int B, int H,

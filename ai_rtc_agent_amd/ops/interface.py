@@ -140,6 +140,19 @@ def conv2d_nhwc(
         b32 = None
         if bias is not None:
             b32 = _cached(bias, "_airtc_b32", lambda: bias.detach().float().contiguous())
+        # persistent zeroed tile counters for the in-kernel split-K
+        # finalize (self-cleaning: the kernel resets each slot after use)
+        B, H, W = x.shape[0], x.shape[1], x.shape[2]
+        HO = (H + 2 * padding - R) // stride + 1
+        WO = (W + 2 * padding - S) // stride + 1
+        slots = B * ((HO * WO + 63) // 64) * ((O + 63) // 64)
+        cnt = getattr(weight, "_airtc_skcnt", None)
+        if cnt is None or cnt.numel() < slots or cnt.device != x.device:
+            cnt = torch.zeros(slots, dtype=torch.int32, device=x.device)
+            try:
+                weight._airtc_skcnt = cnt
+            except AttributeError:
+                pass
         return ext.conv2d(
             x,
             w_perm,
@@ -153,6 +166,7 @@ def conv2d_nhwc(
             act,
             None if in_affine is None else in_affine.contiguous(),
             in_act,
+            cnt,
         )
 
     if in_affine is not None:
