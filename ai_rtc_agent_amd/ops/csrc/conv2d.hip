@@ -858,10 +858,18 @@ extern "C" void airtc_conv2d_mfma(const uint16_t* x, const uint16_t* w,
   const int bm = path > 0 ? 128 : 64;
   const int bn = (IC % 64 == 0) ? conv_bn() : 64;  // BN128 needs BK64 staging
   dim3 grid(ceil_div(M, bm) * ceil_div(OC, bn), 1, B * splitk);
-  // with in-kernel fused finalize (counters) the kernel applies the real
-  // epilogue itself; the separate finalize launch happens only on the
-  // env-gated glds path (which has no counter support)
-  const bool fuse_fin = counters != nullptr && splitk > 1;
+  // In-kernel fused finalize measured 2.2x SLOWER end-to-end (128.5 ->
+  // 57.4 fps): the device-scope __threadfence each split block must issue
+  // before its counter increment is an L2 writeback on CDNA, destroying
+  // the per-XCD L2 reuse that feeds the K-loop — far costlier than the
+  // ~105 finalize launches it removes. Env-gated for experiments
+  // (AIRTC_CONV_FUSED_FIN=1); default stays the two-pass scheme.
+  static int ffin_env = -1;
+  if (ffin_env < 0) {
+    const char* e = getenv("AIRTC_CONV_FUSED_FIN");
+    ffin_env = e ? atoi(e) : 0;
+  }
+  const bool fuse_fin = ffin_env && counters != nullptr && splitk > 1;
   const float* b1 = (splitk == 1 || fuse_fin) ? bias : nullptr;
   const f16* cb1 = (splitk == 1 || fuse_fin) ? cb : nullptr;
   const f16* res1 = (splitk == 1 || fuse_fin) ? res : nullptr;
