@@ -225,6 +225,9 @@ void* airtc_h264enc_create(int w, int h);
 void airtc_h264enc_set_slices(void*, int n);
 void airtc_h264enc_destroy(void*);
 int airtc_h264enc_encode(void*, const uint8_t*, int qp, uint8_t*, int cap);
+void airtc_h264enc_set_mb_mode(void*, int m);
+void airtc_h264_pred4(int, const uint8_t*, const uint8_t*, uint8_t, int, int,
+                      int, uint8_t*);
 void* airtc_h264dec_create();
 void airtc_h264dec_destroy(void*);
 int airtc_h264dec_decode(void*, const uint8_t*, int, uint8_t*, int, int*, int*);
@@ -234,10 +237,12 @@ int airtc_h264sw_table_check();
 
 class H264SwEncoder {
  public:
-  H264SwEncoder(int w, int h, int slices = 4) : w_(w), h_(h) {
+  H264SwEncoder(int w, int h, int slices = 4, int mb_mode = 0)
+      : w_(w), h_(h) {
     handle_ = airtc_h264enc_create(w, h);
     TORCH_CHECK(handle_, "invalid encoder dimensions");
     airtc_h264enc_set_slices(handle_, slices);
+    airtc_h264enc_set_mb_mode(handle_, mb_mode);
   }
   ~H264SwEncoder() {
     if (handle_) airtc_h264enc_destroy(handle_);
@@ -384,9 +389,22 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("h264sw_table_check", []() { return airtc_h264sw_table_check(); },
         "0 iff all CAVLC tables are prefix-free");
   pybind11::class_<H264SwEncoder>(m, "H264SwEncoder")
-      .def(pybind11::init<int, int, int>(), pybind11::arg("w"),
-           pybind11::arg("h"), pybind11::arg("slices") = 4)
+      .def(pybind11::init<int, int, int, int>(), pybind11::arg("w"),
+           pybind11::arg("h"), pybind11::arg("slices") = 4,
+           pybind11::arg("mb_mode") = 0)
       .def("encode", &H264SwEncoder::encode, "RGB24 bytes + QP -> Annex-B IDR");
+  m.def("h264_pred4",
+        [](int mode, pybind11::bytes top, pybind11::bytes left, int tl,
+           bool ht, bool hl, bool htl) {
+          std::string t(top), l(left);
+          TORCH_CHECK(t.size() == 8 && l.size() == 4, "t8/l4 sizes");
+          uint8_t out[16];
+          airtc_h264_pred4(mode, (const uint8_t*)t.data(),
+                           (const uint8_t*)l.data(), (uint8_t)tl, ht, hl, htl,
+                           out);
+          return pybind11::bytes((const char*)out, 16);
+        },
+        "one 4x4 intra prediction (decoder path) for golden tests");
   pybind11::class_<H264SwDecoder>(m, "H264SwDecoder")
       .def(pybind11::init<>())
       .def("decode", &H264SwDecoder::decode,

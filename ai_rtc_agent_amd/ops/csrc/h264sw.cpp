@@ -609,6 +609,141 @@ static void pred_chroma8(const PlaneCtx& pl, int mbx, int mby, int mode,
   }
 }
 
+// ---------------------------------------------------------------------------
+// 4x4 intra luma prediction (all 9 modes) for I_4x4 macroblocks.
+// p_top: 8 samples above (top-right replicated per 8.3.1.2 when absent),
+// p_left: 4 samples left, p_tl: corner. Availability flags gate modes.
+// ---------------------------------------------------------------------------
+static void pred_luma4(int mode, const uint8_t* t /*8*/, const uint8_t* l /*4*/,
+                       uint8_t tl, bool have_t, bool have_l, bool have_tl,
+                       uint8_t pred[16]) {
+  auto P = [&](int x, int y) -> uint8_t& { return pred[y * 4 + x]; };
+  switch (mode) {
+    case 0:  // vertical
+      for (int y = 0; y < 4; ++y)
+        for (int x = 0; x < 4; ++x) P(x, y) = t[x];
+      break;
+    case 1:  // horizontal
+      for (int y = 0; y < 4; ++y)
+        for (int x = 0; x < 4; ++x) P(x, y) = l[y];
+      break;
+    case 2: {  // DC
+      int sum = 0, cnt = 0;
+      if (have_t) { sum += t[0] + t[1] + t[2] + t[3]; cnt += 4; }
+      if (have_l) { sum += l[0] + l[1] + l[2] + l[3]; cnt += 4; }
+      int dc = cnt == 8 ? (sum + 4) >> 3 : cnt == 4 ? (sum + 2) >> 2 : 128;
+      for (int i = 0; i < 16; ++i) pred[i] = (uint8_t)dc;
+      break;
+    }
+    case 3:  // diagonal down-left
+      for (int y = 0; y < 4; ++y)
+        for (int x = 0; x < 4; ++x) {
+          int i = x + y;
+          P(x, y) = (i == 6) ? (uint8_t)((t[6] + 3 * t[7] + 2) >> 2)
+                             : (uint8_t)((t[i] + 2 * t[i + 1] + t[i + 2] + 2) >> 2);
+        }
+      break;
+    case 4: {  // diagonal down-right (spec 8.3.1.2.5)
+      auto T4 = [&](int i) { return i < 0 ? tl : t[i]; };
+      auto L4 = [&](int i) { return i < 0 ? tl : l[i]; };
+      for (int y = 0; y < 4; ++y)
+        for (int x = 0; x < 4; ++x) {
+          if (x > y)
+            P(x, y) = (uint8_t)((T4(x - y - 2) + 2 * T4(x - y - 1) + T4(x - y) + 2) >> 2);
+          else if (x < y)
+            P(x, y) = (uint8_t)((L4(y - x - 2) + 2 * L4(y - x - 1) + L4(y - x) + 2) >> 2);
+          else
+            P(x, y) = (uint8_t)((t[0] + 2 * tl + l[0] + 2) >> 2);
+        }
+      break;
+    }
+    case 5: {  // vertical-right (spec 8.3.1.2.6)
+      auto T4 = [&](int i) { return i < 0 ? tl : t[i]; };
+      auto L4 = [&](int i) { return i < 0 ? tl : l[i]; };
+      for (int y = 0; y < 4; ++y)
+        for (int x = 0; x < 4; ++x) {
+          const int z = 2 * x - y;
+          if (z >= 0 && (z & 1) == 0) {
+            const int i = x - (y >> 1);
+            P(x, y) = (uint8_t)((T4(i - 1) + T4(i) + 1) >> 1);
+          } else if (z >= 0) {
+            const int i = x - (y >> 1);
+            P(x, y) = (uint8_t)((T4(i - 2) + 2 * T4(i - 1) + T4(i) + 2) >> 2);
+          } else if (z == -1) {
+            P(x, y) = (uint8_t)((l[0] + 2 * tl + t[0] + 2) >> 2);
+          } else {  // zVR < -1: pure left-column taps
+            const int base = y - 2 * x;  // 2 or 3
+            P(x, y) = (uint8_t)((L4(base - 1) + 2 * L4(base - 2) + L4(base - 3) + 2) >> 2);
+          }
+        }
+      break;
+    }
+    case 6:  // horizontal-down (mirror of VR)
+      for (int y = 0; y < 4; ++y)
+        for (int x = 0; x < 4; ++x) {
+          int z = 2 * y - x;
+          auto L = [&](int i2) { return i2 < 0 ? tl : l[i2]; };
+          auto T = [&](int i2) { return i2 < 0 ? tl : t[i2]; };
+          if (z >= 0 && (z & 1) == 0) {
+            int i = y - (x >> 1);
+            P(x, y) = (uint8_t)((L(i - 1) + L(i) + 1) >> 1);
+          } else if (z >= 0) {
+            int i = y - (x >> 1);
+            P(x, y) = (uint8_t)((L(i - 2) + 2 * L(i - 1) + L(i) + 2) >> 2);
+          } else if (z == -1) {
+            P(x, y) = (uint8_t)((l[0] + 2 * tl + t[0] + 2) >> 2);
+          } else {
+            int i = x - 2 * y;
+            P(x, y) = (uint8_t)((T(i - 1) + 2 * T(i - 2) + T(i - 3) + 2) >> 2);
+          }
+        }
+      break;
+    case 7:  // vertical-left
+      for (int y = 0; y < 4; ++y)
+        for (int x = 0; x < 4; ++x) {
+          int i = x + (y >> 1);
+          if ((y & 1) == 0)
+            P(x, y) = (uint8_t)((t[i] + t[i + 1] + 1) >> 1);
+          else
+            P(x, y) = (uint8_t)((t[i] + 2 * t[i + 1] + t[i + 2] + 2) >> 2);
+        }
+      break;
+    case 8:  // horizontal-up
+      for (int y = 0; y < 4; ++y)
+        for (int x = 0; x < 4; ++x) {
+          int z = x + 2 * y;
+          if (z > 5)
+            P(x, y) = l[3];
+          else if (z == 5)
+            P(x, y) = (uint8_t)((l[2] + 3 * l[3] + 2) >> 2);
+          else if (z & 1) {
+            int i = y + (x >> 1);
+            P(x, y) = (uint8_t)((l[i] + 2 * l[i + 1] + l[i + 2 > 3 ? 3 : i + 2] + 2) >> 2);
+          } else {
+            int i = y + (x >> 1);
+            P(x, y) = (uint8_t)((l[i] + l[i + 1] + 1) >> 1);
+          }
+        }
+      break;
+  }
+}
+
+// coded_block_pattern me(v) mapping for Intra_4x4 (spec table 9-4)
+static const uint8_t CBP_INTRA[48] = {
+    47, 31, 15, 0,  23, 27, 29, 30, 7,  11, 13, 14, 39, 43, 45, 46,
+    16, 3,  5,  10, 12, 19, 21, 26, 28, 35, 37, 42, 44, 1,  2,  4,
+    8,  17, 18, 20, 24, 6,  9,  22, 25, 32, 33, 34, 36, 40, 38, 41};
+static int cbp_intra_code(int cbp) {  // encoder: cbp -> codeNum
+  for (int i = 0; i < 48; ++i)
+    if (CBP_INTRA[i] == cbp) return i;
+  return 0;
+}
+
+// z-scan index of a 4x4 position (inverse of blk_x4/blk_y4)
+static inline int blk_z(int x4, int y4) {
+  return ((y4 >> 1) * 2 + (x4 >> 1)) * 4 + (y4 & 1) * 2 + (x4 & 1);
+}
+
 // Dequant + inverse transform + add prediction for one I_16x16 luma MB.
 // dc_q: 16 quantized DC levels in RASTER order; ac_q[blk][16] quantized
 // levels in RASTER order with [0] unused.
@@ -774,7 +909,205 @@ struct Encoder {
     return s;
   }
 
+  void write_pcm(BitWriter& wtr, int mbx, int mby) {
+    wtr.ue(25);  // I_PCM
+    wtr.align_byte();
+    const uint8_t* sy = Y.data() + (mby * 16) * pw + mbx * 16;
+    uint8_t* ry = rY.data() + (mby * 16) * pw + mbx * 16;
+    for (int y = 0; y < 16; ++y)
+      for (int x = 0; x < 16; ++x) {
+        wtr.put(sy[y * pw + x], 8);
+        ry[y * pw + x] = sy[y * pw + x];
+      }
+    const uint8_t* sc[2] = {Cb.data() + (mby * 8) * (pw / 2) + mbx * 8,
+                            Cr.data() + (mby * 8) * (pw / 2) + mbx * 8};
+    uint8_t* rc[2] = {rCb.data() + (mby * 8) * (pw / 2) + mbx * 8,
+                      rCr.data() + (mby * 8) * (pw / 2) + mbx * 8};
+    for (int comp = 0; comp < 2; ++comp)
+      for (int y = 0; y < 8; ++y)
+        for (int x = 0; x < 8; ++x) {
+          wtr.put(sc[comp][y * (pw / 2) + x], 8);
+          rc[comp][y * (pw / 2) + x] = sc[comp][y * (pw / 2) + x];
+        }
+    for (int b = 0; b < 16; ++b)
+      nnz.lnz(mbx, mby, blk_x4(b), blk_y4(b)) = 16;
+    for (int comp = 0; comp < 2; ++comp)
+      for (int blk = 0; blk < 4; ++blk)
+        nnz.cnz(comp, mbx, mby, blk & 1, blk >> 1) = 16;
+  }
+
+  // I_4x4 macroblock encode, DC prediction per 4x4 block. Primary purpose:
+  // an in-repo bitstream source for the decoder's I_4x4 path (the syntax a
+  // hardware intra encoder emits); with every neighbourhood mode DC the
+  // predicted mode is always DC, so prev_intra4x4_pred_mode_flag is 1 for
+  // every block.
+  void encode_mb_i4x4(BitWriter& wtr, int mbx, int mby, int qp,
+                      int slice_start) {
+    const bool mb_top = nnz.top_ok(mbx, mby, slice_start);
+    const bool mb_left = nnz.left_ok(mbx, mby, slice_start);
+    const int qm = qp % 6, qbits = 15 + qp / 6, qs = qp / 6;
+    const int32_t fr = (1 << qbits) / 3;
+    uint8_t* base = rY.data();
+    const uint8_t* srcy = Y.data();
+    const int stride = pw;
+
+    int32_t lq[16][16];
+    memset(lq, 0, sizeof(lq));
+    int maxtc = 0;
+    for (int z = 0; z < 16; ++z) {
+      const int x4 = blk_x4(z), y4 = blk_y4(z);
+      const int px = mbx * 16 + x4 * 4, py = mby * 16 + y4 * 4;
+      const bool ht = y4 > 0 || mb_top;
+      const bool hl = x4 > 0 || mb_left;
+      uint8_t tbuf[8] = {128, 128, 128, 128, 128, 128, 128, 128};
+      uint8_t lbuf[4] = {128, 128, 128, 128};
+      if (ht)
+        for (int i = 0; i < 4; ++i) tbuf[i] = base[(py - 1) * stride + px + i];
+      if (hl)
+        for (int i = 0; i < 4; ++i) lbuf[i] = base[(py + i) * stride + px - 1];
+      uint8_t pred[16];
+      pred_luma4(2, tbuf, lbuf, 128, ht, hl, false, pred);  // DC
+      int16_t d[16];
+      for (int y = 0; y < 4; ++y)
+        for (int x = 0; x < 4; ++x)
+          d[y * 4 + x] = (int16_t)((int)srcy[(py + y) * stride + px + x] -
+                                   (int)pred[y * 4 + x]);
+      int32_t W[16];
+      fwd4x4(d, W);
+      int tc = 0;
+      for (int i = 0; i < 16; ++i) {
+        int32_t q =
+            (int32_t)(((int64_t)std::abs(W[i]) * QMF[qm][POSCLS[i]] + fr) >>
+                      qbits);
+        if (q > 2063) q = 2063;
+        lq[z][i] = W[i] < 0 ? -q : q;
+        tc += q != 0;
+      }
+      maxtc = std::max(maxtc, tc);
+      // reconstruct immediately: later blocks predict from these pixels
+      int32_t dq[16], rr[16];
+      for (int i = 0; i < 16; ++i) dq[i] = (lq[z][i] * QV[qm][POSCLS[i]]) << qs;
+      inv4x4(dq, rr);
+      for (int y = 0; y < 4; ++y)
+        for (int x = 0; x < 4; ++x)
+          base[(py + y) * stride + px + x] =
+              clip8(pred[y * 4 + x] + ((rr[y * 4 + x] + 32) >> 6));
+    }
+
+    // chroma (DC mode), same math as the I_16x16 path
+    const int qpc = CHROMA_QP[qp < 0 ? 0 : qp > 51 ? 51 : qp];
+    PlaneCtx rpcb{rCb.data(), pw / 2}, rpcr{rCr.data(), pw / 2};
+    uint8_t cpred[2][64];
+    int32_t cdc_q[2][4], cac_q[2][4][16];
+    memset(cac_q, 0, sizeof(cac_q));
+    const uint8_t* csrc2[2] = {Cb.data() + (mby * 8) * (pw / 2) + mbx * 8,
+                               Cr.data() + (mby * 8) * (pw / 2) + mbx * 8};
+    PlaneCtx* cpl[2] = {&rpcb, &rpcr};
+    int ctc_max = 0;
+    for (int comp = 0; comp < 2; ++comp) {
+      pred_chroma8(*cpl[comp], mbx, mby, 0, mb_top, mb_left, cpred[comp]);
+      int32_t cdc_raw[4];
+      for (int blk = 0; blk < 4; ++blk) {
+        const int x4 = blk & 1, y4 = blk >> 1;
+        int16_t d[16];
+        for (int y = 0; y < 4; ++y)
+          for (int x = 0; x < 4; ++x)
+            d[y * 4 + x] =
+                (int16_t)((int)csrc2[comp][(y4 * 4 + y) * (pw / 2) + x4 * 4 + x] -
+                          (int)cpred[comp][(y4 * 4 + y) * 8 + x4 * 4 + x]);
+        int32_t W[16];
+        fwd4x4(d, W);
+        cdc_raw[blk] = W[0];
+        quant_block(W, qpc, /*skip_dc=*/true, cac_q[comp][blk]);
+        int tc = 0;
+        for (int i = 1; i < 16; ++i) tc += cac_q[comp][blk][i] != 0;
+        ctc_max = std::max(ctc_max, tc);
+      }
+      int32_t a = cdc_raw[0], b2 = cdc_raw[1], c = cdc_raw[2], d2 = cdc_raw[3];
+      int32_t t[4] = {a + b2 + c + d2, a - b2 + c - d2, a + b2 - c - d2,
+                      a - b2 - c + d2};
+      const int qmc = qpc % 6, qbc = 15 + qpc / 6;
+      const int32_t fc = (1 << qbc) / 3;
+      for (int i = 0; i < 4; ++i) {
+        int32_t q = (int32_t)(((int64_t)std::abs(t[i]) * QMF[qmc][0] + 2 * fc) >>
+                              (qbc + 1));
+        if (q > 2063) q = 2063;
+        cdc_q[comp][i] = t[i] < 0 ? -q : q;
+      }
+    }
+    if (maxtc > 9 || ctc_max > 9) {
+      // CAVLC guard (see encode_mb): high-entropy MB -> I_PCM
+      write_pcm(wtr, mbx, mby);
+      return;
+    }
+    int cbp_chroma = 0;
+    for (int comp = 0; comp < 2; ++comp)
+      for (int blk = 0; blk < 4; ++blk)
+        for (int i = 1; i < 16; ++i)
+          if (cac_q[comp][blk][i]) cbp_chroma = 2;
+    if (cbp_chroma == 0)
+      for (int comp = 0; comp < 2; ++comp)
+        for (int i = 0; i < 4; ++i)
+          if (cdc_q[comp][i]) cbp_chroma = 1;
+    int cbp_luma = 0;
+    for (int i8 = 0; i8 < 4; ++i8) {
+      bool any = false;
+      for (int sub = 0; sub < 4; ++sub)
+        for (int i = 0; i < 16; ++i)
+          if (lq[i8 * 4 + sub][i]) any = true;
+      if (any) cbp_luma |= 1 << i8;
+    }
+
+    // --- syntax ---
+    wtr.ue(0);  // mb_type: I_4x4
+    for (int z = 0; z < 16; ++z) wtr.put(1, 1);  // mode == predicted (DC)
+    wtr.ue(0);  // intra_chroma_pred_mode: DC
+    const int cbp = cbp_luma | (cbp_chroma << 4);
+    wtr.ue((uint32_t)cbp_intra_code(cbp));
+    if (cbp) wtr.se(0);  // mb_qp_delta
+    for (int z = 0; z < 16; ++z) {
+      const int x4 = blk_x4(z), y4 = blk_y4(z);
+      if (cbp_luma & (1 << (z >> 2))) {
+        int32_t scan[16];
+        for (int i = 0; i < 16; ++i) scan[i] = lq[z][ZIGZAG[i]];
+        int tc = cavlc_write_block(wtr, scan, 16,
+                                   nnz.luma_nc(mbx, mby, x4, y4, slice_start));
+        nnz.lnz(mbx, mby, x4, y4) = (uint8_t)tc;
+      } else {
+        nnz.lnz(mbx, mby, x4, y4) = 0;
+      }
+    }
+    if (!cbp_chroma) memset(cdc_q, 0, sizeof(cdc_q));
+    if (cbp_chroma < 2) memset(cac_q, 0, sizeof(cac_q));
+    if (cbp_chroma)
+      for (int comp = 0; comp < 2; ++comp)
+        cavlc_write_block(wtr, cdc_q[comp], 4, -1);
+    if (cbp_chroma == 2) {
+      for (int comp = 0; comp < 2; ++comp)
+        for (int blk = 0; blk < 4; ++blk) {
+          const int x2 = blk & 1, y2 = blk >> 1;
+          int32_t scan[15];
+          for (int i = 1; i < 16; ++i) scan[i - 1] = cac_q[comp][blk][ZIGZAG[i]];
+          int tc = cavlc_write_block(
+              wtr, scan, 15, nnz.chroma_nc(comp, mbx, mby, x2, y2, slice_start));
+          nnz.cnz(comp, mbx, mby, x2, y2) = (uint8_t)tc;
+        }
+    } else {
+      for (int comp = 0; comp < 2; ++comp)
+        for (int blk = 0; blk < 4; ++blk)
+          nnz.cnz(comp, mbx, mby, blk & 1, blk >> 1) = 0;
+    }
+    recon_chroma8(rpcb, mbx, mby, cpred[0], cdc_q[0], cac_q[0], qpc);
+    recon_chroma8(rpcr, mbx, mby, cpred[1], cdc_q[1], cac_q[1], qpc);
+  }
+
+  int mb_mode = 0;  // 0 = I_16x16 (default), 1 = I_4x4 DC
+
   void encode_mb(BitWriter& wtr, int mbx, int mby, int qp, int slice_start) {
+    if (mb_mode == 1) {
+      encode_mb_i4x4(wtr, mbx, mby, qp, slice_start);
+      return;
+    }
     PlaneCtx rpy{rY.data(), pw};
     PlaneCtx rpcb{rCb.data(), pw / 2}, rpcr{rCr.data(), pw / 2};
     const uint8_t* src = Y.data() + (mby * 16) * pw + mbx * 16;
@@ -899,29 +1232,7 @@ struct Encoder {
       for (int blk = 0; blk < 4 && !pcm; ++blk)
         pcm = count_nz(cac_q[comp][blk], 1, 16) > 9;
     if (pcm) {
-      wtr.ue(25);  // I_PCM
-      wtr.align_byte();
-      const uint8_t* sy = Y.data() + (mby * 16) * pw + mbx * 16;
-      uint8_t* ry = rY.data() + (mby * 16) * pw + mbx * 16;
-      for (int y = 0; y < 16; ++y)
-        for (int x = 0; x < 16; ++x) {
-          wtr.put(sy[y * pw + x], 8);
-          ry[y * pw + x] = sy[y * pw + x];
-        }
-      const uint8_t* sc[2] = {csrc[0], csrc[1]};
-      uint8_t* rc[2] = {rCb.data() + (mby * 8) * (pw / 2) + mbx * 8,
-                        rCr.data() + (mby * 8) * (pw / 2) + mbx * 8};
-      for (int comp = 0; comp < 2; ++comp)
-        for (int y = 0; y < 8; ++y)
-          for (int x = 0; x < 8; ++x) {
-            wtr.put(sc[comp][y * (pw / 2) + x], 8);
-            rc[comp][y * (pw / 2) + x] = sc[comp][y * (pw / 2) + x];
-          }
-      for (int b = 0; b < 16; ++b)
-        nnz.lnz(mbx, mby, blk_x4(b), blk_y4(b)) = 16;
-      for (int comp = 0; comp < 2; ++comp)
-        for (int blk = 0; blk < 4; ++blk)
-          nnz.cnz(comp, mbx, mby, blk & 1, blk >> 1) = 16;
+      write_pcm(wtr, mbx, mby);
       return;
     }
 
@@ -1080,6 +1391,9 @@ struct Decoder {
   Pps pps;
   std::vector<uint8_t> rY, rCb, rCr;
   NnzCtx nnz;
+  // per-4x4 intra pred modes of I_4x4 MBs (-1 = MB not I_4x4): neighbours'
+  // modes feed the predicted-mode rule; sized mbw*mbh*16, reset per frame
+  std::vector<int8_t> i4modes;
 
   int parse_sps(BitReader& r) {
     Sps s;
@@ -1152,12 +1466,180 @@ struct Decoder {
     return 0;
   }
 
+  // parse + reconstruct the chroma residual of one intra MB (shared by the
+  // I_16x16 and I_4x4 paths; cbp_chroma: 0 none, 1 DC only, 2 DC+AC)
+  int decode_chroma(BitReader& r, int mbx, int mby, int cbp_chroma, int cm,
+                    int qp, int slice_start, bool have_top, bool have_left) {
+    PlaneCtx rpcb{rCb.data(), sps.pw / 2}, rpcr{rCr.data(), sps.pw / 2};
+    int32_t cdc_q[2][4] = {{0}}, cac_q[2][4][16];
+    memset(cac_q, 0, sizeof(cac_q));
+    if (cbp_chroma) {
+      for (int comp = 0; comp < 2; ++comp) {
+        int32_t scan[4];
+        if (cavlc_read_block(r, scan, 4, -1) < 0) return -1;
+        for (int i = 0; i < 4; ++i) cdc_q[comp][i] = scan[i];
+      }
+    }
+    if (cbp_chroma == 2) {
+      for (int comp = 0; comp < 2; ++comp)
+        for (int blk = 0; blk < 4; ++blk) {
+          const int x2 = blk & 1, y2 = blk >> 1;
+          int32_t scan[15];
+          int tc = cavlc_read_block(r, scan, 15,
+                                    nnz.chroma_nc(comp, mbx, mby, x2, y2,
+                                                  slice_start));
+          if (tc < 0) return -1;
+          for (int i = 1; i < 16; ++i) cac_q[comp][blk][ZIGZAG[i]] = scan[i - 1];
+          nnz.cnz(comp, mbx, mby, x2, y2) = (uint8_t)tc;
+        }
+    } else {
+      for (int comp = 0; comp < 2; ++comp)
+        for (int blk = 0; blk < 4; ++blk)
+          nnz.cnz(comp, mbx, mby, blk & 1, blk >> 1) = 0;
+    }
+    const int qpi = qp + pps.chroma_qp_offset;
+    const int qpc = CHROMA_QP[qpi < 0 ? 0 : qpi > 51 ? 51 : qpi];
+    if ((cm == 2 && !have_top) || (cm == 1 && !have_left) ||
+        (cm == 3 && !(have_top && have_left)))
+      return -1;
+    uint8_t cpred[64];
+    pred_chroma8(rpcb, mbx, mby, cm, have_top, have_left, cpred);
+    recon_chroma8(rpcb, mbx, mby, cpred, cdc_q[0], cac_q[0], qpc);
+    pred_chroma8(rpcr, mbx, mby, cm, have_top, have_left, cpred);
+    recon_chroma8(rpcr, mbx, mby, cpred, cdc_q[1], cac_q[1], qpc);
+    return 0;
+  }
+
+  // decode one I_4x4 macroblock (all 9 luma pred modes, full CBP syntax)
+  int decode_mb_i4x4(BitReader& r, int mbx, int mby, int& qp,
+                     int slice_start) {
+    const int mbw = sps.pw / 16;
+    const bool mb_top = nnz.top_ok(mbx, mby, slice_start);
+    const bool mb_left = nnz.left_ok(mbx, mby, slice_start);
+    int8_t* my_modes = &i4modes[((size_t)mby * mbw + mbx) * 16];
+
+    int modes[16];
+    for (int z = 0; z < 16; ++z) {
+      const int x4 = blk_x4(z), y4 = blk_y4(z);
+      int ma = 2, mb_ = 2;  // DC when unavailable / neighbour not I_4x4
+      if (x4 > 0) {
+        ma = modes[blk_z(x4 - 1, y4)];
+      } else if (mb_left) {
+        int8_t v = i4modes[((size_t)mby * mbw + mbx - 1) * 16 + blk_z(3, y4)];
+        ma = v < 0 ? 2 : v;
+      }
+      if (y4 > 0) {
+        mb_ = modes[blk_z(x4, y4 - 1)];
+      } else if (mb_top) {
+        int8_t v = i4modes[((size_t)(mby - 1) * mbw + mbx) * 16 + blk_z(x4, 3)];
+        mb_ = v < 0 ? 2 : v;
+      }
+      const int pred = ma < mb_ ? ma : mb_;
+      if (r.u(1)) {
+        modes[z] = pred;
+      } else {
+        const int rem = (int)r.u(3);
+        modes[z] = rem < pred ? rem : rem + 1;
+      }
+      my_modes[z] = (int8_t)modes[z];
+    }
+    const uint32_t chroma_mode = r.ue();
+    if (chroma_mode > 3 || r.overrun) return -1;
+    const uint32_t cbp_code = r.ue();
+    if (cbp_code >= 48) return -1;
+    const int cbp = CBP_INTRA[cbp_code];
+    const int cbp_luma = cbp & 15, cbp_chroma = cbp >> 4;
+    if (cbp) {
+      qp += r.se();
+      if (qp < 0 || qp > 51) return -1;
+    }
+
+    // parse luma levels (full 16-coeff blocks, zig-zag)
+    int32_t lq[16][16];
+    memset(lq, 0, sizeof(lq));
+    for (int z = 0; z < 16; ++z) {
+      const int x4 = blk_x4(z), y4 = blk_y4(z);
+      if (cbp_luma & (1 << (z >> 2))) {
+        int32_t scan[16];
+        int tc = cavlc_read_block(r, scan, 16,
+                                  nnz.luma_nc(mbx, mby, x4, y4, slice_start));
+        if (tc < 0) return -1;
+        for (int i = 0; i < 16; ++i) lq[z][ZIGZAG[i]] = scan[i];
+        nnz.lnz(mbx, mby, x4, y4) = (uint8_t)tc;
+      } else {
+        nnz.lnz(mbx, mby, x4, y4) = 0;
+      }
+    }
+
+    // reconstruct blocks in z order (later blocks predict from earlier
+    // reconstructed pixels)
+    const int qm = qp % 6, qs = qp / 6;
+    const int stride = sps.pw;
+    uint8_t* base = rY.data();
+    for (int z = 0; z < 16; ++z) {
+      const int x4 = blk_x4(z), y4 = blk_y4(z);
+      const int px = mbx * 16 + x4 * 4, py = mby * 16 + y4 * 4;
+      const bool ht = y4 > 0 || mb_top;
+      const bool hl = x4 > 0 || mb_left;
+      const bool htl = (x4 > 0 && y4 > 0) || (x4 > 0 && mb_top) ||
+                       (y4 > 0 && mb_left) || (mb_top && mb_left);
+      uint8_t tbuf[8] = {128, 128, 128, 128, 128, 128, 128, 128};
+      uint8_t lbuf[4] = {128, 128, 128, 128};
+      uint8_t tlv = 128;
+      if (ht) {
+        for (int i = 0; i < 4; ++i) tbuf[i] = base[(py - 1) * stride + px + i];
+        bool htr;
+        if (y4 == 0) {
+          // above(-right) samples come from the MB row above
+          htr = x4 < 3 ? mb_top
+                       : (mby > 0 && mbx + 1 < mbw &&
+                          ((mby - 1) * mbw + mbx + 1) >= slice_start);
+        } else {
+          htr = x4 < 3 && blk_z(x4 + 1, y4 - 1) < z;
+        }
+        if (htr)
+          for (int i = 0; i < 4; ++i)
+            tbuf[4 + i] = base[(py - 1) * stride + px + 4 + i];
+        else
+          for (int i = 0; i < 4; ++i) tbuf[4 + i] = tbuf[3];
+      }
+      if (hl)
+        for (int i = 0; i < 4; ++i) lbuf[i] = base[(py + i) * stride + px - 1];
+      if (htl) tlv = base[(py - 1) * stride + px - 1];
+
+      const int m = modes[z];
+      // modes that need absent neighbours are illegal in a valid stream
+      const bool needs_t = m == 0 || m == 3 || m == 7;
+      const bool needs_l = m == 1 || m == 8;
+      const bool needs_both = m == 4 || m == 5 || m == 6;
+      if ((needs_t && !ht) || (needs_l && !hl) || (needs_both && !(ht && hl)))
+        return -1;
+      uint8_t pred[16];
+      pred_luma4(m, tbuf, lbuf, tlv, ht, hl, htl, pred);
+      int32_t d[16];
+      for (int i = 0; i < 16; ++i) d[i] = (lq[z][i] * QV[qm][POSCLS[i]]) << qs;
+      int32_t rr[16];
+      inv4x4(d, rr);
+      for (int y = 0; y < 4; ++y)
+        for (int x = 0; x < 4; ++x)
+          base[(py + y) * stride + px + x] =
+              clip8(pred[y * 4 + x] + ((rr[y * 4 + x] + 32) >> 6));
+    }
+    return decode_chroma(r, mbx, mby, cbp_chroma, (int)chroma_mode, qp,
+                         slice_start, mb_top, mb_left);
+  }
+
   // decode one I_16x16 or I_PCM macroblock; returns 0 or negative error
   int decode_mb(BitReader& r, int mbx, int mby, int& qp, int slice_start) {
     PlaneCtx rpy{rY.data(), sps.pw};
-    PlaneCtx rpcb{rCb.data(), sps.pw / 2}, rpcr{rCr.data(), sps.pw / 2};
     uint32_t mb_type = r.ue();
     if (r.overrun) return -1;
+    {  // mark this MB as not-I_4x4 for neighbours' mode prediction
+      const int mbw = sps.pw / 16;
+      int8_t* mm = &i4modes[((size_t)mby * mbw + mbx) * 16];
+      for (int i = 0; i < 16; ++i) mm[i] = -1;
+    }
+    if (mb_type == 0) return decode_mb_i4x4(r, mbx, mby, qp, slice_start);
     if (mb_type == 25) {  // I_PCM
       r.align_byte();
       uint8_t* dst = rY.data() + (mby * 16) * sps.pw + mbx * 16;
@@ -1208,33 +1690,7 @@ struct Decoder {
       for (int b = 0; b < 16; ++b)
         nnz.lnz(mbx, mby, blk_x4(b), blk_y4(b)) = 0;
     }
-    int32_t cdc_q[2][4] = {{0}}, cac_q[2][4][16];
-    memset(cac_q, 0, sizeof(cac_q));
-    if (cbp_chroma) {
-      for (int comp = 0; comp < 2; ++comp) {
-        int32_t scan[4];
-        if (cavlc_read_block(r, scan, 4, -1) < 0) return -1;
-        for (int i = 0; i < 4; ++i) cdc_q[comp][i] = scan[i];
-      }
-    }
-    if (cbp_chroma == 2) {
-      for (int comp = 0; comp < 2; ++comp)
-        for (int blk = 0; blk < 4; ++blk) {
-          const int x2 = blk & 1, y2 = blk >> 1;
-          int32_t scan[15];
-          int tc = cavlc_read_block(
-              r, scan, 15, nnz.chroma_nc(comp, mbx, mby, x2, y2, slice_start));
-          if (tc < 0) return -1;
-          for (int i = 1; i < 16; ++i) cac_q[comp][blk][ZIGZAG[i]] = scan[i - 1];
-          nnz.cnz(comp, mbx, mby, x2, y2) = (uint8_t)tc;
-        }
-    } else {
-      for (int comp = 0; comp < 2; ++comp)
-        for (int blk = 0; blk < 4; ++blk)
-          nnz.cnz(comp, mbx, mby, blk & 1, blk >> 1) = 0;
-    }
-
-    // predict + reconstruct
+    // predict + reconstruct luma, then the shared chroma path
     uint8_t lpred[256];
     const bool have_top = nnz.top_ok(mbx, mby, slice_start);
     const bool have_left = nnz.left_ok(mbx, mby, slice_start);
@@ -1243,19 +1699,8 @@ struct Decoder {
       return -1;
     pred_luma16(rpy, mbx, mby, pred_mode, have_top, have_left, lpred);
     recon_luma16(rpy, mbx, mby, lpred, dc_q, ac_q, qp);
-    // chroma modes: bitstream 0=DC 1=H 2=V 3=Plane (pred_chroma8 order)
-    const int qpi = qp + pps.chroma_qp_offset;
-    const int qpc = CHROMA_QP[qpi < 0 ? 0 : qpi > 51 ? 51 : qpi];
-    int cm = (int)chroma_mode;
-    if ((cm == 2 && !have_top) || (cm == 1 && !have_left) ||
-        (cm == 3 && !(have_top && have_left)))
-      return -1;
-    uint8_t cpred[64];
-    pred_chroma8(rpcb, mbx, mby, cm, have_top, have_left, cpred);
-    recon_chroma8(rpcb, mbx, mby, cpred, cdc_q[0], cac_q[0], qpc);
-    pred_chroma8(rpcr, mbx, mby, cm, have_top, have_left, cpred);
-    recon_chroma8(rpcr, mbx, mby, cpred, cdc_q[1], cac_q[1], qpc);
-    return 0;
+    return decode_chroma(r, mbx, mby, cbp_chroma, (int)chroma_mode, qp,
+                         slice_start, have_top, have_left);
   }
 
   // returns 1 when a frame was reconstructed, 0 for parameter-set-only data,
@@ -1322,6 +1767,7 @@ struct Decoder {
       const int mbw = sps.pw / 16, mbh = sps.ph / 16;
       const int total = mbw * mbh;
       nnz.reset(mbw, mbh);
+      i4modes.assign((size_t)total * 16, -1);
       std::vector<int> rcs(slices.size(), 0), counts(slices.size(), 0);
       auto decode_slice = [&](size_t si) {
         SliceJob& job = slices[si];
@@ -1380,6 +1826,10 @@ void* airtc_h264enc_create(int w, int h) {
 void airtc_h264enc_set_slices(void* h, int n) {
   ((h264sw::Encoder*)h)->n_slices = n < 1 ? 1 : n;
 }
+// macroblock mode: 0 = I_16x16 (default), 1 = I_4x4 DC (decoder-test aid)
+void airtc_h264enc_set_mb_mode(void* h, int m) {
+  ((h264sw::Encoder*)h)->mb_mode = m == 1 ? 1 : 0;
+}
 void airtc_h264enc_destroy(void* h) { delete (h264sw::Encoder*)h; }
 int airtc_h264enc_encode(void* h, const uint8_t* rgb, int qp, uint8_t* out,
                          int cap) {
@@ -1397,6 +1847,14 @@ void airtc_h264dec_dims(void* h, int* w, int* out_h) {
 int airtc_h264dec_decode(void* h, const uint8_t* data, int len, uint8_t* rgb,
                          int cap, int* w, int* out_h) {
   return ((h264sw::Decoder*)h)->decode_au(data, len, rgb, cap, w, out_h);
+}
+
+// test hook: run one 4x4 intra prediction (decoder path) for golden checks
+void airtc_h264_pred4(int mode, const uint8_t* t8, const uint8_t* l4,
+                      uint8_t tl, int have_t, int have_l, int have_tl,
+                      uint8_t* out16) {
+  h264sw::pred_luma4(mode, t8, l4, tl, have_t != 0, have_l != 0, have_tl != 0,
+                     out16);
 }
 
 // sanity: every VLC table must be prefix-free within itself; returns 0 on
