@@ -114,9 +114,13 @@ async def whip(request: web.Request) -> web.Response:
                 await events.stream_ended(stream_id)
 
     st["whip_pc"] = pc
+    st.setdefault("whip_sessions", {})[stream_id] = pc
     await pc.set_remote_description(offer_sdp)
     sdp = await pc.create_answer(host=request.app["host"], direction="recvonly")
-    headers = {"Location": "/whip"}
+    # per-session resource URL (the WHIP spec's DELETE target; OBS follows
+    # the Location header). Bare DELETE /whip still closes the current
+    # publisher for reference-parity clients.
+    headers = {"Location": f"/whip/{stream_id}"}
     for link in get_link_headers(st["ice_servers"]):
         headers.setdefault("Link", link)
     return web.Response(status=201, content_type="application/sdp", text=sdp, headers=headers)
@@ -124,6 +128,17 @@ async def whip(request: web.Request) -> web.Response:
 
 async def whip_delete(request: web.Request) -> web.Response:
     st = _state(request.app)
+    sid = request.match_info.get("sid")
+    if sid is not None:
+        pc = st.get("whip_sessions", {}).pop(sid, None)
+        if pc is None:
+            return web.Response(status=404)
+        await pc.close()
+        if st.get("whip_pc") is pc:
+            st["whip_pc"] = None
+            st["source_track"] = None
+        st["pool"].release(sid)
+        return web.Response(status=200)
     pc = st.get("whip_pc")
     if pc is not None:
         await pc.close()
@@ -275,6 +290,7 @@ def create_app(
     app.router.add_post("/offer", offer)
     app.router.add_post("/whip", whip)
     app.router.add_delete("/whip", whip_delete)
+    app.router.add_delete("/whip/{sid}", whip_delete)
     app.router.add_post("/whep", whep)
     app.router.add_delete("/whep", whep_delete)
     app.router.add_post("/config", update_config)

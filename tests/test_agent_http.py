@@ -86,7 +86,9 @@ def test_whip_then_whep():
         client = await make_client(StubPipeline())
         r = await client.post("/whip", data=OFFER_SDP, headers={"Content-Type": "application/sdp"})
         assert r.status == 201
-        assert r.headers["Location"] == "/whip"
+        # per-session WHIP resource URL (the spec's DELETE target)
+        assert r.headers["Location"].startswith("/whip/")
+        whip_resource = r.headers["Location"]
         assert r.content_type == "application/sdp"
         answer = await r.text()
         assert "H264" in answer and "a=candidate" in answer
@@ -97,8 +99,14 @@ def test_whip_then_whep():
         r2 = await client.post("/whep", data=OFFER_SDP, headers={"Content-Type": "application/sdp"})
         assert r2.status == 401
 
-        r3 = await client.delete("/whip")
+        # DELETE on the per-session resource URL (WHIP spec behaviour)
+        r3 = await client.delete(whip_resource)
         assert r3.status == 200
+        r3b = await client.delete(whip_resource)
+        assert r3b.status == 404  # already gone
+        # bare DELETE /whip stays for reference-parity clients
+        r3c = await client.delete("/whip")
+        assert r3c.status == 200
         await client.close()
 
     run(body())
