@@ -192,7 +192,9 @@ class H264SwCodec:
         arr = frame_u8.detach().to("cpu", torch.uint8).contiguous()
         h, w = int(arr.shape[0]), int(arr.shape[1])
         if self._enc_dims != (w, h):
-            self._enc = self._ext.H264SwEncoder(w, h)
+            # mb_mode=2: per-MB I_16x16 / I_4x4 decision (I_4x4 with full
+            # mode search wins on moderately detailed macroblocks)
+            self._enc = self._ext.H264SwEncoder(w, h, mb_mode=2)
             self._enc_dims = (w, h)
         data = self._enc.encode(arr.numpy().tobytes(), self._qp)
         # QP rate control toward the per-frame byte budget

@@ -872,6 +872,7 @@ struct Encoder {
   std::vector<uint8_t> Y, Cb, Cr;        // source (padded)
   std::vector<uint8_t> rY, rCb, rCr;     // reconstruction
   NnzCtx nnz;
+  std::vector<int8_t> enc_i4modes;       // per-4x4 modes (-1: MB not I_4x4)
   uint32_t idr_id = 0;
 
   Encoder(int width, int height) : w(width), h(height) {
@@ -936,11 +937,10 @@ struct Encoder {
         nnz.cnz(comp, mbx, mby, blk & 1, blk >> 1) = 16;
   }
 
-  // I_4x4 macroblock encode, DC prediction per 4x4 block. Primary purpose:
-  // an in-repo bitstream source for the decoder's I_4x4 path (the syntax a
-  // hardware intra encoder emits); with every neighbourhood mode DC the
-  // predicted mode is always DC, so prev_intra4x4_pred_mode_flag is 1 for
-  // every block.
+  // I_4x4 macroblock encode with per-block mode selection over all 9
+  // prediction modes (SAD decision on the reconstructed neighbourhood,
+  // greedy in z order — the standard intra search shape). Also the
+  // in-repo bitstream source exercising the decoder's I_4x4 path.
   void encode_mb_i4x4(BitWriter& wtr, int mbx, int mby, int qp,
                       int slice_start) {
     const bool mb_top = nnz.top_ok(mbx, mby, slice_start);
@@ -950,28 +950,72 @@ struct Encoder {
     uint8_t* base = rY.data();
     const uint8_t* srcy = Y.data();
     const int stride = pw;
+    int8_t* my_modes = &enc_i4modes[((size_t)mby * mbw + mbx) * 16];
 
     int32_t lq[16][16];
     memset(lq, 0, sizeof(lq));
+    int modes[16];
     int maxtc = 0;
     for (int z = 0; z < 16; ++z) {
       const int x4 = blk_x4(z), y4 = blk_y4(z);
       const int px = mbx * 16 + x4 * 4, py = mby * 16 + y4 * 4;
       const bool ht = y4 > 0 || mb_top;
       const bool hl = x4 > 0 || mb_left;
+      const bool htl = (x4 > 0 && y4 > 0) || (x4 > 0 && mb_top) ||
+                       (y4 > 0 && mb_left) || (mb_top && mb_left);
+      // neighbour samples: EXACTLY the decoder's gather (incl. the
+      // top-right replicate rule) so reconstructions agree bit-for-bit
       uint8_t tbuf[8] = {128, 128, 128, 128, 128, 128, 128, 128};
       uint8_t lbuf[4] = {128, 128, 128, 128};
-      if (ht)
+      uint8_t tlv = 128;
+      if (ht) {
         for (int i = 0; i < 4; ++i) tbuf[i] = base[(py - 1) * stride + px + i];
+        bool htr;
+        if (y4 == 0)
+          htr = x4 < 3 ? mb_top
+                       : (mby > 0 && mbx + 1 < mbw &&
+                          ((mby - 1) * mbw + mbx + 1) >= slice_start);
+        else
+          htr = x4 < 3 && blk_z(x4 + 1, y4 - 1) < z;
+        if (htr)
+          for (int i = 0; i < 4; ++i)
+            tbuf[4 + i] = base[(py - 1) * stride + px + 4 + i];
+        else
+          for (int i = 0; i < 4; ++i) tbuf[4 + i] = tbuf[3];
+      }
       if (hl)
         for (int i = 0; i < 4; ++i) lbuf[i] = base[(py + i) * stride + px - 1];
-      uint8_t pred[16];
-      pred_luma4(2, tbuf, lbuf, 128, ht, hl, false, pred);  // DC
+      if (htl) tlv = base[(py - 1) * stride + px - 1];
+
+      // SAD search over the legal modes
+      uint8_t pred[16], best_pred[16];
+      int best_mode = 2, best_sad = INT32_MAX;
+      for (int m = 0; m < 9; ++m) {
+        const bool needs_t = m == 0 || m == 3 || m == 7;
+        const bool needs_l = m == 1 || m == 8;
+        const bool needs_both = m == 4 || m == 5 || m == 6;
+        if ((needs_t && !ht) || (needs_l && !hl) || (needs_both && !(ht && hl)))
+          continue;
+        pred_luma4(m, tbuf, lbuf, tlv, ht, hl, htl, pred);
+        int sad = 0;
+        for (int y = 0; y < 4; ++y)
+          for (int x = 0; x < 4; ++x)
+            sad += std::abs((int)srcy[(py + y) * stride + px + x] -
+                            (int)pred[y * 4 + x]);
+        if (sad < best_sad) {
+          best_sad = sad;
+          best_mode = m;
+          memcpy(best_pred, pred, 16);
+        }
+      }
+      modes[z] = best_mode;
+      my_modes[z] = (int8_t)best_mode;
+
       int16_t d[16];
       for (int y = 0; y < 4; ++y)
         for (int x = 0; x < 4; ++x)
           d[y * 4 + x] = (int16_t)((int)srcy[(py + y) * stride + px + x] -
-                                   (int)pred[y * 4 + x]);
+                                   (int)best_pred[y * 4 + x]);
       int32_t W[16];
       fwd4x4(d, W);
       int tc = 0;
@@ -991,7 +1035,7 @@ struct Encoder {
       for (int y = 0; y < 4; ++y)
         for (int x = 0; x < 4; ++x)
           base[(py + y) * stride + px + x] =
-              clip8(pred[y * 4 + x] + ((rr[y * 4 + x] + 32) >> 6));
+              clip8(best_pred[y * 4 + x] + ((rr[y * 4 + x] + 32) >> 6));
     }
 
     // chroma (DC mode), same math as the I_16x16 path
@@ -1037,6 +1081,7 @@ struct Encoder {
     }
     if (maxtc > 9 || ctc_max > 9) {
       // CAVLC guard (see encode_mb): high-entropy MB -> I_PCM
+      for (int z = 0; z < 16; ++z) my_modes[z] = -1;
       write_pcm(wtr, mbx, mby);
       return;
     }
@@ -1060,7 +1105,29 @@ struct Encoder {
 
     // --- syntax ---
     wtr.ue(0);  // mb_type: I_4x4
-    for (int z = 0; z < 16; ++z) wtr.put(1, 1);  // mode == predicted (DC)
+    for (int z = 0; z < 16; ++z) {
+      const int x4 = blk_x4(z), y4 = blk_y4(z);
+      int ma = 2, mb_ = 2;
+      if (x4 > 0) {
+        ma = modes[blk_z(x4 - 1, y4)];
+      } else if (mb_left) {
+        int8_t v = enc_i4modes[((size_t)mby * mbw + mbx - 1) * 16 + blk_z(3, y4)];
+        ma = v < 0 ? 2 : v;
+      }
+      if (y4 > 0) {
+        mb_ = modes[blk_z(x4, y4 - 1)];
+      } else if (mb_top) {
+        int8_t v = enc_i4modes[((size_t)(mby - 1) * mbw + mbx) * 16 + blk_z(x4, 3)];
+        mb_ = v < 0 ? 2 : v;
+      }
+      const int predm = ma < mb_ ? ma : mb_;
+      if (modes[z] == predm) {
+        wtr.put(1, 1);  // prev_intra4x4_pred_mode_flag
+      } else {
+        wtr.put(0, 1);
+        wtr.put(modes[z] < predm ? modes[z] : modes[z] - 1, 3);
+      }
+    }
     wtr.ue(0);  // intra_chroma_pred_mode: DC
     const int cbp = cbp_luma | (cbp_chroma << 4);
     wtr.ue((uint32_t)cbp_intra_code(cbp));
@@ -1101,7 +1168,8 @@ struct Encoder {
     recon_chroma8(rpcr, mbx, mby, cpred[1], cdc_q[1], cac_q[1], qpc);
   }
 
-  int mb_mode = 0;  // 0 = I_16x16 (default), 1 = I_4x4 DC
+  // 0 = I_16x16 only; 1 = I_4x4 only; 2 = auto (per-MB SAD decision)
+  int mb_mode = 0;
 
   void encode_mb(BitWriter& wtr, int mbx, int mby, int qp, int slice_start) {
     if (mb_mode == 1) {
@@ -1129,6 +1197,13 @@ struct Encoder {
       }
     }
     const uint8_t* lp = pred[best_mode];
+
+    // auto mode: a detailed MB (mean |residual| > 5 under its best 16x16
+    // prediction) is usually cheaper as I_4x4 with per-block modes
+    if (mb_mode == 2 && best_sad > 256 * 5) {
+      encode_mb_i4x4(wtr, mbx, mby, qp, slice_start);
+      return;
+    }
 
     // --- luma transform/quant ---
     int32_t dc_raw[16];             // raster 4x4 of DC terms
@@ -1314,6 +1389,7 @@ struct Encoder {
       rgb_to_yuv420(rgb, w, h, pw, ph, Y.data(), Cb.data(), Cr.data());
     }
     nnz.reset(mbw, mbh);
+    enc_i4modes.assign((size_t)mbw * mbh * 16, -1);
 
     std::vector<uint8_t> bs;
     {
