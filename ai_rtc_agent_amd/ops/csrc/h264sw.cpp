@@ -1198,9 +1198,13 @@ struct Encoder {
     }
     const uint8_t* lp = pred[best_mode];
 
-    // auto mode: a detailed MB (mean |residual| > 5 under its best 16x16
-    // prediction) is usually cheaper as I_4x4 with per-block modes
-    if (mb_mode == 2 && best_sad > 256 * 5) {
+    // auto mode: a moderately detailed MB (mean |residual| in (5, 24]
+    // under its best 16x16 prediction) is usually cheaper as I_4x4 with
+    // per-block modes. Above that the TotalCoeff guard routes to I_PCM on
+    // either path, so skip the 9-mode search and let the I_16x16 path
+    // reach the same PCM cheaply (noise frames would otherwise pay the
+    // full search for nothing).
+    if (mb_mode == 2 && best_sad > 256 * 5 && best_sad <= 256 * 24) {
       encode_mb_i4x4(wtr, mbx, mby, qp, slice_start);
       return;
     }
