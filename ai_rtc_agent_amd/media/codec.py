@@ -168,7 +168,8 @@ class H264SwCodec:
 
     FPS_ASSUMED = 30
 
-    def __init__(self, cfg: EncoderConfig | None = None):
+    def __init__(self, cfg: EncoderConfig | None = None,
+                 keyframe_interval: int = 60):
         try:
             from .. import ops
 
@@ -178,8 +179,10 @@ class H264SwCodec:
         if self._ext is None or not hasattr(self._ext, "H264SwEncoder"):
             raise CodecUnavailable("native extension with h264sw not built")
         self.cfg = cfg or EncoderConfig()
+        self.keyframe_interval = keyframe_interval
         self._enc = None
         self._enc_dims = None
+        self._enc_count = 0
         self._dec = self._ext.H264SwDecoder()
         self._qp = 30
 
@@ -196,12 +199,20 @@ class H264SwCodec:
             # mode search wins on moderately detailed macroblocks)
             self._enc = self._ext.H264SwEncoder(w, h, mb_mode=2)
             self._enc_dims = (w, h)
-        data = self._enc.encode(arr.numpy().tobytes(), self._qp)
-        # QP rate control toward the per-frame byte budget
+            self._enc_count = 0
+        # GOP cadence: periodic IDR (join-anywhere + loss recovery bound);
+        # PLI sets keyframe=True through the transport (media/rtc.py)
+        force = keyframe or self._enc_count % self.keyframe_interval == 0
+        self._enc_count += 1
+        data = self._enc.encode(arr.numpy().tobytes(), self._qp,
+                                keyframe=bool(force))
+        # QP rate control toward the per-frame byte budget. P frames are
+        # motion-dependent and usually far under budget, so only keyframes
+        # drive the QP down; either frame type can push it up.
         budget = self._budget()
         if len(data) > budget and self._qp < 46:
             self._qp += 2
-        elif len(data) < budget // 2 and self._qp > 14:
+        elif force and len(data) < budget // 2 and self._qp > 14:
             self._qp -= 1
         return data
 

@@ -225,6 +225,8 @@ void* airtc_h264enc_create(int w, int h);
 void airtc_h264enc_set_slices(void*, int n);
 void airtc_h264enc_destroy(void*);
 int airtc_h264enc_encode(void*, const uint8_t*, int qp, uint8_t*, int cap);
+int airtc_h264enc_encode_ex(void*, const uint8_t*, int qp, int force_idr,
+                            uint8_t*, int cap);
 void airtc_h264enc_set_mb_mode(void*, int m);
 void airtc_h264_pred4(int, const uint8_t*, const uint8_t*, uint8_t, int, int,
                       int, uint8_t*);
@@ -248,15 +250,18 @@ class H264SwEncoder {
     if (handle_) airtc_h264enc_destroy(handle_);
   }
   H264SwEncoder(const H264SwEncoder&) = delete;
-  pybind11::bytes encode(pybind11::bytes rgb, int qp) {
+  // keyframe=false emits a P frame (P_Skip + intra refresh) once a
+  // reference exists; true (default) emits an IDR with in-band SPS/PPS
+  pybind11::bytes encode(pybind11::bytes rgb, int qp, bool keyframe = true) {
     std::string s(rgb);
     TORCH_CHECK((int)s.size() == w_ * h_ * 3, "rgb buffer size mismatch");
     std::vector<uint8_t> out((size_t)w_ * h_ * 6 + 4096);
     int n;
     {
       pybind11::gil_scoped_release nogil;
-      n = airtc_h264enc_encode(handle_, (const uint8_t*)s.data(), qp,
-                               out.data(), (int)out.size());
+      n = airtc_h264enc_encode_ex(handle_, (const uint8_t*)s.data(), qp,
+                                  keyframe ? 1 : 0, out.data(),
+                                  (int)out.size());
     }
     TORCH_CHECK(n > 0, "h264 encode failed rc=", n);
     return pybind11::bytes((const char*)out.data(), n);
@@ -392,7 +397,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def(pybind11::init<int, int, int, int>(), pybind11::arg("w"),
            pybind11::arg("h"), pybind11::arg("slices") = 4,
            pybind11::arg("mb_mode") = 0)
-      .def("encode", &H264SwEncoder::encode, "RGB24 bytes + QP -> Annex-B IDR");
+      .def("encode", &H264SwEncoder::encode, pybind11::arg("rgb"),
+           pybind11::arg("qp"), pybind11::arg("keyframe") = true,
+           "RGB24 bytes + QP -> Annex-B (IDR, or P when keyframe=false)");
   m.def("h264_pred4",
         [](int mode, pybind11::bytes top, pybind11::bytes left, int tl,
            bool ht, bool hl, bool htl) {

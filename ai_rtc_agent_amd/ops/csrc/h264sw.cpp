@@ -874,6 +874,17 @@ struct Encoder {
   NnzCtx nnz;
   std::vector<int8_t> enc_i4modes;       // per-4x4 modes (-1: MB not I_4x4)
   uint32_t idr_id = 0;
+  uint32_t frame_num = 0;                // mod 16 (log2_max_frame_num = 4)
+  bool have_ref = false;                 // a decoded-reference frame exists
+  // P_Skip decision: a macroblock whose source-vs-reference SAD is below
+  // the threshold is "unchanged". The floor is the QP's own quantisation
+  // error (at QP q the reconstruction already differs from the source by
+  // roughly qstep/3 per sample), so the threshold scales with QP;
+  // <0 = auto, 0 = bit-exact-only skips.
+  int skip_sad_thresh = -1;
+  int skip_thresh_for(int qp) const {
+    return skip_sad_thresh >= 0 ? skip_sad_thresh : 96 * (qp - 8);
+  }
 
   Encoder(int width, int height) : w(width), h(height) {
     pw = (w + 15) & ~15;
@@ -910,8 +921,8 @@ struct Encoder {
     return s;
   }
 
-  void write_pcm(BitWriter& wtr, int mbx, int mby) {
-    wtr.ue(25);  // I_PCM
+  void write_pcm(BitWriter& wtr, int mbx, int mby, bool in_p_slice = false) {
+    wtr.ue(in_p_slice ? 30 : 25);  // I_PCM (+5 in P slices)
     wtr.align_byte();
     const uint8_t* sy = Y.data() + (mby * 16) * pw + mbx * 16;
     uint8_t* ry = rY.data() + (mby * 16) * pw + mbx * 16;
@@ -942,7 +953,7 @@ struct Encoder {
   // greedy in z order — the standard intra search shape). Also the
   // in-repo bitstream source exercising the decoder's I_4x4 path.
   void encode_mb_i4x4(BitWriter& wtr, int mbx, int mby, int qp,
-                      int slice_start) {
+                      int slice_start, bool in_p_slice = false) {
     const bool mb_top = nnz.top_ok(mbx, mby, slice_start);
     const bool mb_left = nnz.left_ok(mbx, mby, slice_start);
     const int qm = qp % 6, qbits = 15 + qp / 6, qs = qp / 6;
@@ -1082,7 +1093,7 @@ struct Encoder {
     if (maxtc > 9 || ctc_max > 9) {
       // CAVLC guard (see encode_mb): high-entropy MB -> I_PCM
       for (int z = 0; z < 16; ++z) my_modes[z] = -1;
-      write_pcm(wtr, mbx, mby);
+      write_pcm(wtr, mbx, mby, in_p_slice);
       return;
     }
     int cbp_chroma = 0;
@@ -1104,7 +1115,7 @@ struct Encoder {
     }
 
     // --- syntax ---
-    wtr.ue(0);  // mb_type: I_4x4
+    wtr.ue(in_p_slice ? 5 : 0);  // mb_type: I_4x4 (+5 in P slices)
     for (int z = 0; z < 16; ++z) {
       const int x4 = blk_x4(z), y4 = blk_y4(z);
       int ma = 2, mb_ = 2;
@@ -1171,9 +1182,10 @@ struct Encoder {
   // 0 = I_16x16 only; 1 = I_4x4 only; 2 = auto (per-MB SAD decision)
   int mb_mode = 0;
 
-  void encode_mb(BitWriter& wtr, int mbx, int mby, int qp, int slice_start) {
+  void encode_mb(BitWriter& wtr, int mbx, int mby, int qp, int slice_start,
+                 bool in_p_slice = false) {
     if (mb_mode == 1) {
-      encode_mb_i4x4(wtr, mbx, mby, qp, slice_start);
+      encode_mb_i4x4(wtr, mbx, mby, qp, slice_start, in_p_slice);
       return;
     }
     PlaneCtx rpy{rY.data(), pw};
@@ -1205,7 +1217,7 @@ struct Encoder {
     // reach the same PCM cheaply (noise frames would otherwise pay the
     // full search for nothing).
     if (mb_mode == 2 && best_sad > 256 * 5 && best_sad <= 256 * 24) {
-      encode_mb_i4x4(wtr, mbx, mby, qp, slice_start);
+      encode_mb_i4x4(wtr, mbx, mby, qp, slice_start, in_p_slice);
       return;
     }
 
@@ -1311,12 +1323,13 @@ struct Encoder {
       for (int blk = 0; blk < 4 && !pcm; ++blk)
         pcm = count_nz(cac_q[comp][blk], 1, 16) > 9;
     if (pcm) {
-      write_pcm(wtr, mbx, mby);
+      write_pcm(wtr, mbx, mby, in_p_slice);
       return;
     }
 
     // --- write macroblock_layer ---
-    const int mb_type = 1 + best_mode + 4 * cbp_chroma + 12 * (cbp_luma ? 1 : 0);
+    const int mb_type = 1 + best_mode + 4 * cbp_chroma + 12 * (cbp_luma ? 1 : 0)
+                        + (in_p_slice ? 5 : 0);
     wtr.ue(mb_type);
     wtr.ue(0);  // intra_chroma_pred_mode: DC
     wtr.se(0);  // mb_qp_delta (constant QP per frame)
@@ -1373,8 +1386,38 @@ struct Encoder {
   int n_slices = 1;  // MB-row bands, encoded in parallel threads
 
   int encode(const uint8_t* rgb, int qp, uint8_t* out, int cap) {
+    return encode_ex(rgb, qp, /*force_idr=*/1, out, cap);
+  }
+
+  // SAD of a source 16x16 luma MB vs the reconstructed reference
+  int mb_ref_sad(int mbx, int mby) const {
+    const uint8_t* sy = Y.data() + (mby * 16) * pw + mbx * 16;
+    const uint8_t* ry = rY.data() + (mby * 16) * pw + mbx * 16;
+    int sad = 0;
+    for (int y = 0; y < 16; ++y)
+      for (int x = 0; x < 16; ++x)
+        sad += std::abs((int)sy[y * pw + x] - (int)ry[y * pw + x]);
+    const uint8_t* sc[2] = {Cb.data() + (mby * 8) * (pw / 2) + mbx * 8,
+                            Cr.data() + (mby * 8) * (pw / 2) + mbx * 8};
+    const uint8_t* rc[2] = {rCb.data() + (mby * 8) * (pw / 2) + mbx * 8,
+                            rCr.data() + (mby * 8) * (pw / 2) + mbx * 8};
+    for (int comp = 0; comp < 2; ++comp)
+      for (int y = 0; y < 8; ++y)
+        for (int x = 0; x < 8; ++x)
+          sad += std::abs((int)sc[comp][y * (pw / 2) + x] -
+                          (int)rc[comp][y * (pw / 2) + x]);
+    return sad;
+  }
+
+  // force_idr=0 allows a P frame (P_Skip for unchanged MBs + intra-refresh
+  // MBs for changed regions — zero-MV conditional replenishment, the
+  // low-latency WebRTC P shape). Falls back to IDR when no reference
+  // exists yet.
+  int encode_ex(const uint8_t* rgb, int qp, int force_idr, uint8_t* out,
+                int cap) {
     if (qp < 10) qp = 10;
     if (qp > 48) qp = 48;
+    const bool idr = force_idr || !have_ref;
     const int ns = std::max(1, std::min({n_slices, mbh, 16}));
     // parallel colour conversion by row bands
     if (ns > 1) {
@@ -1396,14 +1439,17 @@ struct Encoder {
     enc_i4modes.assign((size_t)mbw * mbh * 16, -1);
 
     std::vector<uint8_t> bs;
-    {
+    if (idr) {
       uint8_t hdr[256];
       int n = airtc_h264_sps_pps(w, h, hdr, sizeof(hdr));
       if (n <= 0) return -1;
       bs.insert(bs.end(), hdr, hdr + n);
+      frame_num = 0;
     }
     const uint32_t pic_idr_id = idr_id & 1;
-    ++idr_id;
+    if (idr) ++idr_id;
+    const uint32_t fnum = frame_num & 15;
+    frame_num = (frame_num + 1) & 15;
 
     // one slice per MB-row band; slices only predict within themselves,
     // so bands encode concurrently (recon rows + nnz rows are disjoint)
@@ -1415,19 +1461,44 @@ struct Encoder {
       const int slice_start = r0 * mbw;
       BitWriter wtr;
       wtr.ue((uint32_t)slice_start);  // first_mb_in_slice
-      wtr.ue(7);                      // slice_type: I (all slices)
+      wtr.ue(idr ? 7 : 5);            // slice_type: I / P (all slices)
       wtr.ue(0);                      // pps_id
-      wtr.put(0, 4);                  // frame_num (log2_max_frame_num = 4)
-      wtr.ue(pic_idr_id);             // idr_pic_id
-      wtr.put(0, 1);                  // no_output_of_prior_pics_flag
-      wtr.put(0, 1);                  // long_term_reference_flag
+      wtr.put(fnum, 4);               // frame_num (log2_max_frame_num = 4)
+      if (idr) {
+        wtr.ue(pic_idr_id);           // idr_pic_id
+        wtr.put(0, 1);                // no_output_of_prior_pics_flag
+        wtr.put(0, 1);                // long_term_reference_flag
+      } else {
+        wtr.put(0, 1);  // num_ref_idx_active_override_flag
+        wtr.put(0, 1);  // ref_pic_list_modification_flag_l0
+        wtr.put(0, 1);  // adaptive_ref_pic_marking_mode_flag
+      }
       wtr.se(qp - 26);                // slice_qp_delta (pic_init_qp = 26)
       wtr.ue(1);                      // disable_deblocking_filter_idc = off
+      uint32_t skip_run = 0;
       for (int mby = r0; mby < r1; ++mby)
-        for (int mbx = 0; mbx < mbw; ++mbx)
-          encode_mb(wtr, mbx, mby, qp, slice_start);
+        for (int mbx = 0; mbx < mbw; ++mbx) {
+          if (!idr && mb_ref_sad(mbx, mby) <= skip_thresh_for(qp)) {
+            // P_Skip: reconstruction = co-located reference (recon planes
+            // already hold it); zero nnz + not-I4x4 for neighbour context
+            ++skip_run;
+            for (int b = 0; b < 16; ++b)
+              nnz.lnz(mbx, mby, blk_x4(b), blk_y4(b)) = 0;
+            for (int comp = 0; comp < 2; ++comp)
+              for (int blk = 0; blk < 4; ++blk)
+                nnz.cnz(comp, mbx, mby, blk & 1, blk >> 1) = 0;
+            continue;
+          }
+          if (!idr) {
+            wtr.ue(skip_run);  // mb_skip_run before every coded MB
+            skip_run = 0;
+          }
+          encode_mb(wtr, mbx, mby, qp, slice_start, /*in_p_slice=*/!idr);
+        }
+      if (!idr) wtr.ue(skip_run);  // trailing skipped MBs
       wtr.rbsp_trailing();
-      emit_nal(&slice_nals[si], 0x65, wtr.bytes);  // nal_ref_idc=3, IDR
+      // nal_ref_idc=3; type 5 (IDR) or 1 (non-IDR)
+      emit_nal(&slice_nals[si], idr ? 0x65 : 0x61, wtr.bytes);
     };
     if (ns > 1) {
       std::vector<std::thread> ts;
@@ -1440,6 +1511,7 @@ struct Encoder {
 
     if ((int)bs.size() > cap) return -2;
     memcpy(out, bs.data(), bs.size());
+    have_ref = true;
     return (int)bs.size();
   }
 };
@@ -1474,6 +1546,10 @@ struct Decoder {
   // per-4x4 intra pred modes of I_4x4 MBs (-1 = MB not I_4x4): neighbours'
   // modes feed the predicted-mode rule; sized mbw*mbh*16, reset per frame
   std::vector<int8_t> i4modes;
+  // P frames reference the previous reconstruction (rY/rCb/rCr persist
+  // between access units; P_Skip = leave the co-located pixels untouched).
+  // A P slice before any IDR has no reference and is refused.
+  bool have_idr = false;
 
   int parse_sps(BitReader& r) {
     Sps s;
@@ -1709,11 +1785,17 @@ struct Decoder {
                          slice_start, mb_top, mb_left);
   }
 
-  // decode one I_16x16 or I_PCM macroblock; returns 0 or negative error
-  int decode_mb(BitReader& r, int mbx, int mby, int& qp, int slice_start) {
+  // decode one intra macroblock (I_4x4 / I_16x16 / I_PCM; +5 type offset
+  // inside P slices); returns 0 or negative error
+  int decode_mb(BitReader& r, int mbx, int mby, int& qp, int slice_start,
+                bool in_p_slice = false) {
     PlaneCtx rpy{rY.data(), sps.pw};
     uint32_t mb_type = r.ue();
     if (r.overrun) return -1;
+    if (in_p_slice) {
+      if (mb_type < 5) return -2;  // explicit inter MBs unsupported
+      mb_type -= 5;
+    }
     {  // mark this MB as not-I_4x4 for neighbours' mode prediction
       const int mbw = sps.pw / 16;
       int8_t* mm = &i4modes[((size_t)mby * mbw + mbx) * 16];
@@ -1790,6 +1872,7 @@ struct Decoder {
     size_t header_bitpos;
     int first_mb;
     int qp;
+    bool is_p;
   };
 
   int decode_au(const uint8_t* data, int len, uint8_t* rgb, int cap, int* ow,
@@ -1810,13 +1893,20 @@ struct Decoder {
         if (!sps.valid || !pps.valid) return -4;
         int first_mb = (int)r.ue();
         uint32_t stype = r.ue();
-        if (stype % 5 != 2) return -6;  // I slices only
+        const bool is_p = stype % 5 == 0;
+        if (stype % 5 != 2 && !is_p) return -6;  // I and P slices only
+        if (is_p && type == 5) return -6;        // IDR must be intra
+        if (is_p && !have_idr) return -10;       // no reference yet
         r.ue();                         // pps_id
         r.u(sps.log2_max_frame_num);    // frame_num
         if (type == 5) r.ue();          // idr_pic_id
         if (sps.poc_type == 0) {
           r.u(sps.log2_max_poc_lsb);
           if (pps.pic_order_present) r.se();
+        }
+        if (is_p) {
+          if (r.u(1)) r.ue();     // num_ref_idx_active_override -> l0 count
+          if (r.u(1)) return -7;  // ref_pic_list_modification unsupported
         }
         if (type == 5) {
           r.u(1);  // no_output_of_prior_pics
@@ -1837,7 +1927,7 @@ struct Decoder {
         if (qp < 0 || qp > 51 || r.overrun) return -1;
         const int total = (sps.pw / 16) * (sps.ph / 16);
         if (first_mb < 0 || first_mb >= total) return -5;
-        slices.push_back({std::move(rbsp), r.pos, first_mb, qp});
+        slices.push_back({std::move(rbsp), r.pos, first_mb, qp, is_p});
         continue;  // rbsp moved into the job
       }
     }
@@ -1856,7 +1946,19 @@ struct Decoder {
         int mb = job.first_mb;
         int qp = job.qp;
         while (mb < total && sr.more_rbsp_data()) {
-          int rc = decode_mb(sr, mb % mbw, mb / mbw, qp, job.first_mb);
+          if (job.is_p) {
+            // mb_skip_run: P_Skip macroblocks keep the co-located
+            // reference pixels (the recon planes persist across AUs)
+            uint32_t run = sr.ue();
+            if (sr.overrun) break;
+            for (; run > 0 && mb < total; --run) {
+              ++counts[si];
+              ++mb;
+            }
+            if (mb >= total || !sr.more_rbsp_data()) break;
+          }
+          int rc = decode_mb(sr, mb % mbw, mb / mbw, qp, job.first_mb,
+                             job.is_p);
           if (rc < 0) {
             rcs[si] = rc;
             return;
@@ -1879,6 +1981,7 @@ struct Decoder {
         covered += counts[si];
       }
       if (covered < total) return -9;  // frame not fully covered
+      if (!slices[0].is_p) have_idr = true;
       got_frame = true;
     }
     if (!got_frame) return 0;
@@ -1914,6 +2017,12 @@ void airtc_h264enc_destroy(void* h) { delete (h264sw::Encoder*)h; }
 int airtc_h264enc_encode(void* h, const uint8_t* rgb, int qp, uint8_t* out,
                          int cap) {
   return ((h264sw::Encoder*)h)->encode(rgb, qp, out, cap);
+}
+// keyframe-controlled encode: force_idr=0 emits a P frame (P_Skip +
+// intra-refresh) when a reference exists
+int airtc_h264enc_encode_ex(void* h, const uint8_t* rgb, int qp,
+                            int force_idr, uint8_t* out, int cap) {
+  return ((h264sw::Encoder*)h)->encode_ex(rgb, qp, force_idr, out, cap);
 }
 
 void* airtc_h264dec_create() { return new h264sw::Decoder(); }

@@ -492,3 +492,70 @@ def test_i4x4_roundtrip_exercises_decoder_path():
             bytearray(buf), dtype=__import__("torch").uint8).reshape(h, w, 3)
         p = _psnr(frame, out)
         assert p > (34 if qp <= 28 else 28), (qp, p)
+
+
+def test_h264sw_p_frames_sequence():
+    """P frames (P_Skip + intra refresh): static content costs almost
+    nothing, a moving region refreshes only its macroblocks, quality holds
+    across the GOP, and a fresh decoder refuses P before any IDR."""
+    import torch
+
+    from ai_rtc_agent_amd.config import EncoderConfig
+    from ai_rtc_agent_amd.media.codec import CodecUnavailable, H264SwCodec
+
+    try:
+        enc = H264SwCodec(EncoderConfig(), keyframe_interval=100)
+    except CodecUnavailable:
+        pytest.skip("extension not built")
+    dec = H264SwCodec(EncoderConfig())
+    late_joiner = H264SwCodec(EncoderConfig())
+
+    w = h = 160
+    ys, xs = __import__("torch").meshgrid(
+        __import__("torch").arange(h), __import__("torch").arange(w),
+        indexing="ij")
+
+    def frame(t):
+        f = __import__("torch").stack(
+            [40 + xs // 4, 60 + ys // 4,
+             __import__("torch").full_like(xs, 90)], dim=-1)
+        sx = 16 + t * 8
+        f[40:88, sx:sx + 48, 0] = 220
+        f[40:88, sx:sx + 48, 1] = 45
+        return f.clamp(0, 255).to(__import__("torch").uint8).contiguous()
+
+    sizes = []
+    for t in range(8):
+        data = enc.encode(frame(t))
+        sizes.append(len(data))
+        if t == 1:
+            # a decoder joining mid-GOP sees a P frame first: no frame
+            assert late_joiner.decode(data) is None
+        out = dec.decode(data)
+        assert out is not None, t
+        err = (out.float() - frame(t).float()).abs().mean().item()
+        assert err < 6.0, (t, err)
+    assert sizes[0] > 4 * max(sizes[1:]), sizes  # P frames far smaller
+    # a forced keyframe (the PLI path) resyncs the late joiner
+    data = enc.encode(frame(8), keyframe=True)
+    assert len(data) > max(sizes[1:])
+    assert late_joiner.decode(data) is not None
+
+
+def test_h264sw_static_p_frames_are_tiny():
+    from ai_rtc_agent_amd.config import EncoderConfig
+    from ai_rtc_agent_amd.media.codec import CodecUnavailable, H264SwCodec
+
+    try:
+        enc = H264SwCodec(EncoderConfig(), keyframe_interval=1000)
+    except CodecUnavailable:
+        pytest.skip("extension not built")
+    import torch
+
+    f = (__import__("torch").arange(128, dtype=__import__("torch").uint8)
+         .view(1, 128, 1).expand(128, 128, 3)).contiguous()
+    first = enc.encode(f)
+    for _ in range(3):
+        p = enc.encode(f)
+    assert len(p) < 64, len(p)      # all-skip P frame
+    assert len(first) > 20 * len(p)
