@@ -561,3 +561,37 @@ def test_h264sw_static_p_frames_are_tiny():
         p = enc.encode(f)
     assert len(p) < 64, len(p)      # all-skip P frame
     assert len(first) > 3 * len(p)  # (smooth ramp: the IDR itself is tiny)
+
+
+def test_h264sw_decoder_fuzz_robustness():
+    """Bit-flipped / truncated / garbage streams must never crash the
+    native decoder — worst case is a refused AU or a garbage frame."""
+    import torch
+
+    ext = _h264_ext()
+    enc = ext.H264SwEncoder(64, 64, mb_mode=2)
+    g = torch.Generator().manual_seed(1)
+    frame = torch.randint(0, 255, (64, 64, 3), generator=g, dtype=torch.uint8)
+    base_i = enc.encode(frame.numpy().tobytes(), 28, keyframe=True)
+    base_p = enc.encode(frame.numpy().tobytes(), 28, keyframe=False)
+    rng = random.Random(7)
+    dec = ext.H264SwDecoder()
+    for trial in range(400):
+        data = bytearray(base_i if trial % 2 == 0 else base_p)
+        kind = trial % 4
+        if kind == 0:  # single bit flip
+            i = rng.randrange(len(data))
+            data[i] ^= 1 << rng.randrange(8)
+        elif kind == 1:  # truncate
+            data = data[: rng.randrange(1, len(data))]
+        elif kind == 2:  # burst corruption
+            i = rng.randrange(len(data))
+            for j in range(i, min(len(data), i + 16)):
+                data[j] = rng.randrange(256)
+        else:  # pure garbage with a start code
+            data = bytearray(b"\x00\x00\x00\x01") + bytearray(
+                rng.randrange(256) for _ in range(rng.randrange(4, 200)))
+        dec.decode(bytes(data))  # any outcome but a crash is fine
+    # the decoder still works after the abuse
+    fresh = ext.H264SwDecoder()
+    assert fresh.decode(base_i) is not None
