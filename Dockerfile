@@ -12,6 +12,14 @@ RUN PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
 
 FROM rocm/pytorch:latest
 
+# VCN hardware codec userspace (VA-API): enables the experimental hardware
+# encode session (AIRTC_VCN_EXPERIMENTAL=1; docs/webrtc.md). The software
+# H.264 codec needs nothing beyond the base image (OpenSSL ships in it for
+# the DTLS-SRTP endpoint).
+RUN apt-get update && apt-get install -y --no-install-recommends \
+        libva2 libva-drm2 mesa-va-drivers && \
+    rm -rf /var/lib/apt/lists/* || true
+
 WORKDIR /app
 COPY --from=build /app/ai_rtc_agent_amd ./ai_rtc_agent_amd
 COPY agent_main.py bench.py build.py download.py ./
@@ -22,7 +30,10 @@ ENV HF_HOME=/models/hf \
     ENGINES_CACHE=/models/engines \
     CIVITAI_CACHE=/models/civitai
 
-# hardware codec toggles (reference Dockerfile:54-56 NVENC/NVDEC)
+# hardware codec toggles (reference Dockerfile:54-56 NVENC/NVDEC).
+# NOTE: the VCN session additionally requires the explicit
+# AIRTC_VCN_EXPERIMENTAL=1 opt-in until hardware-validated; the standard
+# software H.264 codec is the default either way (media/codec.py).
 ENV VCN_ENC=true \
     VCN_DEC=true
 
