@@ -1,20 +1,19 @@
-"""Video codec HAL — VCN hardware H.264 with a software fallback.
+"""Video codec HAL — three tiers, selected per box (select_codec).
 
 Replaces reference components N1/N2: the aiortc fork's
 NVDEC/NVENC H.264 codecs (reference README.md:14-15, env NVENC/NVDEC in
-Dockerfile:54-56). The MI355X path targets the VCN video blocks; this module
-is the selection layer:
+Dockerfile:54-56).
 
-- VcnH264Codec: dlopen-probes the VA-API / rocDecode runtime at import. The
-  build environment ships NO libva/rocDecode (SURVEY.md §7 env note), so the
-  probe fails gracefully here and hardware validation is deferred to a
-  target box that has the VCN userspace stack. Decoded surfaces are HIP
-  device pointers (zero host copies) when active.
-- SoftwareCodec: the no-GPU fallback, mirroring the reference's SW path
-  (lib/pipeline.py:83-94 runs when NVENC unset). PyAV/x264 are also absent
-  offline, so this is a self-contained codec ("RAWZ"): zlib-compressed
-  I-frames plus delta-encoded P-frames with a periodic keyframe interval —
-  real (lossless) compression, good enough for loopback tests and LAN use.
+1. VcnH264Codec — MI355X VCN hardware encode via the VA-API session layer
+   (ops/csrc/vcn.cpp). Hardware-unvalidated this round (no libva anywhere
+   reachable), so it requires the AIRTC_VCN_EXPERIMENTAL=1 opt-in on top
+   of a successful probe; decode is probe-only.
+2. H264SwCodec — the DEFAULT: standard Annex-B H.264 in native C++
+   (ops/csrc/h264sw.cpp): IDR with per-MB I_16x16/I_4x4 mode decision,
+   P frames (P_Skip + intra refresh) between keyframes, multi-slice
+   threading, QP rate control, GOP cadence with PLI-forced IDR.
+3. SoftwareCodec — last-resort "RAWZ" zlib I/P codec for pure-python
+   environments without the built extension (self-interop only).
 
 Encoder knobs (preset/bitrates) mirror the reference's 5 NVENC_* env vars
 (docs/environment.md:17-25) via config.EncoderConfig.
@@ -22,7 +21,6 @@ Encoder knobs (preset/bitrates) mirror the reference's 5 NVENC_* env vars
 from __future__ import annotations
 
 import ctypes
-import ctypes.util
 import struct
 import zlib
 from typing import Optional

@@ -75,7 +75,7 @@ __global__ __launch_bounds__(256) void attention_kernel(
     const f16* __restrict__ q, const f16* __restrict__ k,
     const f16* __restrict__ v, f16* __restrict__ out, int H, int Lq, int Lk,
     long q_sb, long q_sh, long q_row, long k_sb, long k_sh, long k_row,
-    long o_sb, long o_sh, long o_row, float scale) {
+    long o_sb, long o_sh, long o_row, float scale, int n_qt, int xcd_map) {
   constexpr int KPITCH = D + 8;
   constexpr int D8 = D / 8, D32 = D / 32, D16 = D / 16;
   constexpr int QTILE = 64 * QF;  // q rows per workgroup
@@ -83,9 +83,25 @@ __global__ __launch_bounds__(256) void attention_kernel(
   __shared__ f16 ldsV[KVT * KPITCH];  // row-major like K; PV reads via tr_b16
   __shared__ f16 ldsP[4 * QF * 16 * PPITCH];
 
-  const int bh = blockIdx.y;
+  // XCD-chunked mapping (env AIRTC_ATTN_XCD): each XCD walks a CONTIGUOUS
+  // (head, q-tile) range, so one XCD's in-flight blocks share 1-2 heads'
+  // K/V (~1-2 MB) and re-reads hit the 4 MB per-XCD L2 instead of the LLC.
+  // The default round-robin dispatch interleaves ~8 heads per XCD (8 MB of
+  // K/V) and thrashes it.
+  int bh, qtile;
+  if (xcd_map) {
+    const int T = gridDim.x;  // 1-D launch
+    const int x = blockIdx.x & 7, pos = blockIdx.x >> 3;
+    const int q8 = T >> 3, r = T & 7;
+    const int pair = (x < r ? x * (q8 + 1) : r * (q8 + 1) + (x - r) * q8) + pos;
+    bh = pair / n_qt;
+    qtile = pair - bh * n_qt;
+  } else {
+    bh = blockIdx.y;
+    qtile = blockIdx.x;
+  }
   const int b = bh / H, h = bh % H;
-  const int q0 = blockIdx.x * QTILE;
+  const int q0 = qtile * QTILE;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
@@ -272,9 +288,16 @@ extern "C" void airtc_attention(const uint16_t* q, const uint16_t* k,
   // the doubled accumulator/S-fragment footprint costs more occupancy than
   // the halved K/V traffic saves (K/V is LLC-resident at SD sizes). QF=1
   // stays the default; the template is kept for larger-context models.
+  static int xcd_map_env = -1;
+  if (xcd_map_env < 0) {
+    const char* e = getenv("AIRTC_ATTN_XCD");
+    xcd_map_env = e ? atoi(e) : 0;
+  }
   int qf = 1;
   if (qf_force > 0) qf = qf_force;
-  dim3 grid(ceil_div(Lq, 64 * qf), B * H);
+  const int n_qt = ceil_div(Lq, 64 * qf);
+  const int xm = xcd_map_env ? 1 : 0;
+  dim3 grid = xm ? dim3(n_qt * B * H) : dim3(n_qt, B * H);
   const f16* qp = reinterpret_cast<const f16*>(q);
   const f16* kp = reinterpret_cast<const f16*>(k);
   const f16* vp = reinterpret_cast<const f16*>(v);
@@ -282,7 +305,7 @@ extern "C" void airtc_attention(const uint16_t* q, const uint16_t* k,
 #define LAUNCH(D, QF)                                                       \
   hipLaunchKernelGGL((attention_kernel<D, QF>), grid, dim3(256), 0, s, qp,  \
                      kp, vp, op, H, Lq, Lk, q_sb, q_sh, q_row, k_sb, k_sh,  \
-                     k_row, o_sb, o_sh, o_row, scale)
+                     k_row, o_sb, o_sh, o_row, scale, n_qt, xm)
   switch (d * 10 + qf) {
     case 321: LAUNCH(32, 1); break;
     case 322: LAUNCH(32, 2); break;
