@@ -72,21 +72,34 @@ class StreamDiffusionEngine:
         self.vae = vae if vae is not None else TinyVAE()
         # family conventions: SD1.5 = CLIP ViT-L/14 (quick-gelu, last
         # layer); SD2.x = OpenCLIP ViT-H (gelu, PENULTIMATE layer, 23
-        # transformer blocks); sdxl keeps a single encoder stand-in with
-        # the pooled text_projection the addition-embed path consumes
-        if cfg.model_family == "tiny":
-            te_layers, te_act, te_skip = 2, "quick_gelu", 0
-        elif cfg.model_family == "sd21":
-            te_layers, te_act, te_skip = 23, "gelu", 1
+        # blocks); sdxl = the DUAL encoder (ViT-L + OpenCLIP bigG, both
+        # penultimate, per-token concat to 2048, pooled from bigG)
+        if text_encoder is not None:
+            self.text_encoder = text_encoder
+        elif ucfg.addition_embed_dim:
+            from ..models.text_encoder import DualTextEncoder
+
+            if cfg.model_family == "sdxl":
+                self.text_encoder = DualTextEncoder()
+            else:  # tiny_xl: small dual with the same contract
+                half = ucfg.cross_attention_dim // 2
+                self.text_encoder = DualTextEncoder(
+                    hidden1=half, layers1=2,
+                    hidden2=ucfg.cross_attention_dim - half, layers2=2,
+                    vocab_size=512)
         else:
-            te_layers, te_act, te_skip = 12, "quick_gelu", 0
-        self.text_encoder = text_encoder if text_encoder is not None else TextEncoder(
-            hidden=ucfg.cross_attention_dim,
-            layers=te_layers,
-            act=te_act,
-            clip_skip=te_skip,
-            pooled_dim=1280 if ucfg.addition_embed_dim else None,
-        )
+            if cfg.model_family == "tiny":
+                te_layers, te_act, te_skip = 2, "quick_gelu", 0
+            elif cfg.model_family == "sd21":
+                te_layers, te_act, te_skip = 23, "gelu", 1
+            else:
+                te_layers, te_act, te_skip = 12, "quick_gelu", 0
+            self.text_encoder = TextEncoder(
+                hidden=ucfg.cross_attention_dim,
+                layers=te_layers,
+                act=te_act,
+                clip_skip=te_skip,
+            )
         self.ctx_dim = ucfg.cross_attention_dim
         self.addition_embed_dim = ucfg.addition_embed_dim
 

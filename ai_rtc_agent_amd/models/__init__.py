@@ -1,6 +1,6 @@
 from .unet import UNetConfig, UNet2DCondition
 from .taesd import TAESDEncoder, TAESDDecoder, TinyVAE
-from .text_encoder import TextEncoder
+from .text_encoder import DualTextEncoder, TextEncoder
 
 __all__ = [
     "UNetConfig",
@@ -8,5 +8,6 @@ __all__ = [
     "TAESDEncoder",
     "TAESDDecoder",
     "TinyVAE",
+    "DualTextEncoder",
     "TextEncoder",
 ]

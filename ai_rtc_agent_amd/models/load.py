@@ -378,22 +378,26 @@ def load_model_dir(engine, model_dir: str) -> bool:
     if found and n_enc == 0 and n_dec == 0:
         log.warning("model dir %s: UNet loaded but no TAESD weights matched — "
                     "VAE stays random-init", model_dir)
-    # CLIP text encoder + its BPE tokenizer (reference lib/wrapper.py:468-473)
+    # CLIP text encoder(s) + BPE tokenizer(s)
+    # (reference lib/wrapper.py:468-473; sdxl adds text_encoder_2)
     te = getattr(engine, "text_encoder", None)
     if te is not None:
-        for sub in ("text_encoder/model.safetensors",
-                    "text_encoder/pytorch_model.safetensors"):
-            p = os.path.join(model_dir, sub)
-            if os.path.exists(p):
-                try:
-                    n = load_clip_text_encoder(te, load_file(p))
-                    if n:
-                        found = True
-                    else:
-                        log.warning("CLIP checkpoint %s matched 0 tensors", p)
-                except ValueError as e:
-                    log.warning("CLIP checkpoint %s skipped: %s", p, e)
-                break
+        targets = [(te, "text_encoder")]
+        if hasattr(te, "enc1") and hasattr(te, "enc2"):  # sdxl dual
+            targets = [(te.enc1, "text_encoder"), (te.enc2, "text_encoder_2")]
+        for mod, sub in targets:
+            for fname in ("model.safetensors", "pytorch_model.safetensors"):
+                p = os.path.join(model_dir, sub, fname)
+                if os.path.exists(p):
+                    try:
+                        n = load_clip_text_encoder(mod, load_file(p))
+                        if n:
+                            found = True
+                        else:
+                            log.warning("CLIP checkpoint %s matched 0 tensors", p)
+                    except ValueError as e:
+                        log.warning("CLIP checkpoint %s skipped: %s", p, e)
+                    break
         if hasattr(te, "load_tokenizer_dir") and te.load_tokenizer_dir(model_dir):
             log.info("BPE tokenizer loaded from %s", model_dir)
     return found

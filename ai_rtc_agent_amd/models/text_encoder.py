@@ -288,3 +288,54 @@ class TextEncoder(nn.Module):
         if self.pooled_proj is not None:
             p = self.pooled_proj(p.to(self.pooled_proj.weight.dtype))
         return p.to(dtype)
+
+
+class DualTextEncoder(nn.Module):
+    """SDXL text conditioning: CLIP ViT-L/14 (768-d, quick-gelu) and
+    OpenCLIP ViT-bigG (1280-d, gelu, text_projection for the pooled
+    vector), both read at the PENULTIMATE layer per the sdxl convention;
+    per-token features concatenate to the UNet's 2048-d context
+    (reference: diffusers loads text_encoder/ + text_encoder_2/ for
+    stabilityai/sdxl-turbo; the wrapper consumes the same two encoders).
+
+    Exposes the same encode()/pooled() surface as TextEncoder, so the
+    engine treats both interchangeably."""
+
+    def __init__(
+        self,
+        hidden1: int = 768,
+        layers1: int = 12,
+        hidden2: int = 1280,
+        layers2: int = 32,
+        vocab_size: int = 49408,
+        max_length: int = 77,
+        seed: int = 0,
+        pooled_dim: int = 1280,
+    ):
+        super().__init__()
+        self.enc1 = TextEncoder(hidden=hidden1, layers=layers1, act="quick_gelu",
+                                clip_skip=1, vocab_size=vocab_size,
+                                max_length=max_length, seed=seed)
+        self.enc2 = TextEncoder(hidden=hidden2, layers=layers2, act="gelu",
+                                clip_skip=1, vocab_size=vocab_size,
+                                max_length=max_length, seed=seed + 1,
+                                pooled_dim=pooled_dim)
+        self.hidden = hidden1 + hidden2
+
+    def load_tokenizer_dir(self, path: str) -> bool:
+        """sdxl snapshots ship tokenizer/ (ViT-L) and tokenizer_2 (bigG)."""
+        ok1 = self.enc1.load_tokenizer_dir(os.path.join(path, "tokenizer")) \
+            or self.enc1.load_tokenizer_dir(path)
+        ok2 = self.enc2.load_tokenizer_dir(os.path.join(path, "tokenizer_2")) \
+            or self.enc2.load_tokenizer_dir(path)
+        return ok1 or ok2
+
+    @torch.no_grad()
+    def encode(self, prompt: str, device=None, dtype=torch.float32) -> torch.Tensor:
+        e1 = self.enc1.encode(prompt, device=device, dtype=dtype)
+        e2 = self.enc2.encode(prompt, device=device, dtype=dtype)
+        return torch.cat([e1, e2], dim=-1)
+
+    @torch.no_grad()
+    def pooled(self, prompt: str, device=None, dtype=torch.float32) -> torch.Tensor:
+        return self.enc2.pooled(prompt, device=device, dtype=dtype)

@@ -300,3 +300,58 @@ def test_load_model_dir_wires_clip_and_tokenizer(tmp_path):
     assert load_model_dir(eng, str(tmp_path))
     assert not torch.allclose(before, eng.text_encoder.token_emb.weight)
     assert isinstance(eng.text_encoder.tokenizer, ClipBpeTokenizer)
+
+
+def test_sdxl_dual_text_encoder_contract():
+    """sdxl uses two CLIP encoders (ViT-L + bigG): per-token concat context,
+    pooled vector from encoder 2's text_projection."""
+    from ai_rtc_agent_amd.models.text_encoder import DualTextEncoder
+
+    te = DualTextEncoder(hidden1=32, layers1=2, hidden2=48, layers2=2,
+                         vocab_size=128, pooled_dim=64)
+    with torch.no_grad():
+        e = te.encode("a fast car")
+        p = te.pooled("a fast car")
+    assert e.shape == (1, 77, 80)
+    assert p.shape == (1, 64)
+
+
+def test_load_model_dir_loads_both_sdxl_encoders(tmp_path):
+    from safetensors.torch import save_file
+
+    from ai_rtc_agent_amd.models.load import load_model_dir
+    from ai_rtc_agent_amd.models.taesd import TinyVAE
+    from ai_rtc_agent_amd.models.text_encoder import DualTextEncoder
+
+    class _Eng:
+        pass
+
+    eng = _Eng()
+    eng.unet = UNet2DCondition(UNetConfig.tiny())
+    eng.vae = TinyVAE(width=16)
+    eng.text_encoder = DualTextEncoder(hidden1=32, layers1=2, hidden2=48,
+                                       layers2=2, vocab_size=128, pooled_dim=64)
+    for sub, mod in (("text_encoder", eng.text_encoder.enc1),
+                     ("text_encoder_2", eng.text_encoder.enc2)):
+        (tmp_path / sub).mkdir()
+        save_file(_synth_clip_sd(mod), str(tmp_path / sub / "model.safetensors"))
+    b1 = eng.text_encoder.enc1.token_emb.weight.clone()
+    b2 = eng.text_encoder.enc2.token_emb.weight.clone()
+    assert load_model_dir(eng, str(tmp_path))
+    assert not torch.allclose(b1, eng.text_encoder.enc1.token_emb.weight)
+    assert not torch.allclose(b2, eng.text_encoder.enc2.token_emb.weight)
+
+
+def test_tiny_xl_engine_uses_dual_encoder():
+    from ai_rtc_agent_amd.config import EngineConfig
+    from ai_rtc_agent_amd.engine import StreamDiffusionEngine
+    from ai_rtc_agent_amd.models.text_encoder import DualTextEncoder
+
+    cfg = EngineConfig(model_family="tiny_xl", width=64, height=64,
+                       device="cpu", use_hip_graph=False)
+    eng = StreamDiffusionEngine(cfg)
+    assert isinstance(eng.text_encoder, DualTextEncoder)
+    eng.prepare()
+    f = torch.randint(0, 255, (64, 64, 3), dtype=torch.uint8)
+    out = eng(f)
+    assert out.shape == (64, 64, 3)
