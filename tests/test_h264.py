@@ -595,3 +595,33 @@ def test_h264sw_decoder_fuzz_robustness():
     # the decoder still works after the abuse
     fresh = ext.H264SwDecoder()
     assert fresh.decode(base_i) is not None
+
+
+def test_cavlc_coeff_token_kraft_analysis():
+    """Structural check on the coeff_token tables: prefix-free (enforced by
+    h264sw_table_check) AND near-complete Kraft sums, with the only free
+    slots being the all-zeros codewords the spec deliberately leaves
+    unused (long zero runs would risk start-code emulation). Catches
+    future table edits that silently break coverage."""
+    import re
+    from fractions import Fraction
+
+    src = open("ai_rtc_agent_amd/ops/csrc/h264sw.cpp").read()
+    src = re.sub(r"//[^\n]*", "", src)
+
+    def grab(name):
+        m = re.search(name + r"\[3\]\[17\]\[4\] = \{(.*?)\};", src, re.S)
+        nums = [int(x) for x in re.findall(r"\d+", m.group(1))]
+        assert len(nums) == 3 * 17 * 4
+        return [[nums[b * 68 + c * 4:b * 68 + c * 4 + 4] for c in range(17)]
+                for b in range(3)]
+
+    L, B = grab("CT_LEN"), grab("CT_BITS")
+    expect_kraft = {0: Fraction(32767, 32768), 1: Fraction(16381, 16384),
+                    2: Fraction(511, 512)}
+    for b in range(3):
+        codes = [(L[b][c][t], B[b][c][t]) for c in range(17) for t in range(4)
+                 if L[b][c][t]]
+        s = sum(Fraction(1, 2 ** l) for l, _ in codes)
+        assert s == expect_kraft[b], (b, s)
+        assert len(set(codes)) == len(codes), f"duplicate codes in bucket {b}"
