@@ -39,7 +39,10 @@ def init_distributed(backend: str | None = None) -> tuple[int, int, int]:
         if backend is None:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
         if torch.cuda.is_available():
-            torch.cuda.set_device(local)
+            # modulo: oversubscribed launches (more ranks than GPUs, e.g.
+            # the 2-rank RCCL validation on a 1-GPU box) still place every
+            # rank on a real device
+            torch.cuda.set_device(local % torch.cuda.device_count())
         dist.init_process_group(backend=backend)
     return rank, dist.get_world_size(), local
 

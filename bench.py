@@ -108,7 +108,8 @@ def main() -> None:
         args.e2e_frames = min(args.e2e_frames, 2)
     if args.fbs > 1:
         args.e2e_frames = 0  # wire path is per-stream; measured at fbs=1
-    device = f"cuda:{local}" if use_cuda else "cpu"
+    device = (f"cuda:{local % torch.cuda.device_count()}" if use_cuda
+              else "cpu")
 
     graph = not args.no_graph and use_cuda
     if args.model == "sd-turbo":
