@@ -182,16 +182,23 @@ def test_lora_hot_swap_recaptures_graph():
 
 
 @pytest.mark.timeout(280)
-def test_rccl_two_ranks_one_gpu():
-    """REAL RCCL on hardware: two torchrun ranks share cuda:0 (modulo
-    device mapping), broadcast engine weights over RCCL and all-reduce the
-    bench timings — the collective path the 8-GPU scale-out uses, minus
-    the extra GPUs (round-1 verdict: RCCL was unmeasured on a real GPU)."""
+def test_rccl_two_ranks():
+    """REAL RCCL on hardware: two torchrun ranks broadcast engine weights
+    over RCCL and all-reduce the bench timings — the collective path the
+    8-GPU scale-out uses. RCCL requires one DISTINCT device per
+    communicator rank (two ranks on one GPU is "invalid usage" by
+    design), so this runs only on multi-GPU boxes; the single-GPU pool
+    covers the logic with 2-rank gloo tests instead."""
     import json
     import os
     import socket
     import subprocess
     import sys
+
+    import torch
+
+    if torch.cuda.device_count() < 2:
+        pytest.skip("RCCL needs a distinct device per rank")
 
     s = socket.socket()
     s.bind(("127.0.0.1", 0))
