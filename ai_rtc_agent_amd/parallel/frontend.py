@@ -85,6 +85,8 @@ class WorkerFrontend:
         self._load = [0] * n_workers
         self._publisher_worker = 0
         self._http = None
+        self._monitor_task = None
+        self._closing = False
 
     # -- lifecycle -------------------------------------------------------
     def spawn(self) -> None:
@@ -122,12 +124,47 @@ class WorkerFrontend:
                     await asyncio.sleep(0.2)
 
     def shutdown(self) -> None:
+        self._closing = True
+        if self._monitor_task is not None:
+            self._monitor_task.cancel()
+            self._monitor_task = None
         for p in self.procs:
             if p.is_alive():
                 p.terminate()
         for p in self.procs:
             p.join(timeout=10)
         self.procs.clear()
+
+    # -- elastic recovery -------------------------------
+    def _respawn(self, rank: int) -> None:
+        ctx = mp.get_context("spawn")
+        wp = self.udp_ports[rank::self.n] if self.udp_ports else None
+        p = ctx.Process(
+            target=_worker_main,
+            args=(rank, self.ports[rank], self.model_id, self.family,
+                  self.resolution, self.pin_gpu, wp),
+            daemon=True,
+        )
+        p.start()
+        self.procs[rank] = p
+        self._load[rank] = 0  # its sessions died with it
+        logger.warning("worker %d died; respawned as pid %d", rank, p.pid)
+
+    async def _monitor(self, interval: float = 2.0) -> None:
+        """Respawn crashed worker processes (a GPU fault or OOM in one
+        worker must not take down the other GPUs' media planes)."""
+        try:
+            while not self._closing:
+                await asyncio.sleep(interval)
+                for rank, p in enumerate(self.procs):
+                    if not p.is_alive():
+                        self._respawn(rank)
+        except asyncio.CancelledError:
+            pass
+
+    def start_monitor(self) -> None:
+        if self._monitor_task is None:
+            self._monitor_task = asyncio.ensure_future(self._monitor())
 
     # -- proxying --------------------------------------------------------
     def _pick_worker(self) -> int:
@@ -214,11 +251,18 @@ class WorkerFrontend:
                                   "load": list(self._load)}
             return web.json_response(out)
 
+        async def on_startup(app):
+            self.start_monitor()
+
         async def on_shutdown(app):
+            if self._monitor_task is not None:
+                self._monitor_task.cancel()
+                self._monitor_task = None
             if self._http is not None and not self._http.closed:
                 await self._http.close()
 
         app = web.Application()
+        app.on_startup.append(on_startup)
         app.router.add_post("/whip", publish)
         app.router.add_post("/offer", publish)
         app.router.add_delete("/whip", to_publisher)

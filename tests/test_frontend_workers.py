@@ -131,3 +131,43 @@ def test_two_worker_processes_serve_independent_sessions(monkeypatch):
         loop.run_until_complete(asyncio.wait_for(body(), 280))
     finally:
         loop.close()
+
+
+@pytest.mark.timeout(300)
+def test_worker_crash_respawns(monkeypatch):
+    """Elastic recovery: a killed worker process comes
+    back and the front-end keeps serving."""
+    from ai_rtc_agent_amd.parallel.frontend import WorkerFrontend
+
+    async def body():
+        from aiohttp.test_utils import TestClient, TestServer
+
+        fe = WorkerFrontend(2, family="tiny", resolution=64, pin_gpu=False)
+        fe.spawn()
+        try:
+            await fe.wait_ready(timeout=240)
+            http = TestClient(TestServer(fe.create_app()))
+            await http.start_server()  # starts the monitor
+            victim = fe.procs[1]
+            victim.terminate()
+            victim.join(timeout=10)
+            assert not victim.is_alive()
+            # the monitor notices within ~2s and respawns
+            for _ in range(120):
+                await asyncio.sleep(0.5)
+                if fe.procs[1] is not victim and fe.procs[1].is_alive():
+                    break
+            assert fe.procs[1] is not victim and fe.procs[1].is_alive()
+            # the respawned worker serves again
+            await fe.wait_ready(timeout=240)
+            r = await http.get("/")
+            assert r.status == 200
+            await http.close()
+        finally:
+            fe.shutdown()
+
+    loop = asyncio.new_event_loop()
+    try:
+        loop.run_until_complete(asyncio.wait_for(body(), 280))
+    finally:
+        loop.close()
