@@ -215,6 +215,40 @@ async def stats(request: web.Request) -> web.Response:
     return web.json_response(st["pool"].stats())
 
 
+async def metrics(request: web.Request) -> web.Response:
+    """Prometheus exposition of the /stats content (the
+    reference has no metrics endpoint at all; /stats is the JSON form,
+    this is the scrapeable one)."""
+    st = _state(request.app)
+    pool = st["pool"].stats()
+    lines = [
+        "# TYPE airtc_replicas gauge",
+        f"airtc_replicas {pool.get('replicas', 0)}",
+        "# TYPE airtc_sessions gauge",
+        f"airtc_sessions {len(pool.get('sessions', {}))}",
+        "# TYPE airtc_peer_connections gauge",
+        f"airtc_peer_connections {len(st.get('pcs', ()))}",
+        "# TYPE airtc_frames_total counter",
+        "# TYPE airtc_fps gauge",
+        "# TYPE airtc_stage_ms gauge",
+    ]
+    for i, rep in enumerate(pool.get("per_replica", [])):
+        lines.append(f'airtc_frames_total{{replica="{i}"}} {rep.get("frames", 0)}')
+        lines.append(f'airtc_fps{{replica="{i}"}} {rep.get("fps", 0.0)}')
+        for stage, v in rep.get("stages_ms", {}).items():
+            for q in ("p50", "p90"):
+                lines.append(
+                    f'airtc_stage_ms{{replica="{i}",stage="{stage}",q="{q}"}} '
+                    f'{v.get(q, 0.0)}')
+        for si, slot in enumerate(rep.get("per_stream", [])):
+            if slot.get("p50_ms") is not None:
+                lines.append(
+                    f'airtc_stream_latency_ms{{replica="{i}",slot="{si}"}} '
+                    f'{slot["p50_ms"]}')
+    return web.Response(text="\n".join(lines) + "\n",
+                        content_type="text/plain")
+
+
 async def on_startup(app: web.Application) -> None:
     st = _state(app)
     if app["udp_ports"]:
@@ -296,6 +330,7 @@ def create_app(
     app.router.add_post("/config", update_config)
     app.router.add_get("/", health)
     app.router.add_get("/stats", stats)
+    app.router.add_get("/metrics", metrics)
     app.on_startup.append(on_startup)
     app.on_shutdown.append(on_shutdown)
     return app

@@ -143,3 +143,17 @@ def test_pool_assignment():
     pool.release("s1")
     p3 = pool.assign("s3")
     assert p3 is p1, "freed replica is reused"
+
+
+def test_metrics_endpoint():
+    async def body():
+        client = await make_client(StubPipeline())
+        r = await client.get("/metrics")
+        assert r.status == 200
+        text = await r.text()
+        assert "airtc_replicas 1" in text
+        assert "airtc_frames_total" in text
+        assert "airtc_fps" in text
+        await client.close()
+
+    run(body())
