@@ -299,3 +299,20 @@ def test_peerconnection_serves_media_via_relay(monkeypatch):
             t.close()
 
     run(body(), timeout=50)
+
+
+def test_sans_io_stale_nonce_retry():
+    """438 (stale nonce) on Allocate refreshes the nonce and retries."""
+    cli = TurnClient(USER, PASS)
+    cli.realm = REALM
+    cli.nonce = b"old-nonce-0000"
+    first = cli.allocate_request()
+    req = StunMessage.parse(first)
+    err = StunMessage(_mtype(M_ALLOCATE, CLASS_ERROR), req.transaction_id)
+    err.attributes[ATTR_ERROR_CODE] = b"\x00\x00\x04\x26"  # 438
+    err.attributes[ATTR_NONCE] = b"fresh-nonce-111"
+    events, out = cli.feed(err.serialize())
+    assert not events and len(out) == 1
+    retry = StunMessage.parse(out[0])
+    assert retry.attributes[ATTR_NONCE] == b"fresh-nonce-111"
+    assert cli.nonce == b"fresh-nonce-111"
