@@ -53,13 +53,18 @@ class RtpPacket:
                 raise ValueError("short RTP extension")
             ext_len = struct.unpack("!H", data[offset + 2 : offset + 4])[0]
             offset += 4 + 4 * ext_len
+        end = len(data)
+        if b0 & 0x20 and end > offset:  # padding: last byte = pad count
+            pad = data[-1]
+            if 0 < pad <= end - offset:
+                end -= pad
         return RtpPacket(
             payload_type=b1 & 0x7F,
             sequence_number=seq,
             timestamp=ts,
             ssrc=ssrc,
             marker=b1 >> 7,
-            payload=data[offset:],
+            payload=data[offset:end],
         )
 
 

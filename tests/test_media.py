@@ -368,3 +368,19 @@ def test_au_reassembly_across_seq_wrap():
         assert nal in dec.frames[0]
 
     asyncio.get_event_loop_policy().new_event_loop().run_until_complete(body())
+
+
+def test_rtp_parse_csrc_extension_padding():
+    """Browser RTP arrives with CSRC entries, header extensions
+    (abs-send-time etc.) and padding — the parser must locate the payload
+    exactly (RFC 3550)."""
+    import struct as _s
+
+    payload = b"NALDATA"
+    hdr = _s.pack("!BBHII", (2 << 6) | 0x20 | 0x10 | 1, 97, 7, 9000, 5)
+    csrc = _s.pack("!I", 42)
+    ext = _s.pack("!HH", 0xBEDE, 1) + b"\x10\x01\x02\x03"  # one ext word
+    padded = payload + b"\x00\x00\x03"  # 3 pad bytes, count in last byte
+    pkt = RtpPacket.parse(hdr + csrc + ext + padded)
+    assert pkt.payload == payload
+    assert pkt.payload_type == 97 and pkt.sequence_number == 7
