@@ -103,6 +103,7 @@ class PeerConnection:
         self._decode_misses = 0
         self._last_pli = 0.0
         # loss accounting (receiver) + adaptive bitrate (sender)
+        self._rx_lock_ssrc: Optional[int] = None
         self._rx_count = 0
         self._rx_base_seq: Optional[int] = None
         self._rx_high_seq = 0
@@ -402,6 +403,13 @@ class PeerConnection:
         self._remote_addr = addr
         if self.connection_state == "connecting":
             self._set_state("connected")
+        # lock onto the first media SSRC: browsers may carry rtx or
+        # simulcast layers on other SSRCs, which must not interleave into
+        # this stream's access-unit reassembly
+        if self._rx_lock_ssrc is None:
+            self._rx_lock_ssrc = pkt.ssrc
+        elif pkt.ssrc != self._rx_lock_ssrc:
+            return
         self._account_rx(pkt)
         if getattr(self._decoder, "rtp_mode", "raw") == "rfc6184":
             # buffer the access unit's packets per timestamp and reassemble

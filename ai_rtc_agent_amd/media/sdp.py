@@ -54,6 +54,7 @@ class SessionDescription:
     session_id: str = ""
     media: List[MediaSection] = field(default_factory=list)
     fingerprint: Optional[str] = None  # session-level a=fingerprint
+    bundle: Optional[str] = None       # a=group:BUNDLE mids (echoed back)
 
     @staticmethod
     def parse(sdp: str) -> "SessionDescription":
@@ -70,6 +71,8 @@ class SessionDescription:
                     sd.session_id = parts[1]
             elif line.startswith("a=fingerprint:") and cur is None:
                 sd.fingerprint = line.split(":", 1)[1].strip()
+            elif line.startswith("a=group:BUNDLE") and cur is None:
+                sd.bundle = line[len("a=group:"):].strip()
             elif line.startswith("m="):
                 parts = line[2:].split()
                 cur = MediaSection(kind=parts[0], port=int(parts[1]), protocol=parts[2])
@@ -125,6 +128,8 @@ class SessionDescription:
             "s=-",
             "t=0 0",
         ]
+        if self.bundle:
+            lines.append(f"a=group:{self.bundle}")
         for m in self.media:
             pts = " ".join(str(c.payload_type) for c in m.codecs)
             lines.append(f"m={m.kind} {m.port} {m.protocol} {pts}")
@@ -136,6 +141,9 @@ class SessionDescription:
             lines.append(f"a=mid:{m.mid}")
             lines.append(f"a={m.direction}")
             lines.append(f"a=setup:{m.setup}")
+            # RTP and RTCP share the one media socket (RFC 5761); browsers
+            # offer rtcp-mux and expect the answer to accept it
+            lines.append("a=rtcp-mux")
             if m.fingerprint:
                 lines.append(f"a=fingerprint:{m.fingerprint}")
             for c in m.codecs:
@@ -169,7 +177,8 @@ def build_answer(
     """Answer an offer: echo media sections, pick our codec, attach our ICE
     credentials + host candidate (+ DTLS fingerprint, setup:passive — the
     offerer is the DTLS client)."""
-    ans = SessionDescription(session_id=str(random.randint(10**8, 10**9)))
+    ans = SessionDescription(session_id=str(random.randint(10**8, 10**9)),
+                             bundle=offer.bundle)
     for i, m in enumerate(offer.media):
         sec = MediaSection(
             kind=m.kind,

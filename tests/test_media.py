@@ -384,3 +384,50 @@ def test_rtp_parse_csrc_extension_padding():
     pkt = RtpPacket.parse(hdr + csrc + ext + padded)
     assert pkt.payload == payload
     assert pkt.payload_type == 97 and pkt.sequence_number == 7
+
+
+def test_sdp_answer_echoes_bundle_and_rtcp_mux():
+    offer = SessionDescription.parse("\r\n".join([
+        "v=0", "o=- 1 2 IN IP4 127.0.0.1", "s=-", "t=0 0",
+        "a=group:BUNDLE 0",
+        "m=video 9 UDP/TLS/RTP/SAVPF 97",
+        "a=mid:0", "a=rtpmap:97 H264/90000",
+    ]) + "\r\n")
+    assert offer.bundle == "BUNDLE 0"
+    text = build_answer(offer, "127.0.0.1", 40000, "H264", ssrc=1).serialize()
+    assert "a=group:BUNDLE 0" in text
+    assert "a=rtcp-mux" in text
+
+
+def test_rx_ssrc_lock_on():
+    """A second SSRC (rtx/simulcast) must not pollute the locked stream's
+    access-unit reassembly."""
+    import asyncio
+
+    from ai_rtc_agent_amd.media.rtc import PeerConnection
+
+    class _Dec:
+        rtp_mode = "raw"
+
+        def __init__(self):
+            self.calls = []
+
+        def decode(self, data):
+            self.calls.append(data)
+            return torch.zeros(2, 2, 3, dtype=torch.uint8)
+
+    async def body():
+        pc = PeerConnection()
+        dec = _Dec()
+        pc._decoder = dec
+        pk1 = RtpPacketizer(ssrc=101)
+        pk2 = RtpPacketizer(ssrc=202)
+        f1 = b"AAAA" * 50
+        f2 = b"BBBB" * 50
+        for pkt in pk1.packetize(f1, timestamp=3000):
+            pc._on_datagram(pkt.serialize(), ("127.0.0.1", 1))
+        for pkt in pk2.packetize(f2, timestamp=3000):  # foreign SSRC
+            pc._on_datagram(pkt.serialize(), ("127.0.0.1", 1))
+        assert dec.calls == [f1]  # only the locked SSRC's frame decoded
+
+    asyncio.new_event_loop().run_until_complete(body())
