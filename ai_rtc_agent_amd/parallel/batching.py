@@ -132,7 +132,15 @@ class BatchedPipeline:
                 # frames may live on mixed devices (decoder output is CPU,
                 # the engine uploads); normalise before stacking
                 batch = torch.stack([t.cpu() for t in last])
-                out = self.base(batch)  # (K, H, W, 3)
+                try:
+                    out = self.base(batch)  # (K, H, W, 3)
+                except Exception as e:  # engine fault: fail the waiters,
+                    for i in pending:   # keep the loop alive for retries
+                        fut = self._future[i]
+                        if fut is not None and not fut.done():
+                            fut.set_exception(RuntimeError(str(e)))
+                            self._future[i] = None
+                    continue
                 now = time.perf_counter()
                 for i in pending:
                     fut = self._future[i]

@@ -176,20 +176,26 @@ class WorkerFrontend:
         return f"http://127.0.0.1:{self.ports[rank]}{path}"
 
     async def _forward(self, rank: int, request) -> "web.Response":
+        import aiohttp
         from aiohttp import web
 
         body = await request.read()
-        async with self._session().request(
-            request.method, self._url(rank, request.path),
-            data=body, headers={"Content-Type":
-                                request.headers.get("Content-Type", "")},
-        ) as r:
-            payload = await r.read()
-            headers = {}
-            if "Location" in r.headers:
-                headers["Location"] = r.headers["Location"]
-            return web.Response(status=r.status, body=payload,
-                                content_type=r.content_type, headers=headers)
+        try:
+            async with self._session().request(
+                request.method, self._url(rank, request.path),
+                data=body, headers={"Content-Type":
+                                    request.headers.get("Content-Type", "")},
+            ) as r:
+                payload = await r.read()
+                headers = {}
+                if "Location" in r.headers:
+                    headers["Location"] = r.headers["Location"]
+                return web.Response(status=r.status, body=payload,
+                                    content_type=r.content_type,
+                                    headers=headers)
+        except aiohttp.ClientError:
+            # worker mid-restart (see _monitor): tell the client to retry
+            return web.Response(status=503, text=f"worker {rank} restarting")
 
     def _session(self):
         import aiohttp
