@@ -365,6 +365,102 @@ pybind11::dict vcn_probe() {
   return d;
 }
 
+// fp8 conv: MX-scaled MFMA path (conv2d_fp8.hip). Same surface as conv2d
+// but weights are pre-quantized e4m3 bytes + per-OC dequant scales, and the
+// activation scale rides along as a scalar.
+torch::Tensor conv2d_fp8(torch::Tensor x, torch::Tensor w_fp8,
+                         torch::Tensor dq, double a_scale,
+                         c10::optional<torch::Tensor> bias,
+                         c10::optional<torch::Tensor> cbias,
+                         c10::optional<torch::Tensor> residual, int64_t R,
+                         int64_t S, int64_t stride, int64_t pad, int64_t act,
+                         c10::optional<torch::Tensor> in_affine,
+                         int64_t in_act) {
+  CHECK_IN(x);
+  CHECK_IN(w_fp8);
+  CHECK_IN(dq);
+  TORCH_CHECK(x.dtype() == torch::kHalf, "x must be f16");
+  TORCH_CHECK(w_fp8.dtype() == torch::kUInt8, "w_fp8 must be e4m3 bytes");
+  TORCH_CHECK(dq.dtype() == torch::kFloat, "dq must be f32[OC]");
+  const int B = x.size(0), H = x.size(1), W = x.size(2), IC = x.size(3);
+  const int OC = w_fp8.size(0);
+  TORCH_CHECK(IC % 64 == 0, "fp8 conv path requires IC % 64 == 0");
+  TORCH_CHECK(dq.numel() == OC, "dq must have OC entries");
+  TORCH_CHECK(w_fp8.size(1) == (long)R * S * IC, "w_fp8 must be (OC, R*S*IC)");
+  const int HO = (H + 2 * (int)pad - (int)R) / (int)stride + 1;
+  const int WO = (W + 2 * (int)pad - (int)S) / (int)stride + 1;
+  auto out = torch::empty({B, HO, WO, OC}, x.options());
+  const float* bp = nullptr;
+  if (bias.has_value()) {
+    TORCH_CHECK(bias->dtype() == torch::kFloat && bias->is_contiguous());
+    bp = bias->data_ptr<float>();
+  }
+  const uint16_t* cb = nullptr;
+  if (cbias.has_value()) {
+    CHECK_IN((*cbias));
+    TORCH_CHECK(cbias->dtype() == torch::kHalf);
+    cb = h_ptr(*cbias);
+  }
+  const uint16_t* res = nullptr;
+  if (residual.has_value()) {
+    CHECK_IN((*residual));
+    TORCH_CHECK(residual->dtype() == torch::kHalf);
+    TORCH_CHECK(residual->numel() == out.numel(), "residual shape mismatch");
+    res = h_ptr(*residual);
+  }
+  const float* aff = nullptr;
+  if (in_affine.has_value()) {
+    CHECK_IN((*in_affine));
+    TORCH_CHECK(in_affine->dtype() == torch::kFloat &&
+                    in_affine->numel() == (long)B * IC * 2,
+                "in_affine must be (B, IC, 2) f32");
+    aff = in_affine->data_ptr<float>();
+  }
+  int path = airtc_conv2d_splitk_for(B, HO, WO, OC, IC);
+  if (path == 0 || path == 100) path = HO * WO >= 2048 ? 1 : -1;
+  const int splitk = path > 0 ? path : -path;
+  float* wsp = nullptr;
+  torch::Tensor ws;
+  if (splitk > 1) {
+    ws = torch::empty({(long)B * splitk, (long)HO * WO, OC},
+                      x.options().dtype(torch::kFloat));
+    wsp = ws.data_ptr<float>();
+  }
+  airtc_conv2d_fp8_mfma(h_ptr(x), w_fp8.data_ptr<uint8_t>(),
+                        dq.data_ptr<float>(), bp, cb, res, h_ptr_mut(out), wsp,
+                        B, H, W, IC, HO, WO, OC, (int)R, (int)S, (int)stride,
+                        (int)pad, (int)act, path, aff, (int)in_act,
+                        (float)a_scale, cur_stream());
+  return out;
+}
+
+// fp8 hardware probes (fp8.hip): raw MX MFMA tile + fused scale-converts
+torch::Tensor fp8_mx_probe(torch::Tensor A, torch::Tensor B, int64_t sa,
+                           int64_t sb) {
+  CHECK_IN(A);
+  CHECK_IN(B);
+  TORCH_CHECK(A.dtype() == torch::kUInt8 && B.dtype() == torch::kUInt8);
+  TORCH_CHECK(A.numel() == 2048 && B.numel() == 2048, "fragments are 2048B");
+  auto draw = torch::empty({256}, A.options().dtype(torch::kFloat));
+  airtc_fp8_mx_probe(A.data_ptr<uint8_t>(), B.data_ptr<uint8_t>(),
+                     draw.data_ptr<float>(), (int)sa, (int)sb, cur_stream());
+  return draw;
+}
+
+pybind11::tuple fp8_cvt_probe(torch::Tensor fin, double scale,
+                              torch::Tensor enc_in) {
+  CHECK_IN(fin);
+  CHECK_IN(enc_in);
+  TORCH_CHECK(fin.dtype() == torch::kHalf && fin.numel() == 2);
+  TORCH_CHECK(enc_in.dtype() == torch::kUInt8 && enc_in.numel() == 2);
+  auto enc_out = torch::empty({2}, enc_in.options());
+  auto dec_out = torch::empty({2}, fin.options());
+  airtc_fp8_cvt_probe(h_ptr(fin), (float)scale, enc_out.data_ptr<uint8_t>(),
+                      enc_in.data_ptr<uint8_t>(), h_ptr_mut(dec_out),
+                      cur_stream());
+  return pybind11::make_tuple(enc_out, dec_out);
+}
+
 }  // namespace
 
 void airtc_register_dtls(pybind11::module_& m);  // dtls.cpp
@@ -389,6 +485,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("upsample2x", &upsample2x);
   m.def("preprocess_u8", &preprocess_u8);
   m.def("postprocess_u8", &postprocess_u8);
+  m.def("conv2d_fp8", &conv2d_fp8,
+        "fp8 e4m3 implicit-GEMM conv2d on the MX-scaled MFMA (NHWC)",
+        pybind11::arg("x"), pybind11::arg("w_fp8"), pybind11::arg("dq"),
+        pybind11::arg("a_scale"), pybind11::arg("bias"),
+        pybind11::arg("cbias"), pybind11::arg("residual"), pybind11::arg("R"),
+        pybind11::arg("S"), pybind11::arg("stride"), pybind11::arg("pad"),
+        pybind11::arg("act"), pybind11::arg("in_affine") = pybind11::none(),
+        pybind11::arg("in_act") = 0);
+  m.def("fp8_mx_probe", &fp8_mx_probe,
+        "raw-fragment v_mfma_scale_f32_16x16x128_f8f6f4 tile (layout probe)");
+  m.def("fp8_cvt_probe", &fp8_cvt_probe,
+        "v_cvt_scalef32_pk_{fp8_f16,f16_fp8} semantics probe");
   m.def("vcn_probe", &vcn_probe, "probe the VCN VA-API stack");
   m.def("h264_sps_pps", &h264_sps_pps, "Annex-B SPS+PPS for (w, h)");
   m.def("h264sw_table_check", []() { return airtc_h264sw_table_check(); },

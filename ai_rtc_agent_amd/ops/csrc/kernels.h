@@ -58,6 +58,28 @@ void airtc_conv2d_direct(const uint16_t* x, const uint16_t* w,
                          int stride, int pad, int act, const float* in_aff,
                          int in_act, hipStream_t s);
 
+// fp8 (MX-scaled MFMA path) -------------------------------------------------
+// conv2d on v_mfma_scale_f32_16x16x128_f8f6f4 (conv2d_fp8.hip). w_fp8 is
+// [OC][K] OCP e4m3 bytes (per-OC host quantization); dq[oc] = a_scale *
+// w_scale[oc] dequantizes in the epilogue; activations quantize in the
+// staging loads (a_scale). Requires IC % 64 == 0; path from
+// airtc_conv2d_splitk_for (nonzero).
+void airtc_conv2d_fp8_mfma(const uint16_t* x, const uint8_t* w_fp8,
+                           const float* dq, const float* bias,
+                           const uint16_t* cbias, const uint16_t* residual,
+                           uint16_t* out, float* ws, int B, int H, int W,
+                           int IC, int HO, int WO, int OC, int R, int S,
+                           int stride, int pad, int act, int path,
+                           const float* in_aff, int in_act, float a_scale,
+                           hipStream_t s);
+// hardware probes: raw-fragment MX MFMA tile and the fused scale-converts
+// (layout/semantics verified on hardware before the fp8 conv relies on them)
+void airtc_fp8_mx_probe(const uint8_t* A, const uint8_t* B, float* draw,
+                        int sa, int sb, hipStream_t s);
+void airtc_fp8_cvt_probe(const uint16_t* fin, float scale, uint8_t* enc_out,
+                         const uint8_t* enc_in, uint16_t* dec_out,
+                         hipStream_t s);
+
 // attention -----------------------------------------------------------------
 // q: base+strides address (B,H) heads; row stride in elements.
 // all of q/k/v/out share the (b,h) base law: base = b*sb + h*sh

@@ -193,6 +193,115 @@ def conv2d_nhwc(
 
 
 # ---------------------------------------------------------------------------
+# fp8 (OCP e4m3) conv — the MX-scaled MFMA serving tier
+# ---------------------------------------------------------------------------
+
+FP8_MAX = 448.0  # largest finite e4m3fn magnitude
+
+
+def quantize_weight_fp8(weight: torch.Tensor):
+    """(O,I,R,S) f16/f32 -> (uint8 (O, R*S*I) e4m3 codes, f32 (O,) scales).
+
+    Per-out-channel symmetric absmax scaling; codes are w/scale rounded RNE
+    (torch's e4m3fn cast is bit-identical to the gfx950 v_cvt encode at
+    scale 1 — verified by tools/fp8_probe.py). K order matches the f16
+    kernel's (r, s, ic) GEMM layout."""
+    O = weight.shape[0]
+    w = weight.detach().float().permute(0, 2, 3, 1).reshape(O, -1)
+    scale = (w.abs().amax(dim=1) / FP8_MAX).clamp_min(1e-12)
+    q = (w / scale[:, None]).clamp(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8).contiguous(), scale.contiguous()
+
+
+def fp8_roundtrip(x: torch.Tensor, scale: float) -> torch.Tensor:
+    """Emulate the kernel's activation quantization: e4m3(x/scale)*scale.
+    Reference for tests and the CPU path (f32 in/out)."""
+    q = (x.float() / scale).clamp(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
+    return q.to(torch.float32) * scale
+
+
+def conv2d_fp8_nhwc(
+    x: torch.Tensor,
+    weight: torch.Tensor,
+    a_scale: float,
+    bias: torch.Tensor | None = None,
+    stride: int = 1,
+    padding: int = 1,
+    act: int = ACT_NONE,
+    residual: torch.Tensor | None = None,
+    channel_bias: torch.Tensor | None = None,
+    in_affine: torch.Tensor | None = None,
+    in_act: int = ACT_NONE,
+) -> torch.Tensor:
+    """fp8 serving-tier conv: same contract as conv2d_nhwc plus a_scale (the
+    calibrated per-tensor activation scale). GPU: conv2d_fp8.hip — weights
+    pre-quantized per-OC (cached on the parameter), activations quantized in
+    the staging loads AFTER the fused affine/activation. CPU/eager: exact
+    emulation of the same quantization math (the GPU numerics golden).
+
+    Requires IC % 64 == 0 (use conv2d_nhwc for other layers)."""
+    O, I, R, S = weight.shape
+    if _use_hip(x):
+        ext = _require_ext()
+        wq = getattr(weight, "_airtc_wfp8", None)
+        if wq is None:
+            wq = quantize_weight_fp8(weight)
+            weight._airtc_wfp8 = wq
+        w_fp8, w_scale = wq
+        dq = getattr(weight, "_airtc_wdq", None)
+        if dq is None or dq[0] != a_scale:
+            dq = (a_scale, (w_scale * a_scale).contiguous())
+            weight._airtc_wdq = dq
+        b32 = None
+        if bias is not None:
+            b32 = _cached(bias, "_airtc_b32", lambda: bias.detach().float().contiguous())
+        return ext.conv2d_fp8(
+            x,
+            w_fp8,
+            dq[1],
+            a_scale,
+            b32,
+            None if channel_bias is None else channel_bias.contiguous(),
+            None if residual is None else residual.contiguous(),
+            R,
+            S,
+            stride,
+            padding,
+            act,
+            None if in_affine is None else in_affine.contiguous(),
+            in_act,
+        )
+
+    # emulation path (CPU tests + GPU numerics golden): quantize exactly as
+    # the kernel does, then run the f32 reference conv
+    if in_affine is not None:
+        aff = in_affine.float()
+        xf = x.float() * aff[:, None, None, :, 0] + aff[:, None, None, :, 1]
+        if in_act == ACT_SILU:
+            xf = F.silu(xf)
+        elif in_act == ACT_RELU:
+            xf = F.relu(xf)
+    else:
+        xf = x.float()
+    xq = fp8_roundtrip(xf, a_scale)
+    w_fp8, w_scale = quantize_weight_fp8(weight)
+    wdec = w_fp8.view(torch.float8_e4m3fn).to(torch.float32) * w_scale[:, None]
+    wdec = wdec.reshape(O, R, S, I).permute(0, 3, 1, 2)
+    y = F.conv2d(xq.permute(0, 3, 1, 2), wdec,
+                 None if bias is None else bias.float(),
+                 stride=stride, padding=padding).permute(0, 2, 3, 1)
+    if channel_bias is not None:
+        y = y + channel_bias.float()[:, None, None, :]
+    if residual is not None:
+        y = y + residual.float()
+    if act == ACT_SILU:
+        y = F.silu(y)
+    elif act == ACT_RELU:
+        y = F.relu(y)
+    return y.to(x.dtype).contiguous()
+
+
+# ---------------------------------------------------------------------------
 # normalisations
 # ---------------------------------------------------------------------------
 

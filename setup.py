@@ -17,7 +17,7 @@ CSRC = os.path.join(ROOT, "ai_rtc_agent_amd", "ops", "csrc")
 
 sources = [
     os.path.join(CSRC, f)
-    for f in ["ext.cpp", "vcn.cpp", "h264sw.cpp", "dtls.cpp", "elementwise.hip", "norms.hip", "conv2d.hip", "attention.hip"]
+    for f in ["ext.cpp", "vcn.cpp", "h264sw.cpp", "dtls.cpp", "elementwise.hip", "norms.hip", "conv2d.hip", "attention.hip", "fp8.hip", "conv2d_fp8.hip"]
 ]
 
 setup(

@@ -1,0 +1,94 @@
+"""FP8 (OCP e4m3) serving-tier tests.
+
+CPU part: the quantization math (per-OC weight scaling, activation
+round-trip error law, conv emulation accuracy vs f32). GPU part
+(test_fp8_gpu.py): the MX-MFMA kernel against this exact emulation.
+
+The fp8 tier is an MI355X-native addition (no reference counterpart):
+v_mfma_scale_f32_16x16x128_f8f6f4 runs at 2x the f16 MFMA rate and the
+fp8 weights halve the weight-bound layers' HBM traffic (ROADMAP item 1).
+"""
+import math
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from ai_rtc_agent_amd import ops
+
+
+def test_weight_quant_shapes_and_scale_law():
+    g = torch.Generator().manual_seed(0)
+    w = torch.randn(64, 32, 3, 3, generator=g).half()
+    q, s = ops.quantize_weight_fp8(w)
+    assert q.dtype == torch.uint8 and q.shape == (64, 32 * 9)
+    assert s.dtype == torch.float32 and s.shape == (64,)
+    # per-OC absmax maps to +-448: decode magnitude max equals absmax (RNE
+    # of the extreme is exact by construction: absmax/scale = 448 exactly)
+    dec = q.view(torch.float8_e4m3fn).to(torch.float32) * s[:, None]
+    wf = w.float().permute(0, 2, 3, 1).reshape(64, -1)
+    assert torch.allclose(dec.abs().amax(dim=1), wf.abs().amax(dim=1), rtol=1e-6)
+
+
+def test_weight_quant_relative_error_bound():
+    g = torch.Generator().manual_seed(1)
+    w = torch.randn(16, 64, 3, 3, generator=g).half()
+    q, s = ops.quantize_weight_fp8(w)
+    dec = (q.view(torch.float8_e4m3fn).to(torch.float32) * s[:, None])
+    wf = w.float().permute(0, 2, 3, 1).reshape(16, -1)
+    err = (dec - wf).abs()
+    # e4m3: 3 mantissa bits -> rel err <= 2^-4 for normals; subnormal floor
+    # is scale * 2^-10 absolute
+    bound = wf.abs() * 2 ** -4 + s[:, None] * 2 ** -9
+    assert (err <= bound + 1e-9).all()
+
+
+def test_activation_roundtrip_error():
+    g = torch.Generator().manual_seed(2)
+    x = torch.randn(4096, generator=g).float() * 3.0
+    scale = x.abs().max().item() / ops.FP8_MAX
+    y = ops.fp8_roundtrip(x, scale)
+    err = (y - x).abs()
+    assert (err <= x.abs() * 2 ** -4 + scale * 2 ** -9 + 1e-9).all()
+    # SNR sanity: quantization noise well below signal
+    snr = 10 * math.log10((x ** 2).mean().item() / ((y - x) ** 2).mean().item())
+    assert snr > 25, f"e4m3 round-trip SNR {snr:.1f} dB too low"
+
+
+@pytest.mark.parametrize("ic,oc,h,r,act", [
+    (64, 64, 16, 3, ops.ACT_SILU),
+    (128, 64, 8, 1, ops.ACT_NONE),
+])
+def test_conv_fp8_emulation_close_to_f32(ic, oc, h, r, act):
+    g = torch.Generator().manual_seed(3)
+    x = (torch.randn(2, h, h, ic, generator=g)).half()
+    w = (torch.randn(oc, ic, r, r, generator=g) / math.sqrt(ic * r * r)).half()
+    b = torch.randn(oc, generator=g).half()
+    a_scale = x.float().abs().max().item() / ops.FP8_MAX
+    y8 = ops.conv2d_fp8_nhwc(x, w, a_scale, b, padding=r // 2, act=act)
+    yf = ops.conv2d_nhwc(x, w, b, padding=r // 2, act=act)
+    # fp8 noise at K = ic*r*r accumulation: compare in SNR terms
+    num = (yf.float() ** 2).mean().item()
+    den = ((y8.float() - yf.float()) ** 2).mean().item()
+    snr = 10 * math.log10(num / max(den, 1e-20))
+    assert snr > 20, f"fp8 conv emulation SNR {snr:.1f} dB vs f16 path"
+
+
+def test_conv_fp8_emulation_with_affine_residual_cbias():
+    g = torch.Generator().manual_seed(4)
+    ic, oc, h = 64, 64, 8
+    x = torch.randn(2, h, h, ic, generator=g).half()
+    w = (torch.randn(oc, ic, 3, 3, generator=g) / math.sqrt(ic * 9)).half()
+    aff = torch.randn(2, ic, 2, generator=g).float() * 0.2 + 0.6
+    res = torch.randn(2, h, h, oc, generator=g).half()
+    cb = torch.randn(2, oc, generator=g).half()
+    a_scale = 4.0 / ops.FP8_MAX
+    y8 = ops.conv2d_fp8_nhwc(x, w, a_scale, None, act=ops.ACT_SILU,
+                             residual=res, channel_bias=cb, in_affine=aff,
+                             in_act=ops.ACT_SILU)
+    yf = ops.conv2d_nhwc(x, w, None, act=ops.ACT_SILU, residual=res,
+                         channel_bias=cb, in_affine=aff, in_act=ops.ACT_SILU)
+    num = (yf.float() ** 2).mean().item()
+    den = ((y8.float() - yf.float()) ** 2).mean().item()
+    snr = 10 * math.log10(num / max(den, 1e-20))
+    assert snr > 18, f"fused-epilogue fp8 emulation SNR {snr:.1f} dB"

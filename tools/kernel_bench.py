@@ -72,6 +72,22 @@ def main():
         fl = 2.0 * b * ho * wo * oc * ic * k * k
         print(f"  {name:32s} {us:8.1f} us  {fl/us/1e6:7.1f} TF/s")
 
+    print("== conv2d fp8 (MX-scaled MFMA, per-OC weight scales) ==")
+    for name, b, h, w, ic, oc, k, st in CONV_SHAPES:
+        if ic % 64 != 0:
+            continue
+        x = torch.randn(b, h, w, ic, device=dev).half()
+        wt = (torch.randn(oc, ic, k, k, device=dev) * 0.02).half()
+        bias = torch.randn(oc, device=dev).float()
+        pad = k // 2
+        a_scale = x.float().abs().max().item() / ops.FP8_MAX
+        fn = lambda: ops.conv2d_fp8_nhwc(x, wt, a_scale, None, stride=st,
+                                         padding=pad)
+        us = timeit(fn, args.iters)
+        ho, wo = (h + 2 * pad - k) // st + 1, (w + 2 * pad - k) // st + 1
+        fl = 2.0 * b * ho * wo * oc * ic * k * k
+        print(f"  {name:32s} {us:8.1f} us  {fl/us/1e6:7.1f} TF/s")
+
     print("== attention (flash MFMA + tr_b16) ==")
     for name, b, lq, lk, c, hds in ATTN_SHAPES:
         q = torch.randn(b, lq, c, device=dev).half()
