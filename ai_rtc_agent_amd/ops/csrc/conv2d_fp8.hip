@@ -52,15 +52,18 @@ __device__ __forceinline__ f32x4 mfma_mx_fp8(i32x8 a, i32x8 b, f32x4 c) {
 }
 
 // stage 16 fp8 bytes from 16 f16 inputs: optional per-channel affine +
-// activation (f32 math), then quantize by sa (encode divides). OOB -> 0.
-// Values are CLAMPED to +-448*sa first: the HW encode does NOT saturate
-// (overflow -> NaN, measured in tests/test_fp8_gpu.py), and serving frames
-// can exceed the calibrated absmax. clamp is the f16 limit = 448*sa.
+// activation (f32 math), then quantize: q = clamp(v * (1/sa), +-448) via
+// the NON-scaled v_cvt_pk_fp8_f32 — bit-identical to torch's e4m3 RNE.
+// MEASURED HAZARD (tools/fp8_debug.py): v_cvt_scalef32_pk_fp8_* at
+// arbitrary scales rounds every code UP one ULP, turning max-magnitude
+// codes (126) into NaN (127); the scalef32 forms are NOT used here. The
+// clamp also covers post-calibration drift (HW encode overflows to NaN,
+// it does not saturate). OOB -> 0.
 __device__ __forceinline__ i32x4 load_a_fp8(const f16* xb, int ho_s, int wo_s,
                                             int r, int s, int pad, int H,
                                             int W, int IC, int ic,
                                             const float* aff, int in_act,
-                                            float sa, float clampv, bool kok) {
+                                            float inv_sa, bool kok) {
   const int hi = ho_s + r - pad;
   const int wi = wo_s + s - pad;
   const bool ok = kok && (unsigned)hi < (unsigned)H && (unsigned)wi < (unsigned)W;
@@ -70,36 +73,24 @@ __device__ __forceinline__ i32x4 load_a_fp8(const f16* xb, int ho_s, int wo_s,
   const f16x8 hi8 = *reinterpret_cast<const f16x8*>(src + 8);
   i32x4 outv;
 #pragma unroll
-  for (int q = 0; q < 4; ++q) {  // 4 bytes (2 f16x2 pairs) per i32
-    s16x2 packed = {0, 0};
+  for (int q = 0; q < 4; ++q) {  // 4 bytes (2 f32 pairs) per i32
     const int j0 = (q & 1) * 4;
     const f16x8& vsrc = (q < 2) ? lo : hi8;
-    if (aff) {
-      const int icq = ic + (q >> 1) * 8;
-      float a[4];
+    float a[4];
 #pragma unroll
-      for (int h = 0; h < 4; ++h) {
-        float v = apply_act((float)vsrc[j0 + h] * aff[(icq + j0 + h) * 2] +
-                                aff[(icq + j0 + h) * 2 + 1],
-                            in_act);
-        a[h] = fminf(fmaxf(v, -clampv), clampv);
+    for (int h = 0; h < 4; ++h) {
+      float v = (float)vsrc[j0 + h];
+      if (aff) {
+        const int c = ic + (q >> 1) * 8 + j0 + h;
+        v = apply_act(v * aff[c * 2] + aff[c * 2 + 1], in_act);
       }
-      packed =
-          __builtin_amdgcn_cvt_scalef32_pk_fp8_f32(packed, a[0], a[1], sa, false);
-      packed =
-          __builtin_amdgcn_cvt_scalef32_pk_fp8_f32(packed, a[2], a[3], sa, true);
-    } else {
-      const f16 lim = (f16)clampv;
-      hf16x2 limv = {lim, lim}, nlimv = {-lim, -lim};
-      hf16x2 v0 = {vsrc[j0], vsrc[j0 + 1]};
-      hf16x2 v1 = {vsrc[j0 + 2], vsrc[j0 + 3]};
-      v0 = __builtin_elementwise_min(__builtin_elementwise_max(v0, nlimv), limv);
-      v1 = __builtin_elementwise_min(__builtin_elementwise_max(v1, nlimv), limv);
-      packed = __builtin_amdgcn_cvt_scalef32_pk_fp8_f16(packed, v0, sa, false);
-      packed = __builtin_amdgcn_cvt_scalef32_pk_fp8_f16(packed, v1, sa, true);
+      // min(max()) lowers to v_med3_f32; 448 = e4m3 max
+      a[h] = fminf(fmaxf(v * inv_sa, -448.0f), 448.0f);
     }
-    outv[q] = (int)(((unsigned short)packed[0]) |
-                    (((unsigned)(unsigned short)packed[1]) << 16));
+    int p2 = 0;
+    p2 = __builtin_amdgcn_cvt_pk_fp8_f32(a[0], a[1], p2, false);
+    p2 = __builtin_amdgcn_cvt_pk_fp8_f32(a[2], a[3], p2, true);
+    outv[q] = p2;
   }
   return outv;
 }
@@ -163,7 +154,7 @@ __global__ __launch_bounds__(256) void conv2d_mfma_fp8_kernel(
   const int split = blockIdx.z - b * spk;
   const f16* xb = x + (long)b * H * W * IC;
   const float* affb = in_aff ? in_aff + (long)b * IC * 2 : nullptr;
-  const float clampv = 448.0f * sa;  // e4m3 max after the encode's divide
+  const float inv_sa = 1.0f / sa;  // encode multiplies; clamp at +-448
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -213,8 +204,8 @@ __global__ __launch_bounds__(256) void conv2d_mfma_fp8_kernel(
       const KPos8& p = (a_u[i] >= 4) ? p1 : p0;
       const int kg = kbase + a_u[i] * 16;
       regA[i] = load_a_fp8(xb, a_ho[i], a_wo[i], p.r, p.s, pad, H, W, IC,
-                           p.ic0 + (a_u[i] & 3) * 16, affb, in_act, sa,
-                           clampv, kg < K);
+                           p.ic0 + (a_u[i] & 3) * 16, affb, in_act, inv_sa,
+                           kg < K);
     }
 #pragma unroll
     for (int i = 0; i < BLOADS; ++i) {

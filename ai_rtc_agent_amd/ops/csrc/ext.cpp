@@ -447,6 +447,15 @@ torch::Tensor fp8_mx_probe(torch::Tensor A, torch::Tensor B, int64_t sa,
   return draw;
 }
 
+torch::Tensor fp8_quant_probe(torch::Tensor in16, double sa, int64_t variant) {
+  CHECK_IN(in16);
+  TORCH_CHECK(in16.dtype() == torch::kHalf && in16.numel() == 16);
+  auto out = torch::empty({16}, in16.options().dtype(torch::kUInt8));
+  airtc_fp8_quant_probe(h_ptr(in16), (float)sa, out.data_ptr<uint8_t>(),
+                        (int)variant, cur_stream());
+  return out;
+}
+
 pybind11::tuple fp8_cvt_probe(torch::Tensor fin, double scale,
                               torch::Tensor enc_in) {
   CHECK_IN(fin);
@@ -495,6 +504,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("in_act") = 0);
   m.def("fp8_mx_probe", &fp8_mx_probe,
         "raw-fragment v_mfma_scale_f32_16x16x128_f8f6f4 tile (layout probe)");
+  m.def("fp8_quant_probe", &fp8_quant_probe,
+        "conv kernel's exact 16-value activation encode (variant 0/1)");
   m.def("fp8_cvt_probe", &fp8_cvt_probe,
         "v_cvt_scalef32_pk_{fp8_f16,f16_fp8} semantics probe");
   m.def("vcn_probe", &vcn_probe, "probe the VCN VA-API stack");

@@ -64,6 +64,58 @@ __global__ void fp8_cvt_probe_kernel(const _Float16* __restrict__ fin,
   dec_out[1] = dec[1];
 }
 
+// ---------------------------------------------------------------------------
+// Probe 3: the conv kernel's exact activation-encode path (clamp to
+// +-448*sa, two pk_fp8 calls packing 4 bytes per i32) on 16 f16 values —
+// variant 0 = cvt_scalef32_pk_fp8_f16 (divide-by-sa in the instruction),
+// variant 1 = v_mul by 1/sa then non-scaled cvt_pk_fp8_f32. Isolates the
+// encode from the rest of the conv when numerics disagree.
+// ---------------------------------------------------------------------------
+__global__ void fp8_quant_probe_kernel(const _Float16* __restrict__ in16,
+                                       float sa, uint8_t* __restrict__ out16,
+                                       int variant) {
+  if (threadIdx.x != 0) return;
+  const float clampv = 448.0f * sa;
+  const f16 lim = (f16)clampv;
+  const float inv = 1.0f / sa;
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    s16x2 packed = {0, 0};
+    const int j = q * 4;
+    if (variant == 0) {
+      hf16x2 limv = {lim, lim}, nlimv = {(f16)(-lim), (f16)(-lim)};
+      hf16x2 v0 = {in16[j], in16[j + 1]};
+      hf16x2 v1 = {in16[j + 2], in16[j + 3]};
+      v0 = __builtin_elementwise_min(__builtin_elementwise_max(v0, nlimv), limv);
+      v1 = __builtin_elementwise_min(__builtin_elementwise_max(v1, nlimv), limv);
+      packed = __builtin_amdgcn_cvt_scalef32_pk_fp8_f16(packed, v0, sa, false);
+      packed = __builtin_amdgcn_cvt_scalef32_pk_fp8_f16(packed, v1, sa, true);
+    } else {
+      float a[4];
+#pragma unroll
+      for (int h = 0; h < 4; ++h)
+        a[h] = fminf(fmaxf((float)in16[j + h], -clampv), clampv) * inv;
+      int p2 = 0;
+      p2 = __builtin_amdgcn_cvt_pk_fp8_f32(a[0], a[1], p2, false);
+      p2 = __builtin_amdgcn_cvt_pk_fp8_f32(a[2], a[3], p2, true);
+      packed[0] = (short)(p2 & 0xFFFF);
+      packed[1] = (short)((p2 >> 16) & 0xFFFF);
+    }
+    out16[q * 4 + 0] = (uint8_t)(packed[0] & 0xFF);
+    out16[q * 4 + 1] = (uint8_t)((packed[0] >> 8) & 0xFF);
+    out16[q * 4 + 2] = (uint8_t)(packed[1] & 0xFF);
+    out16[q * 4 + 3] = (uint8_t)((packed[1] >> 8) & 0xFF);
+  }
+}
+
+extern "C" void airtc_fp8_quant_probe(const uint16_t* in16, float sa,
+                                      uint8_t* out16, int variant,
+                                      hipStream_t s) {
+  hipLaunchKernelGGL(fp8_quant_probe_kernel, dim3(1), dim3(64), 0, s,
+                     reinterpret_cast<const _Float16*>(in16), sa, out16,
+                     variant);
+}
+
 extern "C" void airtc_fp8_mx_probe(const uint8_t* A, const uint8_t* B,
                                    float* draw, int sa, int sb,
                                    hipStream_t s) {

@@ -79,6 +79,8 @@ void airtc_fp8_mx_probe(const uint8_t* A, const uint8_t* B, float* draw,
 void airtc_fp8_cvt_probe(const uint16_t* fin, float scale, uint8_t* enc_out,
                          const uint8_t* enc_in, uint16_t* dec_out,
                          hipStream_t s);
+void airtc_fp8_quant_probe(const uint16_t* in16, float sa, uint8_t* out16,
+                           int variant, hipStream_t s);
 
 // attention -----------------------------------------------------------------
 // q: base+strides address (B,H) heads; row stride in elements.

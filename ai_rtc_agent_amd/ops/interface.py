@@ -214,9 +214,13 @@ def quantize_weight_fp8(weight: torch.Tensor):
 
 
 def fp8_roundtrip(x: torch.Tensor, scale: float) -> torch.Tensor:
-    """Emulate the kernel's activation quantization: e4m3(x/scale)*scale.
+    """Emulate the kernel's activation quantization:
+    e4m3(clamp(x * (1/scale), +-448)) * scale — the kernel multiplies by the
+    f32 reciprocal (v_cvt_pk_fp8_f32 path), so the emulation does too.
     Reference for tests and the CPU path (f32 in/out)."""
-    q = (x.float() / scale).clamp(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
+    inv = torch.tensor(1.0, dtype=torch.float32) / torch.tensor(
+        scale, dtype=torch.float32)
+    q = (x.float() * inv).clamp(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
     return q.to(torch.float32) * scale
 
 
