@@ -116,6 +116,26 @@ torch::Tensor group_norm_coeffs(torch::Tensor x, int64_t groups,
   return coeffs;
 }
 
+torch::Tensor group_norm_silu_fp8(torch::Tensor x, int64_t groups,
+                                  torch::Tensor gamma, torch::Tensor beta,
+                                  double eps, int64_t act, double a_scale) {
+  CHECK_IN(x);
+  TORCH_CHECK(x.dtype() == torch::kHalf);
+  const int B = x.size(0);
+  const int C = x.size(-1);
+  const long HW = x.numel() / ((long)B * C);
+  auto out = torch::empty_like(x, x.options().dtype(torch::kUInt8));
+  const int nchunk = airtc_group_norm_nchunk(B, (int)groups);
+  auto ws = torch::empty({(long)B * groups * nchunk * 2},
+                         x.options().dtype(torch::kFloat));
+  airtc_group_norm_silu_fp8(h_ptr(x), gamma.data_ptr<float>(),
+                            beta.data_ptr<float>(), out.data_ptr<uint8_t>(),
+                            ws.data_ptr<float>(), B, (int)HW, C, (int)groups,
+                            (float)eps, (int)act, (float)a_scale,
+                            cur_stream());
+  return out;
+}
+
 torch::Tensor group_norm_silu(torch::Tensor x, int64_t groups,
                               torch::Tensor gamma, torch::Tensor beta,
                               double eps, int64_t act) {
@@ -379,7 +399,9 @@ torch::Tensor conv2d_fp8(torch::Tensor x, torch::Tensor w_fp8,
   CHECK_IN(x);
   CHECK_IN(w_fp8);
   CHECK_IN(dq);
-  TORCH_CHECK(x.dtype() == torch::kHalf, "x must be f16");
+  const bool x_q8 = x.dtype() == torch::kUInt8;
+  TORCH_CHECK(x_q8 || x.dtype() == torch::kHalf,
+              "x must be f16 or u8 e4m3 codes");
   TORCH_CHECK(w_fp8.dtype() == torch::kUInt8, "w_fp8 must be e4m3 bytes");
   TORCH_CHECK(dq.dtype() == torch::kFloat, "dq must be f32[OC]");
   const int B = x.size(0), H = x.size(1), W = x.size(2), IC = x.size(3);
@@ -387,9 +409,11 @@ torch::Tensor conv2d_fp8(torch::Tensor x, torch::Tensor w_fp8,
   TORCH_CHECK(IC % 64 == 0, "fp8 conv path requires IC % 64 == 0");
   TORCH_CHECK(dq.numel() == OC, "dq must have OC entries");
   TORCH_CHECK(w_fp8.size(1) == (long)R * S * IC, "w_fp8 must be (OC, R*S*IC)");
+  TORCH_CHECK(!(x_q8 && in_affine.has_value()),
+              "pre-quantized input cannot take a fused input affine");
   const int HO = (H + 2 * (int)pad - (int)R) / (int)stride + 1;
   const int WO = (W + 2 * (int)pad - (int)S) / (int)stride + 1;
-  auto out = torch::empty({B, HO, WO, OC}, x.options());
+  auto out = torch::empty({B, HO, WO, OC}, x.options().dtype(torch::kHalf));
   const float* bp = nullptr;
   if (bias.has_value()) {
     TORCH_CHECK(bias->dtype() == torch::kFloat && bias->is_contiguous());
@@ -426,11 +450,15 @@ torch::Tensor conv2d_fp8(torch::Tensor x, torch::Tensor w_fp8,
                       x.options().dtype(torch::kFloat));
     wsp = ws.data_ptr<float>();
   }
-  airtc_conv2d_fp8_mfma(h_ptr(x), w_fp8.data_ptr<uint8_t>(),
+  const uint16_t* xp = x_q8
+                           ? reinterpret_cast<const uint16_t*>(
+                                 x.data_ptr<uint8_t>())
+                           : h_ptr(x);
+  airtc_conv2d_fp8_mfma(xp, w_fp8.data_ptr<uint8_t>(),
                         dq.data_ptr<float>(), bp, cb, res, h_ptr_mut(out), wsp,
                         B, H, W, IC, HO, WO, OC, (int)R, (int)S, (int)stride,
                         (int)pad, (int)act, path, aff, (int)in_act,
-                        (float)a_scale, cur_stream());
+                        (float)a_scale, x_q8 ? 1 : 0, cur_stream());
   return out;
 }
 
@@ -484,6 +512,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("in_act") = 0,
         pybind11::arg("counters") = pybind11::none());
   m.def("group_norm_silu", &group_norm_silu);
+  m.def("group_norm_silu_fp8", &group_norm_silu_fp8,
+        "GN+act writing e4m3 codes (producer-side fp8 quantization)");
   m.def("group_norm_coeffs", &group_norm_coeffs,
         "(B,C,2) f32 affine pairs for the fused GN->conv input transform");
   m.def("layer_norm", &layer_norm);

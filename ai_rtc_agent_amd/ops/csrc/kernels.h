@@ -28,6 +28,12 @@ void airtc_group_norm_silu(const uint16_t* x, const float* gamma,
 void airtc_layer_norm(const uint16_t* x, const float* gamma, const float* beta,
                       uint16_t* out, long rows, int C, float eps,
                       hipStream_t s);
+// fp8-output GN: apply pass writes e4m3 codes q = clamp(act(gn(x))/a_scale)
+// (producer-side quantization for the fp8 conv path)
+void airtc_group_norm_silu_fp8(const uint16_t* x, const float* gamma,
+                               const float* beta, uint8_t* out, float* ws,
+                               int B, int HW, int C, int G, float eps,
+                               int act, float a_scale, hipStream_t s);
 // (B, C, 2) f32 affine pairs for the fused GN->conv input transform
 void airtc_group_norm_coeffs(const uint16_t* x, const float* gamma,
                              const float* beta, float* coeffs, float* ws,
@@ -64,6 +70,8 @@ void airtc_conv2d_direct(const uint16_t* x, const uint16_t* w,
 // w_scale[oc] dequantizes in the epilogue; activations quantize in the
 // staging loads (a_scale). Requires IC % 64 == 0; path from
 // airtc_conv2d_splitk_for (nonzero).
+// x_is_q8: x already holds e4m3 codes (u8, producer-quantized by e.g.
+// airtc_group_norm_silu_fp8) — staging is a raw byte copy, in_aff unused.
 void airtc_conv2d_fp8_mfma(const uint16_t* x, const uint8_t* w_fp8,
                            const float* dq, const float* bias,
                            const uint16_t* cbias, const uint16_t* residual,
@@ -71,7 +79,7 @@ void airtc_conv2d_fp8_mfma(const uint16_t* x, const uint8_t* w_fp8,
                            int IC, int HO, int WO, int OC, int R, int S,
                            int stride, int pad, int act, int path,
                            const float* in_aff, int in_act, float a_scale,
-                           hipStream_t s);
+                           int x_is_q8, hipStream_t s);
 // hardware probes: raw-fragment MX MFMA tile and the fused scale-converts
 // (layout/semantics verified on hardware before the fp8 conv relies on them)
 void airtc_fp8_mx_probe(const uint8_t* A, const uint8_t* B, float* draw,

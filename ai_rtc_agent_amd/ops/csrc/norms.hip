@@ -106,6 +106,56 @@ __global__ void group_norm_apply_kernel(const f16* __restrict__ x,
   }
 }
 
+// fp8-output variant: the apply pass writes OCP e4m3 CODES (q = clamp(
+// act(gn(x)) * inv_sa, +-448)) so the fp8 conv stages raw bytes with zero
+// encode VALU and half the activation HBM traffic (producer-side
+// quantization — the conv's inline encode would otherwise re-encode every
+// element once per 3x3 tap). Encode uses the non-scaled v_cvt_pk_fp8_f32
+// (the scalef32 forms round up a ULP at arbitrary scales — see
+// conv2d_fp8.hip header).
+__global__ void group_norm_apply_fp8_kernel(
+    const f16* __restrict__ x, const float* __restrict__ ws,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    uint8_t* __restrict__ out, int HW, int C, int G, int nchunk, float eps,
+    int act, float inv_sa) {
+  const int chunk = blockIdx.y;
+  const int b = blockIdx.x / G;
+  const int g = blockIdx.x % G;
+  const int Cg = C / G;
+  const int Cg2 = Cg / 2;
+  const long base = (long)b * HW * C + (long)g * Cg;
+
+  __shared__ float stats[2];
+  if (threadIdx.x == 0) {
+    const float* w = &ws[(long)blockIdx.x * nchunk * 2];
+    float s0 = 0.f, s1 = 0.f;
+    for (int i = 0; i < nchunk; ++i) { s0 += w[2 * i]; s1 += w[2 * i + 1]; }
+    const float n = (float)HW * Cg;
+    const float mean = s0 / n;
+    stats[0] = mean;
+    stats[1] = rsqrtf(s1 / n - mean * mean + eps);
+  }
+  __syncthreads();
+  const float mean = stats[0], rstd = stats[1];
+
+  long lo, hi;
+  gn_chunk_range((long)HW * Cg2, chunk, nchunk, &lo, &hi);
+  for (long i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+    long p = i / Cg2, c2 = i - p * Cg2;
+    long idx = base + p * C + c2 * 2;
+    f16x2 v = *reinterpret_cast<const f16x2*>(&x[idx]);
+    int ch = g * Cg + (int)c2 * 2;
+    float a[2];
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      float t = ((float)v[j] - mean) * rstd * gamma[ch + j] + beta[ch + j];
+      a[j] = fminf(fmaxf(apply_act(t, act) * inv_sa, -448.0f), 448.0f);
+    }
+    int p2 = __builtin_amdgcn_cvt_pk_fp8_f32(a[0], a[1], 0, false);
+    *reinterpret_cast<short*>(&out[idx]) = (short)(p2 & 0xFFFF);
+  }
+}
+
 extern "C" int airtc_group_norm_nchunk(int B, int G) {
   int n = (int)(512 / max(1, B * G));
   if (n < 1) n = 1;
@@ -172,6 +222,21 @@ extern "C" void airtc_group_norm_silu(const uint16_t* x, const float* gamma,
   hipLaunchKernelGGL(group_norm_apply_kernel, grid, dim3(256), 0, s,
                      reinterpret_cast<const f16*>(x), ws, gamma, beta,
                      reinterpret_cast<f16*>(out), HW, C, G, nchunk, eps, act);
+}
+
+extern "C" void airtc_group_norm_silu_fp8(const uint16_t* x,
+                                          const float* gamma,
+                                          const float* beta, uint8_t* out,
+                                          float* ws, int B, int HW, int C,
+                                          int G, float eps, int act,
+                                          float a_scale, hipStream_t s) {
+  const int nchunk = airtc_group_norm_nchunk(B, G);
+  dim3 grid(B * G, nchunk);
+  hipLaunchKernelGGL(group_norm_stats_kernel, grid, dim3(256), 0, s,
+                     reinterpret_cast<const f16*>(x), ws, HW, C, G, nchunk);
+  hipLaunchKernelGGL(group_norm_apply_fp8_kernel, grid, dim3(256), 0, s,
+                     reinterpret_cast<const f16*>(x), ws, gamma, beta, out,
+                     HW, C, G, nchunk, eps, act, 1.0f / a_scale);
 }
 
 // ---------------------------------------------------------------------------

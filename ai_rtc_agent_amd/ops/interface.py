@@ -243,9 +243,15 @@ def conv2d_fp8_nhwc(
     the staging loads AFTER the fused affine/activation. CPU/eager: exact
     emulation of the same quantization math (the GPU numerics golden).
 
+    x may be f16 (the kernel quantizes in its staging loads) or u8 e4m3
+    codes already scaled by a_scale (producer-quantized, e.g. the output of
+    group_norm_silu_nhwc(fp8_scale=...)) — the fast path.
+
     Requires IC % 64 == 0 (use conv2d_nhwc for other layers)."""
     O, I, R, S = weight.shape
-    if _use_hip(x):
+    x_q8 = x.dtype == torch.uint8
+    if _use_hip(x) or (x_q8 and x.is_cuda
+                       and os.environ.get("AIRTC_FORCE_EAGER") != "1"):
         ext = _require_ext()
         wq = getattr(weight, "_airtc_wfp8", None)
         if wq is None:
@@ -278,16 +284,20 @@ def conv2d_fp8_nhwc(
 
     # emulation path (CPU tests + GPU numerics golden): quantize exactly as
     # the kernel does, then run the f32 reference conv
-    if in_affine is not None:
-        aff = in_affine.float()
-        xf = x.float() * aff[:, None, None, :, 0] + aff[:, None, None, :, 1]
-        if in_act == ACT_SILU:
-            xf = F.silu(xf)
-        elif in_act == ACT_RELU:
-            xf = F.relu(xf)
+    if x_q8:
+        assert in_affine is None, "pre-quantized input excludes input affine"
+        xq = x.view(torch.float8_e4m3fn).to(torch.float32) * a_scale
     else:
-        xf = x.float()
-    xq = fp8_roundtrip(xf, a_scale)
+        if in_affine is not None:
+            aff = in_affine.float()
+            xf = x.float() * aff[:, None, None, :, 0] + aff[:, None, None, :, 1]
+            if in_act == ACT_SILU:
+                xf = F.silu(xf)
+            elif in_act == ACT_RELU:
+                xf = F.relu(xf)
+        else:
+            xf = x.float()
+        xq = fp8_roundtrip(xf, a_scale)
     w_fp8, w_scale = quantize_weight_fp8(weight)
     wdec = w_fp8.view(torch.float8_e4m3fn).to(torch.float32) * w_scale[:, None]
     wdec = wdec.reshape(O, R, S, I).permute(0, 3, 1, 2)
@@ -302,7 +312,7 @@ def conv2d_fp8_nhwc(
         y = F.silu(y)
     elif act == ACT_RELU:
         y = F.relu(y)
-    return y.to(x.dtype).contiguous()
+    return y.to(torch.float16 if x_q8 else x.dtype).contiguous()
 
 
 # ---------------------------------------------------------------------------
@@ -345,8 +355,14 @@ def group_norm_silu_nhwc(
     beta: torch.Tensor,
     eps: float = 1e-5,
     silu: bool = True,
+    fp8_scale: float | None = None,
 ) -> torch.Tensor:
-    """Fused GroupNorm(+SiLU) on NHWC. GPU: single kernel, wave-reduced stats."""
+    """Fused GroupNorm(+SiLU) on NHWC. GPU: single kernel, wave-reduced stats.
+
+    fp8_scale: when set, the apply pass QUANTIZES its output to e4m3 codes
+    (u8 tensor, q = clamp(act(gn(x))/fp8_scale, +-448)) for the fp8 conv's
+    pre-quantized input path — producer-side quantization: one encode per
+    element instead of one per 3x3 tap, and half the GN output traffic."""
     if _use_hip(x):
         if (x.shape[-1] // num_groups) % 2 != 0:
             raise ValueError(
@@ -356,13 +372,22 @@ def group_norm_silu_nhwc(
         ext = _require_ext()
         g32 = _cached(gamma, "_airtc_g32", lambda: gamma.detach().float().contiguous())
         b32 = _cached(beta, "_airtc_b32", lambda: beta.detach().float().contiguous())
-        return ext.group_norm_silu(x, num_groups, g32, b32, eps, ACT_SILU if silu else ACT_NONE)
+        act = ACT_SILU if silu else ACT_NONE
+        if fp8_scale is not None:
+            return ext.group_norm_silu_fp8(x, num_groups, g32, b32, eps, act,
+                                           fp8_scale)
+        return ext.group_norm_silu(x, num_groups, g32, b32, eps, act)
     b, h, w, c = x.shape
     xc = x.permute(0, 3, 1, 2).float()
     y = F.group_norm(xc, num_groups, gamma.float(), beta.float(), eps)
     y = y.permute(0, 2, 3, 1)
     if silu:
         y = F.silu(y)
+    if fp8_scale is not None:
+        inv = torch.tensor(1.0, dtype=torch.float32) / torch.tensor(
+            fp8_scale, dtype=torch.float32)
+        q = (y.float() * inv).clamp(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
+        return q.view(torch.uint8).contiguous()
     return y.to(x.dtype).contiguous()
 
 

@@ -74,6 +74,32 @@ def test_conv_fp8_emulation_close_to_f32(ic, oc, h, r, act):
     assert snr > 20, f"fp8 conv emulation SNR {snr:.1f} dB vs f16 path"
 
 
+def test_gn_fp8_output_and_prequantized_conv():
+    """Producer-side quantization: GN writes e4m3 codes, conv consumes u8."""
+    g = torch.Generator().manual_seed(5)
+    ic, oc, h = 64, 64, 8
+    x = torch.randn(2, h, h, ic, generator=g).half()
+    gamma = torch.randn(ic, generator=g).float() * 0.3 + 1.0
+    beta = torch.randn(ic, generator=g).float() * 0.1
+    w = (torch.randn(oc, ic, 3, 3, generator=g) / math.sqrt(ic * 9)).half()
+    ref = ops.group_norm_silu_nhwc(x, 8, gamma, beta).float()
+    sa = ref.abs().max().item() / ops.FP8_MAX  # calibrated scale
+    q = ops.group_norm_silu_nhwc(x, 8, gamma, beta, fp8_scale=sa)
+    assert q.dtype == torch.uint8 and q.shape == x.shape
+    # decode matches the f16 GN output within e4m3 noise
+    dec = q.view(torch.float8_e4m3fn).to(torch.float32) * sa
+    snr = 10 * math.log10((ref ** 2).mean().item() /
+                          ((dec - ref) ** 2).mean().item() + 1e-20)
+    assert snr > 25, f"GN fp8 codes SNR {snr:.1f} dB"
+    # q8-input conv emulation == quantize-then-conv
+    y_q8 = ops.conv2d_fp8_nhwc(q, w, sa)
+    assert y_q8.dtype == torch.float16
+    y_f16 = ops.conv2d_fp8_nhwc(ref.half(), w, sa)
+    d = (y_q8.float() - y_f16.float()).abs().max().item()
+    # same codes modulo the f16 round-trip of ref -> tiny
+    assert d < 0.05, f"pre-quantized vs inline-quantized conv differ by {d}"
+
+
 def test_conv_fp8_emulation_with_affine_residual_cbias():
     g = torch.Generator().manual_seed(4)
     ic, oc, h = 64, 64, 8

@@ -103,7 +103,10 @@ def test_conv2d_fp8_vs_emulation(ic, oc, h, stride, pad, r):
                               stride=stride, padding=pad)
     assert y.shape == ref.shape
     s = snr_db(y.cpu(), ref)
-    assert s > 55, f"kernel vs emulation SNR {s:.1f} dB (accumulation-order only)"
+    # kernel encodes through an f16 intermediate (packed clamp+mul) — up to
+    # 2^-10 relative double-rounding vs the f32 emulation, plus f32
+    # accumulation-order differences
+    assert s > 40, f"kernel vs emulation SNR {s:.1f} dB"
     # and against the plain f32 conv: total quantization noise bound
     f32 = F.conv2d(x.permute(0, 3, 1, 2).float(), w.float(), b.float(),
                    stride=stride, padding=pad).permute(0, 2, 3, 1)
@@ -129,6 +132,29 @@ def test_conv2d_fp8_fused_epilogue_and_affine():
                               in_act=ops.ACT_SILU)
     s = snr_db(y.cpu(), ref)
     assert s > 55, f"fused fp8 epilogue SNR {s:.1f} dB vs emulation"
+
+
+@pytest.mark.parametrize("ic,oc,h", [(320, 320, 16), (64, 64, 64)])
+def test_gn_fp8_to_conv_q8_path(ic, oc, h):
+    """Producer-quantized fast path: GN writes codes, conv stages bytes."""
+    x = rnd(2, h, h, ic, seed=31)
+    g = torch.Generator().manual_seed(32)
+    gamma = (torch.randn(ic, generator=g).float() * 0.3 + 1.0).to(DEV)
+    beta = (torch.randn(ic, generator=g).float() * 0.1).to(DEV)
+    w = rnd(oc, ic, 3, 3, seed=33, scale=1.0 / math.sqrt(ic * 9))
+    gn = ops.group_norm_silu_nhwc(x, 32, gamma, beta)
+    sa = gn.float().abs().max().item() / ops.FP8_MAX
+    q = ops.group_norm_silu_nhwc(x, 32, gamma, beta, fp8_scale=sa)
+    assert q.dtype == torch.uint8
+    y = ops.conv2d_fp8_nhwc(q, w, sa)
+    # reference: CPU emulation fed the same decoded codes
+    ref = ops.conv2d_fp8_nhwc(q.cpu(), w.cpu(), sa)
+    s = snr_db(y.cpu(), ref)
+    assert s > 55, f"q8-path conv vs emulation SNR {s:.1f} dB"
+    # end-to-end noise vs the all-f16 path stays fp8-bounded
+    yf = ops.conv2d_nhwc(gn, w)
+    s2 = snr_db(y.cpu(), yf.cpu())
+    assert s2 > 20, f"GN-fp8->conv vs f16 path SNR {s2:.1f} dB"
 
 
 def test_conv2d_fp8_batch_gt1_and_graph_capture():

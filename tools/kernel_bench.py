@@ -56,12 +56,15 @@ def timeit(fn, iters):
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--iters", type=int, default=50)
+    p.add_argument("--batch", type=int, default=1,
+                   help="scale conv batch (the fbs>1 serving tier)")
     args = p.parse_args()
     assert torch.cuda.is_available(), "GPU microbench"
     dev = "cuda"
 
-    print("== conv2d (NHWC implicit-GEMM MFMA) ==")
-    for name, b, h, w, ic, oc, k, st in CONV_SHAPES:
+    print(f"== conv2d (NHWC implicit-GEMM MFMA) B={args.batch} ==")
+    for name, b0, h, w, ic, oc, k, st in CONV_SHAPES:
+        b = b0 * args.batch
         x = torch.randn(b, h, w, ic, device=dev).half()
         wt = (torch.randn(oc, ic, k, k, device=dev) * 0.02).half()
         bias = torch.randn(oc, device=dev).half()
@@ -72,16 +75,35 @@ def main():
         fl = 2.0 * b * ho * wo * oc * ic * k * k
         print(f"  {name:32s} {us:8.1f} us  {fl/us/1e6:7.1f} TF/s")
 
-    print("== conv2d fp8 (MX-scaled MFMA, per-OC weight scales) ==")
-    for name, b, h, w, ic, oc, k, st in CONV_SHAPES:
+    print(f"== conv2d fp8 (MX-scaled MFMA, per-OC weight scales) B={args.batch} ==")
+    for name, b0, h, w, ic, oc, k, st in CONV_SHAPES:
         if ic % 64 != 0:
             continue
+        b = b0 * args.batch
         x = torch.randn(b, h, w, ic, device=dev).half()
         wt = (torch.randn(oc, ic, k, k, device=dev) * 0.02).half()
         bias = torch.randn(oc, device=dev).float()
         pad = k // 2
         a_scale = x.float().abs().max().item() / ops.FP8_MAX
         fn = lambda: ops.conv2d_fp8_nhwc(x, wt, a_scale, None, stride=st,
+                                         padding=pad)
+        us = timeit(fn, args.iters)
+        ho, wo = (h + 2 * pad - k) // st + 1, (w + 2 * pad - k) // st + 1
+        fl = 2.0 * b * ho * wo * oc * ic * k * k
+        print(f"  {name:32s} {us:8.1f} us  {fl/us/1e6:7.1f} TF/s")
+
+    print(f"== conv2d fp8 pre-quantized input (GN-fp8 -> conv) B={args.batch} ==")
+    for name, b0, h, w, ic, oc, k, st in CONV_SHAPES:
+        if ic % 64 != 0:
+            continue
+        b = b0 * args.batch
+        x = torch.randn(b, h, w, ic, device=dev).half()
+        wt = (torch.randn(oc, ic, k, k, device=dev) * 0.02).half()
+        pad = k // 2
+        a_scale = x.float().abs().max().item() / ops.FP8_MAX
+        xq = ((x.float() / a_scale).clamp(-448, 448)
+              .to(torch.float8_e4m3fn).view(torch.uint8).contiguous())
+        fn = lambda: ops.conv2d_fp8_nhwc(xq, wt, a_scale, None, stride=st,
                                          padding=pad)
         us = timeit(fn, args.iters)
         ho, wo = (h + 2 * pad - k) // st + 1, (w + 2 * pad - k) // st + 1
