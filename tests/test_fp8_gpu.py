@@ -81,7 +81,7 @@ def test_cvt_overflow_nan_and_kernel_clamps():
     assert torch.isfinite(y.float()).all(), "kernel must clamp, not NaN"
     # the clamped row contributes at most 448*a_scale per element
     ref = ops.conv2d_fp8_nhwc(x.cpu(), w.cpu(), a_scale)
-    assert snr_db(y.cpu(), ref) > 50
+    assert snr_db(y.cpu(), ref) > 40  # f16-intermediate inline encode
 
 
 @pytest.mark.parametrize("ic,oc,h,stride,pad,r", [
