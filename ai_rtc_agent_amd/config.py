@@ -138,6 +138,17 @@ class EngineConfig:
     # second HIP stream (pipelined graphs; disabled automatically when the
     # similarity filter needs a per-frame decision)
     pipeline_overlap: bool = True
+    # fp8 (OCP e4m3) serving tier: the resnet GN->conv pairs quantize
+    # producer-side (GN writes e4m3 codes, conv runs on the MX-scaled MFMA
+    # at 2x the f16 rate with half the activation traffic). Per-layer
+    # activation scales calibrate on the first fp8_calib_frames real frames,
+    # then a UNet-forward quality gate must pass fp8_min_snr_db or the
+    # engine falls back to f16. fp16 stays the benchmarked default
+    # (MI355X-native addition; no reference counterpart).
+    use_fp8: bool = field(default_factory=lambda: _env_bool("AIRTC_FP8", False))
+    fp8_calib_frames: int = field(default_factory=lambda: _env_int("AIRTC_FP8_CALIB_FRAMES", 8))
+    fp8_margin: float = field(default_factory=lambda: _env_float("AIRTC_FP8_MARGIN", 1.5))
+    fp8_min_snr_db: float = field(default_factory=lambda: _env_float("AIRTC_FP8_MIN_SNR_DB", 16.0))
 
     @property
     def denoising_steps(self) -> int:

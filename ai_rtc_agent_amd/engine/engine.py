@@ -23,6 +23,10 @@ from __future__ import annotations
 
 from typing import Optional, Sequence
 
+import logging
+import math
+import os
+
 import torch
 
 from ..config import EngineConfig
@@ -143,6 +147,24 @@ class StreamDiffusionEngine:
         self._graph: Optional[torch.cuda.CUDAGraph] = None
         self._prepared = False
         self.prompt: str = cfg.prompt
+
+        # fp8 serving tier (opt-in): calibrate per-layer activation scales
+        # on the first real frames (eager), freeze + quality-gate, then the
+        # graph captures the GN-fp8 -> MX-MFMA conv path. fp16 stays the
+        # default; incompatible with the env-gated AIRTC_FUSE_GN experiment
+        # (that path bypasses the GN apply kernel entirely).
+        self._fp8_norms: list = []
+        self._fp8_calib_left = 0
+        self.fp8_active = False
+        self.fp8_snr_db: Optional[float] = None
+        if cfg.use_fp8 and os.environ.get("AIRTC_FUSE_GN") != "1":
+            from ..models.unet import fp8_eligible_norms
+
+            self._fp8_norms = fp8_eligible_norms(self.unet)
+            for nrm in self._fp8_norms:
+                nrm._fp8_calibrate = True
+                nrm._fp8_amax = 0.0
+            self._fp8_calib_left = max(1, cfg.fp8_calib_frames)
 
     def _load_weights_if_present(self, model_id: str) -> None:
         import glob
@@ -291,7 +313,7 @@ class StreamDiffusionEngine:
                     if hasattr(m, attr):
                         delattr(m, attr)
                 for t in list(m.parameters(recurse=False)):
-                    for attr in ("_airtc_wperm", "_airtc_b32", "_airtc_g32", "_airtc_w16", "_airtc_wpad32"):
+                    for attr in ("_airtc_wperm", "_airtc_b32", "_airtc_g32", "_airtc_w16", "_airtc_wpad32", "_airtc_wfp8", "_airtc_wdq"):
                         if hasattr(t, attr):
                             delattr(t, attr)
         from ..parallel.collectives import broadcast_engine_weights
@@ -376,6 +398,51 @@ class StreamDiffusionEngine:
         if cfg.cfg_type == "initialize" and self.rcfg.active:
             return torch.cat([self._ts_batch[: cfg.frame_buffer_size], self._ts_batch], dim=0)
         return self._ts_batch
+
+    @torch.no_grad()
+    def _fp8_quality_snr(self) -> float:
+        """SNR (dB) of the fp8 UNet forward vs the f16 forward on the same
+        inputs — a pure function of the UNet (no engine state mutated)."""
+        unet_in = self._unet_batch_input(self._init_noise)
+        ts = self._unet_batch_timesteps()
+        emb = self._unet_batch_embeds()
+        added = self._added_cond
+        cfg = self.cfg
+        if added is not None and (cfg.cfg_type in ("full", "initialize") and self.rcfg.active):
+            extra = added.shape[0] if cfg.cfg_type == "full" else cfg.frame_buffer_size
+            added = torch.cat([added[:extra], added], dim=0)
+        scales = [n._fp8_scale for n in self._fp8_norms]
+        for n in self._fp8_norms:
+            n._fp8_scale = None
+        ref = self.unet(unet_in, ts, emb, added_cond=added).float()
+        for n, s in zip(self._fp8_norms, scales):
+            n._fp8_scale = s
+        got = self.unet(unet_in, ts, emb, added_cond=added).float()
+        err = ((got - ref) ** 2).mean().item()
+        sig = (ref ** 2).mean().item()
+        return 10.0 * math.log10(sig / max(err, 1e-20))
+
+    def _fp8_freeze(self) -> None:
+        """End calibration: freeze per-layer scales, run the quality gate,
+        fall back to f16 if it fails (SURVEY.md §6 quality guard)."""
+        log = logging.getLogger("airtc.engine")
+        cfg = self.cfg
+        for nrm in self._fp8_norms:
+            nrm._fp8_calibrate = False
+            if nrm._fp8_amax > 0:
+                nrm._fp8_scale = nrm._fp8_amax * cfg.fp8_margin / 448.0
+        self.fp8_snr_db = self._fp8_quality_snr()
+        if self.fp8_snr_db < cfg.fp8_min_snr_db:
+            for nrm in self._fp8_norms:
+                nrm._fp8_scale = None
+            self.fp8_active = False
+            log.warning(
+                "fp8 quality gate FAILED (%.1f dB < %.1f dB) — serving f16",
+                self.fp8_snr_db, cfg.fp8_min_snr_db)
+        else:
+            self.fp8_active = True
+            log.info("fp8 tier active: %d layers, quality %.1f dB",
+                     len(self._fp8_norms), self.fp8_snr_db)
 
     @torch.no_grad()
     def _denoise_core(self) -> torch.Tensor:
@@ -551,7 +618,8 @@ class StreamDiffusionEngine:
                 return out[0] if squeeze else out
 
         with self.timers.stage("diffusion"):
-            if self.device.type == "cuda" and self.cfg.use_hip_graph:
+            if (self.device.type == "cuda" and self.cfg.use_hip_graph
+                    and not self._fp8_calib_left):
                 self._maybe_capture()
                 if self._pipelined:
                     cur = torch.cuda.current_stream()
@@ -578,6 +646,10 @@ class StreamDiffusionEngine:
             else:
                 self._frame_in.copy_(frame_u8)
                 out = self._step_core()
+                if self._fp8_calib_left:
+                    self._fp8_calib_left -= 1
+                    if self._fp8_calib_left == 0:
+                        self._fp8_freeze()
 
         self._prev_out = out
         self.timers.frame_done()
@@ -599,4 +671,12 @@ class StreamDiffusionEngine:
         return self(dummy)
 
     def stats(self) -> dict:
-        return self.timers.snapshot()
+        d = self.timers.snapshot()
+        if self.cfg.use_fp8:
+            d["fp8"] = {
+                "active": self.fp8_active,
+                "layers": len(self._fp8_norms),
+                "quality_snr_db": self.fp8_snr_db,
+                "calibrating": self._fp8_calib_left > 0,
+            }
+        return d

@@ -141,6 +141,14 @@ class Conv2d(nn.Module):
         in_affine: torch.Tensor | None = None,
         in_act: int = 0,
     ) -> torch.Tensor:
+        if x.dtype == torch.uint8:
+            # producer-quantized e4m3 codes (fp8 tier): the GN attached its
+            # scale to the tensor; run the MX-scaled MFMA conv
+            return ops.conv2d_fp8_nhwc(
+                x, self.weight, x._airtc_fp8_scale, self.bias, self.stride,
+                self.padding, act=(act if act is not None else (ops.ACT_SILU if fuse_silu else ops.ACT_NONE)),
+                residual=residual, channel_bias=channel_bias,
+            )
         return ops.conv2d_nhwc(
             x, self.weight, self.bias, self.stride, self.padding, fuse_silu,
             act=act, residual=residual, channel_bias=channel_bias,
@@ -149,6 +157,14 @@ class Conv2d(nn.Module):
 
 
 class GroupNormSiLU(nn.Module):
+    # fp8 serving tier state (set by the engine's calibrate/freeze pass on
+    # resnet norms whose consumer conv qualifies — see engine._fp8_freeze):
+    #   _fp8_calibrate: record the running absmax of the f16 output
+    #   _fp8_scale: emit e4m3 codes (u8) at this per-layer scale instead
+    _fp8_calibrate = False
+    _fp8_scale: float | None = None
+    _fp8_amax: float = 0.0
+
     def __init__(self, channels: int, groups: int = 32, eps: float = 1e-5, silu: bool = True):
         super().__init__()
         # SD uses 32 groups (channels >= 320 so Cg >= 10). For toy test dims
@@ -163,7 +179,16 @@ class GroupNormSiLU(nn.Module):
         self.bias = nn.Parameter(torch.zeros(channels))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return ops.group_norm_silu_nhwc(x, self.groups, self.weight, self.bias, self.eps, self.silu)
+        if self._fp8_scale is not None and not self._fp8_calibrate:
+            q = ops.group_norm_silu_nhwc(x, self.groups, self.weight,
+                                         self.bias, self.eps, self.silu,
+                                         fp8_scale=self._fp8_scale)
+            q._airtc_fp8_scale = self._fp8_scale
+            return q
+        y = ops.group_norm_silu_nhwc(x, self.groups, self.weight, self.bias, self.eps, self.silu)
+        if self._fp8_calibrate:
+            self._fp8_amax = max(self._fp8_amax, y.float().abs().max().item())
+        return y
 
     def coeffs(self, x: torch.Tensor) -> torch.Tensor:
         """(B, C, 2) input-affine pairs for the fused GN->conv path
@@ -358,6 +383,20 @@ class ResnetBlock(nn.Module):
         h = self.conv1(self.norm1(x), channel_bias=temb_b)
         skip = self.shortcut(x) if self.shortcut is not None else x
         return self.conv2(self.norm2(h), residual=skip)
+
+
+def fp8_eligible_norms(unet: nn.Module) -> list:
+    """The GNs whose consumer conv can run the fp8 MX-MFMA path: resnet
+    norm1/norm2 with conv input channels % 64 == 0 (all SD/SDXL resnets;
+    the engine calibrates + freezes per-layer scales on these)."""
+    out = []
+    for m in unet.modules():
+        if isinstance(m, ResnetBlock):
+            if m.conv1.weight.shape[1] % 64 == 0:
+                out.append(m.norm1)
+            if m.conv2.weight.shape[1] % 64 == 0:
+                out.append(m.norm2)
+    return out
 
 
 class Downsample(nn.Module):

@@ -312,7 +312,8 @@ def conv2d_fp8_nhwc(
         y = F.silu(y)
     elif act == ACT_RELU:
         y = F.relu(y)
-    return y.to(torch.float16 if x_q8 else x.dtype).contiguous()
+    # u8 input carries no output dtype — use the model (weight) dtype
+    return y.to(weight.dtype if x_q8 else x.dtype).contiguous()
 
 
 # ---------------------------------------------------------------------------

@@ -91,6 +91,10 @@ def main() -> None:
                    help="frame_buffer_size: frames batched per engine call "
                         "(multi-stream batched serving; aggregate FPS = "
                         "steps*fbs/elapsed). 1 = headline single-stream")
+    p.add_argument("--fp8", action="store_true",
+                   help="opt-in fp8 e4m3 tier (resnet GN->conv pairs on the "
+                        "MX-scaled MFMA; calibrated + quality-gated; the "
+                        "headline benchmark stays fp16)")
     p.add_argument(
         "--model", default="sd-turbo", choices=["sd-turbo", "sd15", "sdxl"],
         help="sd-turbo 1-step (headline) | sd15 4-step LCM+RCFG+filter "
@@ -132,6 +136,10 @@ def main() -> None:
         cfg.model_family = os.environ["AIRTC_BENCH_FAMILY"]
     if args.fbs > 1:
         cfg.frame_buffer_size = args.fbs
+    if args.fp8:
+        cfg.use_fp8 = True
+        # calibration runs eager on the first frames: keep it inside warmup
+        cfg.fp8_calib_frames = min(8, max(1, args.warmup - 2))
     eng = StreamDiffusionEngine(cfg)
     broadcast_engine_weights(eng)  # RCCL over xGMI; no-op at world=1
     eng.prepare()
@@ -236,7 +244,8 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "fp16" if use_cuda else "fp32",
+            "dtype": (("fp16+fp8-resnet" if getattr(eng, "fp8_active", False)
+                       else "fp16") if use_cuda else "fp32"),
             "data": "synthetic (random frames, random-init SD-Turbo-arch weights)",
             "config": {
                 "model": model_desc,

@@ -100,6 +100,60 @@ def test_gn_fp8_output_and_prequantized_conv():
     assert d < 0.05, f"pre-quantized vs inline-quantized conv differ by {d}"
 
 
+def test_engine_fp8_calibrate_freeze_gate():
+    """Engine lifecycle: calibrate on first frames -> freeze scales ->
+    quality gate -> fp8 active (CPU emulation of the same math)."""
+    from ai_rtc_agent_amd.config import EngineConfig
+    from ai_rtc_agent_amd.engine.engine import StreamDiffusionEngine
+
+    cfg = EngineConfig(
+        model_id="none", model_family="tiny", width=64, height=64,
+        t_index_list=[30], cfg_type="none", use_lcm_lora=False,
+        device="cpu", acceleration="eager", use_hip_graph=False,
+        use_fp8=True, fp8_calib_frames=2, fp8_min_snr_db=10.0,
+    )
+    cfg.similarity_filter.enabled = False
+    eng = StreamDiffusionEngine(cfg)
+    eng.prepare()
+    assert eng._fp8_calib_left == 2 and len(eng._fp8_norms) > 0
+    frame = torch.randint(0, 255, (64, 64, 3), dtype=torch.uint8)
+    eng(frame)
+    assert eng._fp8_calib_left == 1 and not eng.fp8_active
+    out = eng(frame)
+    assert eng._fp8_calib_left == 0
+    assert eng.fp8_snr_db is not None
+    assert eng.fp8_active, f"gate failed at {eng.fp8_snr_db:.1f} dB"
+    # scales frozen on eligible norms, and serving still works
+    assert all(n._fp8_scale is not None and n._fp8_scale > 0
+               for n in eng._fp8_norms)
+    out2 = eng(frame)
+    assert out2.shape == out.shape and out2.dtype == torch.uint8
+    st = eng.stats()
+    assert st["fp8"]["active"] and st["fp8"]["layers"] == len(eng._fp8_norms)
+
+
+def test_engine_fp8_gate_fallback():
+    """An absurd quality threshold must fall back to f16 serving."""
+    from ai_rtc_agent_amd.config import EngineConfig
+    from ai_rtc_agent_amd.engine.engine import StreamDiffusionEngine
+
+    cfg = EngineConfig(
+        model_id="none", model_family="tiny", width=64, height=64,
+        t_index_list=[30], cfg_type="none", use_lcm_lora=False,
+        device="cpu", acceleration="eager", use_hip_graph=False,
+        use_fp8=True, fp8_calib_frames=1, fp8_min_snr_db=200.0,
+    )
+    cfg.similarity_filter.enabled = False
+    eng = StreamDiffusionEngine(cfg)
+    eng.prepare()
+    frame = torch.randint(0, 255, (64, 64, 3), dtype=torch.uint8)
+    eng(frame)
+    assert not eng.fp8_active
+    assert all(n._fp8_scale is None for n in eng._fp8_norms)
+    out = eng(frame)  # still serves, in f16
+    assert out.dtype == torch.uint8
+
+
 def test_conv_fp8_emulation_with_affine_residual_cbias():
     g = torch.Generator().manual_seed(4)
     ic, oc, h = 64, 64, 8
