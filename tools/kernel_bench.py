@@ -110,8 +110,9 @@ def main():
         fl = 2.0 * b * ho * wo * oc * ic * k * k
         print(f"  {name:32s} {us:8.1f} us  {fl/us/1e6:7.1f} TF/s")
 
-    print("== attention (flash MFMA + tr_b16) ==")
-    for name, b, lq, lk, c, hds in ATTN_SHAPES:
+    print(f"== attention (flash MFMA + tr_b16) B={args.batch} ==")
+    for name, b0, lq, lk, c, hds in ATTN_SHAPES:
+        b = b0 * args.batch
         q = torch.randn(b, lq, c, device=dev).half()
         kk = torch.randn(b, lk, c, device=dev).half()
         v = torch.randn(b, lk, c, device=dev).half()
