@@ -65,8 +65,8 @@ def main():
         sa = a.abs().max().float() / 448.0
         sw = w.abs().max().float() / 448.0
         a8 = (a.float() / sa).clamp(-448, 448).to(torch.float8_e4m3fn)
-        w8t = ((w.float() / sw).clamp(-448, 448)
-               .to(torch.float8_e4m3fn).t().contiguous().t())
+        # mat2 must be column-major [K, N]: .t() of the row-major [N, K]
+        w8t = (w.float() / sw).clamp(-448, 448).to(torch.float8_e4m3fn).t()
         us16 = t(lambda: torch.nn.functional.linear(a, w))
         us8 = t(lambda: torch._scaled_mm(a8, w8t, scale_a=sa, scale_b=sw,
                                          out_dtype=torch.float16))
