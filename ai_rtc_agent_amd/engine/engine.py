@@ -458,11 +458,11 @@ class StreamDiffusionEngine:
             # per-frame input buffer
             self._img_proc = ops.preprocess_from_u8(self._frame_in, self.dtype)
             x0 = self.vae.encode(self._img_proc)
-            x_t0 = self.scheduler.add_noise(
+            x_t0 = ops.sched_add_noise(
                 x0,
                 self._init_noise[:fbs],
-                co["alpha_prod_t_sqrt"][:fbs],
-                co["beta_prod_t_sqrt"][:fbs],
+                co["alpha_f32"][:fbs],
+                co["beta_f32"][:fbs],
             ) if cfg.do_add_noise else x0
         else:  # txt2img: stage-0 input is pure noise
             x_t0 = self._init_noise[:fbs]
@@ -487,15 +487,17 @@ class StreamDiffusionEngine:
             )
         eps = self.unet(unet_in, unet_ts, unet_emb, added_cond=added, control=control)
         eps = self.rcfg.apply(eps, fbs)
-        denoised = self.scheduler.step_batch(eps, x_t, co)
+        # fused LCM step (one kernel; same math as scheduler.step_batch)
+        denoised = ops.sched_blend(x_t, eps, co["alpha_f32"], co["beta_f32"],
+                                   co["c_out_f32"], co["c_skip_f32"])
 
         if denoised.shape[0] > fbs:
             # shift: stage i output -> stage i+1 input at tau_{i+1}
-            nxt = self.scheduler.add_noise(
-                denoised[:-fbs],
+            nxt = ops.sched_add_noise(
+                denoised[:-fbs].contiguous(),
                 self._init_noise[fbs:],
-                co["alpha_prod_t_sqrt"][fbs:],
-                co["beta_prod_t_sqrt"][fbs:],
+                co["alpha_f32"][fbs:],
+                co["beta_f32"][fbs:],
             ) if cfg.do_add_noise else denoised[:-fbs]
             self._x_t_buffer.copy_(nxt)
 

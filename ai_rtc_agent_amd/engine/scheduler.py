@@ -82,6 +82,12 @@ class StreamScheduler:
         c_skip = sigma_data ** 2 / (scaled_t ** 2 + sigma_data ** 2)
         c_out = scaled_t / (scaled_t ** 2 + sigma_data ** 2).sqrt()
 
+        def ri32(x: torch.Tensor) -> torch.Tensor:
+            # flat f32 copies for the fused scheduler kernels
+            # (ops.sched_add_noise / ops.sched_blend)
+            return x.repeat_interleave(frame_buffer_size).to(
+                device=device, dtype=torch.float32).contiguous()
+
         return {
             "sub_timesteps": subs,
             "sub_timesteps_tensor": ri(t.to(torch.float32)).to(torch.long),
@@ -89,6 +95,10 @@ class StreamScheduler:
             "beta_prod_t_sqrt": ri(beta_sqrt).view(-1, 1, 1, 1),
             "c_skip": ri(c_skip).view(-1, 1, 1, 1),
             "c_out": ri(c_out).view(-1, 1, 1, 1),
+            "alpha_f32": ri32(alpha_sqrt),
+            "beta_f32": ri32(beta_sqrt),
+            "c_skip_f32": ri32(c_skip),
+            "c_out_f32": ri32(c_out),
         }
 
     # -- core math ----------------------------------------------------------

@@ -385,6 +385,43 @@ pybind11::dict vcn_probe() {
   return d;
 }
 
+// fused LCM scheduler math (elementwise.hip): one kernel per scheduler op
+torch::Tensor sched_add_noise(torch::Tensor x0, torch::Tensor noise,
+                              torch::Tensor a, torch::Tensor bt) {
+  CHECK_IN(x0);
+  CHECK_IN(noise);
+  TORCH_CHECK(x0.dtype() == torch::kHalf && noise.dtype() == torch::kHalf);
+  TORCH_CHECK(a.dtype() == torch::kFloat && bt.dtype() == torch::kFloat);
+  const long B = x0.size(0);
+  TORCH_CHECK(a.numel() == B && bt.numel() == B, "coeffs must be (B,)");
+  const long per_b = x0.numel() / B;
+  TORCH_CHECK(per_b % 8 == 0, "row size must be a multiple of 8");
+  auto out = torch::empty_like(x0);
+  airtc_sched_add_noise(h_ptr(x0), h_ptr(noise), a.data_ptr<float>(),
+                        bt.data_ptr<float>(), h_ptr_mut(out), per_b,
+                        x0.numel(), cur_stream());
+  return out;
+}
+
+torch::Tensor sched_blend(torch::Tensor xt, torch::Tensor eps,
+                          torch::Tensor a, torch::Tensor bt,
+                          torch::Tensor c_out, torch::Tensor c_skip) {
+  CHECK_IN(xt);
+  CHECK_IN(eps);
+  TORCH_CHECK(xt.dtype() == torch::kHalf && eps.dtype() == torch::kHalf);
+  const long B = xt.size(0);
+  TORCH_CHECK(a.numel() == B && bt.numel() == B && c_out.numel() == B &&
+              c_skip.numel() == B, "coeffs must be (B,)");
+  const long per_b = xt.numel() / B;
+  TORCH_CHECK(per_b % 8 == 0, "row size must be a multiple of 8");
+  auto out = torch::empty_like(xt);
+  airtc_sched_blend(h_ptr(xt), h_ptr(eps), a.data_ptr<float>(),
+                    bt.data_ptr<float>(), c_out.data_ptr<float>(),
+                    c_skip.data_ptr<float>(), h_ptr_mut(out), per_b,
+                    xt.numel(), cur_stream());
+  return out;
+}
+
 // fp8 conv: MX-scaled MFMA path (conv2d_fp8.hip). Same surface as conv2d
 // but weights are pre-quantized e4m3 bytes + per-OC dequant scales, and the
 // activation scale rides along as a scalar.
@@ -522,6 +559,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("geglu", &geglu);
   m.def("add_act", &add_act);
   m.def("upsample2x", &upsample2x);
+  m.def("sched_add_noise", &sched_add_noise,
+        "fused q(x_t|x0) sample: a*x0 + b*noise (per-batch-row coeffs)");
+  m.def("sched_blend", &sched_blend,
+        "fused LCM step: c_out*(x_t - b*eps)/a + c_skip*x_t");
   m.def("preprocess_u8", &preprocess_u8);
   m.def("postprocess_u8", &postprocess_u8);
   m.def("conv2d_fp8", &conv2d_fp8,

@@ -18,6 +18,15 @@ void airtc_add_act_f16(const uint16_t* a, const uint16_t* b, uint16_t* out,
                        long n, int act, hipStream_t s);
 void airtc_upsample2x_f16(const uint16_t* in, uint16_t* out, int B, int H,
                           int W, int C, hipStream_t s);
+// fused LCM scheduler math (one kernel each; coefficients are per-batch-row
+// f32 arrays of B entries, per_b = elements per batch row)
+void airtc_sched_add_noise(const uint16_t* x0, const uint16_t* noise,
+                           const float* a, const float* bt, uint16_t* out,
+                           long per_b, long n, hipStream_t s);
+void airtc_sched_blend(const uint16_t* xt, const uint16_t* eps,
+                       const float* a, const float* bt, const float* c_out,
+                       const float* c_skip, uint16_t* out, long per_b, long n,
+                       hipStream_t s);
 
 // norms ---------------------------------------------------------------------
 int airtc_group_norm_nchunk(int B, int G);

@@ -100,6 +100,25 @@ def test_gn_fp8_output_and_prequantized_conv():
     assert d < 0.05, f"pre-quantized vs inline-quantized conv differ by {d}"
 
 
+def test_sched_fused_matches_scheduler_math():
+    """CPU parity of the fused scheduler ops vs StreamScheduler methods."""
+    from ai_rtc_agent_amd.engine.scheduler import StreamScheduler
+
+    sch = StreamScheduler(num_inference_steps=50)
+    co = sch.coefficients([18, 26, 35, 45], 1, torch.device("cpu"), torch.float32)
+    g = torch.Generator().manual_seed(0)
+    x0 = torch.randn(4, 8, 8, 4, generator=g)
+    nz = torch.randn(4, 8, 8, 4, generator=g)
+    eps = torch.randn(4, 8, 8, 4, generator=g)
+    got = ops.sched_add_noise(x0, nz, co["alpha_f32"], co["beta_f32"])
+    ref = sch.add_noise(x0, nz, co["alpha_prod_t_sqrt"], co["beta_prod_t_sqrt"])
+    assert torch.allclose(got, ref, atol=1e-5)
+    got2 = ops.sched_blend(x0, eps, co["alpha_f32"], co["beta_f32"],
+                           co["c_out_f32"], co["c_skip_f32"])
+    ref2 = sch.step_batch(eps, x0, co)
+    assert torch.allclose(got2, ref2, atol=1e-5)
+
+
 def test_engine_fp8_calibrate_freeze_gate():
     """Engine lifecycle: calibrate on first frames -> freeze scales ->
     quality gate -> fp8 active (CPU emulation of the same math)."""

@@ -180,3 +180,27 @@ def test_group_norm_coeffs_vs_cpu():
     got = ops.group_norm_coeffs(x, 32, gamma, beta, 1e-5)
     ref = ops.group_norm_coeffs(x.cpu(), 32, gamma.cpu(), beta.cpu(), 1e-5)
     assert_close(got, ref.to(DEV), 2e-2, "gn coeffs")
+
+
+def test_sched_fused_kernels_vs_fp32():
+    """Fused scheduler kernels (one launch each) vs the fp32 scheduler math."""
+    from ai_rtc_agent_amd.engine.scheduler import StreamScheduler
+
+    sch = StreamScheduler(num_inference_steps=50)
+    co = sch.coefficients([18, 26, 35, 45], 2, torch.device(DEV), torch.float32)
+    B = 8
+    x0 = rnd(B, 16, 16, 4, seed=70)
+    nz = rnd(B, 16, 16, 4, seed=71)
+    eps = rnd(B, 16, 16, 4, seed=72)
+    got = ops.sched_add_noise(x0, nz, co["alpha_f32"], co["beta_f32"])
+    ref = (co["alpha_f32"].view(-1, 1, 1, 1) * x0.float()
+           + co["beta_f32"].view(-1, 1, 1, 1) * nz.float())
+    assert_close(got, ref, 2e-3, "sched_add_noise")
+    got2 = ops.sched_blend(x0, eps, co["alpha_f32"], co["beta_f32"],
+                           co["c_out_f32"], co["c_skip_f32"])
+    a = co["alpha_f32"].view(-1, 1, 1, 1)
+    b = co["beta_f32"].view(-1, 1, 1, 1)
+    x0p = (x0.float() - b * eps.float()) / a
+    ref2 = co["c_out_f32"].view(-1, 1, 1, 1) * x0p \
+        + co["c_skip_f32"].view(-1, 1, 1, 1) * x0.float()
+    assert_close(got2, ref2, 3e-2, "sched_blend")
