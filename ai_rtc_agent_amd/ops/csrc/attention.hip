@@ -209,9 +209,10 @@ __global__ __launch_bounds__(256) void attention_kernel(
       }
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
-#pragma unroll
-        for (int off = 1; off < 16; off <<= 1)
-          mnew[j] = fmaxf(mnew[j], __shfl_xor(mnew[j], off, 64));
+        // 16-lane row reductions on the VALU pipe (DPP) — __shfl_xor's
+        // ds_bpermute contends with this kernel's heavy LDS traffic
+        // (measured: 608 bpermutes/iteration before)
+        mnew[j] = quarter_reduce(mnew[j], MaxOp());
         const float mn = fmaxf(m_i[qi][j], mnew[j]);
         const float alpha = __expf(m_i[qi][j] - mn);
         float rs = 0.f;
@@ -220,8 +221,7 @@ __global__ __launch_bounds__(256) void attention_kernel(
           p[nf][j] = __expf(p[nf][j] - mn);
           rs += p[nf][j];
         }
-#pragma unroll
-        for (int off = 1; off < 16; off <<= 1) rs += __shfl_xor(rs, off, 64);
+        rs = quarter_reduce(rs, SumOp());
         l_i[qi][j] = l_i[qi][j] * alpha + rs;
         m_i[qi][j] = mn;
 #pragma unroll

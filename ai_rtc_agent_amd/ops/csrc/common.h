@@ -53,6 +53,29 @@ __device__ __forceinline__ float wave_reduce(float v, Op op) {
   return v;
 }
 
+// DPP cross-lane move (VALU-pipe, ~free) — __shfl_xor lowers to
+// ds_bpermute (LDS pipe), which contends with staging/fragment traffic in
+// LDS-heavy kernels. ctrl: 0x00-0xFF quad_perm, 0x140 row_mirror,
+// 0x141 row_half_mirror (row = 16 lanes on CDNA).
+template <int CTRL>
+__device__ __forceinline__ float dpp_mov_f32(float v) {
+  int i = __builtin_bit_cast(int, v);
+  i = __builtin_amdgcn_update_dpp(0, i, CTRL, 0xF, 0xF, true);
+  return __builtin_bit_cast(float, i);
+}
+
+// reduce over each 16-lane quarter-wave entirely on the VALU pipe:
+// quad xor1, quad xor2, half-mirror (joins the quads of each 8-group),
+// row-mirror (joins the two 8-groups of the 16-row)
+template <typename Op>
+__device__ __forceinline__ float quarter_reduce(float v, Op op) {
+  v = op(v, dpp_mov_f32<0xB1>(v));   // quad_perm [1,0,3,2] = xor 1
+  v = op(v, dpp_mov_f32<0x4E>(v));   // quad_perm [2,3,0,1] = xor 2
+  v = op(v, dpp_mov_f32<0x141>(v));  // row_half_mirror
+  v = op(v, dpp_mov_f32<0x140>(v));  // row_mirror
+  return v;
+}
+
 struct SumOp {
   __device__ float operator()(float a, float b) const { return a + b; }
 };
