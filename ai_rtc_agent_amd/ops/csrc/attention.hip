@@ -30,9 +30,9 @@
 
 #define KVT 64  // kv tile
 #define QT 64   // q rows per workgroup
-#define PPITCH (KVT + 8)
 
 typedef __attribute__((__vector_size__(2 * sizeof(unsigned)))) unsigned u32x2;
+typedef __attribute__((__vector_size__(4 * sizeof(_Float16)))) _Float16 f16x4;
 typedef const __attribute__((address_space(3))) f16* lds_cptr;
 
 __device__ __forceinline__ unsigned lds_addr(const f16* p) {
@@ -79,9 +79,14 @@ __global__ __launch_bounds__(256) void attention_kernel(
   constexpr int KPITCH = D + 8;
   constexpr int D8 = D / 8, D32 = D / 32, D16 = D / 16;
   constexpr int QTILE = 64 * QF;  // q rows per workgroup
+  // P bridge tile is stored TRANSPOSED [kv][q]: the C-fragment's 4 j-values
+  // are 4 consecutive q rows of one kv column, so the write packs into ONE
+  // b64 (was 16 scalar f16 writes per (qi, tile)); the PV A-fragment comes
+  // back out via the same ds_read_b64_tr_b16 gather the V tile uses.
+  constexpr int PT_PITCH = QF * 16 + 8;  // +16B pad: conflict-free b64 law
   __shared__ f16 ldsK[KVT * KPITCH];
   __shared__ f16 ldsV[KVT * KPITCH];  // row-major like K; PV reads via tr_b16
-  __shared__ f16 ldsP[4 * QF * 16 * PPITCH];
+  __shared__ f16 ldsP[4 * KVT * PT_PITCH];
 
   // XCD-chunked mapping (env AIRTC_ATTN_XCD): each XCD walks a CONTIGUOUS
   // (head, q-tile) range, so one XCD's in-flight blocks share 1-2 heads'
@@ -134,7 +139,7 @@ __global__ __launch_bounds__(256) void attention_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) { m_i[qi][j] = -1e30f; l_i[qi][j] = 0.f; }
 
-  f16* myP = &ldsP[wid * QF * 16 * PPITCH];
+  f16* myP = &ldsP[wid * KVT * PT_PITCH];  // per-wave transposed P tile
   const int fcol = (lane >> 4) * 8;
 
   // T14 async-stage split: each thread owns LOADS_PT row-chunks of the K and
@@ -228,29 +233,35 @@ __global__ __launch_bounds__(256) void attention_kernel(
         for (int f = 0; f < D16; ++f) o_acc[qi][f][j] *= alpha;
       }
 
-      // ---- P -> LDS (C-frag layout -> A-frag layout bridge) ----
+      // ---- P -> LDS transposed [kv][q] (C-frag -> A-frag bridge): the 4
+      // j-values are consecutive q rows of kv column nf*16+(lane&15) ----
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf)
+      for (int nf = 0; nf < 4; ++nf) {
+        f16x4 pv;
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
-          myP[(qi * 16 + (lane >> 4) * 4 + j) * PPITCH + nf * 16 + (lane & 15)] =
-              (f16)p[nf][j];
+        for (int j = 0; j < 4; ++j) pv[j] = (f16)p[nf][j];
+        *reinterpret_cast<f16x4*>(
+            &myP[(nf * 16 + (lane & 15)) * PT_PITCH + qi * 16 +
+                 (lane >> 4) * 4]) = pv;
+      }
     }
     // same-wave LDS write->read: wait for the writes, keep reads below
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
-    // ---- O += P V  (V B-fragments via hardware transpose read) ----
+    // ---- O += P V  (BOTH operands via hardware transpose reads: V from
+    // its row-major tile, P from the transposed bridge tile) ----
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
+      f16x8 afrag[QF];
+#pragma unroll
+      for (int qi = 0; qi < QF; ++qi)
+        afrag[qi] = tr_bfrag(myP, PT_PITCH, ks * 32, qi * 16, lane);
 #pragma unroll
       for (int f = 0; f < D16; ++f) {
         f16x8 bfrag = tr_bfrag(ldsV, KPITCH, ks * 32, f * 16, lane);
 #pragma unroll
-        for (int qi = 0; qi < QF; ++qi) {
-          f16x8 afrag = *reinterpret_cast<const f16x8*>(
-              &myP[(qi * 16 + (lane & 15)) * PPITCH + ks * 32 + fcol]);
-          o_acc[qi][f] = mfma16x16x32(afrag, bfrag, o_acc[qi][f]);
-        }
+        for (int qi = 0; qi < QF; ++qi)
+          o_acc[qi][f] = mfma16x16x32(afrag[qi], bfrag, o_acc[qi][f]);
       }
     }
   }
