@@ -45,13 +45,11 @@ __device__ __forceinline__ float apply_act(float v, int act) {
   return v;
 }
 
-// butterfly reduce over all 64 lanes of a wave
+// butterfly reduce over all 64 lanes of a wave (defined below, after the
+// DPP helpers: the first 4 steps run on the VALU pipe, only the two
+// cross-row steps pay the ds_bpermute cost)
 template <typename Op>
-__device__ __forceinline__ float wave_reduce(float v, Op op) {
-#pragma unroll
-  for (int off = 32; off > 0; off >>= 1) v = op(v, __shfl_xor(v, off, 64));
-  return v;
-}
+__device__ __forceinline__ float wave_reduce(float v, Op op);
 
 // DPP cross-lane move (VALU-pipe, ~free) — __shfl_xor lowers to
 // ds_bpermute (LDS pipe), which contends with staging/fragment traffic in
@@ -73,6 +71,14 @@ __device__ __forceinline__ float quarter_reduce(float v, Op op) {
   v = op(v, dpp_mov_f32<0x4E>(v));   // quad_perm [2,3,0,1] = xor 2
   v = op(v, dpp_mov_f32<0x141>(v));  // row_half_mirror
   v = op(v, dpp_mov_f32<0x140>(v));  // row_mirror
+  return v;
+}
+
+template <typename Op>
+__device__ __forceinline__ float wave_reduce(float v, Op op) {
+  v = quarter_reduce(v, op);  // lanes within each 16-row: VALU-pipe DPP
+  v = op(v, __shfl_xor(v, 16, 64));
+  v = op(v, __shfl_xor(v, 32, 64));
   return v;
 }
 
