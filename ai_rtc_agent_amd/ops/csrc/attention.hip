@@ -70,7 +70,7 @@ __device__ __forceinline__ f16x8 tr_bfrag(const f16* tile, int pitch,
   return cvt.f8;
 }
 
-template <int D, int QF>  // QF = 16-row q-fragments per wave (1 or 2)
+template <int D, int QF, int KT = KVT>  // QF = q-frags/wave; KT = kv tile
 __global__ __launch_bounds__(256) void attention_kernel(
     const f16* __restrict__ q, const f16* __restrict__ k,
     const f16* __restrict__ v, f16* __restrict__ out, int H, int Lq, int Lk,
@@ -78,15 +78,16 @@ __global__ __launch_bounds__(256) void attention_kernel(
     long o_sb, long o_sh, long o_row, float scale, int n_qt, int xcd_map) {
   constexpr int KPITCH = D + 8;
   constexpr int D8 = D / 8, D32 = D / 32, D16 = D / 16;
+  constexpr int NF = KT / 16;     // 16-key S fragments per tile
   constexpr int QTILE = 64 * QF;  // q rows per workgroup
   // P bridge tile is stored TRANSPOSED [kv][q]: the C-fragment's 4 j-values
   // are 4 consecutive q rows of one kv column, so the write packs into ONE
   // b64 (was 16 scalar f16 writes per (qi, tile)); the PV A-fragment comes
   // back out via the same ds_read_b64_tr_b16 gather the V tile uses.
   constexpr int PT_PITCH = QF * 16 + 8;  // +16B pad: conflict-free b64 law
-  __shared__ f16 ldsK[KVT * KPITCH];
-  __shared__ f16 ldsV[KVT * KPITCH];  // row-major like K; PV reads via tr_b16
-  __shared__ f16 ldsP[4 * KVT * PT_PITCH];
+  __shared__ f16 ldsK[KT * KPITCH];
+  __shared__ f16 ldsV[KT * KPITCH];  // row-major like K; PV reads via tr_b16
+  __shared__ f16 ldsP[4 * KT * PT_PITCH];
 
   // XCD-chunked mapping (env AIRTC_ATTN_XCD): each XCD walks a CONTIGUOUS
   // (head, q-tile) range, so one XCD's in-flight blocks share 1-2 heads'
@@ -139,13 +140,13 @@ __global__ __launch_bounds__(256) void attention_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) { m_i[qi][j] = -1e30f; l_i[qi][j] = 0.f; }
 
-  f16* myP = &ldsP[wid * KVT * PT_PITCH];  // per-wave transposed P tile
+  f16* myP = &ldsP[wid * KT * PT_PITCH];  // per-wave transposed P tile
   const int fcol = (lane >> 4) * 8;
 
   // T14 async-stage split: each thread owns LOADS_PT row-chunks of the K and
   // V tiles; tile t+1's global loads are issued right after tile t's LDS
   // image is published, so HBM latency hides under t's MFMA/softmax work.
-  constexpr int LOADS_PT = (KVT * D8 + 255) / 256;
+  constexpr int LOADS_PT = (KT * D8 + 255) / 256;
   f16x8 regK[LOADS_PT], regV[LOADS_PT];
   int st_row[LOADS_PT], st_c8[LOADS_PT];
 #pragma unroll
@@ -158,7 +159,7 @@ __global__ __launch_bounds__(256) void attention_kernel(
   auto stage_load = [&](int t0) {
 #pragma unroll
     for (int i = 0; i < LOADS_PT; ++i) {
-      if (st_row[i] >= KVT) continue;
+      if (st_row[i] >= KT) continue;
       int krow = t0 + st_row[i];
       if (krow >= Lk) krow = Lk - 1;  // masked later
       regK[i] = *reinterpret_cast<const f16x8*>(&kb[(long)krow * k_row + st_c8[i]]);
@@ -167,21 +168,21 @@ __global__ __launch_bounds__(256) void attention_kernel(
   };
 
   stage_load(0);
-  for (int t0 = 0; t0 < Lk; t0 += KVT) {
+  for (int t0 = 0; t0 < Lk; t0 += KT) {
     if (t0) __syncthreads();  // previous tile's reads complete
 #pragma unroll
     for (int i = 0; i < LOADS_PT; ++i) {
-      if (st_row[i] >= KVT) continue;
+      if (st_row[i] >= KT) continue;
       *reinterpret_cast<f16x8*>(&ldsK[st_row[i] * KPITCH + st_c8[i]]) = regK[i];
       *reinterpret_cast<f16x8*>(&ldsV[st_row[i] * KPITCH + st_c8[i]]) = regV[i];
     }
     __syncthreads();
-    if (t0 + KVT < Lk) stage_load(t0 + KVT);
+    if (t0 + KT < Lk) stage_load(t0 + KT);
 
-    // ---- S = scale * Q K^T  (QF x 4 col fragments of 16) ----
-    f32x4 sfrag[QF][4];
+    // ---- S = scale * Q K^T  (QF x NF col fragments of 16) ----
+    f32x4 sfrag[QF][NF];
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf) {
+    for (int nf = 0; nf < NF; ++nf) {
       const f16* kr = &ldsK[(nf * 16 + (lane & 15)) * KPITCH + fcol];
 #pragma unroll
       for (int ds = 0; ds < D32; ++ds) {
@@ -197,12 +198,12 @@ __global__ __launch_bounds__(256) void attention_kernel(
     // ---- online softmax update (rows lane-local), per q-fragment ----
 #pragma unroll
     for (int qi = 0; qi < QF; ++qi) {
-      float p[4][4];  // [nf][reg j]
+      float p[NF][4];  // [nf][reg j]
       float mnew[4];
 #pragma unroll
       for (int j = 0; j < 4; ++j) mnew[j] = -1e30f;
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf) {
+      for (int nf = 0; nf < NF; ++nf) {
         const int kcol = t0 + nf * 16 + (lane & 15);
         const bool valid = kcol < Lk;
 #pragma unroll
@@ -222,7 +223,7 @@ __global__ __launch_bounds__(256) void attention_kernel(
         const float alpha = __expf(m_i[qi][j] - mn);
         float rs = 0.f;
 #pragma unroll
-        for (int nf = 0; nf < 4; ++nf) {
+        for (int nf = 0; nf < NF; ++nf) {
           p[nf][j] = __expf(p[nf][j] - mn);
           rs += p[nf][j];
         }
@@ -236,7 +237,7 @@ __global__ __launch_bounds__(256) void attention_kernel(
       // ---- P -> LDS transposed [kv][q] (C-frag -> A-frag bridge): the 4
       // j-values are consecutive q rows of kv column nf*16+(lane&15) ----
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf) {
+      for (int nf = 0; nf < NF; ++nf) {
         f16x4 pv;
 #pragma unroll
         for (int j = 0; j < 4; ++j) pv[j] = (f16)p[nf][j];
@@ -251,7 +252,7 @@ __global__ __launch_bounds__(256) void attention_kernel(
     // ---- O += P V  (BOTH operands via hardware transpose reads: V from
     // its row-major tile, P from the transposed bridge tile) ----
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
+    for (int ks = 0; ks < KT / 32; ++ks) {
       f16x8 afrag[QF];
 #pragma unroll
       for (int qi = 0; qi < QF; ++qi)
@@ -304,6 +305,14 @@ extern "C" void airtc_attention(const uint16_t* q, const uint16_t* k,
     const char* e = getenv("AIRTC_ATTN_XCD");
     xcd_map_env = e ? atoi(e) : 0;
   }
+  // KVT=128 tiles (half the barriers and m/l bookkeeping per key) —
+  // env-gated experiment AIRTC_ATTN_KVT=128: LDS grows ~2x (occupancy
+  // cliff), only meaningful for Lk >= 128.
+  static int kvt_env = -1;
+  if (kvt_env < 0) {
+    const char* e = getenv("AIRTC_ATTN_KVT");
+    kvt_env = e ? atoi(e) : 64;
+  }
   int qf = 1;
   if (qf_force > 0) qf = qf_force;
   const int n_qt = ceil_div(Lq, 64 * qf);
@@ -313,10 +322,15 @@ extern "C" void airtc_attention(const uint16_t* q, const uint16_t* k,
   const f16* kp = reinterpret_cast<const f16*>(k);
   const f16* vp = reinterpret_cast<const f16*>(v);
   f16* op = reinterpret_cast<f16*>(out);
+#define LAUNCH_T(D, QF, KT)                                                 \
+  hipLaunchKernelGGL((attention_kernel<D, QF, KT>), grid, dim3(256), 0, s,  \
+                     qp, kp, vp, op, H, Lq, Lk, q_sb, q_sh, q_row, k_sb,    \
+                     k_sh, k_row, o_sb, o_sh, o_row, scale, n_qt, xm)
 #define LAUNCH(D, QF)                                                       \
-  hipLaunchKernelGGL((attention_kernel<D, QF>), grid, dim3(256), 0, s, qp,  \
-                     kp, vp, op, H, Lq, Lk, q_sb, q_sh, q_row, k_sb, k_sh,  \
-                     k_row, o_sb, o_sh, o_row, scale, n_qt, xm)
+  do {                                                                      \
+    if (QF == 1 && kvt_env == 128 && Lk >= 128) LAUNCH_T(D, 1, 128);        \
+    else LAUNCH_T(D, QF, 64);                                               \
+  } while (0)
   switch (d * 10 + qf) {
     case 321: LAUNCH(32, 1); break;
     case 322: LAUNCH(32, 2); break;
@@ -331,4 +345,5 @@ extern "C" void airtc_attention(const uint16_t* q, const uint16_t* k,
     default: break;  // host wrapper guarantees one of the above
   }
 #undef LAUNCH
+#undef LAUNCH_T
 }
