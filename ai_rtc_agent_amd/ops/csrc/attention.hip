@@ -305,14 +305,17 @@ extern "C" void airtc_attention(const uint16_t* q, const uint16_t* k,
     const char* e = getenv("AIRTC_ATTN_XCD");
     xcd_map_env = e ? atoi(e) : 0;
   }
-  // KVT=128 tiles (half the barriers and m/l bookkeeping per key) —
-  // env-gated experiment AIRTC_ATTN_KVT=128: LDS grows ~2x (occupancy
-  // cliff), only meaningful for Lk >= 128.
+  // KVT=128 tiles: half the barriers and m/l bookkeeping per key. Measured
+  // (same-box A/B): L4096 B=1 115->92.5us (+25%), B=8 479->454us (+5.5%);
+  // L1024 +24%/+4%; L256 B=8 regresses 7% (occupancy cliff bites when the
+  // grid is small-per-key-work). DEFAULT: auto = 128 for Lk >= 1024, 64
+  // below; AIRTC_ATTN_KVT=64|128 forces.
   static int kvt_env = -1;
   if (kvt_env < 0) {
     const char* e = getenv("AIRTC_ATTN_KVT");
-    kvt_env = e ? atoi(e) : 64;
+    kvt_env = e ? atoi(e) : 0;  // 0 = auto
   }
+  const int kvt_sel = kvt_env ? kvt_env : (Lk >= 1024 ? 128 : 64);
   int qf = 1;
   if (qf_force > 0) qf = qf_force;
   const int n_qt = ceil_div(Lq, 64 * qf);
@@ -328,7 +331,7 @@ extern "C" void airtc_attention(const uint16_t* q, const uint16_t* k,
                      k_sh, k_row, o_sb, o_sh, o_row, scale, n_qt, xm)
 #define LAUNCH(D, QF)                                                       \
   do {                                                                      \
-    if (QF == 1 && kvt_env == 128 && Lk >= 128) LAUNCH_T(D, 1, 128);        \
+    if (QF == 1 && kvt_sel == 128 && Lk >= 128) LAUNCH_T(D, 1, 128);        \
     else LAUNCH_T(D, QF, 64);                                               \
   } while (0)
   switch (d * 10 + qf) {
