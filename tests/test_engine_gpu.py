@@ -220,3 +220,33 @@ def test_rccl_two_ranks():
     d = json.loads(line)
     assert d["n_gpus"] == 2
     assert d["value"] > 0
+
+
+def test_fp8_tier_end_to_end_on_gpu():
+    """fp8 serving tier on real hardware: calibrate on the first frames
+    (eager), freeze + quality gate, capture the GN-fp8 -> MX-MFMA conv
+    graph, and keep serving. Output must track the f16 engine closely
+    (same seed/weights; fp8 noise only)."""
+    cfg16 = sd_turbo_config(device="cuda")
+    e16 = StreamDiffusionEngine(cfg16)
+    e16.prepare()
+    cfg8 = sd_turbo_config(device="cuda", use_fp8=True, fp8_calib_frames=2)
+    e8 = StreamDiffusionEngine(cfg8)
+    e8.prepare()
+    outs16, outs8 = [], []
+    for i in range(5):
+        f = frame(seed=40 + i)
+        o16 = e16(f)
+        e16.sync_output()
+        outs16.append(o16.float().clone())
+        o8 = e8(f)
+        e8.sync_output()
+        outs8.append(o8.float().clone())
+    assert e8.fp8_active, f"gate failed: {e8.fp8_snr_db} dB"
+    assert e8.fp8_snr_db > cfg8.fp8_min_snr_db
+    # after capture (frame >= calib+1) the fp8 graph serves; outputs are
+    # u8 images whose difference from f16 is bounded quantization noise
+    d = (outs8[-1] - outs16[-1]).abs().mean().item()
+    assert d < 24.0, f"fp8 output drifted {d} u8 steps from f16"
+    st = e8.stats()
+    assert st["fp8"]["active"] and not st["fp8"]["calibrating"]
