@@ -157,13 +157,21 @@ class StreamDiffusionEngine:
         self._fp8_calib_left = 0
         self.fp8_active = False
         self.fp8_snr_db: Optional[float] = None
+        self._fp8_vae: dict = {"convs": [], "outs": []}
         if cfg.use_fp8 and os.environ.get("AIRTC_FUSE_GN") != "1":
+            from ..models.taesd import fp8_flag_convs
             from ..models.unet import fp8_eligible_norms
 
             self._fp8_norms = fp8_eligible_norms(self.unet)
             for nrm in self._fp8_norms:
                 nrm._fp8_calibrate = True
                 nrm._fp8_amax = 0.0
+            if cfg.use_tiny_vae:
+                self._fp8_vae = fp8_flag_convs(self.vae)
+                for c in self._fp8_vae["convs"]:
+                    c._fp8_calibrate = True
+                    c._fp8_in_amax = 0.0
+                    c._fp8_out_amax = 0.0
             self._fp8_calib_left = max(1, cfg.fp8_calib_frames)
 
     def _load_weights_if_present(self, model_id: str) -> None:
@@ -422,27 +430,61 @@ class StreamDiffusionEngine:
         sig = (ref ** 2).mean().item()
         return 10.0 * math.log10(sig / max(err, 1e-20))
 
+    @torch.no_grad()
+    def _fp8_vae_snr(self) -> float:
+        """SNR of the fp8 TAESD decode vs f16 on a random latent (pure
+        function of the VAE; calibration-domain input)."""
+        g = torch.Generator(device="cpu").manual_seed(self.cfg.seed + 7)
+        lat = torch.randn((1, self.cfg.latent_height, self.cfg.latent_width, 4),
+                          generator=g).to(self.device, self.dtype)
+        scales = [(c, c._fp8_in_scale, c._fp8_out_scale)
+                  for c in self._fp8_vae["convs"]]
+        for c, _, _ in scales:
+            c._fp8_in_scale = c._fp8_out_scale = None
+        ref = self.vae.decode(lat).float()
+        for c, si, so in scales:
+            c._fp8_in_scale, c._fp8_out_scale = si, so
+        got = self.vae.decode(lat).float()
+        err = ((got - ref) ** 2).mean().item()
+        sig = (ref ** 2).mean().item()
+        return 10.0 * math.log10(sig / max(err, 1e-20))
+
     def _fp8_freeze(self) -> None:
-        """End calibration: freeze per-layer scales, run the quality gate,
-        fall back to f16 if it fails (SURVEY.md §6 quality guard)."""
+        """End calibration: freeze per-layer scales, run the quality gates
+        (UNet forward + TAESD decode), fall back to f16 if either fails
+        (SURVEY.md §6 quality guard)."""
         log = logging.getLogger("airtc.engine")
         cfg = self.cfg
         for nrm in self._fp8_norms:
             nrm._fp8_calibrate = False
             if nrm._fp8_amax > 0:
                 nrm._fp8_scale = nrm._fp8_amax * cfg.fp8_margin / 448.0
+        for c in self._fp8_vae["convs"]:
+            c._fp8_calibrate = False
+            if c._fp8_in_amax > 0:
+                c._fp8_in_scale = c._fp8_in_amax * cfg.fp8_margin / 448.0
+        for c in self._fp8_vae["outs"]:
+            if c._fp8_out_amax > 0:
+                c._fp8_out_scale = c._fp8_out_amax * cfg.fp8_margin / 448.0
         self.fp8_snr_db = self._fp8_quality_snr()
-        if self.fp8_snr_db < cfg.fp8_min_snr_db:
+        vae_snr = self._fp8_vae_snr() if self._fp8_vae["convs"] else 1e9
+        if self.fp8_snr_db < cfg.fp8_min_snr_db or vae_snr < cfg.fp8_min_snr_db:
             for nrm in self._fp8_norms:
                 nrm._fp8_scale = None
+            for c in self._fp8_vae["convs"]:
+                c._fp8_in_scale = c._fp8_out_scale = None
             self.fp8_active = False
             log.warning(
-                "fp8 quality gate FAILED (%.1f dB < %.1f dB) — serving f16",
-                self.fp8_snr_db, cfg.fp8_min_snr_db)
+                "fp8 quality gate FAILED (unet %.1f dB, vae %.1f dB, "
+                "min %.1f) — serving f16",
+                self.fp8_snr_db, vae_snr, cfg.fp8_min_snr_db)
         else:
             self.fp8_active = True
-            log.info("fp8 tier active: %d layers, quality %.1f dB",
-                     len(self._fp8_norms), self.fp8_snr_db)
+            log.info(
+                "fp8 tier active: %d GN->conv pairs + %d VAE convs, "
+                "unet %.1f dB / vae %.1f dB",
+                len(self._fp8_norms), len(self._fp8_vae["convs"]),
+                self.fp8_snr_db, vae_snr)
 
     @torch.no_grad()
     def _denoise_core(self) -> torch.Tensor:

@@ -41,6 +41,20 @@ class _Block(nn.Module):
         return self.c3(h, residual=x, act=ops.ACT_RELU)
 
 
+def fp8_flag_convs(vae) -> dict:
+    """fp8 chain roles inside each _Block (all convs are 64ch, IC%64==0):
+    c1/c2 emit e4m3 codes (out scales), every conv gets an input scale so
+    split-K shapes that refuse q8 OUTPUT still consume f16 via the inline
+    encode. conv_in/out, downs and upsamples stay f16 (tiny or IC%64!=0).
+    The engine calibrates the amaxes on its first frames and freezes."""
+    convs, outs = [], []
+    for m in vae.modules():
+        if isinstance(m, _Block):
+            convs += [m.c1, m.c2, m.c3]
+            outs += [m.c1, m.c2]
+    return {"convs": convs, "outs": outs}
+
+
 class TAESDEncoder(nn.Module):
     """(B,H,W,3) in [-1,1] -> (B,H/8,W/8,4) scaled latent."""
 

@@ -122,7 +122,23 @@ class Linear(nn.Module):
 
 
 class Conv2d(nn.Module):
-    """3x3/1x1 conv over NHWC via ops.conv2d_nhwc. Weight kept OIHW."""
+    """3x3/1x1 conv over NHWC via ops.conv2d_nhwc. Weight kept OIHW.
+
+    fp8 tier state (set by the engine's calibrate/freeze pass on chained
+    fp8 layers, e.g. the TAESD conv stacks):
+      _fp8_calibrate  record input/output absmax on the f16 path
+      _fp8_in_scale   quantize an f16 INPUT inline at this scale (chain
+                      heads whose producer is not fp8)
+      _fp8_out_scale  ask the epilogue for e4m3 codes at this scale (the
+                      GPU may still return f16 on split-K shapes — callers
+                      dispatch on the returned dtype)
+    """
+
+    _fp8_calibrate = False
+    _fp8_in_scale: float | None = None
+    _fp8_out_scale: float | None = None
+    _fp8_in_amax: float = 0.0
+    _fp8_out_amax: float = 0.0
 
     def __init__(self, cin: int, cout: int, k: int = 3, stride: int = 1, bias: bool = True):
         super().__init__()
@@ -141,19 +157,33 @@ class Conv2d(nn.Module):
         in_affine: torch.Tensor | None = None,
         in_act: int = 0,
     ) -> torch.Tensor:
+        act_r = act if act is not None else (ops.ACT_SILU if fuse_silu else ops.ACT_NONE)
+        a_scale = None
         if x.dtype == torch.uint8:
-            # producer-quantized e4m3 codes (fp8 tier): the GN attached its
-            # scale to the tensor; run the MX-scaled MFMA conv
-            return ops.conv2d_fp8_nhwc(
-                x, self.weight, x._airtc_fp8_scale, self.bias, self.stride,
-                self.padding, act=(act if act is not None else (ops.ACT_SILU if fuse_silu else ops.ACT_NONE)),
-                residual=residual, channel_bias=channel_bias,
+            # producer-quantized e4m3 codes: the producer attached its scale
+            a_scale = x._airtc_fp8_scale
+        elif self._fp8_in_scale is not None and not self._fp8_calibrate:
+            a_scale = self._fp8_in_scale  # chain head: inline encode
+        if a_scale is not None:
+            y = ops.conv2d_fp8_nhwc(
+                x, self.weight, a_scale, self.bias, self.stride, self.padding,
+                act=act_r, residual=residual, channel_bias=channel_bias,
+                out_fp8_scale=self._fp8_out_scale,
             )
-        return ops.conv2d_nhwc(
+            if y.dtype == torch.uint8:
+                y._airtc_fp8_scale = self._fp8_out_scale
+            return y
+        y = ops.conv2d_nhwc(
             x, self.weight, self.bias, self.stride, self.padding, fuse_silu,
             act=act, residual=residual, channel_bias=channel_bias,
             in_affine=in_affine, in_act=in_act,
         )
+        if self._fp8_calibrate:
+            self._fp8_in_amax = max(self._fp8_in_amax,
+                                    x.float().abs().max().item())
+            self._fp8_out_amax = max(self._fp8_out_amax,
+                                     y.float().abs().max().item())
+        return y
 
 
 class GroupNormSiLU(nn.Module):

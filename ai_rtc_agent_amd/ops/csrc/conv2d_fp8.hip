@@ -134,15 +134,33 @@ __device__ __forceinline__ i32x4 load_a_q8(const uint8_t* xb, int ho_s,
 }
 
 // fused epilogue (dequant -> bias -> cbias -> residual -> act), f32 math
-__device__ __forceinline__ f16 epilogue_fp8(float acc, float dq,
-                                            const float* bias, const f16* cbias,
-                                            long cb_off, const f16* residual,
-                                            long idx, int oc, int act) {
+__device__ __forceinline__ float epilogue_fp8_val(float acc, float dq,
+                                                  const float* bias,
+                                                  const f16* cbias, long cb_off,
+                                                  const f16* residual, long idx,
+                                                  int oc, int act) {
   float v = acc * dq;
   if (bias) v += bias[oc];
   if (cbias) v += (float)cbias[cb_off + oc];
   if (residual) v += (float)residual[idx];
-  return (f16)apply_act(v, act);
+  return apply_act(v, act);
+}
+
+__device__ __forceinline__ f16 epilogue_fp8(float acc, float dq,
+                                            const float* bias, const f16* cbias,
+                                            long cb_off, const f16* residual,
+                                            long idx, int oc, int act) {
+  return (f16)epilogue_fp8_val(acc, dq, bias, cbias, cb_off, residual, idx, oc,
+                               act);
+}
+
+// quantized-output store: code = e4m3(clamp(v * out_inv, +-448)) — the
+// CONSUMING fp8 conv then stages these bytes directly (chained fp8 layers,
+// e.g. the TAESD conv stacks)
+__device__ __forceinline__ uint8_t q8_of(float v, float out_inv) {
+  const float a = fminf(fmaxf(v * out_inv, -448.0f), 448.0f);
+  const int p = __builtin_amdgcn_cvt_pk_fp8_f32(a, a, 0, false);
+  return (uint8_t)(p & 0xFF);
 }
 
 // XCD-aware mapping shared with conv2d.hip (same dispatcher law)
@@ -168,7 +186,8 @@ __global__ __launch_bounds__(256) void conv2d_mfma_fp8_kernel(
     const f16* __restrict__ cbias, const f16* __restrict__ residual,
     f16* __restrict__ out, float* __restrict__ ws, int H, int W, int IC,
     int HO, int WO, int OC, int R, int S, int stride, int pad, int act, int K,
-    int splitk, const float* __restrict__ in_aff, int in_act, float sa) {
+    int splitk, const float* __restrict__ in_aff, int in_act, float sa,
+    uint8_t* __restrict__ out_q8, float out_inv) {
   constexpr int BM = MFRAG * 32;
   constexpr int KPITCH = BK8 + 16;            // bytes; same 144B law as f16
   constexpr int ALOADS = MFRAG;               // 16B units: BM*8/256
@@ -308,6 +327,7 @@ __global__ __launch_bounds__(256) void conv2d_mfma_fp8_kernel(
 
   if (spk == 1) {
     f16* ob = out + (long)b * M * OC;
+    uint8_t* oq = out_q8 ? out_q8 + (long)b * M * OC : nullptr;
     const long cb_off = (long)b * OC;
 #pragma unroll
     for (int ni = 0; ni < 2; ++ni) {
@@ -321,8 +341,12 @@ __global__ __launch_bounds__(256) void conv2d_mfma_fp8_kernel(
           const int m = m0 + wm * (MFRAG * 16) + mi * 16 + (lane >> 4) * 4 + j;
           if (m < M) {
             const long idx = (long)b * M * OC + (long)m * OC + col;
-            ob[(long)m * OC + col] = epilogue_fp8(
-                acc[mi][ni][j], d, bias, cbias, cb_off, residual, idx, col, act);
+            const float v = epilogue_fp8_val(acc[mi][ni][j], d, bias, cbias,
+                                             cb_off, residual, idx, col, act);
+            if (oq)
+              oq[(long)m * OC + col] = q8_of(v, out_inv);
+            else
+              ob[(long)m * OC + col] = (f16)v;
           }
         }
     }
@@ -372,7 +396,7 @@ extern "C" void airtc_conv2d_fp8_mfma(
     uint16_t* out, float* ws, int B, int H, int W, int IC, int HO, int WO,
     int OC, int R, int S, int stride, int pad, int act, int path,
     const float* in_aff, int in_act, float a_scale, int x_is_q8,
-    hipStream_t s) {
+    uint8_t* out_q8, float out_inv, hipStream_t s) {
   const int K = R * S * IC;
   const f16* cb = reinterpret_cast<const f16*>(cbias);
   const f16* res = reinterpret_cast<const f16*>(residual);
@@ -391,7 +415,7 @@ extern "C" void airtc_conv2d_fp8_mfma(
   hipLaunchKernelGGL((conv2d_mfma_fp8_kernel<MF, XT>), grid, dim3(256), 0, s, \
                      XP, w_fp8, dq, b1, cb1, res1, op, ws, H, W, IC, HO, WO,  \
                      OC, R, S, stride, pad, act, K, -splitk, in_aff, in_act,  \
-                     a_scale)
+                     a_scale, splitk == 1 ? out_q8 : nullptr, out_inv)
   if (x_is_q8) {
     const uint8_t* xq = reinterpret_cast<const uint8_t*>(x);
     if (path > 0) FP8_LAUNCH(4, uint8_t, xq);

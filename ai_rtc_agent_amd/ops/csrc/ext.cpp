@@ -432,7 +432,7 @@ torch::Tensor conv2d_fp8(torch::Tensor x, torch::Tensor w_fp8,
                          c10::optional<torch::Tensor> residual, int64_t R,
                          int64_t S, int64_t stride, int64_t pad, int64_t act,
                          c10::optional<torch::Tensor> in_affine,
-                         int64_t in_act) {
+                         int64_t in_act, double out_scale) {
   CHECK_IN(x);
   CHECK_IN(w_fp8);
   CHECK_IN(dq);
@@ -450,7 +450,16 @@ torch::Tensor conv2d_fp8(torch::Tensor x, torch::Tensor w_fp8,
               "pre-quantized input cannot take a fused input affine");
   const int HO = (H + 2 * (int)pad - (int)R) / (int)stride + 1;
   const int WO = (W + 2 * (int)pad - (int)S) / (int)stride + 1;
-  auto out = torch::empty({B, HO, WO, OC}, x.options().dtype(torch::kHalf));
+  int path0 = airtc_conv2d_splitk_for(B, HO, WO, OC, IC);
+  if (path0 == 0 || path0 == 100) path0 = HO * WO >= 2048 ? 1 : -1;
+  const int splitk0 = path0 > 0 ? path0 : -path0;
+  // q8 output only on the split-K-free path (the finalize pass stays
+  // fp8-agnostic); callers get f16 back otherwise and must dispatch on
+  // the returned dtype
+  const bool want_q8 = out_scale > 0 && splitk0 == 1;
+  auto out = torch::empty({B, HO, WO, OC},
+                          x.options().dtype(want_q8 ? torch::kUInt8
+                                                    : torch::kHalf));
   const float* bp = nullptr;
   if (bias.has_value()) {
     TORCH_CHECK(bias->dtype() == torch::kFloat && bias->is_contiguous());
@@ -477,9 +486,8 @@ torch::Tensor conv2d_fp8(torch::Tensor x, torch::Tensor w_fp8,
                 "in_affine must be (B, IC, 2) f32");
     aff = in_affine->data_ptr<float>();
   }
-  int path = airtc_conv2d_splitk_for(B, HO, WO, OC, IC);
-  if (path == 0 || path == 100) path = HO * WO >= 2048 ? 1 : -1;
-  const int splitk = path > 0 ? path : -path;
+  const int path = path0;
+  const int splitk = splitk0;
   float* wsp = nullptr;
   torch::Tensor ws;
   if (splitk > 1) {
@@ -491,11 +499,15 @@ torch::Tensor conv2d_fp8(torch::Tensor x, torch::Tensor w_fp8,
                            ? reinterpret_cast<const uint16_t*>(
                                  x.data_ptr<uint8_t>())
                            : h_ptr(x);
+  uint16_t* op = want_q8 ? nullptr : h_ptr_mut(out);
+  uint8_t* oq = want_q8 ? out.data_ptr<uint8_t>() : nullptr;
   airtc_conv2d_fp8_mfma(xp, w_fp8.data_ptr<uint8_t>(),
-                        dq.data_ptr<float>(), bp, cb, res, h_ptr_mut(out), wsp,
+                        dq.data_ptr<float>(), bp, cb, res, op, wsp,
                         B, H, W, IC, HO, WO, OC, (int)R, (int)S, (int)stride,
                         (int)pad, (int)act, path, aff, (int)in_act,
-                        (float)a_scale, x_q8 ? 1 : 0, cur_stream());
+                        (float)a_scale, x_q8 ? 1 : 0, oq,
+                        want_q8 ? (float)(1.0 / out_scale) : 0.f,
+                        cur_stream());
   return out;
 }
 
@@ -572,7 +584,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("cbias"), pybind11::arg("residual"), pybind11::arg("R"),
         pybind11::arg("S"), pybind11::arg("stride"), pybind11::arg("pad"),
         pybind11::arg("act"), pybind11::arg("in_affine") = pybind11::none(),
-        pybind11::arg("in_act") = 0);
+        pybind11::arg("in_act") = 0, pybind11::arg("out_scale") = 0.0);
   m.def("fp8_mx_probe", &fp8_mx_probe,
         "raw-fragment v_mfma_scale_f32_16x16x128_f8f6f4 tile (layout probe)");
   m.def("fp8_quant_probe", &fp8_quant_probe,

@@ -81,6 +81,8 @@ void airtc_conv2d_direct(const uint16_t* x, const uint16_t* w,
 // airtc_conv2d_splitk_for (nonzero).
 // x_is_q8: x already holds e4m3 codes (u8, producer-quantized by e.g.
 // airtc_group_norm_silu_fp8) — staging is a raw byte copy, in_aff unused.
+// out_q8/out_inv: when non-null and split-K == 1, the epilogue writes e4m3
+// codes q = e4m3(clamp(v / out_scale)) instead of f16 (chained fp8 layers).
 void airtc_conv2d_fp8_mfma(const uint16_t* x, const uint8_t* w_fp8,
                            const float* dq, const float* bias,
                            const uint16_t* cbias, const uint16_t* residual,
@@ -88,7 +90,8 @@ void airtc_conv2d_fp8_mfma(const uint16_t* x, const uint8_t* w_fp8,
                            int IC, int HO, int WO, int OC, int R, int S,
                            int stride, int pad, int act, int path,
                            const float* in_aff, int in_act, float a_scale,
-                           int x_is_q8, hipStream_t s);
+                           int x_is_q8, uint8_t* out_q8, float out_inv,
+                           hipStream_t s);
 // hardware probes: raw-fragment MX MFMA tile and the fused scale-converts
 // (layout/semantics verified on hardware before the fp8 conv relies on them)
 void airtc_fp8_mx_probe(const uint8_t* A, const uint8_t* B, float* draw,

@@ -236,6 +236,7 @@ def conv2d_fp8_nhwc(
     channel_bias: torch.Tensor | None = None,
     in_affine: torch.Tensor | None = None,
     in_act: int = ACT_NONE,
+    out_fp8_scale: float | None = None,
 ) -> torch.Tensor:
     """fp8 serving-tier conv: same contract as conv2d_nhwc plus a_scale (the
     calibrated per-tensor activation scale). GPU: conv2d_fp8.hip — weights
@@ -246,6 +247,10 @@ def conv2d_fp8_nhwc(
     x may be f16 (the kernel quantizes in its staging loads) or u8 e4m3
     codes already scaled by a_scale (producer-quantized, e.g. the output of
     group_norm_silu_nhwc(fp8_scale=...)) — the fast path.
+
+    out_fp8_scale: ask the epilogue to emit e4m3 codes at that scale
+    (chained fp8 layers, e.g. TAESD conv stacks). The GPU honours it only
+    on split-K-free shapes — dispatch on the RETURNED dtype.
 
     Requires IC % 64 == 0 (use conv2d_nhwc for other layers)."""
     O, I, R, S = weight.shape
@@ -280,6 +285,7 @@ def conv2d_fp8_nhwc(
             act,
             None if in_affine is None else in_affine.contiguous(),
             in_act,
+            0.0 if out_fp8_scale is None else out_fp8_scale,
         )
 
     # emulation path (CPU tests + GPU numerics golden): quantize exactly as
@@ -312,6 +318,11 @@ def conv2d_fp8_nhwc(
         y = F.silu(y)
     elif act == ACT_RELU:
         y = F.relu(y)
+    if out_fp8_scale is not None:
+        inv = torch.tensor(1.0, dtype=torch.float32) / torch.tensor(
+            out_fp8_scale, dtype=torch.float32)
+        q = (y.float() * inv).clamp(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
+        return q.view(torch.uint8).contiguous()
     # u8 input carries no output dtype — use the model (weight) dtype
     return y.to(weight.dtype if x_q8 else x.dtype).contiguous()
 
