@@ -253,6 +253,8 @@ class StreamDiffusionEngine:
         self._last_done = None
         self._prepared = True
         self._refresh_static_kv()
+        self._ts_unet_cache = None
+        self._refresh_temb_static()
 
     def _cross_attn_modules(self):
         from ..models.unet import CrossAttention
@@ -329,6 +331,7 @@ class StreamDiffusionEngine:
         broadcast_engine_weights(self)  # no-op at world_size 1
         if self._prepared:
             self._refresh_static_kv()
+            self._refresh_temb_static()  # time-emb MLP weights changed
         if self.device.type == "cuda":
             # quiesce in-flight replays before their graphs are dropped
             torch.cuda.synchronize()
@@ -350,6 +353,7 @@ class StreamDiffusionEngine:
             self._embeds_full[B:].copy_(self._embeds_batch)
         if self._added_cond is not None:
             self._refresh_added_cond()
+            self._refresh_temb_static()  # added-cond feeds the static temb
         self._refresh_static_kv()
 
     @torch.no_grad()
@@ -377,6 +381,11 @@ class StreamDiffusionEngine:
             self._coeff[k].copy_(new[k])
         self._coeff["sub_timesteps"] = new["sub_timesteps"]
         self._ts_batch.copy_(new["sub_timesteps_tensor"].to(self.device))
+        for k in ("alpha_f32", "beta_f32", "c_skip_f32", "c_out_f32"):
+            self._coeff[k].copy_(new[k].to(self.device))
+        # static time-embedding caches follow the new timesteps (in place —
+        # any captured graph reads the same storage)
+        self._refresh_temb_static()
 
     # ------------------------------------------------------------------
     # core step (graph-capturable: static shapes, static buffers)
@@ -400,12 +409,33 @@ class StreamDiffusionEngine:
         return self._embeds_batch
 
     def _unet_batch_timesteps(self) -> torch.Tensor:
+        # cached: the static-timestep UNet fast path identity-checks this
+        # exact tensor object (prepare()/t-index updates reset the cache)
+        t = getattr(self, "_ts_unet_cache", None)
+        if t is not None:
+            return t
         cfg = self.cfg
         if cfg.cfg_type == "full" and self.rcfg.active:
-            return torch.cat([self._ts_batch, self._ts_batch], dim=0)
-        if cfg.cfg_type == "initialize" and self.rcfg.active:
-            return torch.cat([self._ts_batch[: cfg.frame_buffer_size], self._ts_batch], dim=0)
-        return self._ts_batch
+            t = torch.cat([self._ts_batch, self._ts_batch], dim=0)
+        elif cfg.cfg_type == "initialize" and self.rcfg.active:
+            t = torch.cat([self._ts_batch[: cfg.frame_buffer_size], self._ts_batch], dim=0)
+        else:
+            t = self._ts_batch
+        self._ts_unet_cache = t
+        return t
+
+    def _refresh_temb_static(self) -> None:
+        """(Re)compute the UNet's static time-embedding caches — call after
+        anything that changes timesteps, added-cond or weights. In-place
+        when shapes match, so captured graphs keep reading the same
+        storage."""
+        cfg = self.cfg
+        ts = self._unet_batch_timesteps()
+        added = self._added_cond
+        if added is not None and (cfg.cfg_type in ("full", "initialize") and self.rcfg.active):
+            extra = added.shape[0] if cfg.cfg_type == "full" else cfg.frame_buffer_size
+            added = torch.cat([added[:extra], added], dim=0)
+        self.unet.precompute_time_embeddings(ts, added, dtype=self.dtype)
 
     @torch.no_grad()
     def _fp8_quality_snr(self) -> float:

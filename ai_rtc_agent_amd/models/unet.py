@@ -385,6 +385,10 @@ class SpatialTransformer(nn.Module):
 
 
 class ResnetBlock(nn.Module):
+    # per-resnet precomputed time-emb projection (static-timestep serving
+    # path; UNet2DCondition.precompute_time_embeddings fills it)
+    _temb_b_static: torch.Tensor | None = None
+
     def __init__(self, cin: int, cout: int, temb_dim: int):
         super().__init__()
         self.norm1 = GroupNormSiLU(cin)
@@ -394,7 +398,8 @@ class ResnetBlock(nn.Module):
         self.conv2 = Conv2d(cout, cout, 3)
         self.shortcut = Conv2d(cin, cout, 1) if cin != cout else None
 
-    def forward(self, x: torch.Tensor, temb: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, temb: torch.Tensor,
+                temb_b: torch.Tensor | None = None) -> torch.Tensor:
         # time-emb add fused into conv1's epilogue; skip add fused into
         # conv2's. The DEEPER fusion (GN apply + SiLU inside the conv's
         # A-load, in_affine) measured SLOWER end-to-end on MI355X
@@ -403,7 +408,10 @@ class ResnetBlock(nn.Module):
         # critical path of kernels that are already issue/latency-bound —
         # same regime as the rejected halo-tiled loads (ladder). Kept
         # env-gated for experiments: AIRTC_FUSE_GN=1.
-        temb_b = self.time_emb_proj(ops.silu(temb))
+        # temb_b: the precomputed per-resnet projection (static timesteps
+        # — see UNet2DCondition.precompute_time_embeddings).
+        if temb_b is None:
+            temb_b = self.time_emb_proj(ops.silu(temb))
         if _fuse_gn():
             h = self.conv1(x, channel_bias=temb_b,
                            in_affine=self.norm1.coeffs(x), in_act=ops.ACT_SILU)
@@ -448,6 +456,10 @@ class Upsample(nn.Module):
 
 
 class UNet2DCondition(nn.Module):
+    # static-timestep serving caches (see precompute_time_embeddings)
+    _temb_static: torch.Tensor | None = None
+    _temb_src: torch.Tensor | None = None
+
     def __init__(self, cfg: UNetConfig):
         super().__init__()
         self.cfg = cfg
@@ -523,6 +535,38 @@ class UNet2DCondition(nn.Module):
         self.conv_out = Conv2d(chans[0], cfg.out_channels, 3)
 
     # ------------------------------------------------------------------
+    @torch.no_grad()
+    def precompute_time_embeddings(self, timesteps: torch.Tensor,
+                                   added_cond: torch.Tensor | None = None,
+                                   dtype: torch.dtype = torch.float16) -> None:
+        """Serving fast path: the engine's timesteps are STATIC between
+        prepare()/updates, so the whole time-embedding pipeline (~40 small
+        kernels per frame: sinusoid + MLP + one silu+linear per resnet)
+        runs ONCE here. Buffers refresh IN PLACE when shapes match so a
+        captured hipGraph keeps reading the same storage; callers must
+        re-invoke after weight refresh / t-index / added-cond updates.
+        forward() uses the cache only when handed the SAME timesteps
+        tensor object (identity check)."""
+        temb = timestep_embedding(timesteps, self.time_proj_dim).to(dtype)
+        temb = self.time_embed[1](ops.silu(self.time_embed[0](temb)))
+        if self.add_embed is not None and added_cond is not None:
+            temb = temb + self.add_embed[1](
+                ops.silu(self.add_embed[0](added_cond.to(dtype))))
+
+        def keep(buf, val):
+            if buf is not None and buf.shape == val.shape and buf.dtype == val.dtype:
+                buf.copy_(val)
+                return buf
+            return val
+
+        self._temb_static = keep(self._temb_static, temb)
+        self._temb_src = timesteps
+        st = ops.silu(self._temb_static)
+        for m in self.modules():
+            if isinstance(m, ResnetBlock):
+                m._temb_b_static = keep(m._temb_b_static,
+                                        m.time_emb_proj(st).contiguous())
+
     def forward(
         self,
         sample: torch.Tensor,
@@ -534,17 +578,28 @@ class UNet2DCondition(nn.Module):
         """sample: (B,H,W,C_in) NHWC latent; timesteps: (B,);
         encoder_hidden_states: (B,77,ctx)."""
         cfg = self.cfg
-        temb = timestep_embedding(timesteps, self.time_proj_dim).to(sample.dtype)
-        temb = self.time_embed[1](ops.silu(self.time_embed[0](temb)))
-        if self.add_embed is not None and added_cond is not None:
-            temb = temb + self.add_embed[1](ops.silu(self.add_embed[0](added_cond.to(sample.dtype))))
+        # static-timestep fast path: the serving engine precomputes the
+        # whole time-embedding pipeline (and each resnet's projection) once
+        # per prepare/update — identity-checked so direct callers with
+        # other timesteps still compute normally
+        use_static = (getattr(self, "_temb_src", None) is timesteps
+                      and self._temb_static is not None)
+        if use_static:
+            temb = self._temb_static
+        else:
+            temb = timestep_embedding(timesteps, self.time_proj_dim).to(sample.dtype)
+            temb = self.time_embed[1](ops.silu(self.time_embed[0](temb)))
+            if self.add_embed is not None and added_cond is not None:
+                temb = temb + self.add_embed[1](ops.silu(self.add_embed[0](added_cond.to(sample.dtype))))
 
         x = self.conv_in(sample)
         skips = [x]
         ri = 0
         for bi in range(len(cfg.block_out_channels)):
             for _ in range(cfg.layers_per_block):
-                x = self.down_resnets[ri](x, temb)
+                x = self.down_resnets[ri](
+                    x, temb,
+                    self.down_resnets[ri]._temb_b_static if use_static else None)
                 if self.down_attns[ri] is not None:
                     x = self.down_attns[ri](x, encoder_hidden_states)
                 skips.append(x)
@@ -557,9 +612,11 @@ class UNet2DCondition(nn.Module):
             # ControlNet residuals: one per skip entry + one for mid
             skip_res, mid_res = control
             skips = [s + c for s, c in zip(skips, skip_res)]
-        x = self.mid_res1(x, temb)
+        x = self.mid_res1(x, temb,
+                          self.mid_res1._temb_b_static if use_static else None)
         x = self.mid_attn(x, encoder_hidden_states)
-        x = self.mid_res2(x, temb)
+        x = self.mid_res2(x, temb,
+                          self.mid_res2._temb_b_static if use_static else None)
         if control is not None:
             x = x + mid_res
 
@@ -568,7 +625,9 @@ class UNet2DCondition(nn.Module):
             for _ in range(cfg.layers_per_block + 1):
                 skip = skips.pop()
                 x = torch.cat([x, skip], dim=-1)
-                x = self.up_resnets[ri](x, temb)
+                x = self.up_resnets[ri](
+                    x, temb,
+                    self.up_resnets[ri]._temb_b_static if use_static else None)
                 if self.up_attns[ri] is not None:
                     x = self.up_attns[ri](x, encoder_hidden_states)
                 ri += 1

@@ -160,3 +160,32 @@ def test_frame_buffer_batching(tiny_cfg):
     frames = torch.randint(0, 256, (2, 64, 64, 3), dtype=torch.uint8)
     out = e(frames)
     assert out.shape == (2, 64, 64, 3)
+
+
+def test_t_index_same_length_update_refreshes_fused_coeffs_and_temb():
+    """Same-length t-index updates write IN PLACE (no re-capture): the
+    fused-scheduler f32 coefficient arrays and the static time-embedding
+    caches must follow."""
+    from ai_rtc_agent_amd.config import EngineConfig
+    from ai_rtc_agent_amd.engine.engine import StreamDiffusionEngine
+
+    cfg = EngineConfig(model_id="none", model_family="tiny", width=64,
+                       height=64, t_index_list=[30], cfg_type="none",
+                       use_lcm_lora=False, device="cpu",
+                       acceleration="eager", use_hip_graph=False)
+    cfg.similarity_filter.enabled = False
+    eng = StreamDiffusionEngine(cfg)
+    eng.prepare()
+    temb_before = eng.unet._temb_static.clone()
+    eng.update_t_index_list([10])
+    fresh = eng.scheduler.coefficients([10], cfg.frame_buffer_size,
+                                       eng.device, eng.dtype)
+    for k in ("alpha_f32", "beta_f32", "c_skip_f32", "c_out_f32"):
+        assert torch.allclose(eng._coeff[k], fresh[k]), k
+    assert not torch.equal(eng.unet._temb_static, temb_before), \
+        "static temb must refresh with the new timesteps"
+    assert eng.unet._temb_src is eng._unet_batch_timesteps()
+    # serving still works after the in-place update
+    frame = torch.randint(0, 255, (64, 64, 3), dtype=torch.uint8)
+    out = eng(frame)
+    assert out.shape == (64, 64, 3)
