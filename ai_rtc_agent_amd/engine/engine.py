@@ -384,7 +384,9 @@ class StreamDiffusionEngine:
         for k in ("alpha_f32", "beta_f32", "c_skip_f32", "c_out_f32"):
             self._coeff[k].copy_(new[k].to(self.device))
         # static time-embedding caches follow the new timesteps (in place —
-        # any captured graph reads the same storage)
+        # any captured graph reads the same storage). The batched-ts cache
+        # is a SEPARATE cat buffer for cfg full/initialize: rebuild it.
+        self._ts_unet_cache = None
         self._refresh_temb_static()
 
     # ------------------------------------------------------------------

@@ -189,3 +189,26 @@ def test_t_index_same_length_update_refreshes_fused_coeffs_and_temb():
     frame = torch.randint(0, 255, (64, 64, 3), dtype=torch.uint8)
     out = eng(frame)
     assert out.shape == (64, 64, 3)
+
+
+def test_t_index_update_with_cfg_full_rebuilds_batched_ts():
+    """cfg full doubles the ts batch via a separate cat buffer — a
+    same-length t-index update must rebuild it (stale-cache regression)."""
+    from ai_rtc_agent_amd.config import EngineConfig
+    from ai_rtc_agent_amd.engine.engine import StreamDiffusionEngine
+
+    cfg = EngineConfig(model_id="none", model_family="tiny", width=64,
+                       height=64, t_index_list=[30], cfg_type="full",
+                       guidance_scale=1.5, use_lcm_lora=False, device="cpu",
+                       acceleration="eager", use_hip_graph=False)
+    cfg.similarity_filter.enabled = False
+    eng = StreamDiffusionEngine(cfg)
+    eng.prepare()
+    assert eng.rcfg.active
+    eng.update_t_index_list([10])
+    ts = eng._unet_batch_timesteps()
+    want = int(eng.scheduler.timesteps[10])
+    assert (ts == want).all(), f"batched ts stale: {ts.tolist()} != {want}"
+    frame = torch.randint(0, 255, (64, 64, 3), dtype=torch.uint8)
+    out = eng(frame)
+    assert out.shape == (64, 64, 3)
