@@ -332,6 +332,19 @@ class StreamDiffusionEngine:
         if self._prepared:
             self._refresh_static_kv()
             self._refresh_temb_static()  # time-emb MLP weights changed
+        # fp8 tier: new weights shift the activation distributions — drop
+        # back to calibration (a few eager frames) and re-gate
+        if self.cfg.use_fp8 and (self._fp8_norms or self._fp8_vae["convs"]):
+            for nrm in self._fp8_norms:
+                nrm._fp8_calibrate = True
+                nrm._fp8_scale = None
+                nrm._fp8_amax = 0.0
+            for c in self._fp8_vae["convs"]:
+                c._fp8_calibrate = True
+                c._fp8_in_scale = c._fp8_out_scale = None
+                c._fp8_in_amax = c._fp8_out_amax = 0.0
+            self.fp8_active = False
+            self._fp8_calib_left = max(1, self.cfg.fp8_calib_frames)
         if self.device.type == "cuda":
             # quiesce in-flight replays before their graphs are dropped
             torch.cuda.synchronize()

@@ -219,3 +219,29 @@ def test_conv_fp8_emulation_with_affine_residual_cbias():
     den = ((y8.float() - yf.float()) ** 2).mean().item()
     snr = 10 * math.log10(num / max(den, 1e-20))
     assert snr > 18, f"fused-epilogue fp8 emulation SNR {snr:.1f} dB"
+
+
+def test_fp8_recalibrates_after_weight_refresh():
+    """LoRA hot-swap shifts activation stats: refresh_weights must drop the
+    fp8 tier back to calibration and re-gate on fresh frames."""
+    from ai_rtc_agent_amd.config import EngineConfig
+    from ai_rtc_agent_amd.engine.engine import StreamDiffusionEngine
+    from ai_rtc_agent_amd.models.lora import fuse_lora_state_dict, make_random_lora
+
+    cfg = EngineConfig(
+        model_id="none", model_family="tiny", width=64, height=64,
+        t_index_list=[30], cfg_type="none", use_lcm_lora=False,
+        device="cpu", acceleration="eager", use_hip_graph=False,
+        use_fp8=True, fp8_calib_frames=1, fp8_min_snr_db=10.0,
+    )
+    cfg.similarity_filter.enabled = False
+    eng = StreamDiffusionEngine(cfg)
+    eng.prepare()
+    frame = torch.randint(0, 255, (64, 64, 3), dtype=torch.uint8)
+    eng(frame)
+    assert eng.fp8_active
+    fuse_lora_state_dict(eng.unet, make_random_lora(eng.unet, rank=2, seed=9), scale=0.5)
+    eng.refresh_weights()
+    assert not eng.fp8_active and eng._fp8_calib_left == 1
+    eng(frame)  # one calibration frame -> freeze + gate again
+    assert eng.fp8_active, f"re-gate failed at {eng.fp8_snr_db} dB"
