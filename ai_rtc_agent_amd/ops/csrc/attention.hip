@@ -219,35 +219,19 @@ __global__ __launch_bounds__(256) void attention_kernel(
         // ds_bpermute contends with this kernel's heavy LDS traffic
         // (measured: 608 bpermutes/iteration before)
         mnew[j] = quarter_reduce(mnew[j], MaxOp());
-        // EXACT defer-rescale (guide T13, safe form): when no lane's max
-        // grew, alpha == 1 wave-wide — skip the alpha exp, the l blend
-        // and the O rescale with bitwise-identical results (this is not
-        // the thresholded approximation; no numerics change)
-        if (__ballot(mnew[j] > m_i[qi][j]) == 0ull) {
-          const float mn = m_i[qi][j];
-          float rs = 0.f;
+        const float mn = fmaxf(m_i[qi][j], mnew[j]);
+        const float alpha = __expf(m_i[qi][j] - mn);
+        float rs = 0.f;
 #pragma unroll
-          for (int nf = 0; nf < NF; ++nf) {
-            p[nf][j] = __expf(p[nf][j] - mn);
-            rs += p[nf][j];
-          }
-          rs = quarter_reduce(rs, SumOp());
-          l_i[qi][j] += rs;
-        } else {
-          const float mn = fmaxf(m_i[qi][j], mnew[j]);
-          const float alpha = __expf(m_i[qi][j] - mn);
-          float rs = 0.f;
-#pragma unroll
-          for (int nf = 0; nf < NF; ++nf) {
-            p[nf][j] = __expf(p[nf][j] - mn);
-            rs += p[nf][j];
-          }
-          rs = quarter_reduce(rs, SumOp());
-          l_i[qi][j] = l_i[qi][j] * alpha + rs;
-          m_i[qi][j] = mn;
-#pragma unroll
-          for (int f = 0; f < D16; ++f) o_acc[qi][f][j] *= alpha;
+        for (int nf = 0; nf < NF; ++nf) {
+          p[nf][j] = __expf(p[nf][j] - mn);
+          rs += p[nf][j];
         }
+        rs = quarter_reduce(rs, SumOp());
+        l_i[qi][j] = l_i[qi][j] * alpha + rs;
+        m_i[qi][j] = mn;
+#pragma unroll
+        for (int f = 0; f < D16; ++f) o_acc[qi][f][j] *= alpha;
       }
 
       // ---- P -> LDS transposed [kv][q] (C-frag -> A-frag bridge): the 4
