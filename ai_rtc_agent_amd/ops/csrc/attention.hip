@@ -112,6 +112,10 @@ __global__ __launch_bounds__(256) void attention_kernel(
   const int lane = tid & 63;
   const int wid = tid >> 6;
 
+  // softmax runs in the BASE-2 domain: folding log2(e) into the logit
+  // scale turns every exp into a bare v_exp_f32 (exp2) — __expf otherwise
+  // pays a v_mul per call, and this kernel is VALU-bound (PMC: 53% busy)
+  const float scale2 = scale * 1.44269504088896340736f;
   const f16* qb = q + b * q_sb + h * q_sh;
   const f16* kb = k + b * k_sb + h * k_sh;
   const f16* vb = v + b * k_sb + h * k_sh;  // v shares k's layout
@@ -208,7 +212,7 @@ __global__ __launch_bounds__(256) void attention_kernel(
         const bool valid = kcol < Lk;
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
-          float val = valid ? sfrag[qi][nf][j] * scale : -1e30f;
+          float val = valid ? sfrag[qi][nf][j] * scale2 : -1e30f;
           p[nf][j] = val;
           mnew[j] = fmaxf(mnew[j], val);
         }
@@ -220,11 +224,11 @@ __global__ __launch_bounds__(256) void attention_kernel(
         // (measured: 608 bpermutes/iteration before)
         mnew[j] = quarter_reduce(mnew[j], MaxOp());
         const float mn = fmaxf(m_i[qi][j], mnew[j]);
-        const float alpha = __expf(m_i[qi][j] - mn);
+        const float alpha = __builtin_amdgcn_exp2f(m_i[qi][j] - mn);
         float rs = 0.f;
 #pragma unroll
         for (int nf = 0; nf < NF; ++nf) {
-          p[nf][j] = __expf(p[nf][j] - mn);
+          p[nf][j] = __builtin_amdgcn_exp2f(p[nf][j] - mn);
           rs += p[nf][j];
         }
         rs = quarter_reduce(rs, SumOp());
