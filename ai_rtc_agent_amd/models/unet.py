@@ -161,7 +161,11 @@ class Conv2d(nn.Module):
         a_scale = None
         if x.dtype == torch.uint8:
             # producer-quantized e4m3 codes: the producer attached its scale
-            a_scale = x._airtc_fp8_scale
+            a_scale = getattr(x, "_airtc_fp8_scale", None)
+            if a_scale is None:
+                raise ValueError(
+                    "u8 conv input must carry _airtc_fp8_scale (e4m3 codes "
+                    "from a producer-quantized fp8 layer)")
         elif self._fp8_in_scale is not None and not self._fp8_calibrate:
             a_scale = self._fp8_in_scale  # chain head: inline encode
         if a_scale is not None:
