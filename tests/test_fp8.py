@@ -245,3 +245,53 @@ def test_fp8_recalibrates_after_weight_refresh():
     assert not eng.fp8_active and eng._fp8_calib_left == 1
     eng(frame)  # one calibration frame -> freeze + gate again
     assert eng.fp8_active, f"re-gate failed at {eng.fp8_snr_db} dB"
+
+
+def test_fp8_composes_with_sdxl_added_cond_path():
+    """tiny_xl engine (addition-embedding path) with fp8 + the static temb
+    cache: calibrate, gate, serve — the added-cond contribution must live
+    inside the precomputed temb and survive a prompt update."""
+    from ai_rtc_agent_amd.config import EngineConfig
+    from ai_rtc_agent_amd.engine.engine import StreamDiffusionEngine
+
+    cfg = EngineConfig(
+        model_id="none", model_family="tiny_xl", width=64, height=64,
+        t_index_list=[30], cfg_type="none", use_lcm_lora=False,
+        device="cpu", acceleration="eager", use_hip_graph=False,
+        use_fp8=True, fp8_calib_frames=1, fp8_min_snr_db=8.0,
+    )
+    cfg.similarity_filter.enabled = False
+    eng = StreamDiffusionEngine(cfg)
+    eng.prepare()
+    assert eng._added_cond is not None
+    frame = torch.randint(0, 255, (64, 64, 3), dtype=torch.uint8)
+    eng(frame)
+    assert eng.fp8_active
+    temb0 = eng.unet._temb_static.clone()
+    eng.update_prompt("a different prompt")
+    assert not torch.equal(eng.unet._temb_static, temb0), \
+        "prompt update changes pooled added-cond -> static temb must refresh"
+    out = eng(frame)
+    assert out.shape == (64, 64, 3) and out.dtype == torch.uint8
+
+
+def test_fp8_with_frame_buffer_batching():
+    """fp8 tier under the multi-stream batched serving shape (fbs=2)."""
+    from ai_rtc_agent_amd.config import EngineConfig
+    from ai_rtc_agent_amd.engine.engine import StreamDiffusionEngine
+
+    cfg = EngineConfig(
+        model_id="none", model_family="tiny", width=64, height=64,
+        t_index_list=[30], cfg_type="none", use_lcm_lora=False,
+        device="cpu", acceleration="eager", use_hip_graph=False,
+        frame_buffer_size=2, use_fp8=True, fp8_calib_frames=1,
+        fp8_min_snr_db=8.0,
+    )
+    cfg.similarity_filter.enabled = False
+    eng = StreamDiffusionEngine(cfg)
+    eng.prepare()
+    batch = torch.randint(0, 255, (2, 64, 64, 3), dtype=torch.uint8)
+    eng(batch)
+    assert eng.fp8_active
+    out = eng(batch)
+    assert out.shape == (2, 64, 64, 3) and out.dtype == torch.uint8
