@@ -453,6 +453,7 @@ class StreamDiffusionEngine:
         self.unet.precompute_time_embeddings(ts, added, dtype=self.dtype)
 
     @torch.no_grad()
+    @torch.no_grad()
     def _fp8_quality_snr(self) -> float:
         """SNR (dB) of the fp8 UNet forward vs the f16 forward on the same
         inputs — a pure function of the UNet (no engine state mutated)."""
@@ -477,22 +478,29 @@ class StreamDiffusionEngine:
 
     @torch.no_grad()
     def _fp8_vae_snr(self) -> float:
-        """SNR of the fp8 TAESD decode vs f16 on a random latent (pure
-        function of the VAE; calibration-domain input)."""
+        """min SNR of the fp8 TAESD decode (random latent) and encode
+        (random frame) vs f16 — pure functions of the VAE."""
         g = torch.Generator(device="cpu").manual_seed(self.cfg.seed + 7)
         lat = torch.randn((1, self.cfg.latent_height, self.cfg.latent_width, 4),
                           generator=g).to(self.device, self.dtype)
+        img = (torch.rand((1, self.cfg.height, self.cfg.width, 3),
+                          generator=g) * 2 - 1).to(self.device, self.dtype)
         scales = [(c, c._fp8_in_scale, c._fp8_out_scale)
                   for c in self._fp8_vae["convs"]]
         for c, _, _ in scales:
             c._fp8_in_scale = c._fp8_out_scale = None
-        ref = self.vae.decode(lat).float()
+        ref_d = self.vae.decode(lat).float()
+        ref_e = self.vae.encode(img).float()
         for c, si, so in scales:
             c._fp8_in_scale, c._fp8_out_scale = si, so
-        got = self.vae.decode(lat).float()
-        err = ((got - ref) ** 2).mean().item()
-        sig = (ref ** 2).mean().item()
-        return 10.0 * math.log10(sig / max(err, 1e-20))
+        got_d = self.vae.decode(lat).float()
+        got_e = self.vae.encode(img).float()
+
+        def snr(got, ref):
+            err = ((got - ref) ** 2).mean().item()
+            return 10.0 * math.log10((ref ** 2).mean().item() / max(err, 1e-20))
+
+        return min(snr(got_d, ref_d), snr(got_e, ref_e))
 
     def _fp8_freeze(self) -> None:
         """End calibration: freeze per-layer scales, run the quality gates
