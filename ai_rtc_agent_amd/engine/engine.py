@@ -163,6 +163,9 @@ class StreamDiffusionEngine:
             from ..models.unet import fp8_eligible_norms
 
             self._fp8_norms = fp8_eligible_norms(self.unet)
+            if self.controlnet is not None:
+                # ControlNet shares the resnet blocks — same tier applies
+                self._fp8_norms += fp8_eligible_norms(self.controlnet)
             for nrm in self._fp8_norms:
                 nrm._fp8_calibrate = True
                 nrm._fp8_amax = 0.0

@@ -295,3 +295,31 @@ def test_fp8_with_frame_buffer_batching():
     assert eng.fp8_active
     out = eng(batch)
     assert out.shape == (2, 64, 64, 3) and out.dtype == torch.uint8
+
+
+def test_fp8_covers_controlnet_resnets():
+    """ControlNet shares ResnetBlock — its GN->conv pairs join the tier."""
+    from ai_rtc_agent_amd.config import EngineConfig
+    from ai_rtc_agent_amd.engine.engine import StreamDiffusionEngine
+    from ai_rtc_agent_amd.models.unet import fp8_eligible_norms
+
+    cfg = EngineConfig(
+        model_id="none", model_family="tiny", width=64, height=64,
+        t_index_list=[30], cfg_type="none", use_lcm_lora=False,
+        device="cpu", acceleration="eager", use_hip_graph=False,
+        use_controlnet=True, use_fp8=True, fp8_calib_frames=1,
+        fp8_min_snr_db=8.0,
+    )
+    cfg.similarity_filter.enabled = False
+    eng = StreamDiffusionEngine(cfg)
+    eng.prepare()
+    n_unet = len(fp8_eligible_norms(eng.unet))
+    assert len(eng._fp8_norms) > n_unet, "controlnet norms must be included"
+    frame = torch.randint(0, 255, (64, 64, 3), dtype=torch.uint8)
+    eng(frame)
+    assert eng.fp8_active
+    out = eng(frame)
+    assert out.shape == (64, 64, 3)
+    # controlnet norms that saw activations have frozen scales
+    cn = [n for n in eng._fp8_norms[n_unet:] if n._fp8_scale is not None]
+    assert len(cn) > 0
