@@ -194,8 +194,14 @@ class H264SwCodec:
         h, w = int(arr.shape[0]), int(arr.shape[1])
         if self._enc_dims != (w, h):
             # mb_mode=2: per-MB I_16x16 / I_4x4 decision (I_4x4 with full
-            # mode search wins on moderately detailed macroblocks)
-            self._enc = self._ext.H264SwEncoder(w, h, mb_mode=2)
+            # mode search wins on moderately detailed macroblocks).
+            # slices = threads: measured 2.88 -> 2.28 ms/frame (worst-case
+            # noise 512²) going 4 -> 8 slices on an 8-core box; slice-header
+            # overhead is a few bytes each
+            import os as _os
+
+            slices = max(4, min(8, _os.cpu_count() or 4))
+            self._enc = self._ext.H264SwEncoder(w, h, slices, 2)
             self._enc_dims = (w, h)
             self._enc_count = 0
         # GOP cadence: periodic IDR (join-anywhere + loss recovery bound);
