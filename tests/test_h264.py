@@ -559,8 +559,10 @@ def test_h264sw_static_p_frames_are_tiny():
     first = enc.encode(f)
     for _ in range(3):
         p = enc.encode(f)
-    assert len(p) < 64, len(p)      # all-skip P frame
-    assert len(first) > 3 * len(p)  # (smooth ramp: the IDR itself is tiny)
+    # all-skip P frame: a few bytes per slice NAL (slice count adapts to
+    # cores, <= 8 slices -> <= ~12 B each)
+    assert len(p) < 128, len(p)
+    assert len(first) > 2 * len(p)  # (smooth ramp: the IDR itself is tiny)
 
 
 def test_h264sw_decoder_fuzz_robustness():
