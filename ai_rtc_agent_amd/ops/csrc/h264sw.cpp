@@ -444,24 +444,42 @@ static int cavlc_read_block(BitReader& r, int32_t* coeffs, int n, int nC) {
 static void rgb_to_yuv420_rows(const uint8_t* rgb, int w, int h, int pw,
                                int y0, int y1, uint8_t* Y, uint8_t* Cb,
                                uint8_t* Cr) {
-  // write padded planes (pw wide), edge-replicated, luma rows [y0, y1)
+  // write padded planes (pw wide), edge-replicated, luma rows [y0, y1).
+  // The interior loop is branch-free (no edge clamps) so the compiler can
+  // vectorize it; edge replication runs in separate tail loops.
   for (int y = y0; y < y1; ++y) {
-    int sy = y < h ? y : h - 1;
-    for (int x = 0; x < pw; ++x) {
-      int sx = x < w ? x : w - 1;
-      const uint8_t* p = rgb + (sy * w + sx) * 3;
-      int R = p[0], G = p[1], B = p[2];
-      Y[y * pw + x] = clip8(((66 * R + 129 * G + 25 * B + 128) >> 8) + 16);
+    const int sy = y < h ? y : h - 1;
+    const uint8_t* row = rgb + (size_t)sy * w * 3;
+    uint8_t* yo = &Y[(size_t)y * pw];
+    for (int x = 0; x < w; ++x) {
+      const uint8_t* p = row + x * 3;
+      yo[x] = clip8(((66 * p[0] + 129 * p[1] + 25 * p[2] + 128) >> 8) + 16);
     }
+    for (int x = w; x < pw; ++x) yo[x] = yo[w - 1];
   }
-  int cw = pw / 2;
+  const int cw = pw / 2;
+  const int cwi = w / 2;  // full 2x2 blocks (w, h are even for 16|dims)
   for (int y = y0 / 2; y < y1 / 2; ++y) {
-    for (int x = 0; x < cw; ++x) {
+    uint8_t* cbo = &Cb[(size_t)y * cw];
+    uint8_t* cro = &Cr[(size_t)y * cw];
+    const int sy0 = std::min(2 * y, h - 1), sy1 = std::min(2 * y + 1, h - 1);
+    const uint8_t* r0 = rgb + (size_t)sy0 * w * 3;
+    const uint8_t* r1 = rgb + (size_t)sy1 * w * 3;
+    for (int x = 0; x < cwi; ++x) {
+      const uint8_t* a = r0 + 2 * x * 3;
+      const uint8_t* b = r1 + 2 * x * 3;
+      const int R = (a[0] + a[3] + b[0] + b[3] + 2) >> 2;
+      const int G = (a[1] + a[4] + b[1] + b[4] + 2) >> 2;
+      const int B = (a[2] + a[5] + b[2] + b[5] + 2) >> 2;
+      cbo[x] = clip8(((-38 * R - 74 * G + 112 * B + 128) >> 8) + 128);
+      cro[x] = clip8(((112 * R - 94 * G - 18 * B + 128) >> 8) + 128);
+    }
+    for (int x = cwi; x < cw; ++x) {
       int R = 0, G = 0, B = 0;
       for (int dy = 0; dy < 2; ++dy)
         for (int dx = 0; dx < 2; ++dx) {
           int sy = std::min(2 * y + dy, h - 1), sx = std::min(2 * x + dx, w - 1);
-          const uint8_t* p = rgb + (sy * w + sx) * 3;
+          const uint8_t* p = rgb + ((size_t)sy * w + sx) * 3;
           R += p[0];
           G += p[1];
           B += p[2];
@@ -469,8 +487,8 @@ static void rgb_to_yuv420_rows(const uint8_t* rgb, int w, int h, int pw,
       R = (R + 2) >> 2;
       G = (G + 2) >> 2;
       B = (B + 2) >> 2;
-      Cb[y * cw + x] = clip8(((-38 * R - 74 * G + 112 * B + 128) >> 8) + 128);
-      Cr[y * cw + x] = clip8(((112 * R - 94 * G - 18 * B + 128) >> 8) + 128);
+      cbo[x] = clip8(((-38 * R - 74 * G + 112 * B + 128) >> 8) + 128);
+      cro[x] = clip8(((112 * R - 94 * G - 18 * B + 128) >> 8) + 128);
     }
   }
 }
