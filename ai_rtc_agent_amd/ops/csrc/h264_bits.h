@@ -19,14 +19,25 @@ struct BitWriter {
   int nbits = 0;
 
   void put(uint32_t value, int width) {
-    for (int i = width - 1; i >= 0; --i) {
-      cur = (cur << 1) | ((value >> i) & 1);
-      if (++nbits == 8) {
+    // bulk bit packing (bit-exact with the bit-at-a-time form): fill the
+    // current byte, then whole bytes, then the remainder
+    while (width > 0) {
+      const int take = (8 - nbits) < width ? (8 - nbits) : width;
+      const uint32_t mask = (take == 32) ? 0xFFFFFFFFu : ((1u << take) - 1u);
+      cur = (cur << take) | ((value >> (width - take)) & mask);
+      nbits += take;
+      width -= take;
+      if (nbits == 8) {
         bytes.push_back((uint8_t)cur);
         cur = 0;
         nbits = 0;
       }
     }
+  }
+  // raw byte run; caller must be byte-aligned (nbits == 0), e.g. right
+  // after align_byte() for I_PCM samples
+  void put_aligned_bytes(const uint8_t* p, size_t n) {
+    bytes.insert(bytes.end(), p, p + n);
   }
   void ue(uint32_t v) {  // unsigned exp-Golomb
     uint32_t vp1 = v + 1;
