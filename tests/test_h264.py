@@ -627,3 +627,29 @@ def test_cavlc_coeff_token_kraft_analysis():
         s = sum(Fraction(1, 2 ** l) for l, _ in codes)
         assert s == expect_kraft[b], (b, s)
         assert len(set(codes)) == len(codes), f"duplicate codes in bucket {b}"
+
+
+def test_encoder_slice_count_changes_reuse_pool():
+    """Changing n_slices on a live encoder resizes the persistent worker
+    pool; every configuration must still produce decodable frames."""
+    ext = _h264_ext()
+    import torch as _t
+    g = _t.Generator().manual_seed(11)
+    f1 = _t.randint(0, 255, (64, 64, 3), generator=g, dtype=_t.uint8)
+    f2 = _t.randint(0, 255, (64, 64, 3), generator=g, dtype=_t.uint8)
+    dec = ext.H264SwDecoder()
+    for slices in (2, 8, 1, 4):
+        enc = ext.H264SwEncoder(64, 64, slices, 2)
+        au = enc.encode(f1.numpy().tobytes(), 28, keyframe=True)
+        r = dec.decode(au)
+        assert r is not None and r[1] == 64 and r[2] == 64
+        p = enc.encode(f2.numpy().tobytes(), 28, keyframe=False)
+        assert dec.decode(p) is not None
+    # same encoder object across many frames (pool longevity)
+    enc = ext.H264SwEncoder(64, 64, 8, 2)
+    dec2 = ext.H264SwDecoder()
+    assert dec2.decode(enc.encode(f1.numpy().tobytes(), 28, keyframe=True))
+    for i in range(50):
+        au = enc.encode((f1 if i % 2 else f2).numpy().tobytes(), 28,
+                        keyframe=(i % 10 == 0))
+        assert dec2.decode(au) is not None or len(au) > 0
