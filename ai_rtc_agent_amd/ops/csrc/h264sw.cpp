@@ -847,7 +847,7 @@ static void recon_luma16(PlaneCtx& pl, int mbx, int mby, const uint8_t pred[256]
   const int32_t ls00 = QV[qm][0] * 16;
   for (int i = 0; i < 16; ++i) {
     if (qp >= 36)
-      dcd[i] = (dct[i] * ls00) << (qs - 6);
+      dcd[i] = (dct[i] * ls00) * (1 << (qs - 6));
     else
       dcd[i] = (dct[i] * ls00 + (1 << (5 - qs))) >> (6 - qs);
   }
@@ -855,7 +855,7 @@ static void recon_luma16(PlaneCtx& pl, int mbx, int mby, const uint8_t pred[256]
     const int x4 = blk_x4(b), y4 = blk_y4(b);
     int32_t d[16];
     d[0] = dcd[y4 * 4 + x4];
-    for (int i = 1; i < 16; ++i) d[i] = (ac_q[b][i] * QV[qm][POSCLS[i]]) << qs;
+    for (int i = 1; i < 16; ++i) d[i] = (ac_q[b][i] * QV[qm][POSCLS[i]]) * (1 << qs);
     int32_t r[16];
     inv4x4(d, r);
     uint8_t* dst = pl.data + (mby * 16 + y4 * 4) * pl.stride + mbx * 16 + x4 * 4;
@@ -876,12 +876,12 @@ static void recon_chroma8(PlaneCtx& pl, int mbx, int mby, const uint8_t pred[64]
   int32_t t[4] = {a + b + c + d, a - b + c - d, a + b - c - d, a - b - c + d};
   int32_t dcd[4];
   // spec 8.5.11: dcC = ((f * LevelScale(0,0)) << qP/6) >> 5, LevelScale = 16*V
-  for (int i = 0; i < 4; ++i) dcd[i] = ((t[i] * QV[qm][0] * 16) << qs) >> 5;
+  for (int i = 0; i < 4; ++i) dcd[i] = (t[i] * QV[qm][0] * 16 * (1 << qs)) >> 5;
   for (int blk = 0; blk < 4; ++blk) {
     const int x4 = blk & 1, y4 = blk >> 1;
     int32_t dq[16];
     dq[0] = dcd[blk];
-    for (int i = 1; i < 16; ++i) dq[i] = (ac_q[blk][i] * QV[qm][POSCLS[i]]) << qs;
+    for (int i = 1; i < 16; ++i) dq[i] = (ac_q[blk][i] * QV[qm][POSCLS[i]]) * (1 << qs);
     int32_t r[16];
     inv4x4(dq, r);
     uint8_t* dst = pl.data + (mby * 8 + y4 * 4) * pl.stride + mbx * 8 + x4 * 4;
@@ -1131,7 +1131,7 @@ struct Encoder {
       maxtc = std::max(maxtc, tc);
       // reconstruct immediately: later blocks predict from these pixels
       int32_t dq[16], rr[16];
-      for (int i = 0; i < 16; ++i) dq[i] = (lq[z][i] * QV[qm][POSCLS[i]]) << qs;
+      for (int i = 0; i < 16; ++i) dq[i] = (lq[z][i] * QV[qm][POSCLS[i]]) * (1 << qs);
       inv4x4(dq, rr);
       for (int y = 0; y < 4; ++y)
         for (int x = 0; x < 4; ++x)
@@ -1860,7 +1860,7 @@ struct Decoder {
       uint8_t pred[16];
       pred_luma4(m, tbuf, lbuf, tlv, ht, hl, htl, pred);
       int32_t d[16];
-      for (int i = 0; i < 16; ++i) d[i] = (lq[z][i] * QV[qm][POSCLS[i]]) << qs;
+      for (int i = 0; i < 16; ++i) d[i] = (lq[z][i] * QV[qm][POSCLS[i]]) * (1 << qs);
       int32_t rr[16];
       inv4x4(d, rr);
       for (int y = 0; y < 4; ++y)
