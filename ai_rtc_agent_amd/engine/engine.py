@@ -776,6 +776,7 @@ class StreamDiffusionEngine:
             d["fp8"] = {
                 "active": self.fp8_active,
                 "layers": len(self._fp8_norms),
+                "vae_convs": len(self._fp8_vae["convs"]),
                 "quality_snr_db": self.fp8_snr_db,
                 "calibrating": self._fp8_calib_left > 0,
             }
