@@ -355,3 +355,47 @@ def test_tiny_xl_engine_uses_dual_encoder():
     f = torch.randint(0, 255, (64, 64, 3), dtype=torch.uint8)
     out = eng(f)
     assert out.shape == (64, 64, 3)
+
+
+def test_bpe_matches_transformers_cliptokenizer(tmp_path):
+    """Independent validation: our CLIP BPE produces EXACTLY the ids
+    transformers.CLIPTokenizer does on the same vocab/merges files
+    (lowercasing, apostrophe splits, whitespace collapse, truncation)."""
+    import json
+
+    pytest.importorskip("transformers")
+    from transformers import CLIPTokenizer
+
+    from ai_rtc_agent_amd.models.text_encoder import (
+        ClipBpeTokenizer,
+        bytes_to_unicode,
+    )
+
+    symbols = sorted(set(bytes_to_unicode().values()))
+    vocab = {}
+    for s in symbols:
+        vocab[s] = len(vocab)
+    for s in symbols:
+        vocab[s + "</w>"] = len(vocab)
+    merges = [("h", "e"), ("l", "l"), ("he", "ll"), ("hell", "o</w>"),
+              ("w", "o"), ("r", "l"), ("wo", "rl"), ("worl", "d</w>"),
+              ("r", "e"), ("f", "i"), ("k", "y</w>"), ("s", "ky</w>")]
+    for a, b in merges:
+        vocab.setdefault(a + b, len(vocab))
+    vocab["<|startoftext|>"] = len(vocab)
+    vocab["<|endoftext|>"] = len(vocab)
+    (tmp_path / "vocab.json").write_text(json.dumps(vocab))
+    (tmp_path / "merges.txt").write_text(
+        "#version: 0.2\n" + "\n".join(f"{a} {b}" for a, b in merges) + "\n")
+
+    ours = ClipBpeTokenizer.from_dir(str(tmp_path))
+    ref = CLIPTokenizer(str(tmp_path / "vocab.json"),
+                        str(tmp_path / "merges.txt"))
+    cases = ["hello world", "Hello, WORLD!", "fire in the sky", "a b c",
+             "hello  world\n", "don't stop", "x" * 100, "", "2 cats 4 dogs",
+             "hello-world... (really)"]
+    for t in cases:
+        want = ref(t, padding="max_length", truncation=True,
+                   max_length=77)["input_ids"]
+        got = ours(t).flatten().tolist()
+        assert got == list(want), f"BPE diverged on {t!r}"
