@@ -363,6 +363,8 @@ def test_bpe_matches_transformers_cliptokenizer(tmp_path):
     (lowercasing, apostrophe splits, whitespace collapse, truncation)."""
     import json
 
+    import pytest
+
     pytest.importorskip("transformers")
     from transformers import CLIPTokenizer
 
