@@ -308,11 +308,15 @@ def load_clip_text_encoder(model, sd: Dict[str, torch.Tensor],
     missing: List[str] = []
     for src, dst in clip_text_key_map(len(model.blocks)):
         optional = dst.startswith("pooled_proj")
-        if src not in sd or dst not in own:
+        # transformers <= 4.x serializes CLIPTextModel with a "text_model."
+        # prefix (what shipped diffusers snapshots contain); 5.x dropped it
+        v = sd.get(src)
+        if v is None:
+            v = sd.get(src.removeprefix("text_model."))
+        if v is None or dst not in own:
             if not optional:
                 missing.append(src)
             continue
-        v = sd[src]
         if v.shape != own[dst].shape:
             raise ValueError(f"shape mismatch {src}: {v.shape} vs {own[dst].shape}")
         own[dst].copy_(v.to(own[dst].dtype))

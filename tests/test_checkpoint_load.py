@@ -401,3 +401,45 @@ def test_bpe_matches_transformers_cliptokenizer(tmp_path):
                    max_length=77)["input_ids"]
         got = ours(t).flatten().tolist()
         assert got == list(want), f"BPE diverged on {t!r}"
+
+
+def test_text_encoder_matches_transformers_clip_bitexact():
+    """Independent architecture validation: with the same weights, our
+    TextEncoder reproduces transformers.CLIPTextModel's hidden states
+    EXACTLY (causal mask, quick-gelu, LN placement, penultimate layer,
+    EOS pooling)."""
+    import pytest
+
+    pytest.importorskip("transformers")
+    from transformers import CLIPTextConfig, CLIPTextModel
+
+    from ai_rtc_agent_amd.models.load import load_clip_text_encoder
+    from ai_rtc_agent_amd.models.text_encoder import TextEncoder
+
+    cfg = CLIPTextConfig(vocab_size=512, hidden_size=64,
+                         intermediate_size=256, num_hidden_layers=2,
+                         num_attention_heads=4, max_position_embeddings=77,
+                         hidden_act="quick_gelu", bos_token_id=510,
+                         eos_token_id=511)
+    torch.manual_seed(0)
+    hf = CLIPTextModel(cfg).eval()
+    ours = TextEncoder(hidden=64, layers=2, heads=4, vocab_size=512,
+                       act="quick_gelu").eval()
+    n = load_clip_text_encoder(ours, hf.state_dict(), strict=True)
+    assert n == 36
+    ids = torch.randint(1, 500, (2, 77))
+    ids[:, 0] = 510
+    ids[0, 40:] = 511
+    with torch.no_grad():
+        hf_last = hf(input_ids=ids).last_hidden_state
+        hf_hidden = hf(input_ids=ids, output_hidden_states=True).hidden_states
+        our_states = ours._hidden_states(ids)
+        our_last = ours.final_ln(our_states[-1])
+    assert torch.equal(our_last, hf_last), "last hidden diverged"
+    assert torch.equal(our_states[-2], hf_hidden[-2]), "penultimate diverged"
+    # EOS pooling convention (highest token id == eos, real CLIP vocab law)
+    eos_pos = ids.argmax(dim=-1)
+    want = hf_last[torch.arange(2), eos_pos]
+    with torch.no_grad():
+        got = our_last[torch.arange(2), eos_pos]
+    assert torch.equal(got, want)
