@@ -109,3 +109,29 @@ def test_ice_helpers_offline(monkeypatch):
     ])
     assert any('rel="ice-server"' in l and 'username="u"' in l for l in links)
     assert any("stun.example.com" in l for l in links)
+
+
+def test_stage_timers_cpu_accounting():
+    """StageTimers: per-stage percentiles and frame FPS accounting (the
+    /stats + rocTX tracing layer) — CPU clock path."""
+    import time
+
+    from ai_rtc_agent_amd.utils.timers import StageTimers, _Percentile
+
+    p = _Percentile()
+    for v in [5.0, 1.0, 3.0, 2.0, 4.0]:
+        p.add(v)
+    assert p.percentile(50) == 3.0
+    assert p.percentile(0) == 1.0 and p.percentile(100) == 5.0
+    assert abs(p.mean() - 3.0) < 1e-9
+
+    t = StageTimers(use_cuda=False)
+    for _ in range(4):
+        with t.stage("work"):
+            time.sleep(0.002)
+        t.frame_done()
+    snap = t.snapshot()
+    assert snap["frames"] == 4
+    assert "work" in snap["stages_ms"]
+    assert snap["stages_ms"]["work"]["p50"] >= 1.5  # ~2ms sleeps
+    assert snap["fps"] > 0
