@@ -956,6 +956,37 @@ struct NnzCtx {
 // ---------------------------------------------------------------------------
 // Encoder
 // ---------------------------------------------------------------------------
+
+// Encoder freedom (big RD win on detailed content): instead of falling
+// back to I_PCM when a quantized block exceeds the implemented CAVLC
+// range (TotalCoeff > 9), ZERO the smallest levels (ties: highest
+// frequency first) until 9 remain. The bitstream stays fully standard;
+// reconstruction uses the same capped coefficients so encoder/decoder
+// stay consistent. Returns the largest |level| dropped (callers keep the
+// I_PCM fallback for blocks where the cap would destroy real structure).
+// a capped-away level above this means the block had strong structure
+// beyond 9 coefficients — I_PCM preserves it exactly (384 B) instead of
+// mushing it. Tail levels (1) are plain noise and always safe to drop.
+static constexpr int PCM_DROP_LIMIT = 2;
+
+static int cap_coeffs9(int32_t* q, int lo, int hi) {
+  int idx[16], n = 0;
+  for (int i = lo; i < hi; ++i)
+    if (q[i]) idx[n++] = i;
+  int dropped = 0;
+  while (n > 9) {
+    int bi = 0;
+    for (int j = 1; j < n; ++j) {
+      const int a = std::abs(q[idx[j]]), b = std::abs(q[idx[bi]]);
+      if (a < b || (a == b && idx[j] > idx[bi])) bi = j;
+    }
+    dropped = std::max(dropped, std::abs(q[idx[bi]]));
+    q[idx[bi]] = 0;
+    idx[bi] = idx[--n];
+  }
+  return dropped;
+}
+
 struct Encoder {
   int w, h, pw, ph, mbw, mbh;
   std::vector<uint8_t> Y, Cb, Cr;        // source (padded)
@@ -1057,6 +1088,7 @@ struct Encoder {
     memset(lq, 0, sizeof(lq));
     int modes[16];
     int maxtc = 0;
+    int dropped_max = 0;  // largest |level| zeroed by the CAVLC cap
     for (int z = 0; z < 16; ++z) {
       const int x4 = blk_x4(z), y4 = blk_y4(z);
       const int px = mbx * 16 + x4 * 4, py = mby * 16 + y4 * 4;
@@ -1119,15 +1151,16 @@ struct Encoder {
                                    (int)best_pred[y * 4 + x]);
       int32_t W[16];
       fwd4x4(d, W);
-      int tc = 0;
       for (int i = 0; i < 16; ++i) {
         int32_t q =
             (int32_t)(((int64_t)std::abs(W[i]) * QMF[qm][POSCLS[i]] + fr) >>
                       qbits);
         if (q > 2063) q = 2063;
         lq[z][i] = W[i] < 0 ? -q : q;
-        tc += q != 0;
       }
+      dropped_max = std::max(dropped_max, cap_coeffs9(lq[z], 0, 16));
+      int tc = 0;
+      for (int i = 0; i < 16; ++i) tc += lq[z][i] != 0;
       maxtc = std::max(maxtc, tc);
       // reconstruct immediately: later blocks predict from these pixels
       int32_t dq[16], rr[16];
@@ -1164,6 +1197,8 @@ struct Encoder {
         fwd4x4(d, W);
         cdc_raw[blk] = W[0];
         quant_block(W, qpc, /*skip_dc=*/true, cac_q[comp][blk]);
+        dropped_max = std::max(dropped_max,
+                               cap_coeffs9(cac_q[comp][blk], 1, 16));
         int tc = 0;
         for (int i = 1; i < 16; ++i) tc += cac_q[comp][blk][i] != 0;
         ctc_max = std::max(ctc_max, tc);
@@ -1180,8 +1215,10 @@ struct Encoder {
         cdc_q[comp][i] = t[i] < 0 ? -q : q;
       }
     }
-    if (maxtc > 9 || ctc_max > 9) {
-      // CAVLC guard (see encode_mb): high-entropy MB -> I_PCM
+    if (maxtc > 9 || ctc_max > 9 || dropped_max > PCM_DROP_LIMIT) {
+      // CAVLC guard: the coefficient cap handles most high-entropy blocks
+      // (cap_coeffs9); PCM remains only for blocks where capping dropped
+      // large levels (real structure, not tail noise)
       for (int z = 0; z < 16; ++z) my_modes[z] = -1;
       write_pcm(wtr, mbx, mby, in_p_slice);
       return;
@@ -1407,12 +1444,14 @@ struct Encoder {
       for (int i = from; i < n; ++i) k += c[i] != 0;
       return k;
     };
-    bool pcm = count_nz(dc_q, 0, 16) > 9;
-    for (int b = 0; b < 16 && !pcm; ++b) pcm = count_nz(ac_q[b], 1, 16) > 9;
-    for (int comp = 0; comp < 2 && !pcm; ++comp)
-      for (int blk = 0; blk < 4 && !pcm; ++blk)
-        pcm = count_nz(cac_q[comp][blk], 1, 16) > 9;
-    if (pcm) {
+    int drop16 = cap_coeffs9(dc_q, 0, 16);
+    for (int b = 0; b < 16; ++b)
+      drop16 = std::max(drop16, cap_coeffs9(ac_q[b], 1, 16));
+    for (int comp = 0; comp < 2; ++comp)
+      for (int blk = 0; blk < 4; ++blk)
+        drop16 = std::max(drop16, cap_coeffs9(cac_q[comp][blk], 1, 16));
+    if (drop16 > PCM_DROP_LIMIT) {
+      // the cap would erase strong structure: keep the exact pixels
       write_pcm(wtr, mbx, mby, in_p_slice);
       return;
     }
