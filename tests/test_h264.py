@@ -653,3 +653,26 @@ def test_encoder_slice_count_changes_reuse_pool():
         au = enc.encode((f1 if i % 2 else f2).numpy().tobytes(), 28,
                         keyframe=(i % 10 == 0))
         assert dec2.decode(au) is not None or len(au) > 0
+
+
+def test_detailed_content_rate_falls_with_qp():
+    """The CAVLC coefficient cap (TotalCoeff<=9 by zeroing smallest levels)
+    must keep R(QP) monotone on detailed content — the old always-I_PCM
+    fallback put a ~384 B/MB floor under it, so the rate controller could
+    never reach low budgets on busy scenes."""
+    import torch as _t
+
+    ext = _h264_ext()
+    y, x = _t.meshgrid(_t.arange(128), _t.arange(128), indexing="ij")
+    checks = (((x // 9) + (y // 7)) % 2 * 200 + 28).to(_t.uint8)
+    f = _t.stack([checks, checks, checks], -1).contiguous()
+    dec = ext.H264SwDecoder()
+    sizes = []
+    for qp in (22, 30, 38, 46):
+        enc = ext.H264SwEncoder(128, 128, 4, 2)
+        au = enc.encode(f.numpy().tobytes(), qp, keyframe=True)
+        assert dec.decode(au) is not None
+        sizes.append(len(au))
+    assert sizes[0] > sizes[-1] * 2, f"rate barely responds to QP: {sizes}"
+    for a, b in zip(sizes, sizes[1:]):
+        assert b <= a * 1.05, f"rate not ~monotone in QP: {sizes}"
