@@ -185,7 +185,10 @@ def build_answer(
             port=port,
             protocol=m.protocol,
             mid=m.mid or str(i),
-            direction=direction,
+            # browsers typically offer audio+video; this agent serves video
+            # only, so every other kind is answered a=inactive (the section
+            # must still be echoed for BUNDLE mid alignment)
+            direction=direction if m.kind == "video" else "inactive",
             ice_ufrag=_rand(8),
             ice_pwd=_rand(24),
             ssrc=ssrc,

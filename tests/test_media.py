@@ -431,3 +431,25 @@ def test_rx_ssrc_lock_on():
         assert dec.calls == [f1]  # only the locked SSRC's frame decoded
 
     asyncio.new_event_loop().run_until_complete(body())
+
+
+def test_answer_marks_non_video_sections_inactive():
+    """Browsers offer audio+video; the agent serves video only — the audio
+    m-section must be echoed (BUNDLE mid alignment) but answered
+    a=inactive so the browser stops expecting audio flow."""
+    from ai_rtc_agent_amd.media.sdp import SessionDescription, build_answer
+
+    offer = SessionDescription.parse(
+        "v=0\r\no=- 1 2 IN IP4 0.0.0.0\r\ns=-\r\nt=0 0\r\n"
+        "a=group:BUNDLE 0 1\r\n"
+        "m=audio 9 UDP/TLS/RTP/SAVPF 111\r\na=mid:0\r\n"
+        "a=rtpmap:111 opus/48000/2\r\n"
+        "m=video 9 UDP/TLS/RTP/SAVPF 96\r\na=mid:1\r\n"
+        "a=rtpmap:96 H264/90000\r\n")
+    ans = build_answer(offer, "127.0.0.1", 5004, "H264", ssrc=7)
+    assert [m.kind for m in ans.media] == ["audio", "video"]
+    assert ans.media[0].direction == "inactive"
+    assert ans.media[1].direction == "sendrecv"
+    assert ans.media[0].mid == "0" and ans.media[1].mid == "1"
+    txt = ans.serialize()
+    assert "a=inactive" in txt and "a=group:BUNDLE 0 1" in txt
