@@ -179,6 +179,10 @@ def build_answer(
     offerer is the DTLS client)."""
     ans = SessionDescription(session_id=str(random.randint(10**8, 10**9)),
                              bundle=offer.bundle)
+    # one transport for the whole (bundled) answer -> ONE ICE credential
+    # pair shared by every m-section (RFC 8843: same transport, same
+    # ufrag/pwd; browsers reject per-section credentials within a bundle)
+    ufrag, pwd = _rand(8), _rand(24)
     for i, m in enumerate(offer.media):
         sec = MediaSection(
             kind=m.kind,
@@ -189,8 +193,8 @@ def build_answer(
             # only, so every other kind is answered a=inactive (the section
             # must still be echoed for BUNDLE mid alignment)
             direction=direction if m.kind == "video" else "inactive",
-            ice_ufrag=_rand(8),
-            ice_pwd=_rand(24),
+            ice_ufrag=ufrag,
+            ice_pwd=pwd,
             ssrc=ssrc,
             setup="passive",
             fingerprint=fingerprint,

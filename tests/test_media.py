@@ -453,3 +453,6 @@ def test_answer_marks_non_video_sections_inactive():
     assert ans.media[0].mid == "0" and ans.media[1].mid == "1"
     txt = ans.serialize()
     assert "a=inactive" in txt and "a=group:BUNDLE 0 1" in txt
+    # bundled sections share ONE transport -> identical ICE credentials
+    assert ans.media[0].ice_ufrag == ans.media[1].ice_ufrag
+    assert ans.media[0].ice_pwd == ans.media[1].ice_pwd
