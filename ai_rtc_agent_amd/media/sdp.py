@@ -55,6 +55,7 @@ class SessionDescription:
     media: List[MediaSection] = field(default_factory=list)
     fingerprint: Optional[str] = None  # session-level a=fingerprint
     bundle: Optional[str] = None       # a=group:BUNDLE mids (echoed back)
+    ice_lite: bool = False             # a=ice-lite (we answer as lite agent)
 
     @staticmethod
     def parse(sdp: str) -> "SessionDescription":
@@ -73,6 +74,8 @@ class SessionDescription:
                 sd.fingerprint = line.split(":", 1)[1].strip()
             elif line.startswith("a=group:BUNDLE") and cur is None:
                 sd.bundle = line[len("a=group:"):].strip()
+            elif line == "a=ice-lite" and cur is None:
+                sd.ice_lite = True
             elif line.startswith("m="):
                 parts = line[2:].split()
                 cur = MediaSection(kind=parts[0], port=int(parts[1]), protocol=parts[2])
@@ -128,6 +131,10 @@ class SessionDescription:
             "s=-",
             "t=0 0",
         ]
+        if self.ice_lite:
+            # RFC 8445 ICE-lite: tells the full agent (the browser) it owns
+            # the connectivity checks and the controlling role
+            lines.append("a=ice-lite")
         if self.bundle:
             lines.append(f"a=group:{self.bundle}")
         for m in self.media:
@@ -178,7 +185,7 @@ def build_answer(
     credentials + host candidate (+ DTLS fingerprint, setup:passive — the
     offerer is the DTLS client)."""
     ans = SessionDescription(session_id=str(random.randint(10**8, 10**9)),
-                             bundle=offer.bundle)
+                             bundle=offer.bundle, ice_lite=True)
     # one transport for the whole (bundled) answer -> ONE ICE credential
     # pair shared by every m-section (RFC 8843: same transport, same
     # ufrag/pwd; browsers reject per-section credentials within a bundle)

@@ -456,3 +456,5 @@ def test_answer_marks_non_video_sections_inactive():
     # bundled sections share ONE transport -> identical ICE credentials
     assert ans.media[0].ice_ufrag == ans.media[1].ice_ufrag
     assert ans.media[0].ice_pwd == ans.media[1].ice_pwd
+    # lite answerer must say so (RFC 8445): the browser runs the checks
+    assert "a=ice-lite" in txt
